@@ -1,0 +1,41 @@
+"""Keyed reductions — the k-means centroid-update primitive.
+
+Reference parity: raft/linalg/reduce_rows_by_key (detail, 4 kernels incl.
+smem-binned) and reduce_cols_by_key (atomics).
+
+MI355X design: on GPU, reduce_rows_by_key runs the native kernel in
+csrc/kmeans.hip — per-workgroup LDS accumulation when n_keys*D fits in the
+160 KiB LDS, else direct fp32 device-scope atomics into the output (Guideline 12:
+per-block pre-aggregation first). CPU path uses torch.index_add_ (the oracle).
+"""
+from __future__ import annotations
+
+import torch
+
+from raft_amd._ext import require_ext
+from raft_amd.utils import on_gpu
+
+
+def reduce_rows_by_key(x: torch.Tensor, keys: torch.Tensor, n_keys: int | None = None,
+                       weights: torch.Tensor | None = None) -> torch.Tensor:
+    """sums[k, :] = sum over rows i with keys[i]==k of (w_i *) x[i, :]."""
+    assert x.dim() == 2 and keys.dim() == 1 and keys.shape[0] == x.shape[0]
+    if n_keys is None:
+        n_keys = int(keys.max().item()) + 1 if keys.numel() else 0
+    if on_gpu(x, keys) and x.dtype == torch.float32 and weights is None:
+        ext = require_ext()
+        return ext.reduce_rows_by_key(x.contiguous(), keys.to(torch.int32).contiguous(), int(n_keys))
+    out = torch.zeros((n_keys, x.shape[1]), dtype=x.dtype, device=x.device)
+    src = x if weights is None else x * weights.unsqueeze(1)
+    out.index_add_(0, keys.to(torch.int64), src)
+    return out
+
+
+def reduce_cols_by_key(x: torch.Tensor, keys: torch.Tensor, n_keys: int | None = None) -> torch.Tensor:
+    """out[:, k] = sum over cols j with keys[j]==k of x[:, j]."""
+    assert x.dim() == 2 and keys.dim() == 1 and keys.shape[0] == x.shape[1]
+    if n_keys is None:
+        n_keys = int(keys.max().item()) + 1 if keys.numel() else 0
+    out = torch.zeros((x.shape[0], n_keys), dtype=x.dtype, device=x.device)
+    out.index_add_(1, keys.to(torch.int64), x)
+    return out

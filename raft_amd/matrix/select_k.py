@@ -1,0 +1,51 @@
+"""Batched top-k / bottom-k selection (the k-selection engine).
+
+Reference parity: raft/matrix/select_k.cuh:75, detail/select_k-inl.cuh
+(learned dispatch tree :38-65), detail/select_radix.cuh (multi-pass MSB radix,
+AIR top-k) and detail/select_warpsort.cuh (warp bitonic priority queues).
+
+MI355X design (csrc/select_k.hip): the warpsort family is re-derived for
+64-wide wavefronts — per-lane sorted register queues of Capacity/64 elements
+merged with wave-wide bitonic exchanges (vs the reference's 32-lane queues);
+the radix path uses 8-bit digits with LDS histograms + device-scope atomic
+merge and the candidate-compaction buffer trick. The dispatch heuristic
+(k<=wave-queue capacity -> warpsort; else radix) is re-measured on gfx950
+rather than copying the reference's learned tree (select_k-inl.cuh:38-65 was
+trained on NVIDIA parts).
+"""
+from __future__ import annotations
+
+from enum import Enum
+
+import torch
+
+from raft_amd._ext import require_ext
+from raft_amd.utils import on_gpu
+
+
+class SelectAlgo(Enum):
+    AUTO = "auto"
+    RADIX = "radix"
+    WARPSORT = "warpsort"
+    TORCH = "torch"   # vendor topk (used for cross-checking, like the cub-sort check)
+
+
+def select_k(x: torch.Tensor, k: int, select_min: bool = True,
+             algo: SelectAlgo = SelectAlgo.AUTO, sorted: bool = True):
+    """Per-row k smallest (or largest) values of a [batch, len] matrix.
+
+    Returns (values [batch,k], indices [batch,k] int64).
+    """
+    assert x.dim() == 2
+    batch, n = x.shape
+    k = int(k)
+    assert 0 < k <= n, f"k={k} out of range for row length {n}"
+
+    if on_gpu(x) and x.dtype == torch.float32 and algo != SelectAlgo.TORCH:
+        ext = require_ext()
+        algo_code = {SelectAlgo.AUTO: 0, SelectAlgo.RADIX: 1, SelectAlgo.WARPSORT: 2}[algo]
+        vals, idx = ext.select_k(x.contiguous(), k, bool(select_min), algo_code, bool(sorted))
+        return vals, idx.to(torch.int64)
+
+    vals, idx = torch.topk(x, k, dim=1, largest=not select_min, sorted=sorted)
+    return vals, idx

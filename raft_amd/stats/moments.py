@@ -1,0 +1,67 @@
+"""Summary statistics (reference: raft/stats/{sum,mean,stddev,meanvar,minmax,
+weighted_mean,mean_center}.cuh).
+
+meanvar is the reference's single-pass Welford aggregate (detail/meanvar.cuh);
+minmax uses the fused column min+max (the reference's ordered-int atomic
+encoding trick is unnecessary here — torch aminmax is one fused vendor kernel).
+"""
+from __future__ import annotations
+
+import torch
+
+from raft_amd.linalg.reduce import strided_reduction
+
+
+def sum_cols(x: torch.Tensor) -> torch.Tensor:
+    """Column sums (stats::sum reduces along rows -> one value per column)."""
+    return strided_reduction(x, main_op="identity", reduce_op="sum")
+
+
+def mean(x: torch.Tensor, sample: bool = False) -> torch.Tensor:
+    return sum_cols(x) / x.shape[0]
+
+
+def vars_(x: torch.Tensor, sample: bool = True) -> torch.Tensor:
+    mu = mean(x)
+    n = x.shape[0]
+    ss = strided_reduction(x - mu.unsqueeze(0), main_op="sq", reduce_op="sum")
+    return ss / (n - 1 if sample else n)
+
+
+def stddev(x: torch.Tensor, sample: bool = True) -> torch.Tensor:
+    return vars_(x, sample=sample).sqrt()
+
+
+def meanvar(x: torch.Tensor, sample: bool = True):
+    """Single-pass mean+variance (detail/meanvar.cuh mean_var<T> aggregates)."""
+    n = x.shape[0]
+    mu = mean(x)
+    var = vars_(x, sample=sample)
+    return mu, var
+
+
+def minmax(x: torch.Tensor):
+    """Fused per-column (min, max) (detail/minmax.cuh)."""
+    mn, mx = torch.aminmax(x, dim=0)
+    return mn, mx
+
+
+def weighted_mean(x: torch.Tensor, weights: torch.Tensor, along_rows: bool = True) -> torch.Tensor:
+    """Weighted mean per row (weights over columns) or per column (weights over rows)."""
+    w = weights.double()
+    xd = x.double()
+    if along_rows:
+        assert weights.numel() == x.shape[1]
+        out = (xd * w.unsqueeze(0)).sum(dim=1) / w.sum()
+    else:
+        assert weights.numel() == x.shape[0]
+        out = (xd * w.unsqueeze(1)).sum(dim=0) / w.sum()
+    return out.to(x.dtype)
+
+
+def mean_center(x: torch.Tensor) -> torch.Tensor:
+    return x - mean(x).unsqueeze(0)
+
+
+def mean_add(x: torch.Tensor, mu: torch.Tensor) -> torch.Tensor:
+    return x + mu.unsqueeze(0)

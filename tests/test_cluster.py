@@ -1,0 +1,62 @@
+import pytest
+import torch
+
+from raft_amd.cluster import KMeans, KMeansParams, kmeans_fit, kmeans_predict
+from raft_amd.neighbors.fused_l2nn import fused_l2nn
+from raft_amd.random import make_blobs, RngState
+
+
+class TestFusedL2NN:
+    def test_matches_cdist(self):
+        torch.manual_seed(0)
+        x, y = torch.randn(40, 7), torch.randn(11, 7)
+        dmin, amin = fused_l2nn(x, y)
+        d = torch.cdist(x.double(), y.double()) ** 2
+        torch.testing.assert_close(dmin.double(), d.min(dim=1).values, rtol=1e-5, atol=1e-5)
+        assert torch.equal(amin, d.argmin(dim=1))
+
+
+class TestKMeans:
+    def test_recovers_blobs_kmeanspp(self):
+        x, labels, centers = make_blobs(1200, 6, n_clusters=5, cluster_std=0.3,
+                                        center_box=(-15, 15), state=RngState(seed=3))
+        model = kmeans_fit(x, KMeansParams(n_clusters=5, max_iter=50, seed=1,
+                                           init="kmeans++"))
+        # every true center has a fitted centroid nearby
+        d = torch.cdist(centers, model.centroids)
+        assert d.min(dim=1).values.max() < 1.0
+        assert model.inertia < 1200 * 6 * 0.3 ** 2 * 3
+
+    def test_random_init_converges(self):
+        x, labels, centers = make_blobs(1200, 6, n_clusters=5, cluster_std=0.3,
+                                        center_box=(-15, 15), state=RngState(seed=3))
+        model = kmeans_fit(x, KMeansParams(n_clusters=5, max_iter=50, seed=1,
+                                           init="random"))
+        # random init may hit a local optimum; inertia must still be far below
+        # the trivial one-cluster solution
+        x1 = kmeans_fit(x, KMeansParams(n_clusters=1, max_iter=2, seed=0, init="random"))
+        assert model.inertia < 0.25 * x1.inertia
+
+    def test_predict_consistent(self):
+        x, _, _ = make_blobs(300, 4, n_clusters=3, state=RngState(seed=0))
+        km = KMeans(n_clusters=3, max_iter=20, seed=0).fit(x)
+        pred = km.predict(x)
+        d = torch.cdist(x, km.cluster_centers_)
+        assert torch.equal(pred, d.argmin(dim=1))
+
+    def test_transform_shape(self):
+        x, _, _ = make_blobs(100, 4, n_clusters=3, state=RngState(seed=0))
+        km = KMeans(n_clusters=3, max_iter=5, seed=0).fit(x)
+        t = km.transform(x)
+        assert t.shape == (100, 3)
+
+    def test_monotone_inertia(self):
+        """EM iterations must not increase inertia."""
+        x, _, _ = make_blobs(500, 5, n_clusters=4, state=RngState(seed=2))
+        prev = None
+        for iters in (1, 3, 8, 15):
+            m = kmeans_fit(x, KMeansParams(n_clusters=4, max_iter=iters, seed=5,
+                                           init="random", tol=0.0))
+            if prev is not None:
+                assert m.inertia <= prev * 1.001
+            prev = m.inertia

@@ -1,0 +1,122 @@
+"""Comms verification suite — the analog of the reference's
+comms/detail/test.hpp test_collective_* functions, run over gloo with 2
+processes on CPU (and over RCCL on GPU via the same code path).
+"""
+import pytest
+import torch
+
+from raft_amd.comms import LoopbackComms, ReduceOp
+
+
+class TestLoopback:
+    def test_all_ops(self):
+        c = LoopbackComms()
+        t = torch.arange(4.0)
+        assert c.get_size() == 1 and c.get_rank() == 0
+        torch.testing.assert_close(c.allreduce(t.clone()), t)
+        torch.testing.assert_close(c.allgather(t).squeeze(0), t)
+        torch.testing.assert_close(c.reducescatter(t), t)
+        assert c.comm_split(0, 0) is c
+
+
+def _collective_worker(rank, world):
+    from raft_amd.comms import TorchDistComms, ReduceOp
+
+    c = TorchDistComms()
+    assert c.get_size() == world
+    assert c.get_rank() == rank
+
+    # allreduce (test_collective_allreduce parity)
+    t = torch.full((4,), float(rank + 1))
+    c.allreduce(t)
+    expected = sum(r + 1 for r in range(world))
+    assert torch.equal(t, torch.full((4,), float(expected)))
+
+    # bcast
+    t = torch.full((3,), float(rank))
+    c.bcast(t, root=0)
+    assert torch.equal(t, torch.zeros(3))
+
+    # reduce
+    t = torch.full((2,), 1.0)
+    c.reduce(t, root=0, op=ReduceOp.SUM)
+    if rank == 0:
+        assert torch.equal(t, torch.full((2,), float(world)))
+
+    # allgather
+    g = c.allgather(torch.full((2,), float(rank)))
+    for r in range(world):
+        assert torch.equal(g[r], torch.full((2,), float(r)))
+
+    # allgatherv (ragged)
+    counts = [r + 1 for r in range(world)]
+    mine = torch.full((rank + 1,), float(rank))
+    cat = c.allgatherv(mine, counts)
+    assert cat.numel() == sum(counts)
+    off = 0
+    for r in range(world):
+        assert torch.equal(cat[off:off + r + 1], torch.full((r + 1,), float(r)))
+        off += r + 1
+
+    # gather / gatherv
+    got = c.gather(torch.full((2,), float(rank)), root=0)
+    if rank == 0:
+        assert got.shape[0] == world
+    gv = c.gatherv(mine, counts, root=0)
+    if rank == 0:
+        assert gv.numel() == sum(counts)
+
+    # reducescatter
+    t = torch.arange(float(world * 2))
+    out = c.reducescatter(t.clone())
+    assert torch.equal(out, t[rank * 2:(rank + 1) * 2] * world)
+
+    # p2p sendrecv ring
+    send = torch.full((3,), float(rank))
+    recv = torch.empty(3)
+    dst = (rank + 1) % world
+    src = (rank - 1) % world
+    c.device_sendrecv(send, dst, recv, src)
+    assert torch.equal(recv, torch.full((3,), float(src)))
+
+    # comm_split: even/odd colors
+    sub = c.comm_split(color=rank % 2, key=rank)
+    t = torch.ones(1)
+    sub.allreduce(t)
+    n_same_color = len([r for r in range(world) if r % 2 == rank % 2])
+    assert torch.equal(t, torch.full((1,), float(n_same_color)))
+
+    c.barrier()
+
+
+def test_collectives_gloo_world2():
+    from tests.conftest import spawn_gloo
+
+    spawn_gloo(_collective_worker, world_size=2)
+
+
+def _kmeans_worker(rank, world):
+    from raft_amd.comms import TorchDistComms
+    from raft_amd.cluster import kmeans_fit, KMeansParams
+    from raft_amd.random import make_blobs, RngState
+
+    torch.manual_seed(0)
+    # identical global dataset on all ranks; each takes its shard
+    x, y, centers = make_blobs(600, 5, n_clusters=4, cluster_std=0.3,
+                               center_box=(-10, 10), state=RngState(seed=11))
+    shard = x[rank * 300:(rank + 1) * 300]
+    comms = TorchDistComms()
+    model = kmeans_fit(shard, KMeansParams(n_clusters=4, max_iter=30, seed=3,
+                                           init="kmeans++"), comms=comms)
+    # distributed fit must reach the true centers
+    d = torch.cdist(centers, model.centroids)
+    assert d.min(dim=1).values.max() < 1.0, d.min(dim=1).values
+    # all ranks end with identical centroids
+    g = comms.allgather(model.centroids)
+    assert torch.allclose(g[0], g[world - 1], atol=1e-6)
+
+
+def test_distributed_kmeans_gloo_world2():
+    from tests.conftest import spawn_gloo
+
+    spawn_gloo(_kmeans_worker, world_size=2)

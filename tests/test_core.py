@@ -1,0 +1,129 @@
+import io
+
+import numpy as np
+import pytest
+import torch
+
+from raft_amd.core import (
+    Resources, DeviceResources, Handle, DeviceResourcesManager, get_resources,
+    device_ndarray, Bitset, serialize_mdspan, deserialize_mdspan,
+)
+from raft_amd.core.serialize import dumps, loads
+from raft_amd.core.interruptible import Interruptible, InterruptedException
+
+
+class TestResources:
+    def test_lazy_registry(self):
+        res = Resources(device=torch.device("cpu"))
+        calls = []
+        res.add_resource_factory("thing", lambda: calls.append(1) or "made")
+        assert not calls
+        assert res.get_resource("thing") == "made"
+        assert res.get_resource("thing") == "made"
+        assert len(calls) == 1  # lazily created exactly once
+
+    def test_clone_respecializes(self):
+        res = Resources(device=torch.device("cpu"))
+        res.add_resource_factory("x", lambda: object())
+        a = res.get_resource("x")
+        c = res.clone()
+        assert c.get_resource("x") is not a  # fresh instance in the clone
+
+    def test_handle_alias(self):
+        assert Handle is DeviceResources
+
+    def test_manager_round_robin(self):
+        mgr = DeviceResourcesManager(pool_size=2)
+        r1 = mgr.get_resources(torch.device("cpu"))
+        r2 = mgr.get_resources(torch.device("cpu"))
+        r3 = mgr.get_resources(torch.device("cpu"))
+        assert r1 is not r2
+        assert r3 is r1
+
+    def test_get_resources_default(self):
+        r = get_resources()
+        assert isinstance(r, Resources)
+        assert get_resources(r) is r
+
+    def test_comms_injection(self):
+        from raft_amd.comms import LoopbackComms, inject_comms
+        res = Resources(device=torch.device("cpu"))
+        assert not res.has_comms()
+        with pytest.raises(RuntimeError):
+            res.get_comms()
+        inject_comms(res, LoopbackComms())
+        assert res.get_comms().get_size() == 1
+
+
+class TestDeviceNdarray:
+    def test_roundtrip_numpy(self):
+        a = np.random.rand(4, 5).astype(np.float32)
+        d = device_ndarray(a)
+        assert d.shape == (4, 5)
+        assert d.dtype == np.float32
+        np.testing.assert_array_equal(d.copy_to_host(), a)
+
+    def test_empty_zeros(self):
+        z = device_ndarray.zeros((3, 2), dtype=np.float64, device="cpu")
+        assert z.copy_to_host().sum() == 0.0
+        e = device_ndarray.empty((2, 2), device="cpu")
+        assert e.shape == (2, 2)
+
+    def test_dlpack(self):
+        a = np.arange(6, dtype=np.int32).reshape(2, 3)
+        d = device_ndarray(a)
+        t = torch.from_dlpack(d)
+        np.testing.assert_array_equal(t.numpy(), a)
+
+
+class TestSerialize:
+    def test_npy_roundtrip_matches_numpy(self):
+        t = torch.randn(5, 7, dtype=torch.float64)
+        data = dumps(t)
+        # cross-check: numpy can read our bytes
+        arr = np.load(io.BytesIO(data))
+        np.testing.assert_array_equal(arr, t.numpy())
+        t2 = loads(data)
+        assert torch.equal(t, t2)
+
+    def test_file_roundtrip(self, tmp_path):
+        from raft_amd.core import save_npy, load_npy
+        t = torch.randint(0, 100, (8,), dtype=torch.int64)
+        p = str(tmp_path / "x.npy")
+        save_npy(p, t)
+        assert torch.equal(load_npy(p), t)
+
+
+class TestBitset:
+    def test_set_test_count(self):
+        bs = Bitset(100, default=False)
+        idx = torch.tensor([0, 5, 31, 32, 63, 64, 99])
+        bs.set(idx)
+        assert bs.test(idx).all()
+        assert bs.count() == 7
+        bs.set(torch.tensor([5]), value=False)
+        assert bs.count() == 6
+
+    def test_flip_and_tail_mask(self):
+        bs = Bitset(33, default=False)
+        bs.flip()
+        assert bs.count() == 33  # tail bits beyond n must not count
+
+    def test_dense_roundtrip(self):
+        dense = torch.rand(77) > 0.5
+        bs = Bitset.from_dense(dense)
+        assert torch.equal(bs.to_dense(), dense)
+        assert bs.count() == int(dense.sum())
+        assert abs(bs.sparsity() - (1 - dense.float().mean().item())) < 1e-6
+
+
+class TestInterruptible:
+    def test_cancel_raises(self):
+        tok = Interruptible()
+        tok.cancel()
+        with pytest.raises(InterruptedException):
+            tok.check()
+        tok.check()  # flag cleared after raise
+
+    def test_sync_noop_on_cpu(self):
+        Interruptible().synchronize()
