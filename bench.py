@@ -1,0 +1,153 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: k-means fit (BASELINE config 3) on 1..8 MI355X.
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`; for N>1 the
+driver launches via torch.distributed.run with one rank per GPU over RCCL.
+Rank 0 prints ONE JSON line with the whole-job metric.
+
+Workload: kmeans::fit on synthetic make_blobs data, 10M x 256 fp32, k=1024 —
+the BASELINE.json config. Strong scaling: the 10M rows are sharded across
+ranks; metric = Lloyd iterations/sec of the WHOLE job (one iteration = full
+assignment of all 10M rows + centroid update + allreduce).
+
+fp32 on CDNA4 has no MFMA; the default engine is split-bf16 fp32 emulation
+(bf16x3: fp32-class accuracy, validated in tests/test_linalg.py and below via
+an inertia cross-check) on the 2.5 PF bf16 matrix cores. --fp32-mode native
+uses rocBLAS SGEMM (157 TF vector ALU) for comparison.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=3)
+    p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--rows", type=int, default=10_000_000, help="total rows (global)")
+    p.add_argument("--dim", type=int, default=256)
+    p.add_argument("--k", type=int, default=1024)
+    p.add_argument("--fp32-mode", default="bf16x3",
+                   choices=["bf16x3", "bf16x2", "native"],
+                   help="fp32 GEMM engine for the assignment step")
+    p.add_argument("--chunk-rows", type=int, default=262144)
+    p.add_argument("--seed", type=int, default=42)
+    p.add_argument("--check", action="store_true",
+                   help="cross-check final inertia vs native-fp32 assignment")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    use_gpu = torch.cuda.is_available()
+    if use_gpu:
+        torch.cuda.set_device(local_rank % max(torch.cuda.device_count(), 1))
+        device = torch.device("cuda")
+    else:
+        device = torch.device("cpu")
+
+    from raft_amd.comms import init_comms as comms_init
+    from raft_amd.comms import LoopbackComms
+    from raft_amd.cluster.kmeans import kmeans_iterate
+    from raft_amd.random import make_blobs, RngState
+    from raft_amd.neighbors.fused_l2nn import fused_l2nn
+
+    if world > 1:
+        comms = comms_init()
+    else:
+        comms = LoopbackComms()
+
+    # ---- synthetic data: this rank's shard of the global 10M x 256 ----------
+    rows_local = args.rows // world + (1 if rank < args.rows % world else 0)
+    state = RngState(seed=args.seed + rank)  # independent shards, same blob centers
+    centers_state = RngState(seed=args.seed)
+    from raft_amd.random.rng import uniform
+    true_centers = uniform((args.k, args.dim), -10.0, 10.0, state=centers_state,
+                           device=device)
+    x, _, _ = make_blobs(rows_local, args.dim, n_clusters=args.k, cluster_std=1.0,
+                         centers=true_centers, state=state, device=device)
+    x = x.contiguous()
+
+    # ---- init centroids: rank 0's first k rows of blob centers + jitter -----
+    centroids = true_centers.clone()
+    jit = RngState(seed=args.seed + 777)
+    centroids += uniform((args.k, args.dim), -0.5, 0.5, state=jit, device=device)
+    if world > 1:
+        comms.bcast(centroids, root=0)
+
+    def sync():
+        if world > 1:
+            comms.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    # ---- warmup -------------------------------------------------------------
+    c = centroids.clone()
+    c, _ = kmeans_iterate(x, c, max(args.warmup, 0), comms=comms,
+                          fp32_mode=args.fp32_mode, chunk_rows=args.chunk_rows)
+    sync()
+
+    # ---- timed: EXACTLY args.steps iterations -------------------------------
+    t0 = time.perf_counter()
+    c, inertia = kmeans_iterate(x, c, args.steps, comms=comms,
+                                fp32_mode=args.fp32_mode, chunk_rows=args.chunk_rows)
+    sync()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if torch.distributed.get_backend() == "nccl" else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    iters_per_sec = args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    result = {
+        "metric": f"kmeans-fit iters/sec ({args.rows} x {args.dim} fp32, k={args.k})",
+        "value": round(iters_per_sec, 6),
+        "unit": "iters/sec",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(ms_per_step, 3),
+        "higher_is_better": True,
+        "scaling": "strong",
+        "vs_baseline": None,
+        "dtype": "fp32",
+        "data": "synthetic make_blobs (random-init centers, generated on-device)",
+        "config": {
+            "model": "kmeans-lloyd",
+            "global_batch": args.rows,
+            "seq_len": args.dim,
+            "n_clusters": args.k,
+            "parallelism": f"dp{world}",
+            "fp32_engine": args.fp32_mode,
+            "inertia": inertia,
+        },
+    }
+
+    if args.check and use_gpu:
+        # assignment agreement between emulated and native fp32 engines
+        d_emul, a_emul = fused_l2nn(x[:65536], c, fp32_mode=args.fp32_mode)
+        d_nat, a_nat = fused_l2nn(x[:65536], c, fp32_mode="native")
+        agree = float((a_emul == a_nat).float().mean().item())
+        result["config"]["assign_agreement_vs_native_fp32"] = agree
+
+    if rank == 0:
+        print(json.dumps(result), flush=True)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,256 @@
+// Python bindings for the raft_amd native extension (raft_amd._C).
+// Torch tensors in, torch tensors out; every launch goes onto the current
+// PyTorch HIP stream so the ops compose with torch's own kernels.
+
+#include <torch/extension.h>
+
+#include <hip/hip_runtime.h>
+#include <c10/hip/HIPStream.h>
+
+namespace raft_amd {
+
+// from reductions.hip
+template <int OP, typename T>
+void launch_reduce_rows(const T*, T*, long long, long long, hipStream_t);
+template <int OP, typename T>
+void launch_reduce_cols(const T*, T*, long long, long long, hipStream_t);
+void launch_row_argmin(const float*, int*, long long, long long, hipStream_t);
+void launch_row_normalize_l2(const float*, float*, long long, long long, float, hipStream_t);
+// from pairwise.hip
+void launch_l2_epilogue(float*, const float*, const float*, long long, long long, hipStream_t);
+void launch_l2nn_epilogue(const float*, const float*, const float*, float*, int*,
+                          long long, long long, hipStream_t);
+void launch_pairwise_unexpanded(const float*, const float*, float*, long long, long long,
+                                long long, int, float, hipStream_t);
+// from rng.hip
+void launch_rng_uniform(float*, long long, uint64_t, uint64_t, hipStream_t);
+void launch_rng_normal(float*, long long, uint64_t, uint64_t, hipStream_t);
+void launch_make_blobs(float*, int*, const float*, long long, long long, int, float,
+                       uint64_t, uint64_t, hipStream_t);
+// from spmv.hip
+template <typename T>
+void launch_csr_spmv(const int*, const int*, const T*, const T*, T*, long long,
+                     long long, hipStream_t);
+// from kmeans.hip
+void launch_reduce_rows_by_key(const float*, const int*, float*, long long, long long,
+                               hipStream_t);
+// from select_k.hip
+void launch_select_k(const float*, float*, int*, long long, long long, int, bool, bool,
+                     hipStream_t);
+// from gemm_rocblas.cpp
+void gemm_bf16_f32_rowmajor(const void*, const void*, float*, long long, long long,
+                            long long, float, void*);
+void gemm_f32_rowmajor(const float*, const float*, float*, long long, long long,
+                       long long, float, void*);
+}  // namespace raft_amd
+
+namespace {
+
+hipStream_t cur_stream() {
+  return (hipStream_t)c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_f32_2d(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.dim() == 2, name, " must be 2D");
+}
+
+torch::Tensor reduce_rows(torch::Tensor x, int64_t op) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 2);
+  auto out = torch::empty({x.size(0)}, x.options());
+  auto s = cur_stream();
+  const long long m = x.size(0), d = x.size(1);
+#define CASE(OPC)                                                                     \
+  case OPC:                                                                           \
+    if (x.scalar_type() == torch::kFloat32)                                           \
+      raft_amd::launch_reduce_rows<OPC, float>(x.data_ptr<float>(), out.data_ptr<float>(), m, d, s); \
+    else                                                                              \
+      raft_amd::launch_reduce_rows<OPC, double>(x.data_ptr<double>(), out.data_ptr<double>(), m, d, s); \
+    break;
+  switch (op) { CASE(0) CASE(1) CASE(2) CASE(3) CASE(4) CASE(5)
+    default: TORCH_CHECK(false, "bad op code"); }
+#undef CASE
+  return out;
+}
+
+torch::Tensor reduce_cols(torch::Tensor x, int64_t op) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 2);
+  auto out = torch::empty({x.size(1)}, x.options());
+  auto s = cur_stream();
+  const long long m = x.size(0), d = x.size(1);
+#define CASE(OPC)                                                                     \
+  case OPC:                                                                           \
+    if (x.scalar_type() == torch::kFloat32)                                           \
+      raft_amd::launch_reduce_cols<OPC, float>(x.data_ptr<float>(), out.data_ptr<float>(), m, d, s); \
+    else                                                                              \
+      raft_amd::launch_reduce_cols<OPC, double>(x.data_ptr<double>(), out.data_ptr<double>(), m, d, s); \
+    break;
+  switch (op) { CASE(0) CASE(1) CASE(2) CASE(3) CASE(4) CASE(5)
+    default: TORCH_CHECK(false, "bad op code"); }
+#undef CASE
+  return out;
+}
+
+torch::Tensor row_argmin(torch::Tensor x) {
+  check_f32_2d(x, "x");
+  auto out = torch::empty({x.size(0)}, x.options().dtype(torch::kInt32));
+  raft_amd::launch_row_argmin(x.data_ptr<float>(), out.data_ptr<int>(), x.size(0),
+                              x.size(1), cur_stream());
+  return out;
+}
+
+torch::Tensor row_normalize_l2(torch::Tensor x, double eps) {
+  check_f32_2d(x, "x");
+  auto out = torch::empty_like(x);
+  raft_amd::launch_row_normalize_l2(x.data_ptr<float>(), out.data_ptr<float>(),
+                                    x.size(0), x.size(1), (float)eps, cur_stream());
+  return out;
+}
+
+torch::Tensor l2_epilogue_(torch::Tensor g, torch::Tensor xn, torch::Tensor yn) {
+  check_f32_2d(g, "g");
+  raft_amd::launch_l2_epilogue(g.data_ptr<float>(), xn.data_ptr<float>(),
+                               yn.data_ptr<float>(), g.size(0), g.size(1), cur_stream());
+  return g;
+}
+
+void l2nn_epilogue(torch::Tensor g, torch::Tensor xn, torch::Tensor yn,
+                   torch::Tensor dmin, torch::Tensor amin) {
+  check_f32_2d(g, "g");
+  raft_amd::launch_l2nn_epilogue(g.data_ptr<float>(), xn.data_ptr<float>(),
+                                 yn.data_ptr<float>(), dmin.data_ptr<float>(),
+                                 amin.data_ptr<int>(), g.size(0), g.size(1), cur_stream());
+}
+
+torch::Tensor pairwise_unexpanded(torch::Tensor x, torch::Tensor y, int64_t code,
+                                  double p) {
+  check_f32_2d(x, "x");
+  check_f32_2d(y, "y");
+  auto out = torch::empty({x.size(0), y.size(0)}, x.options());
+  raft_amd::launch_pairwise_unexpanded(x.data_ptr<float>(), y.data_ptr<float>(),
+                                       out.data_ptr<float>(), x.size(0), y.size(0),
+                                       x.size(1), (int)code, (float)p, cur_stream());
+  return out;
+}
+
+torch::Tensor rng_uniform(int64_t n, int64_t seed, int64_t subseq, int64_t device) {
+  auto out = torch::empty({n}, torch::dtype(torch::kFloat32).device(torch::kCUDA, device));
+  raft_amd::launch_rng_uniform(out.data_ptr<float>(), n, (uint64_t)seed,
+                               (uint64_t)subseq, cur_stream());
+  return out;
+}
+
+torch::Tensor rng_normal(int64_t n, int64_t seed, int64_t subseq, int64_t device) {
+  auto out = torch::empty({n}, torch::dtype(torch::kFloat32).device(torch::kCUDA, device));
+  raft_amd::launch_rng_normal(out.data_ptr<float>(), n, (uint64_t)seed,
+                              (uint64_t)subseq, cur_stream());
+  return out;
+}
+
+std::tuple<torch::Tensor, torch::Tensor> make_blobs(int64_t n_rows, int64_t d,
+                                                    torch::Tensor centers, double std_,
+                                                    int64_t seed, int64_t subseq) {
+  check_f32_2d(centers, "centers");
+  auto x = torch::empty({n_rows, d}, centers.options());
+  auto labels = torch::empty({n_rows}, centers.options().dtype(torch::kInt32));
+  raft_amd::launch_make_blobs(x.data_ptr<float>(), labels.data_ptr<int>(),
+                              centers.data_ptr<float>(), n_rows, d,
+                              (int)centers.size(0), (float)std_, (uint64_t)seed,
+                              (uint64_t)subseq, cur_stream());
+  return {x, labels};
+}
+
+torch::Tensor csr_spmv(torch::Tensor indptr, torch::Tensor indices, torch::Tensor values,
+                       torch::Tensor x, int64_t n_rows) {
+  TORCH_CHECK(indptr.is_cuda() && indptr.scalar_type() == torch::kInt32);
+  TORCH_CHECK(indices.is_cuda() && indices.scalar_type() == torch::kInt32);
+  auto y = torch::empty({n_rows}, values.options());
+  if (values.scalar_type() == torch::kFloat32) {
+    raft_amd::launch_csr_spmv<float>(indptr.data_ptr<int>(), indices.data_ptr<int>(),
+                                     values.data_ptr<float>(), x.data_ptr<float>(),
+                                     y.data_ptr<float>(), n_rows, values.numel(),
+                                     cur_stream());
+  } else {
+    raft_amd::launch_csr_spmv<double>(indptr.data_ptr<int>(), indices.data_ptr<int>(),
+                                      values.data_ptr<double>(), x.data_ptr<double>(),
+                                      y.data_ptr<double>(), n_rows, values.numel(),
+                                      cur_stream());
+  }
+  return y;
+}
+
+torch::Tensor reduce_rows_by_key(torch::Tensor x, torch::Tensor keys, int64_t n_keys) {
+  check_f32_2d(x, "x");
+  TORCH_CHECK(keys.scalar_type() == torch::kInt32);
+  auto sums = torch::zeros({n_keys, x.size(1)}, x.options());
+  raft_amd::launch_reduce_rows_by_key(x.data_ptr<float>(), keys.data_ptr<int>(),
+                                      sums.data_ptr<float>(), x.size(0), x.size(1),
+                                      cur_stream());
+  return sums;
+}
+
+std::tuple<torch::Tensor, torch::Tensor> select_k(torch::Tensor x, int64_t k,
+                                                  bool select_min, int64_t algo,
+                                                  bool do_sort) {
+  check_f32_2d(x, "x");
+  auto vals = torch::empty({x.size(0), k}, x.options());
+  auto idx = torch::empty({x.size(0), k}, x.options().dtype(torch::kInt32));
+  raft_amd::launch_select_k(x.data_ptr<float>(), vals.data_ptr<float>(),
+                            idx.data_ptr<int>(), x.size(0), x.size(1), (int)k,
+                            select_min, do_sort, cur_stream());
+  return {vals, idx};
+}
+
+torch::Tensor gemm_bf16_f32(torch::Tensor a, torch::Tensor b,
+                            c10::optional<torch::Tensor> out, double beta) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16 && a.is_contiguous());
+  TORCH_CHECK(b.is_cuda() && b.scalar_type() == torch::kBFloat16 && b.is_contiguous());
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(0));
+  torch::Tensor c;
+  float bt = (float)beta;
+  if (out.has_value()) {
+    c = out.value();
+    TORCH_CHECK(c.scalar_type() == torch::kFloat32 && c.is_contiguous());
+  } else {
+    c = torch::empty({a.size(0), b.size(1)},
+                     a.options().dtype(torch::kFloat32));
+    bt = 0.0f;
+  }
+  raft_amd::gemm_bf16_f32_rowmajor(a.data_ptr(), b.data_ptr(), c.data_ptr<float>(),
+                                   a.size(0), b.size(1), a.size(1), bt,
+                                   (void*)cur_stream());
+  return c;
+}
+
+torch::Tensor gemm_f32(torch::Tensor a, torch::Tensor b) {
+  check_f32_2d(a, "a");
+  check_f32_2d(b, "b");
+  auto c = torch::empty({a.size(0), b.size(1)}, a.options());
+  raft_amd::gemm_f32_rowmajor(a.data_ptr<float>(), b.data_ptr<float>(),
+                              c.data_ptr<float>(), a.size(0), b.size(1), a.size(1),
+                              0.0f, (void*)cur_stream());
+  return c;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("reduce_rows", &reduce_rows, "rowwise reduction (op code)");
+  m.def("reduce_cols", &reduce_cols, "columnwise reduction (op code)");
+  m.def("row_argmin", &row_argmin, "rowwise argmin");
+  m.def("row_normalize_l2", &row_normalize_l2, "fused L2 row normalize");
+  m.def("l2_epilogue_", &l2_epilogue_, "in-place L2 distance epilogue");
+  m.def("l2nn_epilogue", &l2nn_epilogue, "fused argmin epilogue over GEMM tile");
+  m.def("pairwise_unexpanded", &pairwise_unexpanded, "tiled unexpanded distances");
+  m.def("rng_uniform", &rng_uniform, "PCG32 uniform [0,1)");
+  m.def("rng_normal", &rng_normal, "PCG32 + Box-Muller standard normal");
+  m.def("make_blobs", &make_blobs, "fused gaussian blob generator");
+  m.def("csr_spmv", &csr_spmv, "CSR SpMV (sub-wave per row)");
+  m.def("reduce_rows_by_key", &reduce_rows_by_key, "keyed row accumulation");
+  m.def("select_k", &select_k, "batched top-k (radix)");
+  m.def("gemm_bf16_f32", &gemm_bf16_f32, "bf16 x bf16 -> f32 rocBLAS gemm_ex",
+        pybind11::arg("a"), pybind11::arg("b"), pybind11::arg("out") = pybind11::none(),
+        pybind11::arg("beta") = 0.0);
+  m.def("gemm_f32", &gemm_f32, "fp32 rocBLAS sgemm");
+}

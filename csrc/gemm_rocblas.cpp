@@ -1,0 +1,65 @@
+// rocBLAS gemm_ex wrapper: bf16 x bf16 -> fp32 with fp32 accumulate.
+//
+// Reference parity: raft/linalg/detail/cublas_wrappers.hpp cublasgemmEx usage.
+// This is the building block of the split-bf16 fp32 GEMM emulation
+// (raft_amd/linalg/gemm.py): torch.matmul on bf16 rounds the output to bf16,
+// which destroys the splitting scheme — rocblas_gemm_ex keeps C in fp32.
+// The MFMA path (2.5 PF dense bf16) is what rocBLAS dispatches to on gfx950.
+
+#include <rocblas/rocblas.h>
+
+#include <mutex>
+#include <stdexcept>
+#include <string>
+
+#define ROCBLAS_CHECK(expr)                                                   \
+  do {                                                                        \
+    rocblas_status _s = (expr);                                               \
+    if (_s != rocblas_status_success) {                                       \
+      throw std::runtime_error(std::string("rocBLAS error ") +                \
+                               std::to_string((int)_s) + " at " __FILE__ ":" + \
+                               std::to_string(__LINE__));                     \
+    }                                                                         \
+  } while (0)
+
+namespace raft_amd {
+
+static rocblas_handle get_handle() {
+  static rocblas_handle h = nullptr;
+  static std::once_flag flag;
+  std::call_once(flag, [] { ROCBLAS_CHECK(rocblas_create_handle(&h)); });
+  return h;
+}
+
+// C[m,n] (row-major, fp32) = A[m,k] (row-major bf16) @ B[k,n] (row-major bf16)
+//                            + beta * C
+// Row-major is expressed as the transposed column-major problem:
+// C^T = B^T A^T with everything column-major.
+void gemm_bf16_f32_rowmajor(const void* a, const void* b, float* c, long long m,
+                            long long n, long long k, float beta, void* stream) {
+  rocblas_handle h = get_handle();
+  ROCBLAS_CHECK(rocblas_set_stream(h, (hipStream_t)stream));
+  const float alpha = 1.0f;
+  ROCBLAS_CHECK(rocblas_gemm_ex(
+      h, rocblas_operation_none, rocblas_operation_none,
+      (rocblas_int)n, (rocblas_int)m, (rocblas_int)k, &alpha,
+      b, rocblas_datatype_bf16_r, (rocblas_int)n,
+      a, rocblas_datatype_bf16_r, (rocblas_int)k, &beta,
+      c, rocblas_datatype_f32_r, (rocblas_int)n,
+      c, rocblas_datatype_f32_r, (rocblas_int)n,
+      rocblas_datatype_f32_r, rocblas_gemm_algo_standard, 0, 0));
+}
+
+// fp32 SGEMM (row-major) — the native fp32 vector-ALU path for comparison.
+void gemm_f32_rowmajor(const float* a, const float* b, float* c, long long m,
+                       long long n, long long k, float beta, void* stream) {
+  rocblas_handle h = get_handle();
+  ROCBLAS_CHECK(rocblas_set_stream(h, (hipStream_t)stream));
+  const float alpha = 1.0f;
+  ROCBLAS_CHECK(rocblas_sgemm(h, rocblas_operation_none, rocblas_operation_none,
+                              (rocblas_int)n, (rocblas_int)m, (rocblas_int)k,
+                              &alpha, b, (rocblas_int)n, a, (rocblas_int)k,
+                              &beta, c, (rocblas_int)n));
+}
+
+}  // namespace raft_amd

@@ -1,3 +1,4 @@
-from .kmeans import KMeans, KMeansParams, kmeans_fit, kmeans_predict, kmeans_transform
+from .kmeans import (KMeans, KMeansParams, kmeans_fit, kmeans_predict,
+                     kmeans_transform, kmeans_iterate)
 
-__all__ = ["KMeans", "KMeansParams", "kmeans_fit", "kmeans_predict", "kmeans_transform"]
+__all__ = ["KMeans", "KMeansParams", "kmeans_fit", "kmeans_predict", "kmeans_transform", "kmeans_iterate"]

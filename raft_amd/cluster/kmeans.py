@@ -180,6 +180,36 @@ def kmeans_fit(x: torch.Tensor, params: KMeansParams,
     return KMeansModel(centroids=centroids, inertia=inertia, n_iter=it, labels=labels)
 
 
+def kmeans_iterate(x: torch.Tensor, centroids: torch.Tensor, n_iters: int,
+                   comms: Comms | None = None, fp32_mode: str = "auto",
+                   chunk_rows: int = 262144):
+    """Run EXACTLY n_iters Lloyd iterations (no convergence early-exit).
+
+    The benchmark entry point: every iteration performs the full assignment
+    (fused L2-NN over all rows) + update (keyed reduction, counts, ONE packed
+    allreduce, centroid recompute). Returns (centroids, inertia).
+    """
+    comms = comms or LoopbackComms()
+    k, d = centroids.shape
+    inertia = float("inf")
+    for _ in range(n_iters):
+        dmin, labels = fused_l2nn(x, centroids, fp32_mode=fp32_mode,
+                                  chunk_rows=chunk_rows)
+        sums = reduce_rows_by_key(x, labels, n_keys=k)
+        counts = torch.bincount(labels, minlength=k).to(x.dtype)
+        local_inertia = dmin.double().sum().to(x.dtype)
+        packed = torch.cat([sums.reshape(-1), counts, local_inertia.reshape(1)])
+        if comms.get_size() > 1:
+            comms.allreduce(packed, op=ReduceOp.SUM)
+        sums = packed[: k * d].reshape(k, d)
+        counts = packed[k * d: k * d + k]
+        inertia = float(packed[-1].item())
+        nonzero = counts > 0
+        centroids = torch.where(nonzero.unsqueeze(1),
+                                sums / counts.clamp_min(1).unsqueeze(1), centroids)
+    return centroids, inertia
+
+
 def kmeans_predict(model_or_centroids, x: torch.Tensor, fp32_mode: str = "auto") -> torch.Tensor:
     c = getattr(model_or_centroids, "centroids", model_or_centroids)
     return fused_l2nn(x, c, fp32_mode=fp32_mode)[1]

@@ -101,23 +101,24 @@ def gemm_fp32_emulated(a: torch.Tensor, b: torch.Tensor, mode: str = "bf16x3") -
     mode "bf16x2": 3 slice-products -> TF32-class accuracy (~2^-16 rel).
     """
     assert a.dtype == torch.float32 and b.dtype == torch.float32
+    fused = on_gpu(a, b)  # rocBLAS beta=1 accumulates in-GEMM (no extra passes)
     if mode == "bf16x2":
         ah, al = _split_bf16(a, 2)
         bh, bl = _split_bf16(b, 2)
         c = gemm_bf16_f32(ah, bh)
-        c += gemm_bf16_f32(ah, bl)
-        c += gemm_bf16_f32(al, bh)
-        return c
-    if mode == "bf16x3":
+        terms = [(ah, bl), (al, bh)]
+    elif mode == "bf16x3":
         ah, am, al = _split_bf16(a, 3)
         bh, bm, bl = _split_bf16(b, 3)
         c = gemm_bf16_f32(ah, bh)
-        c += gemm_bf16_f32(ah, bm)
-        c += gemm_bf16_f32(am, bh)
-        c += gemm_bf16_f32(am, bm)
-        c += gemm_bf16_f32(ah, bl)
-        c += gemm_bf16_f32(al, bh)
-        return c
-    if mode == "native":
+        terms = [(ah, bm), (am, bh), (am, bm), (ah, bl), (al, bh)]
+    elif mode == "native":
         return torch.matmul(a, b)
-    raise ValueError(f"unknown fp32 emulation mode {mode}")
+    else:
+        raise ValueError(f"unknown fp32 emulation mode {mode}")
+    for ta, tb in terms:
+        if fused:
+            gemm_bf16_f32(ta, tb, out=c, beta=1.0)
+        else:
+            c += gemm_bf16_f32(ta, tb)
+    return c
