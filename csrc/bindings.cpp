@@ -37,6 +37,9 @@ void launch_reduce_rows_by_key(const float*, const int*, float*, long long, long
 // from select_k.hip
 void launch_select_k(const float*, float*, int*, long long, long long, int, bool, bool,
                      hipStream_t);
+// from fused_l2nn.hip
+void launch_fused_l2nn_split(const void**, const void**, const float*, const float*,
+                             float*, int*, long long, int, int, int, hipStream_t);
 // from gemm_rocblas.cpp
 void gemm_bf16_f32_rowmajor(const void*, const void*, float*, long long, long long,
                             long long, float, void*);
@@ -202,6 +205,37 @@ std::tuple<torch::Tensor, torch::Tensor> select_k(torch::Tensor x, int64_t k,
   return {vals, idx};
 }
 
+std::tuple<torch::Tensor, torch::Tensor> fused_l2nn_split(
+    std::vector<torch::Tensor> x_slices, std::vector<torch::Tensor> c_slices,
+    torch::Tensor xn, torch::Tensor cn) {
+  const int nslice = (int)x_slices.size();
+  TORCH_CHECK(nslice >= 1 && nslice <= 3 && c_slices.size() == x_slices.size());
+  const void* xsl[3];
+  const void* csl[3];
+  for (int s = 0; s < nslice; s++) {
+    auto& xt = x_slices[s];
+    auto& ct = c_slices[s];
+    TORCH_CHECK(xt.is_cuda() && xt.scalar_type() == torch::kBFloat16 && xt.is_contiguous());
+    TORCH_CHECK(ct.is_cuda() && ct.scalar_type() == torch::kBFloat16 && ct.is_contiguous());
+    xsl[s] = xt.data_ptr();
+    csl[s] = ct.data_ptr();
+  }
+  const long long m = x_slices[0].size(0);
+  const long long n = c_slices[0].size(0);
+  const long long d = x_slices[0].size(1);
+  TORCH_CHECK(c_slices[0].size(1) == d, "dim mismatch");
+  TORCH_CHECK(n % 128 == 0, "fused_l2nn: n must be a multiple of 128");
+  TORCH_CHECK(d % 64 == 0, "fused_l2nn: d must be a multiple of 64");
+  TORCH_CHECK(cn.numel() == n && xn.numel() == m);
+  auto dmin = torch::empty({m}, xn.options());
+  auto amin = torch::empty({m}, xn.options().dtype(torch::kInt32));
+  raft_amd::launch_fused_l2nn_split(xsl, csl, xn.data_ptr<float>(),
+                                    cn.data_ptr<float>(), dmin.data_ptr<float>(),
+                                    amin.data_ptr<int>(), m, (int)n, (int)d,
+                                    nslice, cur_stream());
+  return {dmin, amin};
+}
+
 torch::Tensor gemm_bf16_f32(torch::Tensor a, torch::Tensor b,
                             c10::optional<torch::Tensor> out, double beta) {
   TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16 && a.is_contiguous());
@@ -249,6 +283,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("csr_spmv", &csr_spmv, "CSR SpMV (sub-wave per row)");
   m.def("reduce_rows_by_key", &reduce_rows_by_key, "keyed row accumulation");
   m.def("select_k", &select_k, "batched top-k (radix)");
+  m.def("fused_l2nn_split", &fused_l2nn_split,
+        "fused split-bf16 MFMA L2-NN (distance + argmin, no materialization)");
   m.def("gemm_bf16_f32", &gemm_bf16_f32, "bf16 x bf16 -> f32 rocBLAS gemm_ex",
         pybind11::arg("a"), pybind11::arg("b"), pybind11::arg("out") = pybind11::none(),
         pybind11::arg("beta") = 0.0);

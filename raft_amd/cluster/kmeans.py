@@ -189,12 +189,25 @@ def kmeans_iterate(x: torch.Tensor, centroids: torch.Tensor, n_iters: int,
     (fused L2-NN over all rows) + update (keyed reduction, counts, ONE packed
     allreduce, centroid recompute). Returns (centroids, inertia).
     """
+    from raft_amd.neighbors.fused_l2nn import (_MODE_NSLICE, fused_l2nn_presplit,
+                                               split_bf16_slices)
+
     comms = comms or LoopbackComms()
     k, d = centroids.shape
     inertia = float("inf")
+    # X is iteration-invariant: pre-split the bf16 slices and row norms ONCE
+    # (the same caching the reference does for row norms in its kmeans)
+    use_fused = (x.is_cuda and x.dtype == torch.float32
+                 and fp32_mode in _MODE_NSLICE and d % 64 == 0)
+    if use_fused:
+        x_slices = split_bf16_slices(x, _MODE_NSLICE[fp32_mode])
+        xn = (x * x).sum(dim=1)
     for _ in range(n_iters):
-        dmin, labels = fused_l2nn(x, centroids, fp32_mode=fp32_mode,
-                                  chunk_rows=chunk_rows)
+        if use_fused:
+            dmin, labels = fused_l2nn_presplit(x_slices, xn, centroids)
+        else:
+            dmin, labels = fused_l2nn(x, centroids, fp32_mode=fp32_mode,
+                                      chunk_rows=chunk_rows)
         sums = reduce_rows_by_key(x, labels, n_keys=k)
         counts = torch.bincount(labels, minlength=k).to(x.dtype)
         local_inertia = dmin.double().sum().to(x.dtype)

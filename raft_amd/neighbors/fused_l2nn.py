@@ -4,13 +4,16 @@ Reference parity: RAFT's historical fused-L2-NN (the contraction engine with a
 key-value argmin epilogue) — required by BASELINE.json config 5 and the
 k-means EM loop.
 
-MI355X design (csrc/fused_l2nn.hip): each workgroup owns a row tile of X and
-loops over ALL of Y in N-tiles; the running (min, argmin) pair per row lives
-in registers, so the m x n distance matrix is never written to HBM. For fp32
-inputs the dot-product term uses in-kernel split-bf16 MFMA accumulation
-(fp32-class accuracy, 2.5 PF matrix cores) with the norm epilogue fused.
-The GEMM+epilogue chunked path below is the fallback/reference engine on GPU
-until the fused kernel covers the shape; CPU is the torch oracle.
+MI355X design (csrc/fused_l2nn.hip): each workgroup owns a 128-row tile of X
+and loops over Y in 128-column tiles; the running (min, argmin) pair per row
+lives in registers, so the m x n distance matrix never touches HBM. fp32
+inputs are pre-split into bf16 slices and the dot term runs on the 2.5 PF
+bf16 matrix cores:
+    nslice=2 (fp32 mode "bf16x2"): 3 MFMA slice-products, ~2^-16 accuracy
+    nslice=3 (fp32 mode "bf16x3"): 6 products, fp32-class accuracy
+bf16 inputs run nslice=1 directly. A rocBLAS-GEMM + epilogue chunked path
+remains as the fallback for shapes the fused kernel does not cover
+(d % 64 != 0) and for the explicit "native" fp32 engine.
 """
 from __future__ import annotations
 
@@ -20,26 +23,76 @@ from raft_amd._ext import ext_or_none, require_ext
 from raft_amd.utils import on_gpu, row_chunks
 from raft_amd.linalg.gemm import gemm_fp32_emulated
 
+_MODE_NSLICE = {"bf16x2": 2, "bf16x3": 3, "auto": 3, "fused": 3}
+
+
+def split_bf16_slices(t: torch.Tensor, nslice: int):
+    """fp32 -> bf16 slices with t ≈ sum(slices). Iteration-invariant for the
+    k-means X matrix, so callers may precompute (see kmeans_iterate)."""
+    slices = []
+    resid = t
+    for i in range(nslice):
+        s = resid.to(torch.bfloat16)
+        slices.append(s.contiguous())
+        if i + 1 < nslice:
+            resid = resid - s.to(torch.float32)
+    return slices
+
+
+def _pad_cols(y: torch.Tensor, yn: torch.Tensor, mult: int = 128):
+    n = y.shape[0]
+    pad = (-n) % mult
+    if pad == 0:
+        return y, yn, n
+    yp = torch.cat([y, torch.zeros(pad, y.shape[1], dtype=y.dtype, device=y.device)])
+    ynp = torch.cat([yn, torch.full((pad,), float("inf"), dtype=yn.dtype, device=yn.device)])
+    return yp, ynp, n
+
+
+def fused_l2nn_presplit(x_slices, xn: torch.Tensor, y: torch.Tensor,
+                        sqrt: bool = False):
+    """Fused kernel entry with precomputed X slices (k-means hot loop)."""
+    ext = require_ext()
+    nslice = len(x_slices)
+    yn = (y * y).sum(dim=1)
+    yp, ynp, n_true = _pad_cols(y, yn)
+    y_slices = split_bf16_slices(yp, nslice)
+    dmin, amin = ext.fused_l2nn_split(list(x_slices), list(y_slices),
+                                      xn.contiguous(), ynp.contiguous())
+    if sqrt:
+        dmin = dmin.clamp_min(0).sqrt()
+    return dmin, amin.to(torch.int64)
+
 
 def fused_l2nn(x: torch.Tensor, y: torch.Tensor, sqrt: bool = False,
-               fp32_mode: str = "auto", chunk_rows: int = 65536):
+               fp32_mode: str = "auto", chunk_rows: int = 262144):
     """For each row of x [m,d]: (min L2 distance to rows of y [n,d], argmin).
 
     Returns (min_dists [m], argmins [m] int64). Distances are squared L2
     unless sqrt=True.
     """
     assert x.dim() == 2 and y.dim() == 2 and x.shape[1] == y.shape[1]
-    m, d = x.shape
-    n = y.shape[0]
+    d = x.shape[1]
 
-    if on_gpu(x, y) and x.dtype == torch.float32:
-        ext = require_ext()
-        if hasattr(ext, "fused_l2nn") and fp32_mode in ("auto", "bf16x3", "fused"):
-            dmin, amin = ext.fused_l2nn(x.contiguous(), y.contiguous())
+    if on_gpu(x, y):
+        if x.dtype == torch.bfloat16 and d % 64 == 0:
+            ext = require_ext()
+            xn = x.to(torch.float32).pow(2).sum(dim=1)
+            yf = y.to(torch.float32)
+            yn = (yf * yf).sum(dim=1)
+            yp, ynp, _ = _pad_cols(y.contiguous(), yn)
+            dmin, amin = ext.fused_l2nn_split([x.contiguous()], [yp.contiguous()],
+                                              xn.contiguous(), ynp.contiguous())
             if sqrt:
                 dmin = dmin.clamp_min(0).sqrt()
             return dmin, amin.to(torch.int64)
-        return _chunked_gpu(x, y, sqrt, fp32_mode, chunk_rows)
+        if x.dtype == torch.float32:
+            if fp32_mode in _MODE_NSLICE and d % 64 == 0:
+                nslice = _MODE_NSLICE[fp32_mode]
+                xs = split_bf16_slices(x, nslice)
+                xn = (x * x).sum(dim=1)
+                return fused_l2nn_presplit(xs, xn, y, sqrt=sqrt)
+            return _chunked_gpu(x, y, sqrt, fp32_mode, chunk_rows)
 
     # CPU oracle
     d2 = torch.cdist(x.double(), y.double(), p=2) ** 2

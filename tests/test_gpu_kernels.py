@@ -126,6 +126,60 @@ class TestPairwiseGpu:
         assert (a3 == an).float().mean() > 0.999
 
 
+class TestFusedL2NNMfma:
+    """The fused split-bf16 MFMA kernel vs torch/cdist references."""
+
+    @pytest.mark.parametrize("mode,min_agree,dtol", [("bf16x2", 0.999, 2e-3),
+                                                     ("bf16x3", 0.9999, 2e-4)])
+    def test_vs_fp64_reference(self, dev, ext, mode, min_agree, dtol):
+        torch.manual_seed(0)
+        x = torch.randn(4096, 128, device=dev)
+        y = torch.randn(512, 128, device=dev)
+        from raft_amd.neighbors.fused_l2nn import fused_l2nn
+        dmin, amin = fused_l2nn(x, y, fp32_mode=mode)
+        ref = torch.cdist(x.double(), y.double()) ** 2
+        rd, ra = ref.min(dim=1)
+        assert (amin == ra).float().mean() > min_agree
+        rel = ((dmin.double() - rd).abs() / rd.clamp_min(1e-2)).median()
+        assert float(rel) < dtol
+
+    def test_unaligned_rows_and_padded_cols(self, dev, ext):
+        """m not multiple of 128, n not multiple of 128 (padding path)."""
+        torch.manual_seed(1)
+        x = torch.randn(1111, 64, device=dev)
+        y = torch.randn(200, 64, device=dev)
+        from raft_amd.neighbors.fused_l2nn import fused_l2nn
+        dmin, amin = fused_l2nn(x, y, fp32_mode="bf16x3")
+        ref = torch.cdist(x.double(), y.double()) ** 2
+        rd, ra = ref.min(dim=1)
+        assert (amin == ra).float().mean() > 0.999
+        assert (amin < 200).all()  # padded columns never win
+
+    def test_bf16_input_path(self, dev, ext):
+        torch.manual_seed(2)
+        x = torch.randn(2048, 128, device=dev).bfloat16()
+        y = torch.randn(256, 128, device=dev).bfloat16()
+        from raft_amd.neighbors.fused_l2nn import fused_l2nn
+        dmin, amin = fused_l2nn(x, y)
+        ref = torch.cdist(x.double(), y.double()) ** 2
+        rd, ra = ref.min(dim=1)
+        assert (amin == ra).float().mean() > 0.99
+
+    def test_blob_data_agreement_is_exact(self, dev, ext):
+        """On clustered data (the bench workload) assignments must match
+        native fp32 exactly for both split modes."""
+        from raft_amd.random import make_blobs, RngState
+        from raft_amd.neighbors.fused_l2nn import fused_l2nn
+        x, _, centers = make_blobs(100000, 256, n_clusters=1024, cluster_std=1.0,
+                                   state=RngState(seed=7), device=dev)
+        c = centers + 0.3
+        _, an = fused_l2nn(x, c, fp32_mode="native")
+        for mode in ("bf16x2", "bf16x3"):
+            _, am = fused_l2nn(x, c, fp32_mode=mode)
+            agree = float((am == an).float().mean())
+            assert agree > 0.9999, (mode, agree)
+
+
 class TestRngGpu:
     def test_uniform_bitwise_matches_cpu(self, dev, ext):
         from raft_amd.random import uniform, RngState
