@@ -32,8 +32,8 @@ template <typename T>
 void launch_csr_spmv(const int*, const int*, const T*, const T*, T*, long long,
                      long long, hipStream_t);
 // from kmeans.hip
-void launch_reduce_rows_by_key(const float*, const int*, float*, long long, long long,
-                               hipStream_t);
+void launch_reduce_rows_by_key(const float*, const int*, float*, float*, long long,
+                               long long, long long, int, hipStream_t);
 // from select_k.hip
 void launch_select_k(const float*, float*, int*, long long, long long, int, bool, bool,
                      hipStream_t);
@@ -186,9 +186,14 @@ torch::Tensor csr_spmv(torch::Tensor indptr, torch::Tensor indices, torch::Tenso
 torch::Tensor reduce_rows_by_key(torch::Tensor x, torch::Tensor keys, int64_t n_keys) {
   check_f32_2d(x, "x");
   TORCH_CHECK(keys.scalar_type() == torch::kInt32);
-  auto sums = torch::zeros({n_keys, x.size(1)}, x.options());
+  const long long kd = n_keys * x.size(1);
+  // replica count: bound the workspace at ~32 MB, contention drop ~= replicas
+  int replicas = (int)std::min<long long>(16, std::max<long long>(1, (32ll << 20) / (kd * 4)));
+  auto work = torch::zeros({replicas, n_keys, x.size(1)}, x.options());
+  auto sums = torch::empty({n_keys, x.size(1)}, x.options());
   raft_amd::launch_reduce_rows_by_key(x.data_ptr<float>(), keys.data_ptr<int>(),
-                                      sums.data_ptr<float>(), x.size(0), x.size(1),
+                                      work.data_ptr<float>(), sums.data_ptr<float>(),
+                                      x.size(0), x.size(1), n_keys, replicas,
                                       cur_stream());
   return sums;
 }

