@@ -45,6 +45,8 @@ void gemm_bf16_f32_rowmajor(const void*, const void*, float*, long long, long lo
                             long long, float, void*);
 void gemm_f32_rowmajor(const float*, const float*, float*, long long, long long,
                        long long, float, void*);
+void gemm_bf16_f32_nt_rowmajor(const void*, const void*, float*, long long, long long,
+                               long long, float, void*);
 }  // namespace raft_amd
 
 namespace {
@@ -262,6 +264,26 @@ torch::Tensor gemm_bf16_f32(torch::Tensor a, torch::Tensor b,
   return c;
 }
 
+torch::Tensor gemm_bf16_f32_nt(torch::Tensor a, torch::Tensor b,
+                               c10::optional<torch::Tensor> out, double beta) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16 && a.is_contiguous());
+  TORCH_CHECK(b.is_cuda() && b.scalar_type() == torch::kBFloat16 && b.is_contiguous());
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2 && a.size(1) == b.size(1));
+  torch::Tensor c;
+  float bt = (float)beta;
+  if (out.has_value()) {
+    c = out.value();
+    TORCH_CHECK(c.scalar_type() == torch::kFloat32 && c.is_contiguous());
+  } else {
+    c = torch::empty({a.size(0), b.size(0)}, a.options().dtype(torch::kFloat32));
+    bt = 0.0f;
+  }
+  raft_amd::gemm_bf16_f32_nt_rowmajor(a.data_ptr(), b.data_ptr(), c.data_ptr<float>(),
+                                      a.size(0), b.size(0), a.size(1), bt,
+                                      (void*)cur_stream());
+  return c;
+}
+
 torch::Tensor gemm_f32(torch::Tensor a, torch::Tensor b) {
   check_f32_2d(a, "a");
   check_f32_2d(b, "b");
@@ -291,6 +313,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_l2nn_split", &fused_l2nn_split,
         "fused split-bf16 MFMA L2-NN (distance + argmin, no materialization)");
   m.def("gemm_bf16_f32", &gemm_bf16_f32, "bf16 x bf16 -> f32 rocBLAS gemm_ex",
+        pybind11::arg("a"), pybind11::arg("b"), pybind11::arg("out") = pybind11::none(),
+        pybind11::arg("beta") = 0.0);
+  m.def("gemm_bf16_f32_nt", &gemm_bf16_f32_nt,
+        "bf16 A @ B^T -> f32 (both row-major) rocBLAS gemm_ex",
         pybind11::arg("a"), pybind11::arg("b"), pybind11::arg("out") = pybind11::none(),
         pybind11::arg("beta") = 0.0);
   m.def("gemm_f32", &gemm_f32, "fp32 rocBLAS sgemm");

@@ -50,6 +50,24 @@ void gemm_bf16_f32_rowmajor(const void* a, const void* b, float* c, long long m,
       rocblas_datatype_f32_r, rocblas_gemm_algo_standard, 0, 0));
 }
 
+// C[m,n] (row-major fp32) = A[m,k] @ B[n,k]^T + beta*C  (both row-major) —
+// the natural pairwise-distance form: no transposed copies.
+// Column-major: C_cm[n,m] = B_cm^T[n,k] * A_cm[k,m].
+void gemm_bf16_f32_nt_rowmajor(const void* a, const void* b, float* c, long long m,
+                               long long n, long long k, float beta, void* stream) {
+  rocblas_handle h = get_handle();
+  ROCBLAS_CHECK(rocblas_set_stream(h, (hipStream_t)stream));
+  const float alpha = 1.0f;
+  ROCBLAS_CHECK(rocblas_gemm_ex(
+      h, rocblas_operation_transpose, rocblas_operation_none,
+      (rocblas_int)n, (rocblas_int)m, (rocblas_int)k, &alpha,
+      b, rocblas_datatype_bf16_r, (rocblas_int)k,
+      a, rocblas_datatype_bf16_r, (rocblas_int)k, &beta,
+      c, rocblas_datatype_f32_r, (rocblas_int)n,
+      c, rocblas_datatype_f32_r, (rocblas_int)n,
+      rocblas_datatype_f32_r, rocblas_gemm_algo_standard, 0, 0));
+}
+
 // fp32 SGEMM (row-major) — the native fp32 vector-ALU path for comparison.
 void gemm_f32_rowmajor(const float* a, const float* b, float* c, long long m,
                        long long n, long long k, float beta, void* stream) {

@@ -105,6 +105,14 @@ def _gemm_xyt(x, y, fp32_mode):
 
 def _l2_squared(x, y, fp32_mode):
     """||x||^2 + ||y||^2 - 2 x.y with fused epilogue on GPU."""
+    if on_gpu(x, y) and x.dtype == torch.bfloat16:
+        # bf16 path: MFMA GEMM with fp32 accumulate/output + fused epilogue;
+        # distances stay fp32 (select_k consumes them directly)
+        ext = require_ext()
+        xn = x.float().pow(2).sum(dim=1)
+        yn = y.float().pow(2).sum(dim=1)
+        g = ext.gemm_bf16_f32_nt(x.contiguous(), y.contiguous())
+        return ext.l2_epilogue_(g, xn.contiguous(), yn.contiguous())
     xn = (x.double() * x.double()).sum(dim=1) if x.device.type == "cpu" else (x * x).sum(dim=1)
     yn = (y.double() * y.double()).sum(dim=1) if y.device.type == "cpu" else (y * y).sum(dim=1)
     g = _gemm_xyt(x, y, fp32_mode)

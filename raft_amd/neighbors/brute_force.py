@@ -27,7 +27,10 @@ def knn(x: torch.Tensor, queries: torch.Tensor, k: int,
     """
     q = queries.shape[0]
     n = x.shape[0]
-    out_d = torch.empty((q, k), dtype=queries.dtype, device=queries.device)
+    # distances are fp32 on the GPU bf16 path (MFMA fp32 accumulate)
+    out_dtype = torch.float32 if (queries.is_cuda and queries.dtype == torch.bfloat16) \
+        else queries.dtype
+    out_d = torch.empty((q, k), dtype=out_dtype, device=queries.device)
     out_i = torch.empty((q, k), dtype=torch.int64, device=queries.device)
     for qs, qe in row_chunks(q, query_chunk):
         cand_d, cand_i = [], []
