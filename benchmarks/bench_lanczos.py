@@ -12,9 +12,13 @@ from __future__ import annotations
 
 import argparse
 import json
+import os
+import sys
 import time
 
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
 def make_symmetric_csr(n, nnz_per_row, device, seed=0):

@@ -34,6 +34,8 @@ void launch_csr_spmv(const int*, const int*, const T*, const T*, T*, long long,
 // from kmeans.hip
 void launch_reduce_rows_by_key(const float*, const int*, float*, float*, long long,
                                long long, long long, int, hipStream_t);
+void launch_reduce_rows_by_key_sorted(const float*, const int*, const int*, float*,
+                                      long long, long long, hipStream_t);
 // from select_k.hip
 void launch_select_k(const float*, float*, int*, long long, long long, int, bool, bool,
                      hipStream_t);
@@ -185,6 +187,19 @@ torch::Tensor csr_spmv(torch::Tensor indptr, torch::Tensor indices, torch::Tenso
   return y;
 }
 
+torch::Tensor reduce_rows_by_key_sorted(torch::Tensor x, torch::Tensor perm,
+                                        torch::Tensor keys_sorted, int64_t n_keys) {
+  check_f32_2d(x, "x");
+  TORCH_CHECK(perm.scalar_type() == torch::kInt32 && perm.is_contiguous());
+  TORCH_CHECK(keys_sorted.scalar_type() == torch::kInt32 && keys_sorted.is_contiguous());
+  auto sums = torch::zeros({n_keys, x.size(1)}, x.options());
+  raft_amd::launch_reduce_rows_by_key_sorted(x.data_ptr<float>(), perm.data_ptr<int>(),
+                                             keys_sorted.data_ptr<int>(),
+                                             sums.data_ptr<float>(), x.size(0),
+                                             x.size(1), cur_stream());
+  return sums;
+}
+
 torch::Tensor reduce_rows_by_key(torch::Tensor x, torch::Tensor keys, int64_t n_keys) {
   check_f32_2d(x, "x");
   TORCH_CHECK(keys.scalar_type() == torch::kInt32);
@@ -309,6 +324,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("make_blobs", &make_blobs, "fused gaussian blob generator");
   m.def("csr_spmv", &csr_spmv, "CSR SpMV (sub-wave per row)");
   m.def("reduce_rows_by_key", &reduce_rows_by_key, "keyed row accumulation");
+  m.def("reduce_rows_by_key_sorted", &reduce_rows_by_key_sorted,
+        "keyed row accumulation over a key-sorted permutation");
   m.def("select_k", &select_k, "batched top-k (radix)");
   m.def("fused_l2nn_split", &fused_l2nn_split,
         "fused split-bf16 MFMA L2-NN (distance + argmin, no materialization)");

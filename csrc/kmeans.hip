@@ -41,6 +41,78 @@ __global__ void reduce_rows_by_key_kernel(const float* __restrict__ x,
   }
 }
 
+// sort-based segmented accumulation: rows pre-ordered by key (perm from a
+// radix argsort). Each wave walks a contiguous chunk of the sorted order,
+// accumulating rows of the SAME key in registers (4 f32/lane for d<=256-ish,
+// strided for larger d) and issuing atomics only at run boundaries —
+// ~k + n/chunk atomic bursts instead of n*d single-element atomics
+// (measured: the naive kernel is atomic-issue-rate bound at ~77 G/s,
+// 33 ms @ 10M x 256; this design needs ~2 passes of HBM instead).
+template <int MAX_DREG>  // registers per lane for the accumulator
+__global__ void reduce_rows_by_key_sorted_kernel(const float* __restrict__ x,
+                                                 const int* __restrict__ perm,
+                                                 const int* __restrict__ keys_sorted,
+                                                 float* __restrict__ sums,
+                                                 long long n_rows, long long d,
+                                                 long long chunk) {
+  const int lane = threadIdx.x % RAFT_AMD_WAVE;
+  const long long wave_id = ((long long)blockIdx.x * blockDim.x + threadIdx.x) / RAFT_AMD_WAVE;
+  const long long start = wave_id * chunk;
+  if (start >= n_rows) return;
+  const long long stop = min(start + chunk, n_rows);
+
+  const int dreg = (int)((d + RAFT_AMD_WAVE - 1) / RAFT_AMD_WAVE);
+  float acc[MAX_DREG];
+#pragma unroll
+  for (int j = 0; j < MAX_DREG; j++) acc[j] = 0.f;
+
+  int cur_key = keys_sorted[start];
+  for (long long i = start; i < stop; i++) {
+    const int key = keys_sorted[i];
+    if (key != cur_key) {
+      float* sp = sums + (long long)cur_key * d;
+      for (int j = 0; j < dreg; j++) {
+        const long long col = (long long)j * RAFT_AMD_WAVE + lane;
+        if (col < d) atomicAdd(&sp[col], acc[j]);
+        acc[j] = 0.f;
+      }
+      cur_key = key;
+    }
+    const float* rp = x + (long long)perm[i] * d;
+    for (int j = 0; j < dreg; j++) {
+      const long long col = (long long)j * RAFT_AMD_WAVE + lane;
+      if (col < d) acc[j] += rp[col];
+    }
+  }
+  float* sp = sums + (long long)cur_key * d;
+  for (int j = 0; j < dreg; j++) {
+    const long long col = (long long)j * RAFT_AMD_WAVE + lane;
+    if (col < d) atomicAdd(&sp[col], acc[j]);
+  }
+}
+
+void launch_reduce_rows_by_key_sorted(const float* x, const int* perm,
+                                      const int* keys_sorted, float* sums,
+                                      long long n_rows, long long d,
+                                      hipStream_t stream) {
+  // chunk sized so the grid fills the chip (~2048 blocks * 4 waves)
+  const long long n_waves_target = 2048 * 4;
+  long long chunk = (n_rows + n_waves_target - 1) / n_waves_target;
+  if (chunk < 8) chunk = 8;
+  const long long n_waves = (n_rows + chunk - 1) / chunk;
+  const int grid = (int)((n_waves * RAFT_AMD_WAVE + 255) / 256);
+  if (d <= 256) {
+    hipLaunchKernelGGL((reduce_rows_by_key_sorted_kernel<4>), dim3(grid), dim3(256),
+                       0, stream, x, perm, keys_sorted, sums, n_rows, d, chunk);
+  } else if (d <= 1024) {
+    hipLaunchKernelGGL((reduce_rows_by_key_sorted_kernel<16>), dim3(grid), dim3(256),
+                       0, stream, x, perm, keys_sorted, sums, n_rows, d, chunk);
+  } else {
+    hipLaunchKernelGGL((reduce_rows_by_key_sorted_kernel<64>), dim3(grid), dim3(256),
+                       0, stream, x, perm, keys_sorted, sums, n_rows, d, chunk);
+  }
+}
+
 __global__ void fold_replicas_kernel(const float* __restrict__ work,
                                      float* __restrict__ out, long long kd,
                                      int replicas) {

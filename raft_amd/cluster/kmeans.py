@@ -135,7 +135,7 @@ def kmeans_fit(x: torch.Tensor, params: KMeansParams,
         dmin, labels = fused_l2nn(x, centroids, fp32_mode=params.fp32_mode)
         sums = reduce_rows_by_key(x, labels, n_keys=k)
         counts = torch.bincount(labels, minlength=k).to(x.dtype)
-        local_inertia = dmin.double().sum().to(x.dtype)
+        local_inertia = torch.sum(dmin, dtype=torch.float64).to(x.dtype)
         # ONE packed allreduce: [k, d] sums | [k] counts | [1] inertia
         packed = torch.cat([sums.reshape(-1), counts, local_inertia.reshape(1)])
         if comms.get_size() > 1:
@@ -210,7 +210,7 @@ def kmeans_iterate(x: torch.Tensor, centroids: torch.Tensor, n_iters: int,
                                       chunk_rows=chunk_rows)
         sums = reduce_rows_by_key(x, labels, n_keys=k)
         counts = torch.bincount(labels, minlength=k).to(x.dtype)
-        local_inertia = dmin.double().sum().to(x.dtype)
+        local_inertia = torch.sum(dmin, dtype=torch.float64).to(x.dtype)
         packed = torch.cat([sums.reshape(-1), counts, local_inertia.reshape(1)])
         if comms.get_size() > 1:
             comms.allreduce(packed, op=ReduceOp.SUM)

@@ -24,6 +24,14 @@ def reduce_rows_by_key(x: torch.Tensor, keys: torch.Tensor, n_keys: int | None =
         n_keys = int(keys.max().item()) + 1 if keys.numel() else 0
     if on_gpu(x, keys) and x.dtype == torch.float32 and weights is None:
         ext = require_ext()
+        if x.shape[0] >= 65536:
+            # sort-based: atomics only at run boundaries (the naive atomic
+            # kernel is issue-rate bound — measured 33 ms @ 10M x 256)
+            k32 = keys.to(torch.int32)
+            keys_sorted, perm = torch.sort(k32)
+            return ext.reduce_rows_by_key_sorted(x.contiguous(),
+                                                 perm.to(torch.int32).contiguous(),
+                                                 keys_sorted.contiguous(), int(n_keys))
         return ext.reduce_rows_by_key(x.contiguous(), keys.to(torch.int32).contiguous(), int(n_keys))
     out = torch.zeros((n_keys, x.shape[1]), dtype=x.dtype, device=x.device)
     src = x if weights is None else x * weights.unsqueeze(1)
