@@ -27,7 +27,8 @@ def main():
     p = argparse.ArgumentParser()
     p.add_argument("--rows", type=int, default=1_000_000)
     p.add_argument("--dim", type=int, default=128)
-    p.add_argument("--mode", default="native", choices=["native", "bf16x2", "bf16x3"])
+    p.add_argument("--mode", default="native",
+                   choices=["native", "bf16x2", "bf16x3", "mfma2", "mfma3"])
     p.add_argument("--tile-q", type=int, default=50000)
     p.add_argument("--tile-i", type=int, default=100000)
     p.add_argument("--max-tiles", type=int, default=0, help="0 = full matrix")
@@ -49,7 +50,7 @@ def main():
 
     out = torch.empty((args.tile_q, args.tile_i), dtype=torch.float32, device=dev)
     if args.mode != "native":
-        nsl = 2 if args.mode == "bf16x2" else 3
+        nsl = 2 if args.mode in ("bf16x2", "mfma2") else 3
         x_slices = _split_bf16(x, nsl)
 
     def tile(q0, q1, i0, i1):
@@ -58,6 +59,12 @@ def main():
         g = out[: q1 - q0, : i1 - i0]
         if args.mode == "native":
             torch.matmul(xq, xi.t(), out=g)
+        elif args.mode in ("mfma2", "mfma3"):
+            sq = [s[q0:q1] for s in x_slices]
+            si = [s[i0:i1] for s in x_slices]
+            ext.pairwise_l2_mfma(sq, si, xn[q0:q1].contiguous(),
+                                 xn[i0:i1].contiguous(), g)
+            return
         else:
             sq = [s[q0:q1] for s in x_slices]
             si = [s[i0:i1] for s in x_slices]

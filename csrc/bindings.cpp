@@ -42,6 +42,10 @@ void launch_select_k(const float*, float*, int*, long long, long long, int, bool
 // from fused_l2nn.hip
 void launch_fused_l2nn_split(const void**, const void**, const float*, const float*,
                              float*, int*, long long, int, int, int, hipStream_t);
+// from pairwise_mfma.hip
+void launch_pairwise_l2_mfma(const void**, const void**, const float*, const float*,
+                             float*, long long, long long, int, long long, int, bool,
+                             hipStream_t);
 // from gemm_rocblas.cpp
 void gemm_bf16_f32_rowmajor(const void*, const void*, float*, long long, long long,
                             long long, float, void*);
@@ -258,6 +262,39 @@ std::tuple<torch::Tensor, torch::Tensor> fused_l2nn_split(
   return {dmin, amin};
 }
 
+torch::Tensor pairwise_l2_mfma(std::vector<torch::Tensor> x_slices,
+                               std::vector<torch::Tensor> y_slices,
+                               torch::Tensor xn, torch::Tensor yn,
+                               c10::optional<torch::Tensor> out, bool sqrt_out) {
+  const int nslice = (int)x_slices.size();
+  TORCH_CHECK(nslice >= 1 && nslice <= 3 && y_slices.size() == x_slices.size());
+  const void* xsl[3];
+  const void* csl[3];
+  for (int s = 0; s < nslice; s++) {
+    TORCH_CHECK(x_slices[s].is_cuda() && x_slices[s].scalar_type() == torch::kBFloat16
+                && x_slices[s].is_contiguous());
+    TORCH_CHECK(y_slices[s].is_cuda() && y_slices[s].scalar_type() == torch::kBFloat16
+                && y_slices[s].is_contiguous());
+    xsl[s] = x_slices[s].data_ptr();
+    csl[s] = y_slices[s].data_ptr();
+  }
+  const long long m = x_slices[0].size(0);
+  const long long n = y_slices[0].size(0);
+  const long long d = x_slices[0].size(1);
+  TORCH_CHECK(d % 64 == 0, "pairwise_l2_mfma: d must be a multiple of 64");
+  torch::Tensor o;
+  if (out.has_value()) {
+    o = out.value();
+    TORCH_CHECK(o.is_contiguous() && o.size(0) >= m && o.size(1) == n);
+  } else {
+    o = torch::empty({m, n}, xn.options());
+  }
+  raft_amd::launch_pairwise_l2_mfma(xsl, csl, xn.data_ptr<float>(), yn.data_ptr<float>(),
+                                    o.data_ptr<float>(), m, n, (int)d, o.size(1),
+                                    nslice, sqrt_out, cur_stream());
+  return o;
+}
+
 torch::Tensor gemm_bf16_f32(torch::Tensor a, torch::Tensor b,
                             c10::optional<torch::Tensor> out, double beta) {
   TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16 && a.is_contiguous());
@@ -327,6 +364,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reduce_rows_by_key_sorted", &reduce_rows_by_key_sorted,
         "keyed row accumulation over a key-sorted permutation");
   m.def("select_k", &select_k, "batched top-k (radix)");
+  m.def("pairwise_l2_mfma", &pairwise_l2_mfma,
+        "fused split-bf16 MFMA pairwise L2 tile (single-write epilogue)",
+        pybind11::arg("x_slices"), pybind11::arg("y_slices"), pybind11::arg("xn"),
+        pybind11::arg("yn"), pybind11::arg("out") = pybind11::none(),
+        pybind11::arg("sqrt_out") = false);
   m.def("fused_l2nn_split", &fused_l2nn_split,
         "fused split-bf16 MFMA L2-NN (distance + argmin, no materialization)");
   m.def("gemm_bf16_f32", &gemm_bf16_f32, "bf16 x bf16 -> f32 rocBLAS gemm_ex",

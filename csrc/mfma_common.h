@@ -1,0 +1,106 @@
+// Shared machinery for split-bf16 MFMA contraction kernels (gfx950).
+// See fused_l2nn.hip header comment for the design rationale (tile geometry,
+// XOR swizzle + global_load_lds both-sides rule, slice-product emulation).
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include "common.h"
+
+namespace raft_amd {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define GLOAD_LDS(gp, lp)                                                      \
+  __builtin_amdgcn_global_load_lds(                                           \
+      (const __attribute__((address_space(1))) void*)(gp),                    \
+      (__attribute__((address_space(3))) void*)(lp), 16, 0, 0)
+
+// swizzle: flip byte-offset bit4 by row bits (rows are 128 B = 64 bf16)
+__device__ __forceinline__ int mfma_swz(int byte_off) {
+  return byte_off ^ (((byte_off >> 7) & 7) << 4);
+}
+
+// stage a [128][64] bf16 tile from row-major global (leading dim ld elements)
+// into a swizzled LDS tile (16 KiB). 256 threads, 4 gload rounds.
+__device__ __forceinline__ void mfma_stage_tile128(const __bf16* __restrict__ g,
+                                                   __bf16* lds, long long row0,
+                                                   long long k0, long long ld,
+                                                   long long max_row) {
+  const int t = threadIdx.x;
+  const int w = t / RAFT_AMD_WAVE;
+#pragma unroll
+  for (int j = 0; j < 4; j++) {
+    const int o = j * 4096 + t * 16;  // linear dest byte
+    const int o_src = mfma_swz(o);    // fetch what belongs here
+    long long r = row0 + (o_src >> 7);
+    if (r > max_row) r = max_row;
+    const long long goff = r * ld + k0 + ((o_src & 127) >> 1);
+    __bf16* lbase = lds + (j * 4096 + w * 1024) / 2;
+    GLOAD_LDS(g + goff, lbase);
+  }
+}
+
+template <int NSLICE>
+__device__ __forceinline__ constexpr int mfma_n_products() {
+  return NSLICE == 1 ? 1 : (NSLICE == 2 ? 3 : 6);
+}
+
+__device__ constexpr int MFMA_PROD_A[6] = {0, 0, 1, 1, 0, 2};
+__device__ constexpr int MFMA_PROD_B[6] = {0, 1, 0, 1, 2, 0};
+
+// Run the K loop for one 128x128 tile pair: stages slices, MFMAs into acc.
+// acc is the per-wave [4][4] fragment grid (wave (wr,wc) of a 2x2 wave grid).
+template <int NSLICE>
+__device__ __forceinline__ void mfma_tile_kloop(
+    const __bf16* const (&xg)[3], const __bf16* const (&cg)[3],
+    __bf16* (&xs)[NSLICE], __bf16* (&cs)[NSLICE],
+    f32x4 (&acc)[4][4], long long row0, long long col0, int d,
+    long long m_max, long long n_max, int wr, int wc, int lane) {
+  const int k_tiles = d / 64;
+  for (int kt = 0; kt < k_tiles; kt++) {
+#pragma unroll
+    for (int s = 0; s < NSLICE; s++) {
+      mfma_stage_tile128(xg[s], xs[s], row0, (long long)kt * 64, d, m_max);
+      mfma_stage_tile128(cg[s], cs[s], col0, (long long)kt * 64, d, n_max);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
+#pragma unroll
+    for (int kf = 0; kf < 2; kf++) {
+      bf16x8 a_frag[NSLICE][4], b_frag[NSLICE][4];
+#pragma unroll
+      for (int fr = 0; fr < 4; fr++) {
+        const int r = wr * 64 + fr * 16 + (lane & 15);
+        const int byte = mfma_swz(r * 128 + (kf * 32 + (lane >> 4) * 8) * 2);
+#pragma unroll
+        for (int s = 0; s < NSLICE; s++)
+          a_frag[s][fr] = *reinterpret_cast<const bf16x8*>((const char*)xs[s] + byte);
+      }
+#pragma unroll
+      for (int fc = 0; fc < 4; fc++) {
+        const int c = wc * 64 + fc * 16 + (lane & 15);
+        const int byte = mfma_swz(c * 128 + (kf * 32 + (lane >> 4) * 8) * 2);
+#pragma unroll
+        for (int s = 0; s < NSLICE; s++)
+          b_frag[s][fc] = *reinterpret_cast<const bf16x8*>((const char*)cs[s] + byte);
+      }
+#pragma unroll
+      for (int fr = 0; fr < 4; fr++)
+#pragma unroll
+        for (int fc = 0; fc < 4; fc++) {
+#pragma unroll
+          for (int p = 0; p < mfma_n_products<NSLICE>(); p++) {
+            acc[fr][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_frag[MFMA_PROD_A[p]][fr], b_frag[MFMA_PROD_B[p]][fc],
+                acc[fr][fc], 0, 0, 0);
+          }
+        }
+    }
+    __syncthreads();
+  }
+}
+
+}  // namespace raft_amd

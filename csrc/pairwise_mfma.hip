@@ -1,0 +1,118 @@
+// Pairwise L2-expanded distance tile kernel — split-bf16 MFMA with the
+// epilogue fused into the C-write (BASELINE config 2's engine).
+//
+// Reference parity (WHAT): RAFT's historical pairwise-distance L2-expanded
+// epilogue over the contraction engine. Versus the rocBLAS 3/6-GEMM
+// beta-accumulate path, this writes the fp32 distance tile EXACTLY ONCE
+// (d2 = xn + yn - 2 x.c computed in registers from the accumulator), cutting
+// the dominant HBM cost of large distance matrices by ~5x
+// (3-GEMM path: 3 writes + 2 reads + epilogue read/write = 7 tile passes).
+//
+// Same geometry/staging as fused_l2nn.hip (see mfma_common.h).
+
+#include <hip/hip_runtime.h>
+
+#include "mfma_common.h"
+
+namespace raft_amd {
+
+template <int NSLICE>
+__launch_bounds__(256, 2)
+__global__ void pairwise_l2_kernel(const __bf16* __restrict__ x0,
+                                   const __bf16* __restrict__ x1,
+                                   const __bf16* __restrict__ x2,
+                                   const __bf16* __restrict__ c0,
+                                   const __bf16* __restrict__ c1,
+                                   const __bf16* __restrict__ c2,
+                                   const float* __restrict__ xn,
+                                   const float* __restrict__ yn,
+                                   float* __restrict__ out,
+                                   long long m, long long n, int d,
+                                   long long ldo, int sqrt_out) {
+  extern __shared__ __bf16 smem[];
+  __bf16* xs[NSLICE];
+  __bf16* cs[NSLICE];
+  const __bf16* const xg[3] = {x0, x1, x2};
+  const __bf16* const cg[3] = {c0, c1, c2};
+#pragma unroll
+  for (int s = 0; s < NSLICE; s++) {
+    xs[s] = smem + s * 8192;
+    cs[s] = smem + (NSLICE + s) * 8192;
+  }
+
+  const int lane = threadIdx.x % RAFT_AMD_WAVE;
+  const int w = threadIdx.x / RAFT_AMD_WAVE;
+  const int wr = w >> 1, wc = w & 1;
+  const long long row0 = (long long)blockIdx.y * 128;
+  const long long col0 = (long long)blockIdx.x * 128;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int a = 0; a < 4; a++)
+#pragma unroll
+    for (int b = 0; b < 4; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  mfma_tile_kloop<NSLICE>(xg, cg, xs, cs, acc, row0, col0, d, m - 1, n - 1,
+                          wr, wc, lane);
+
+  // fused epilogue + write: d2 = max(xn[r] + yn[c] - 2 acc, 0)
+  // C/D layout: col = lane&15 (+fc*16), row = (lane>>4)*4 + reg (+fr*16).
+#pragma unroll
+  for (int fr = 0; fr < 4; fr++) {
+#pragma unroll
+    for (int reg = 0; reg < 4; reg++) {
+      const long long row = row0 + wr * 64 + fr * 16 + (lane >> 4) * 4 + reg;
+      if (row >= m) continue;
+      const float xv = xn[row];
+#pragma unroll
+      for (int fc = 0; fc < 4; fc++) {
+        const long long col = col0 + wc * 64 + fc * 16 + (lane & 15);
+        if (col < n) {
+          float v = fmaxf(xv + yn[col] - 2.f * acc[fr][fc][reg], 0.f);
+          if (sqrt_out) v = sqrtf(v);
+          out[row * ldo + col] = v;
+        }
+      }
+    }
+  }
+}
+
+void launch_pairwise_l2_mfma(const void** xsl, const void** csl, const float* xn,
+                             const float* yn, float* out, long long m, long long n,
+                             int d, long long ldo, int nslice, bool sqrt_out,
+                             hipStream_t stream) {
+  dim3 grid((unsigned)((n + 127) / 128), (unsigned)((m + 127) / 128));
+  const size_t lds = (size_t)nslice * 2 * 8192 * sizeof(__bf16);
+  const __bf16* x0 = (const __bf16*)xsl[0];
+  const __bf16* x1 = (const __bf16*)(nslice > 1 ? xsl[1] : xsl[0]);
+  const __bf16* x2 = (const __bf16*)(nslice > 2 ? xsl[2] : xsl[0]);
+  const __bf16* c0 = (const __bf16*)csl[0];
+  const __bf16* c1 = (const __bf16*)(nslice > 1 ? csl[1] : csl[0]);
+  const __bf16* c2 = (const __bf16*)(nslice > 2 ? csl[2] : csl[0]);
+  switch (nslice) {
+    case 1:
+      hipLaunchKernelGGL((pairwise_l2_kernel<1>), grid, dim3(256), lds, stream,
+                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out);
+      break;
+    case 2:
+      hipLaunchKernelGGL((pairwise_l2_kernel<2>), grid, dim3(256), lds, stream,
+                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out);
+      break;
+    case 3: {
+      static bool attr_set = false;
+      if (!attr_set) {
+        HIP_CHECK(hipFuncSetAttribute((const void*)&pairwise_l2_kernel<3>,
+                                      hipFuncAttributeMaxDynamicSharedMemorySize,
+                                      96 * 1024));
+        attr_set = true;
+      }
+      hipLaunchKernelGGL((pairwise_l2_kernel<3>), grid, dim3(256), lds, stream,
+                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out);
+      break;
+    }
+    default:
+      throw std::runtime_error("pairwise_l2_mfma: nslice must be 1, 2 or 3");
+  }
+}
+
+}  // namespace raft_amd

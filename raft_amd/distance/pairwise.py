@@ -106,13 +106,26 @@ def _gemm_xyt(x, y, fp32_mode):
 def _l2_squared(x, y, fp32_mode):
     """||x||^2 + ||y||^2 - 2 x.y with fused epilogue on GPU."""
     if on_gpu(x, y) and x.dtype == torch.bfloat16:
-        # bf16 path: MFMA GEMM with fp32 accumulate/output + fused epilogue;
-        # distances stay fp32 (select_k consumes them directly)
         ext = require_ext()
         xn = x.float().pow(2).sum(dim=1)
         yn = y.float().pow(2).sum(dim=1)
+        if x.shape[1] % 64 == 0:
+            # single-write MFMA tile kernel (epilogue fused into the C-write)
+            return ext.pairwise_l2_mfma([x.contiguous()], [y.contiguous()],
+                                        xn.contiguous(), yn.contiguous())
         g = ext.gemm_bf16_f32_nt(x.contiguous(), y.contiguous())
         return ext.l2_epilogue_(g, xn.contiguous(), yn.contiguous())
+    if (on_gpu(x, y) and x.dtype == torch.float32 and x.shape[1] % 64 == 0
+            and fp32_mode in ("bf16x2", "bf16x3", "mfma")):
+        # split-bf16 MFMA tile kernel: distance tile written exactly once
+        from raft_amd.neighbors.fused_l2nn import split_bf16_slices
+        ext = require_ext()
+        nsl = 2 if fp32_mode == "bf16x2" else 3
+        xs = split_bf16_slices(x, nsl)
+        ys = split_bf16_slices(y, nsl)
+        xn = (x * x).sum(dim=1)
+        yn = (y * y).sum(dim=1)
+        return ext.pairwise_l2_mfma(xs, ys, xn.contiguous(), yn.contiguous())
     xn = (x.double() * x.double()).sum(dim=1) if x.device.type == "cpu" else (x * x).sum(dim=1)
     yn = (y.double() * y.double()).sum(dim=1) if y.device.type == "cpu" else (y * y).sum(dim=1)
     g = _gemm_xyt(x, y, fp32_mode)

@@ -180,6 +180,36 @@ class TestFusedL2NNMfma:
             assert agree > 0.9999, (mode, agree)
 
 
+class TestPairwiseMfma:
+    @pytest.mark.parametrize("mode,tol", [("bf16x2", 3e-3), ("bf16x3", 3e-4)])
+    def test_fp32_tile_vs_fp64(self, dev, ext, mode, tol):
+        from raft_amd.distance import pairwise_distance, DistanceType
+        torch.manual_seed(0)
+        x = torch.randn(777, 192, device=dev)   # non-multiple of 128 rows/cols
+        y = torch.randn(333, 192, device=dev)
+        d = pairwise_distance(x, y, DistanceType.L2Expanded, fp32_mode=mode)
+        ref = torch.cdist(x.double(), y.double()) ** 2
+        err = (d.double() - ref).abs().max() / ref.abs().max()
+        assert float(err) < tol, float(err)
+
+    def test_bf16_tile(self, dev, ext):
+        from raft_amd.distance import pairwise_distance, DistanceType
+        torch.manual_seed(1)
+        x = torch.randn(500, 128, device=dev).bfloat16()
+        y = torch.randn(300, 128, device=dev).bfloat16()
+        d = pairwise_distance(x, y, DistanceType.L2Expanded)
+        ref = torch.cdist(x.double(), y.double()) ** 2
+        err = (d.double() - ref).abs().max() / ref.abs().max()
+        assert float(err) < 2e-2, float(err)
+
+    def test_sqrt_variant(self, dev, ext):
+        from raft_amd.distance import pairwise_distance, DistanceType
+        x = torch.randn(256, 64, device=dev)
+        d = pairwise_distance(x, x, DistanceType.L2SqrtExpanded, fp32_mode="bf16x3")
+        assert d.diagonal().abs().max() < 1e-2
+        torch.testing.assert_close(d, d.t(), rtol=1e-3, atol=1e-3)
+
+
 class TestRngGpu:
     def test_uniform_bitwise_matches_cpu(self, dev, ext):
         from raft_amd.random import uniform, RngState
