@@ -42,6 +42,10 @@ void launch_select_k(const float*, float*, int*, long long, long long, int, bool
 // from fused_l2nn.hip
 void launch_fused_l2nn_split(const void**, const void**, const float*, const float*,
                              float*, int*, long long, int, int, int, hipStream_t);
+// from fused_l2nn_v2.hip (persistent-X variant)
+bool fused_l2nn_persist_supported(int nslice, int d);
+void launch_fused_l2nn_persist(const void**, const void**, const float*, const float*,
+                               float*, int*, long long, int, int, int, hipStream_t);
 // from pairwise_mfma.hip
 void launch_pairwise_l2_mfma(const void**, const void**, const float*, const float*,
                              float*, long long, long long, int, long long, int, bool,
@@ -255,10 +259,17 @@ std::tuple<torch::Tensor, torch::Tensor> fused_l2nn_split(
   TORCH_CHECK(cn.numel() == n && xn.numel() == m);
   auto dmin = torch::empty({m}, xn.options());
   auto amin = torch::empty({m}, xn.options().dtype(torch::kInt32));
-  raft_amd::launch_fused_l2nn_split(xsl, csl, xn.data_ptr<float>(),
-                                    cn.data_ptr<float>(), dmin.data_ptr<float>(),
-                                    amin.data_ptr<int>(), m, (int)n, (int)d,
-                                    nslice, cur_stream());
+  if (raft_amd::fused_l2nn_persist_supported(nslice, (int)d)) {
+    raft_amd::launch_fused_l2nn_persist(xsl, csl, xn.data_ptr<float>(),
+                                        cn.data_ptr<float>(), dmin.data_ptr<float>(),
+                                        amin.data_ptr<int>(), m, (int)n, (int)d,
+                                        nslice, cur_stream());
+  } else {
+    raft_amd::launch_fused_l2nn_split(xsl, csl, xn.data_ptr<float>(),
+                                      cn.data_ptr<float>(), dmin.data_ptr<float>(),
+                                      amin.data_ptr<int>(), m, (int)n, (int)d,
+                                      nslice, cur_stream());
+  }
   return {dmin, amin};
 }
 

@@ -39,9 +39,13 @@ void launch_csr_spmv(const int* indptr, const int* indices, const T* values,
   constexpr int BLOCK = 256;
   auto launch = [&](auto swc) {
     constexpr int SW = decltype(swc)::value;
-    int grid = grid_1d(n_rows * SW, BLOCK);
-    hipLaunchKernelGGL((csr_spmv_kernel<SW, T>), dim3(grid), dim3(BLOCK), 0, stream,
-                       indptr, indices, values, x, y, n_rows);
+    // NOT grid-capped: short rows make this latency-bound (indptr loads per
+    // row); one sub-wave per row maximizes memory-level parallelism
+    // (capped grid measured 746 GB/s effective @ 10 nnz/row)
+    long long blocks = (n_rows * SW + BLOCK - 1) / BLOCK;
+    if (blocks > 2147483647ll) blocks = 2147483647ll;
+    hipLaunchKernelGGL((csr_spmv_kernel<SW, T>), dim3((unsigned)blocks), dim3(BLOCK),
+                       0, stream, indptr, indices, values, x, y, n_rows);
   };
   if (mean <= 4) launch(std::integral_constant<int, 2>{});
   else if (mean <= 8) launch(std::integral_constant<int, 4>{});
