@@ -259,7 +259,11 @@ std::tuple<torch::Tensor, torch::Tensor> fused_l2nn_split(
   TORCH_CHECK(cn.numel() == n && xn.numel() == m);
   auto dmin = torch::empty({m}, xn.options());
   auto amin = torch::empty({m}, xn.options().dtype(torch::kInt32));
-  if (raft_amd::fused_l2nn_persist_supported(nslice, (int)d)) {
+  static const bool use_persist = [] {
+    const char* e = getenv("RAFT_AMD_PERSIST_L2NN");
+    return e && e[0] == '1';
+  }();  // measured slower than v1 at 10M x 256 (occupancy); kept for tuning
+  if (use_persist && raft_amd::fused_l2nn_persist_supported(nslice, (int)d)) {
     raft_amd::launch_fused_l2nn_persist(xsl, csl, xn.data_ptr<float>(),
                                         cn.data_ptr<float>(), dmin.data_ptr<float>(),
                                         amin.data_ptr<int>(), m, (int)n, (int)d,

@@ -202,9 +202,11 @@ def kmeans_iterate(x: torch.Tensor, centroids: torch.Tensor, n_iters: int,
     if use_fused:
         x_slices = split_bf16_slices(x, _MODE_NSLICE[fp32_mode])
         xn = (x * x).sum(dim=1)
-    for _ in range(n_iters):
+    inertia_t = None
+    for it in range(n_iters):
         if use_fused:
-            dmin, labels = fused_l2nn_presplit(x_slices, xn, centroids)
+            dmin, labels = fused_l2nn_presplit(x_slices, xn, centroids,
+                                               int32_labels=True)
         else:
             dmin, labels = fused_l2nn(x, centroids, fp32_mode=fp32_mode,
                                       chunk_rows=chunk_rows)
@@ -216,10 +218,12 @@ def kmeans_iterate(x: torch.Tensor, centroids: torch.Tensor, n_iters: int,
             comms.allreduce(packed, op=ReduceOp.SUM)
         sums = packed[: k * d].reshape(k, d)
         counts = packed[k * d: k * d + k]
-        inertia = float(packed[-1].item())
+        inertia_t = packed[-1]
         nonzero = counts > 0
         centroids = torch.where(nonzero.unsqueeze(1),
                                 sums / counts.clamp_min(1).unsqueeze(1), centroids)
+    # single host sync at the end (a per-iter .item() serializes the pipeline)
+    inertia = float(inertia_t.item()) if inertia_t is not None else float("inf")
     return centroids, inertia
 
 

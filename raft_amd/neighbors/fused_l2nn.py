@@ -50,7 +50,7 @@ def _pad_cols(y: torch.Tensor, yn: torch.Tensor, mult: int = 128):
 
 
 def fused_l2nn_presplit(x_slices, xn: torch.Tensor, y: torch.Tensor,
-                        sqrt: bool = False):
+                        sqrt: bool = False, int32_labels: bool = False):
     """Fused kernel entry with precomputed X slices (k-means hot loop)."""
     ext = require_ext()
     nslice = len(x_slices)
@@ -61,7 +61,7 @@ def fused_l2nn_presplit(x_slices, xn: torch.Tensor, y: torch.Tensor,
                                       xn.contiguous(), ynp.contiguous())
     if sqrt:
         dmin = dmin.clamp_min(0).sqrt()
-    return dmin, amin.to(torch.int64)
+    return dmin, (amin if int32_labels else amin.to(torch.int64))
 
 
 def fused_l2nn(x: torch.Tensor, y: torch.Tensor, sqrt: bool = False,
