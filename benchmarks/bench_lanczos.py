@@ -24,9 +24,8 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 def make_symmetric_csr(n, nnz_per_row, device, seed=0):
     from raft_amd.sparse.types import COO
     from raft_amd.sparse.convert import coo_to_csr
-    g = torch.Generator(device="cpu").manual_seed(seed)
+    torch.manual_seed(seed)   # deterministic graph
     m = n * nnz_per_row // 2
-    # generate on device with torch RNG (fast path for the generator itself)
     rows = torch.randint(0, n, (m,), device=device, dtype=torch.int64)
     cols = torch.randint(0, n, (m,), device=device, dtype=torch.int64)
     vals = torch.rand(m, device=device, dtype=torch.float32) + 0.1
@@ -76,6 +75,17 @@ def main():
     spmv_dt = (time.perf_counter() - t0) / args.spmv_iters
     spmv_gbps = (nnz * 12 + args.n * 16) / spmv_dt / 1e9
 
+    # vendor comparison: rocSPARSE csrmv via torch.sparse
+    ts = csr.to_torch_sparse()
+    xu = x.unsqueeze(1)
+    _ = ts @ xu
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.spmv_iters):
+        _ = ts @ xu
+    torch.cuda.synchronize()
+    rocsparse_dt = (time.perf_counter() - t0) / args.spmv_iters
+
     # ---- Lanczos ------------------------------------------------------------
     ncv = max(2 * args.k + 1, 32)
     cfg = LanczosConfig(n_components=args.k, max_iterations=args.lanczos_restarts,
@@ -93,6 +103,7 @@ def main():
         "spmv_ms": round(spmv_dt * 1e3, 3),
         "spmv_sweeps_per_sec": round(1.0 / spmv_dt, 2),
         "spmv_effective_GBps": round(spmv_gbps, 1),
+        "rocsparse_spmv_ms": round(rocsparse_dt * 1e3, 3),
         "lanczos_steps_per_sec": round(steps / lancz_dt, 2),
         "lanczos_elapsed_s": round(lancz_dt, 2),
         "lanczos_steps": steps,
