@@ -24,6 +24,47 @@ __device__ __forceinline__ uint32_t f32_to_ord(float f, bool select_min) {
 }
 
 constexpr int SELECT_K_MAX = 2048;
+constexpr int CAND_CAP = 8192;
+
+
+// sort (optional) + write the k selected pairs for one row
+template <int BLOCK>
+__device__ __forceinline__ void select_k_finish(const float* __restrict__ rp,
+                                                float* __restrict__ out_v,
+                                                int* __restrict__ out_i,
+                                                long long row, int k, bool do_sort,
+                                                uint32_t* pair_u, int* pair_i) {
+  if (do_sort) {
+    int kp = 1;
+    while (kp < k) kp <<= 1;
+    for (int j = threadIdx.x + k; j < kp; j += BLOCK) {
+      if (j < SELECT_K_MAX) { pair_u[j] = 0xFFFFFFFFu; pair_i[j] = -1; }
+    }
+    __syncthreads();
+    for (int size = 2; size <= kp; size <<= 1) {
+      for (int strd = size >> 1; strd > 0; strd >>= 1) {
+        for (int t = threadIdx.x; t < kp / 2; t += BLOCK) {
+          const int i0 = 2 * t - (t & (strd - 1));
+          const int i1 = i0 + strd;
+          const bool up = ((i0 & size) == 0);
+          const uint32_t a = pair_u[i0], b = pair_u[i1];
+          const bool swap_ = (a > b || (a == b && pair_i[i0] > pair_i[i1])) == up;
+          if (swap_) {
+            pair_u[i0] = b; pair_u[i1] = a;
+            const int ti = pair_i[i0]; pair_i[i0] = pair_i[i1]; pair_i[i1] = ti;
+          }
+        }
+        __syncthreads();
+      }
+    }
+  }
+  for (int j = threadIdx.x; j < k; j += BLOCK) {
+    const int src = pair_i[j];
+    out_v[row * k + j] = rp[src];
+    out_i[row * k + j] = src;
+  }
+  __syncthreads();
+}
 
 template <int BLOCK = 256>
 __global__ void select_k_radix_kernel(const float* __restrict__ x,
@@ -35,16 +76,99 @@ __global__ void select_k_radix_kernel(const float* __restrict__ x,
   __shared__ unsigned int sh_prefix, sh_below, sh_cnt_lt, sh_cnt_eq;
   __shared__ uint32_t pair_u[SELECT_K_MAX];
   __shared__ int pair_i[SELECT_K_MAX];
+  __shared__ uint32_t cand_u[CAND_CAP];
+  __shared__ int cand_i[CAND_CAP];
+  __shared__ unsigned int sh_cand_cnt;
 
   for (long long row = blockIdx.x; row < batch; row += gridDim.x) {
     const float* rp = x + row * len;
 
-    // ---- 4 MSB-first passes to find the k-th smallest transformed key ----
-    uint32_t prefix = 0;          // high bits decided so far
-    uint32_t prefix_mask = 0;     // which bits are decided
-    unsigned int below = 0;       // count strictly below current prefix bucket
-    int remaining = k;
-    for (int pass = 0; pass < 4; pass++) {
+    // ---- pass 1: MSB histogram (full scan) -------------------------------
+    if (threadIdx.x < 256) hist[threadIdx.x] = 0;
+    if (threadIdx.x == 0) { sh_cnt_lt = 0; sh_cand_cnt = 0; }
+    __syncthreads();
+    for (long long j = threadIdx.x; j < len; j += BLOCK) {
+      const uint32_t u = f32_to_ord(rp[j], select_min);
+      atomicAdd(&hist[u >> 24], 1u);
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      unsigned int cum = 0;
+      int bucket = 255;
+      for (int b = 0; b < 256; b++) {
+        const unsigned int c = hist[b];
+        if (cum + c >= (unsigned int)k) { bucket = b; break; }
+        cum += c;
+      }
+      sh_prefix = (unsigned int)bucket;
+      sh_below = cum;
+    }
+    __syncthreads();
+    const uint32_t b1 = sh_prefix;
+    const unsigned int below1 = sh_below;
+    const int remaining1 = k - (int)below1;
+
+    // ---- pass 2 (full scan): emit sure winners, compact the kth bucket ---
+    for (long long j = threadIdx.x; j < len; j += BLOCK) {
+      const uint32_t u = f32_to_ord(rp[j], select_min);
+      const uint32_t byte0 = u >> 24;
+      if (byte0 < b1) {
+        const unsigned int slot = atomicAdd(&sh_cnt_lt, 1u);
+        pair_u[slot] = u;
+        pair_i[slot] = (int)j;
+      } else if (byte0 == b1) {
+        const unsigned int c = atomicAdd(&sh_cand_cnt, 1u);
+        if (c < CAND_CAP) { cand_u[c] = u; cand_i[c] = (int)j; }
+      }
+    }
+    __syncthreads();
+    const unsigned int cand_cnt = sh_cand_cnt;
+
+    if (cand_cnt <= CAND_CAP) {
+      // ---- fast path: sort the candidate bucket in LDS, take remaining1 --
+      unsigned int cp = 1;
+      while (cp < cand_cnt) cp <<= 1;
+      if (cp < 1) cp = 1;
+      for (unsigned int j = threadIdx.x + cand_cnt; j < cp; j += BLOCK) {
+        cand_u[j] = 0xFFFFFFFFu;
+        cand_i[j] = -1;
+      }
+      __syncthreads();
+      for (unsigned int size = 2; size <= cp; size <<= 1) {
+        for (unsigned int strd = size >> 1; strd > 0; strd >>= 1) {
+          for (unsigned int t = threadIdx.x; t < cp / 2; t += BLOCK) {
+            const unsigned int i0 = 2 * t - (t & (strd - 1));
+            const unsigned int i1 = i0 + strd;
+            const bool up = ((i0 & size) == 0);
+            const uint32_t a = cand_u[i0], b = cand_u[i1];
+            // tie-break on index for deterministic output
+            const bool swap_ = (a > b || (a == b && cand_i[i0] > cand_i[i1])) == up;
+            if (swap_) {
+              cand_u[i0] = b; cand_u[i1] = a;
+              const int ti = cand_i[i0]; cand_i[i0] = cand_i[i1]; cand_i[i1] = ti;
+            }
+          }
+          __syncthreads();
+        }
+      }
+      for (int j = threadIdx.x; j < remaining1; j += BLOCK) {
+        pair_u[below1 + j] = cand_u[j];
+        pair_i[below1 + j] = cand_i[j];
+      }
+      __syncthreads();
+      select_k_finish<BLOCK>(rp, out_v, out_i, row, k, do_sort, pair_u, pair_i);
+      continue;
+    }
+
+    // ---- fallback (skewed data overflowed the candidate buffer):
+    // classic multi-pass refinement over the full row ----------------------
+    uint32_t prefix = b1 << 24;
+    uint32_t prefix_mask = 0xFFu << 24;
+    unsigned int below = below1;
+    int remaining = remaining1;
+    if (threadIdx.x == 0) { sh_cnt_lt = below1; }
+    __syncthreads();
+    for (int pass = 1; pass < 4; pass++) {
       const int shift = 8 * (3 - pass);
       if (threadIdx.x < 256) hist[threadIdx.x] = 0;
       __syncthreads();
@@ -94,39 +218,7 @@ __global__ void select_k_radix_kernel(const float* __restrict__ x,
       }
     }
     __syncthreads();
-
-    // ---- optional in-LDS bitonic sort of the k pairs (ascending u) ----
-    if (do_sort) {
-      int kp = 1;
-      while (kp < k) kp <<= 1;
-      for (int j = threadIdx.x + k; j < kp; j += BLOCK) {
-        if (j < SELECT_K_MAX) { pair_u[j] = 0xFFFFFFFFu; pair_i[j] = -1; }
-      }
-      __syncthreads();
-      for (int size = 2; size <= kp; size <<= 1) {
-        for (int strd = size >> 1; strd > 0; strd >>= 1) {
-          for (int t = threadIdx.x; t < kp / 2; t += BLOCK) {
-            const int i0 = 2 * t - (t & (strd - 1));
-            const int i1 = i0 + strd;
-            const bool up = ((i0 & size) == 0);
-            const uint32_t a = pair_u[i0], b = pair_u[i1];
-            if ((a > b) == up) {
-              pair_u[i0] = b; pair_u[i1] = a;
-              const int ti = pair_i[i0]; pair_i[i0] = pair_i[i1]; pair_i[i1] = ti;
-            }
-          }
-          __syncthreads();
-        }
-      }
-    }
-
-    // ---- write results (value re-read from source by index: no inverse map) --
-    for (int j = threadIdx.x; j < k; j += BLOCK) {
-      const int src = pair_i[j];
-      out_v[row * k + j] = rp[src];
-      out_i[row * k + j] = src;
-    }
-    __syncthreads();
+    select_k_finish<BLOCK>(rp, out_v, out_i, row, k, do_sort, pair_u, pair_i);
   }
 }
 

@@ -41,8 +41,36 @@ def select_k(x: torch.Tensor, k: int, select_min: bool = True,
     k = int(k)
     assert 0 < k <= n, f"k={k} out of range for row length {n}"
 
-    if on_gpu(x) and x.dtype == torch.float32 and algo != SelectAlgo.TORCH:
+    if on_gpu(x) and x.dtype == torch.float32 and algo != SelectAlgo.TORCH and k <= 2048:
         ext = require_ext()
+        # two-level split: a small batch over a huge row leaves the chip idle
+        # (one workgroup per row); split rows into S segments, select per
+        # segment, then select over the S*k candidates (the same multi-block
+        # merge idea as the reference's two-pass warpsort).
+        if batch < 256 and n >= 262144 and n >= 8 * k:
+            s = 1
+            while batch * s * 2 <= 2048 and (n // (s * 2)) >= max(4 * k, 4096):
+                s *= 2
+            if s > 1:
+                seg = -(-n // s)          # ceil
+                pad = seg * s - n
+                if pad:
+                    fill = float("inf") if select_min else float("-inf")
+                    xp = torch.cat([x, torch.full((batch, pad), fill, dtype=x.dtype,
+                                                  device=x.device)], dim=1)
+                else:
+                    xp = x
+                segs = xp.reshape(batch * s, seg).contiguous()
+                sv, si = ext.select_k(segs, min(k, seg), bool(select_min), 0, False)
+                kk = sv.shape[1]
+                cand_v = sv.reshape(batch, s * kk)
+                base = (torch.arange(s, device=x.device, dtype=torch.int64) * seg)
+                cand_i = (si.to(torch.int64).reshape(batch, s, kk)
+                          + base.view(1, s, 1)).reshape(batch, s * kk)
+                fv, fpos = ext.select_k(cand_v.contiguous(), k, bool(select_min),
+                                        0, bool(sorted))
+                gi = torch.gather(cand_i, 1, fpos.to(torch.int64))
+                return fv, gi
         algo_code = {SelectAlgo.AUTO: 0, SelectAlgo.RADIX: 1, SelectAlgo.WARPSORT: 2}[algo]
         vals, idx = ext.select_k(x.contiguous(), k, bool(select_min), algo_code, bool(sorted))
         return vals, idx.to(torch.int64)
