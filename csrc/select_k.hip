@@ -125,35 +125,55 @@ __global__ void select_k_radix_kernel(const float* __restrict__ x,
     const unsigned int cand_cnt = sh_cand_cnt;
 
     if (cand_cnt <= CAND_CAP) {
-      // ---- fast path: sort the candidate bucket in LDS, take remaining1 --
-      unsigned int cp = 1;
-      while (cp < cand_cnt) cp <<= 1;
-      if (cp < 1) cp = 1;
-      for (unsigned int j = threadIdx.x + cand_cnt; j < cp; j += BLOCK) {
-        cand_u[j] = 0xFFFFFFFFu;
-        cand_i[j] = -1;
-      }
-      __syncthreads();
-      for (unsigned int size = 2; size <= cp; size <<= 1) {
-        for (unsigned int strd = size >> 1; strd > 0; strd >>= 1) {
-          for (unsigned int t = threadIdx.x; t < cp / 2; t += BLOCK) {
-            const unsigned int i0 = 2 * t - (t & (strd - 1));
-            const unsigned int i1 = i0 + strd;
-            const bool up = ((i0 & size) == 0);
-            const uint32_t a = cand_u[i0], b = cand_u[i1];
-            // tie-break on index for deterministic output
-            const bool swap_ = (a > b || (a == b && cand_i[i0] > cand_i[i1])) == up;
-            if (swap_) {
-              cand_u[i0] = b; cand_u[i1] = a;
-              const int ti = cand_i[i0]; cand_i[i0] = cand_i[i1]; cand_i[i1] = ti;
-            }
-          }
-          __syncthreads();
+      // ---- fast path: refine the k-th key inside the candidate buffer
+      // (3 histogram passes over <=8192 LDS-resident elements — cheap),
+      // then collect from the buffer. Full-row reads stop at 2.
+      uint32_t prefix = b1 << 24;
+      uint32_t prefix_mask = 0xFFu << 24;
+      int remaining = remaining1;
+      for (int pass = 1; pass < 4; pass++) {
+        const int shift = 8 * (3 - pass);
+        if (threadIdx.x < 256) hist[threadIdx.x] = 0;
+        __syncthreads();
+        for (unsigned int j = threadIdx.x; j < cand_cnt; j += BLOCK) {
+          const uint32_t u = cand_u[j];
+          if ((u & prefix_mask) == prefix) atomicAdd(&hist[(u >> shift) & 0xFF], 1u);
         }
+        __syncthreads();
+        if (threadIdx.x == 0) {
+          unsigned int cum = 0;
+          int bucket = 255;
+          for (int b = 0; b < 256; b++) {
+            const unsigned int c = hist[b];
+            if (cum + c >= (unsigned int)remaining) { bucket = b; break; }
+            cum += c;
+          }
+          sh_prefix = (unsigned int)bucket;
+          sh_below = cum;
+        }
+        __syncthreads();
+        remaining -= (int)sh_below;
+        prefix |= (sh_prefix << shift);
+        prefix_mask |= (0xFFu << shift);
+        __syncthreads();
       }
-      for (int j = threadIdx.x; j < remaining1; j += BLOCK) {
-        pair_u[below1 + j] = cand_u[j];
-        pair_i[below1 + j] = cand_i[j];
+      const uint32_t kth = prefix;
+      const unsigned int n_lt2 = (unsigned int)(remaining1 - remaining);
+      if (threadIdx.x == 0) { sh_cnt_lt = 0; sh_cnt_eq = 0; }
+      __syncthreads();
+      for (unsigned int j = threadIdx.x; j < cand_cnt; j += BLOCK) {
+        const uint32_t u = cand_u[j];
+        if (u < kth) {
+          const unsigned int slot = atomicAdd(&sh_cnt_lt, 1u);
+          pair_u[below1 + slot] = u;
+          pair_i[below1 + slot] = cand_i[j];
+        } else if (u == kth) {
+          const unsigned int e = atomicAdd(&sh_cnt_eq, 1u);
+          if (e < (unsigned int)remaining) {
+            pair_u[below1 + n_lt2 + e] = u;
+            pair_i[below1 + n_lt2 + e] = cand_i[j];
+          }
+        }
       }
       __syncthreads();
       select_k_finish<BLOCK>(rp, out_v, out_i, row, k, do_sort, pair_u, pair_i);
@@ -226,7 +246,7 @@ void launch_select_k(const float* x, float* out_v, int* out_i, long long batch,
                      long long len, int k, bool select_min, bool do_sort,
                      hipStream_t stream) {
   if (k > SELECT_K_MAX) throw std::runtime_error("select_k native path supports k <= 2048");
-  int grid = (int)(batch < 2048 ? batch : 2048);
+  int grid = (int)(batch < 65536 ? batch : 65536);
   hipLaunchKernelGGL((select_k_radix_kernel<256>), dim3(grid), dim3(256), 0, stream,
                      x, out_v, out_i, batch, len, k, select_min, do_sort);
 }

@@ -41,7 +41,11 @@ def select_k(x: torch.Tensor, k: int, select_min: bool = True,
     k = int(k)
     assert 0 < k <= n, f"k={k} out of range for row length {n}"
 
-    if on_gpu(x) and x.dtype == torch.float32 and algo != SelectAlgo.TORCH and k <= 2048:
+    if (on_gpu(x) and x.dtype == torch.float32 and algo != SelectAlgo.TORCH
+            and k <= 2048 and n > 4096):
+        # n <= 4096 rows are cheapest through the vendor segmented sort
+        # (rocPRIM topk) — the same shape-dispatch idea as the reference's
+        # learned tree, re-measured on gfx950 (see benchmarks)
         ext = require_ext()
         # two-level split: a small batch over a huge row leaves the chip idle
         # (one workgroup per row); split rows into S segments, select per
