@@ -37,8 +37,9 @@ void launch_reduce_rows_by_key(const float*, const int*, float*, float*, long lo
 void launch_reduce_rows_by_key_sorted(const float*, const int*, const int*, float*,
                                       long long, long long, hipStream_t);
 // from select_k.hip
-void launch_select_k(const float*, float*, int*, long long, long long, int, bool, bool,
-                     hipStream_t);
+long long select_k_workspace_bytes(long long batch);
+void launch_select_k(const float*, float*, int*, void*, long long, long long, int,
+                     bool, bool, hipStream_t);
 // from fused_l2nn.hip
 void launch_fused_l2nn_split(const void**, const void**, const float*, const float*,
                              float*, int*, long long, int, int, int, hipStream_t);
@@ -229,9 +230,11 @@ std::tuple<torch::Tensor, torch::Tensor> select_k(torch::Tensor x, int64_t k,
   check_f32_2d(x, "x");
   auto vals = torch::empty({x.size(0), k}, x.options());
   auto idx = torch::empty({x.size(0), k}, x.options().dtype(torch::kInt32));
+  auto ws = torch::empty({raft_amd::select_k_workspace_bytes(x.size(0))},
+                         x.options().dtype(torch::kUInt8));
   raft_amd::launch_select_k(x.data_ptr<float>(), vals.data_ptr<float>(),
-                            idx.data_ptr<int>(), x.size(0), x.size(1), (int)k,
-                            select_min, do_sort, cur_stream());
+                            idx.data_ptr<int>(), ws.data_ptr(), x.size(0), x.size(1),
+                            (int)k, select_min, do_sort, cur_stream());
   return {vals, idx};
 }
 

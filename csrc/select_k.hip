@@ -1,15 +1,20 @@
 // Batched top-k selection — block-per-row MSB radix select (wave64 CDNA4).
 //
 // Reference parity (WHAT): raft/matrix/detail/select_radix.cuh (multi-pass
-// MSB-first radix with histogram + bucket choose; the one-block-per-row
-// fully-in-kernel variant radix_topk_one_block_kernel:1040) and the optional
-// post-sort. The reference's cross-block Counter machinery is replaced by a
-// one-block-per-row design: MI355X's 256 CUs × grid-stride cover large
-// batches, and per-row data streams from HBM/L2 at full width (4+1 passes).
+// MSB-first radix with histogram + bucket choose, candidate compaction, and
+// the one-block-per-row variant radix_topk_one_block_kernel:1040) and the
+// optional post-sort.
 //
-// Monotone bit transform: ascending float order == ascending transformed-u32
-// order; select_max runs the same kernel on bit-complemented keys.
-// Final top-k pairs are bitonic-sorted in LDS (k <= 2048).
+// MI355X design:
+//  * one block per row, float4-vectorized scans, per-wave LDS histograms
+//    (merged after) to avoid LDS-atomic serialization;
+//  * 2 full-row passes: (1) MSB histogram, (2) emit sure winners + compact
+//    the boundary bucket into a per-block GLOBAL workspace (keeping LDS at
+//    ~21 KiB for multi-block occupancy — an LDS candidate buffer measured
+//    1 block/CU and 5x slower); remaining radix passes refine inside the
+//    compacted candidates (expected len/256 elements, L2-resident);
+//  * skewed rows overflowing the workspace fall back to full-row passes;
+//  * final top-k pairs bitonic-sorted in LDS (k <= 2048).
 
 #include <hip/hip_runtime.h>
 
@@ -24,8 +29,7 @@ __device__ __forceinline__ uint32_t f32_to_ord(float f, bool select_min) {
 }
 
 constexpr int SELECT_K_MAX = 2048;
-constexpr int CAND_CAP = 8192;
-
+constexpr int CAND_CAP = 16384;  // per-block gmem candidate slots
 
 // sort (optional) + write the k selected pairs for one row
 template <int BLOCK>
@@ -66,37 +70,60 @@ __device__ __forceinline__ void select_k_finish(const float* __restrict__ rp,
   __syncthreads();
 }
 
+// per-wave histogram helper: each wave owns hist[wave][256]; merged by caller
+template <int BLOCK>
+__device__ __forceinline__ void merge_hists(unsigned int (*hist)[256]) {
+  constexpr int NW = BLOCK / RAFT_AMD_WAVE;
+  __syncthreads();
+  for (int b = threadIdx.x; b < 256; b += BLOCK) {
+    unsigned int s = 0;
+#pragma unroll
+    for (int w = 0; w < NW; w++) s += hist[w][b];
+    hist[0][b] = s;
+  }
+  __syncthreads();
+}
+
 template <int BLOCK = 256>
 __global__ void select_k_radix_kernel(const float* __restrict__ x,
                                       float* __restrict__ out_v,
                                       int* __restrict__ out_i,
+                                      uint2* __restrict__ cand_ws,
                                       long long batch, long long len, int k,
                                       bool select_min, bool do_sort) {
-  __shared__ unsigned int hist[256];
-  __shared__ unsigned int sh_prefix, sh_below, sh_cnt_lt, sh_cnt_eq;
+  constexpr int NW = BLOCK / RAFT_AMD_WAVE;
+  __shared__ unsigned int hist[NW][256];
+  __shared__ unsigned int sh_prefix, sh_below, sh_cnt_lt, sh_cnt_eq, sh_cand_cnt;
   __shared__ uint32_t pair_u[SELECT_K_MAX];
   __shared__ int pair_i[SELECT_K_MAX];
-  __shared__ uint32_t cand_u[CAND_CAP];
-  __shared__ int cand_i[CAND_CAP];
-  __shared__ unsigned int sh_cand_cnt;
+  uint2* cand = cand_ws + (long long)blockIdx.x * CAND_CAP;
+  const int wid = threadIdx.x / RAFT_AMD_WAVE;
 
   for (long long row = blockIdx.x; row < batch; row += gridDim.x) {
     const float* rp = x + row * len;
+    const long long len4 = len / 4;
+    const float4* rp4 = reinterpret_cast<const float4*>(rp);
 
-    // ---- pass 1: MSB histogram (full scan) -------------------------------
-    if (threadIdx.x < 256) hist[threadIdx.x] = 0;
+    // ---- pass 1: MSB histogram (vectorized full scan, per-wave hists) ----
+    for (int b = threadIdx.x; b < NW * 256; b += BLOCK)
+      reinterpret_cast<unsigned int*>(hist)[b] = 0;
     if (threadIdx.x == 0) { sh_cnt_lt = 0; sh_cand_cnt = 0; }
     __syncthreads();
-    for (long long j = threadIdx.x; j < len; j += BLOCK) {
-      const uint32_t u = f32_to_ord(rp[j], select_min);
-      atomicAdd(&hist[u >> 24], 1u);
+    for (long long j = threadIdx.x; j < len4; j += BLOCK) {
+      const float4 v = rp4[j];
+      atomicAdd(&hist[wid][f32_to_ord(v.x, select_min) >> 24], 1u);
+      atomicAdd(&hist[wid][f32_to_ord(v.y, select_min) >> 24], 1u);
+      atomicAdd(&hist[wid][f32_to_ord(v.z, select_min) >> 24], 1u);
+      atomicAdd(&hist[wid][f32_to_ord(v.w, select_min) >> 24], 1u);
     }
-    __syncthreads();
+    for (long long j = len4 * 4 + threadIdx.x; j < len; j += BLOCK)
+      atomicAdd(&hist[wid][f32_to_ord(rp[j], select_min) >> 24], 1u);
+    merge_hists<BLOCK>(hist);
     if (threadIdx.x == 0) {
       unsigned int cum = 0;
       int bucket = 255;
       for (int b = 0; b < 256; b++) {
-        const unsigned int c = hist[b];
+        const unsigned int c = hist[0][b];
         if (cum + c >= (unsigned int)k) { bucket = b; break; }
         cum += c;
       }
@@ -108,9 +135,8 @@ __global__ void select_k_radix_kernel(const float* __restrict__ x,
     const unsigned int below1 = sh_below;
     const int remaining1 = k - (int)below1;
 
-    // ---- pass 2 (full scan): emit sure winners, compact the kth bucket ---
-    for (long long j = threadIdx.x; j < len; j += BLOCK) {
-      const uint32_t u = f32_to_ord(rp[j], select_min);
+    // ---- pass 2: emit sure winners (LDS), compact boundary bucket (gmem) --
+    auto classify = [&](uint32_t u, long long j) {
       const uint32_t byte0 = u >> 24;
       if (byte0 < b1) {
         const unsigned int slot = atomicAdd(&sh_cnt_lt, 1u);
@@ -118,33 +144,43 @@ __global__ void select_k_radix_kernel(const float* __restrict__ x,
         pair_i[slot] = (int)j;
       } else if (byte0 == b1) {
         const unsigned int c = atomicAdd(&sh_cand_cnt, 1u);
-        if (c < CAND_CAP) { cand_u[c] = u; cand_i[c] = (int)j; }
+        if (c < CAND_CAP) cand[c] = uint2{u, (unsigned int)j};
       }
+    };
+    for (long long j = threadIdx.x; j < len4; j += BLOCK) {
+      const float4 v = rp4[j];
+      classify(f32_to_ord(v.x, select_min), j * 4 + 0);
+      classify(f32_to_ord(v.y, select_min), j * 4 + 1);
+      classify(f32_to_ord(v.z, select_min), j * 4 + 2);
+      classify(f32_to_ord(v.w, select_min), j * 4 + 3);
     }
+    for (long long j = len4 * 4 + threadIdx.x; j < len; j += BLOCK)
+      classify(f32_to_ord(rp[j], select_min), j);
     __syncthreads();
     const unsigned int cand_cnt = sh_cand_cnt;
 
+    uint32_t prefix = b1 << 24;
+    uint32_t prefix_mask = 0xFFu << 24;
+    int remaining = remaining1;
+
     if (cand_cnt <= CAND_CAP) {
-      // ---- fast path: refine the k-th key inside the candidate buffer
-      // (3 histogram passes over <=8192 LDS-resident elements — cheap),
-      // then collect from the buffer. Full-row reads stop at 2.
-      uint32_t prefix = b1 << 24;
-      uint32_t prefix_mask = 0xFFu << 24;
-      int remaining = remaining1;
+      // ---- fast path: refine + collect inside the compacted candidates ---
       for (int pass = 1; pass < 4; pass++) {
         const int shift = 8 * (3 - pass);
-        if (threadIdx.x < 256) hist[threadIdx.x] = 0;
+        for (int b = threadIdx.x; b < NW * 256; b += BLOCK)
+          reinterpret_cast<unsigned int*>(hist)[b] = 0;
         __syncthreads();
         for (unsigned int j = threadIdx.x; j < cand_cnt; j += BLOCK) {
-          const uint32_t u = cand_u[j];
-          if ((u & prefix_mask) == prefix) atomicAdd(&hist[(u >> shift) & 0xFF], 1u);
+          const uint32_t u = cand[j].x;
+          if ((u & prefix_mask) == prefix)
+            atomicAdd(&hist[wid][(u >> shift) & 0xFF], 1u);
         }
-        __syncthreads();
+        merge_hists<BLOCK>(hist);
         if (threadIdx.x == 0) {
           unsigned int cum = 0;
           int bucket = 255;
           for (int b = 0; b < 256; b++) {
-            const unsigned int c = hist[b];
+            const unsigned int c = hist[0][b];
             if (cum + c >= (unsigned int)remaining) { bucket = b; break; }
             cum += c;
           }
@@ -162,16 +198,16 @@ __global__ void select_k_radix_kernel(const float* __restrict__ x,
       if (threadIdx.x == 0) { sh_cnt_lt = 0; sh_cnt_eq = 0; }
       __syncthreads();
       for (unsigned int j = threadIdx.x; j < cand_cnt; j += BLOCK) {
-        const uint32_t u = cand_u[j];
+        const uint32_t u = cand[j].x;
         if (u < kth) {
           const unsigned int slot = atomicAdd(&sh_cnt_lt, 1u);
           pair_u[below1 + slot] = u;
-          pair_i[below1 + slot] = cand_i[j];
+          pair_i[below1 + slot] = (int)cand[j].y;
         } else if (u == kth) {
           const unsigned int e = atomicAdd(&sh_cnt_eq, 1u);
           if (e < (unsigned int)remaining) {
             pair_u[below1 + n_lt2 + e] = u;
-            pair_i[below1 + n_lt2 + e] = cand_i[j];
+            pair_i[below1 + n_lt2 + e] = (int)cand[j].y;
           }
         }
       }
@@ -180,29 +216,24 @@ __global__ void select_k_radix_kernel(const float* __restrict__ x,
       continue;
     }
 
-    // ---- fallback (skewed data overflowed the candidate buffer):
-    // classic multi-pass refinement over the full row ----------------------
-    uint32_t prefix = b1 << 24;
-    uint32_t prefix_mask = 0xFFu << 24;
-    unsigned int below = below1;
-    int remaining = remaining1;
-    if (threadIdx.x == 0) { sh_cnt_lt = below1; }
-    __syncthreads();
+    // ---- fallback (boundary bucket overflowed the workspace): classic
+    // multi-pass refinement + collection over the full row -----------------
     for (int pass = 1; pass < 4; pass++) {
       const int shift = 8 * (3 - pass);
-      if (threadIdx.x < 256) hist[threadIdx.x] = 0;
+      for (int b = threadIdx.x; b < NW * 256; b += BLOCK)
+        reinterpret_cast<unsigned int*>(hist)[b] = 0;
       __syncthreads();
       for (long long j = threadIdx.x; j < len; j += BLOCK) {
         const uint32_t u = f32_to_ord(rp[j], select_min);
         if ((u & prefix_mask) == prefix)
-          atomicAdd(&hist[(u >> shift) & 0xFF], 1u);
+          atomicAdd(&hist[wid][(u >> shift) & 0xFF], 1u);
       }
-      __syncthreads();
+      merge_hists<BLOCK>(hist);
       if (threadIdx.x == 0) {
         unsigned int cum = 0;
         int bucket = 255;
         for (int b = 0; b < 256; b++) {
-          const unsigned int c = hist[b];
+          const unsigned int c = hist[0][b];
           if (cum + c >= (unsigned int)remaining) { bucket = b; break; }
           cum += c;
         }
@@ -210,18 +241,15 @@ __global__ void select_k_radix_kernel(const float* __restrict__ x,
         sh_below = cum;
       }
       __syncthreads();
-      below += sh_below;
       remaining -= (int)sh_below;
       prefix |= (sh_prefix << shift);
       prefix_mask |= (0xFFu << shift);
       __syncthreads();
     }
-    const uint32_t kth = prefix;  // exact k-th smallest transformed key
-
-    // ---- collection pass: all u < kth, then u == kth up to k ----
+    const uint32_t kth = prefix;
     if (threadIdx.x == 0) { sh_cnt_lt = 0; sh_cnt_eq = 0; }
     __syncthreads();
-    const unsigned int n_lt = (unsigned int)(k - remaining);  // = count of u < kth
+    const unsigned int n_lt = (unsigned int)(k - remaining);
     for (long long j = threadIdx.x; j < len; j += BLOCK) {
       const uint32_t u = f32_to_ord(rp[j], select_min);
       if (u < kth) {
@@ -231,9 +259,8 @@ __global__ void select_k_radix_kernel(const float* __restrict__ x,
       } else if (u == kth) {
         const unsigned int e = atomicAdd(&sh_cnt_eq, 1u);
         if (e < (unsigned int)remaining) {
-          const unsigned int slot = n_lt + e;
-          pair_u[slot] = u;
-          pair_i[slot] = (int)j;
+          pair_u[n_lt + e] = u;
+          pair_i[n_lt + e] = (int)j;
         }
       }
     }
@@ -242,13 +269,20 @@ __global__ void select_k_radix_kernel(const float* __restrict__ x,
   }
 }
 
-void launch_select_k(const float* x, float* out_v, int* out_i, long long batch,
-                     long long len, int k, bool select_min, bool do_sort,
-                     hipStream_t stream) {
+int select_k_grid(long long batch) { return (int)(batch < 4096 ? batch : 4096); }
+
+long long select_k_workspace_bytes(long long batch) {
+  return (long long)select_k_grid(batch) * CAND_CAP * sizeof(uint2);
+}
+
+void launch_select_k(const float* x, float* out_v, int* out_i, void* cand_ws,
+                     long long batch, long long len, int k, bool select_min,
+                     bool do_sort, hipStream_t stream) {
   if (k > SELECT_K_MAX) throw std::runtime_error("select_k native path supports k <= 2048");
-  int grid = (int)(batch < 65536 ? batch : 65536);
+  const int grid = select_k_grid(batch);
   hipLaunchKernelGGL((select_k_radix_kernel<256>), dim3(grid), dim3(256), 0, stream,
-                     x, out_v, out_i, batch, len, k, select_min, do_sort);
+                     x, out_v, out_i, (uint2*)cand_ws, batch, len, k, select_min,
+                     do_sort);
 }
 
 }  // namespace raft_amd
