@@ -36,17 +36,21 @@ void launch_reduce_rows_by_key(const float*, const int*, float*, float*, long lo
                                long long, long long, int, hipStream_t);
 void launch_reduce_rows_by_key_sorted(const float*, const int*, const int*, float*,
                                       long long, long long, hipStream_t);
+void launch_l2nn_verify_repair(const float*, const float*, const float*, float*, int*,
+                               const float*, float, long long, int, int, hipStream_t);
 // from select_k.hip
 long long select_k_workspace_bytes(long long batch);
 void launch_select_k(const float*, float*, int*, void*, long long, long long, int,
                      bool, bool, hipStream_t);
 // from fused_l2nn.hip
 void launch_fused_l2nn_split(const void**, const void**, const float*, const float*,
-                             float*, int*, long long, int, int, int, hipStream_t);
+                             float*, int*, float*, long long, int, int, int,
+                             hipStream_t);
 // from fused_l2nn_v2.hip (persistent-X variant)
 bool fused_l2nn_persist_supported(int nslice, int d);
 void launch_fused_l2nn_persist(const void**, const void**, const float*, const float*,
-                               float*, int*, long long, int, int, int, hipStream_t);
+                               float*, int*, float*, long long, int, int, int,
+                               hipStream_t);
 // from pairwise_mfma.hip
 void launch_pairwise_l2_mfma(const void**, const void**, const float*, const float*,
                              float*, long long, long long, int, long long, int, bool,
@@ -238,7 +242,7 @@ std::tuple<torch::Tensor, torch::Tensor> select_k(torch::Tensor x, int64_t k,
   return {vals, idx};
 }
 
-std::tuple<torch::Tensor, torch::Tensor> fused_l2nn_split(
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> fused_l2nn_split(
     std::vector<torch::Tensor> x_slices, std::vector<torch::Tensor> c_slices,
     torch::Tensor xn, torch::Tensor cn) {
   const int nslice = (int)x_slices.size();
@@ -262,6 +266,7 @@ std::tuple<torch::Tensor, torch::Tensor> fused_l2nn_split(
   TORCH_CHECK(cn.numel() == n && xn.numel() == m);
   auto dmin = torch::empty({m}, xn.options());
   auto amin = torch::empty({m}, xn.options().dtype(torch::kInt32));
+  auto dmin2 = torch::empty({m}, xn.options());
   static const bool use_persist = [] {
     const char* e = getenv("RAFT_AMD_PERSIST_L2NN");
     return e && e[0] == '1';
@@ -269,15 +274,15 @@ std::tuple<torch::Tensor, torch::Tensor> fused_l2nn_split(
   if (use_persist && raft_amd::fused_l2nn_persist_supported(nslice, (int)d)) {
     raft_amd::launch_fused_l2nn_persist(xsl, csl, xn.data_ptr<float>(),
                                         cn.data_ptr<float>(), dmin.data_ptr<float>(),
-                                        amin.data_ptr<int>(), m, (int)n, (int)d,
-                                        nslice, cur_stream());
+                                        amin.data_ptr<int>(), dmin2.data_ptr<float>(),
+                                        m, (int)n, (int)d, nslice, cur_stream());
   } else {
     raft_amd::launch_fused_l2nn_split(xsl, csl, xn.data_ptr<float>(),
                                       cn.data_ptr<float>(), dmin.data_ptr<float>(),
-                                      amin.data_ptr<int>(), m, (int)n, (int)d,
-                                      nslice, cur_stream());
+                                      amin.data_ptr<int>(), dmin2.data_ptr<float>(),
+                                      m, (int)n, (int)d, nslice, cur_stream());
   }
-  return {dmin, amin};
+  return {dmin, amin, dmin2};
 }
 
 torch::Tensor pairwise_l2_mfma(std::vector<torch::Tensor> x_slices,
@@ -311,6 +316,18 @@ torch::Tensor pairwise_l2_mfma(std::vector<torch::Tensor> x_slices,
                                     o.data_ptr<float>(), m, n, (int)d, o.size(1),
                                     nslice, sqrt_out, cur_stream());
   return o;
+}
+
+void l2nn_verify_repair(torch::Tensor x, torch::Tensor c, torch::Tensor xn,
+                        torch::Tensor dmin, torch::Tensor amin, torch::Tensor dmin2,
+                        double cn_max) {
+  check_f32_2d(x, "x");
+  check_f32_2d(c, "c");
+  raft_amd::launch_l2nn_verify_repair(x.data_ptr<float>(), c.data_ptr<float>(),
+                                      xn.data_ptr<float>(), dmin.data_ptr<float>(),
+                                      amin.data_ptr<int>(), dmin2.data_ptr<float>(),
+                                      (float)cn_max, x.size(0), (int)c.size(0),
+                                      (int)x.size(1), cur_stream());
 }
 
 torch::Tensor gemm_bf16_f32(torch::Tensor a, torch::Tensor b,
@@ -387,6 +404,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("x_slices"), pybind11::arg("y_slices"), pybind11::arg("xn"),
         pybind11::arg("yn"), pybind11::arg("out") = pybind11::none(),
         pybind11::arg("sqrt_out") = false);
+  m.def("l2nn_verify_repair", &l2nn_verify_repair,
+        "exact-fp32 verification/repair of split-bf16 L2-NN results");
   m.def("fused_l2nn_split", &fused_l2nn_split,
         "fused split-bf16 MFMA L2-NN (distance + argmin, no materialization)");
   m.def("gemm_bf16_f32", &gemm_bf16_f32, "bf16 x bf16 -> f32 rocBLAS gemm_ex",

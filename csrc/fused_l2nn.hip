@@ -42,6 +42,7 @@ __global__ void fused_l2nn_kernel(const __bf16* __restrict__ x0,
                                   const float* __restrict__ xn,
                                   const float* __restrict__ cn,
                                   float* __restrict__ dmin, int* __restrict__ amin,
+                                  float* __restrict__ dmin2,
                                   long long m, int n, int d) {
   extern __shared__ __bf16 smem[];
   __bf16* xs[NSLICE];
@@ -59,12 +60,12 @@ __global__ void fused_l2nn_kernel(const __bf16* __restrict__ x0,
   const int wr = w >> 1, wc = w & 1;  // 2x2 wave grid
   const long long row0 = (long long)blockIdx.x * 128;
 
-  float best[4][4];
+  float best[4][4], best2[4][4];
   int bidx[4][4];
 #pragma unroll
   for (int a = 0; a < 4; a++)
 #pragma unroll
-    for (int b = 0; b < 4; b++) { best[a][b] = INFINITY; bidx[a][b] = 0; }
+    for (int b = 0; b < 4; b++) { best[a][b] = INFINITY; best2[a][b] = INFINITY; bidx[a][b] = 0; }
 
   const int n_tiles = n / 128;
 
@@ -78,30 +79,39 @@ __global__ void fused_l2nn_kernel(const __bf16* __restrict__ x0,
     mfma_tile_kloop<NSLICE>(xg, cg, xs, cs, acc, row0, (long long)nt * 128, d,
                             m - 1, n - 1, wr, wc, lane);
 
-    // epilogue: fold this tile's 64 columns-per-wave into the running best.
+    // epilogue: fold this tile's 64 columns-per-wave into the running
+    // (best, second-best) pair — second-best feeds the exact-argmin
+    // verification margin (see fused_l2nn.py verify path).
     const int col_base = nt * 128 + wc * 64;
 #pragma unroll
     for (int fr = 0; fr < 4; fr++) {
 #pragma unroll
       for (int reg = 0; reg < 4; reg++) {
-        float v = INFINITY;
+        float v = INFINITY, v2 = INFINITY;
         int vi = 0;
 #pragma unroll
         for (int fc = 0; fc < 4; fc++) {
           const int col = col_base + fc * 16 + (lane & 15);
           const float s = cn[col] - 2.f * acc[fr][fc][reg];
-          if (s < v) { v = s; vi = col; }
+          if (s < v) { v2 = v; v = s; vi = col; }
+          else if (s < v2) { v2 = s; }
         }
 #pragma unroll
         for (int off = 8; off > 0; off >>= 1) {
           const float ov = __shfl_xor(v, off, RAFT_AMD_WAVE);
+          const float ov2 = __shfl_xor(v2, off, RAFT_AMD_WAVE);
           const int oi = __shfl_xor(vi, off, RAFT_AMD_WAVE);
+          const float hi = fmaxf(v, ov);
+          v2 = fminf(fminf(v2, ov2), hi);
           if (ov < v || (ov == v && oi < vi)) { v = ov; vi = oi; }
         }
+        const float hi = fmaxf(best[fr][reg], v);
+        const float merged2 = fminf(fminf(best2[fr][reg], v2), hi);
         if (v < best[fr][reg] || (v == best[fr][reg] && vi < bidx[fr][reg])) {
           best[fr][reg] = v;
           bidx[fr][reg] = vi;
         }
+        best2[fr][reg] = merged2;
       }
     }
   }
@@ -111,6 +121,7 @@ __global__ void fused_l2nn_kernel(const __bf16* __restrict__ x0,
   __syncthreads();
   float* comb_v = reinterpret_cast<float*>(smem);          // [2][128]
   int* comb_i = reinterpret_cast<int*>(comb_v + 256);      // [2][128]
+  float* comb_v2 = reinterpret_cast<float*>(comb_i + 256); // [2][128]
   if ((lane & 15) == 0) {
 #pragma unroll
     for (int fr = 0; fr < 4; fr++)
@@ -119,6 +130,7 @@ __global__ void fused_l2nn_kernel(const __bf16* __restrict__ x0,
         const int rl = wr * 64 + fr * 16 + (lane >> 4) * 4 + reg;  // 0..127
         comb_v[wc * 128 + rl] = best[fr][reg];
         comb_i[wc * 128 + rl] = bidx[fr][reg];
+        comb_v2[wc * 128 + rl] = best2[fr][reg];
       }
   }
   __syncthreads();
@@ -130,6 +142,8 @@ __global__ void fused_l2nn_kernel(const __bf16* __restrict__ x0,
         const int rl = wr * 64 + fr * 16 + (lane >> 4) * 4 + reg;
         float v0 = comb_v[rl], v1 = comb_v[128 + rl];
         int i0 = comb_i[rl], i1 = comb_i[128 + rl];
+        const float s2 = fminf(fminf(comb_v2[rl], comb_v2[128 + rl]),
+                               fmaxf(v0, v1));
         const bool take1 = (v1 < v0) || (v1 == v0 && i1 < i0);
         const float v = take1 ? v1 : v0;
         const int vi = take1 ? i1 : i0;
@@ -137,14 +151,16 @@ __global__ void fused_l2nn_kernel(const __bf16* __restrict__ x0,
         if (row < m) {
           dmin[row] = fmaxf(v + xn[row], 0.f);
           amin[row] = vi;
+          if (dmin2) dmin2[row] = v + xn[row];  // unclamped: margin math
         }
       }
   }
 }
 
 void launch_fused_l2nn_split(const void** xsl, const void** csl, const float* xn,
-                             const float* cn, float* dmin, int* amin, long long m,
-                             int n, int d, int nslice, hipStream_t stream) {
+                             const float* cn, float* dmin, int* amin, float* dmin2,
+                             long long m, int n, int d, int nslice,
+                             hipStream_t stream) {
   const int grid = (int)((m + 127) / 128);
   const size_t lds = (size_t)nslice * 2 * 8192 * sizeof(__bf16);
   const __bf16* x0 = (const __bf16*)xsl[0];
@@ -156,11 +172,11 @@ void launch_fused_l2nn_split(const void** xsl, const void** csl, const float* xn
   switch (nslice) {
     case 1:
       hipLaunchKernelGGL((fused_l2nn_kernel<1>), dim3(grid), dim3(256), lds, stream,
-                         x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, m, n, d);
+                         x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d);
       break;
     case 2:
       hipLaunchKernelGGL((fused_l2nn_kernel<2>), dim3(grid), dim3(256), lds, stream,
-                         x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, m, n, d);
+                         x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d);
       break;
     case 3: {
       static bool attr_set = false;
@@ -171,7 +187,7 @@ void launch_fused_l2nn_split(const void** xsl, const void** csl, const float* xn
         attr_set = true;
       }
       hipLaunchKernelGGL((fused_l2nn_kernel<3>), dim3(grid), dim3(256), lds, stream,
-                         x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, m, n, d);
+                         x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d);
       break;
     }
     default:

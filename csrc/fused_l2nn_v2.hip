@@ -46,7 +46,8 @@ __global__ void fused_l2nn_persist_kernel(
     const __bf16* __restrict__ x2, const __bf16* __restrict__ c0,
     const __bf16* __restrict__ c1, const __bf16* __restrict__ c2,
     const float* __restrict__ xn, const float* __restrict__ cn,
-    float* __restrict__ dmin, int* __restrict__ amin, long long m, int n, int d) {
+    float* __restrict__ dmin, int* __restrict__ amin, float* __restrict__ dmin2,
+    long long m, int n, int d) {
   constexpr int BLOCK = WR * WC * 64;
   constexpr int RF = BM / (16 * WR);   // row fragments per wave
   constexpr int CF = 128 / (16 * WC);  // col fragments per wave
@@ -191,6 +192,7 @@ __global__ void fused_l2nn_persist_kernel(
         if (row < m) {
           dmin[row] = fmaxf(v + xn[row], 0.f);
           amin[row] = vi;
+          if (dmin2) dmin2[row] = v + xn[row];  // best2 not tracked here: mark
         }
       }
   }
@@ -200,7 +202,7 @@ template <int NSLICE, int BM, int WR, int WC>
 static void launch_persist(const __bf16* x0, const __bf16* x1, const __bf16* x2,
                            const __bf16* c0, const __bf16* c1, const __bf16* c2,
                            const float* xn, const float* cn, float* dmin, int* amin,
-                           long long m, int n, int d, hipStream_t stream) {
+                           float* dmin2, long long m, int n, int d, hipStream_t stream) {
   const int grid = (int)((m + BM - 1) / BM);
   const size_t lds =
       (size_t)NSLICE * (d / 64) * BM * 64 * 2 + (size_t)NSLICE * 8192 * 2;
@@ -213,7 +215,7 @@ static void launch_persist(const __bf16* x0, const __bf16* x1, const __bf16* x2,
   }
   hipLaunchKernelGGL((fused_l2nn_persist_kernel<NSLICE, BM, WR, WC>), dim3(grid),
                      dim3(WR * WC * 64), lds, stream, x0, x1, x2, c0, c1, c2, xn,
-                     cn, dmin, amin, m, n, d);
+                     cn, dmin, amin, dmin2, m, n, d);
 }
 
 // returns true when a persistent variant covers (nslice, d)
@@ -224,8 +226,9 @@ bool fused_l2nn_persist_supported(int nslice, int d) {
 }
 
 void launch_fused_l2nn_persist(const void** xsl, const void** csl, const float* xn,
-                               const float* cn, float* dmin, int* amin, long long m,
-                               int n, int d, int nslice, hipStream_t stream) {
+                               const float* cn, float* dmin, int* amin, float* dmin2,
+                               long long m, int n, int d, int nslice,
+                               hipStream_t stream) {
   const __bf16* x0 = (const __bf16*)xsl[0];
   const __bf16* x1 = (const __bf16*)(nslice > 1 ? xsl[1] : xsl[0]);
   const __bf16* x2 = (const __bf16*)(nslice > 2 ? xsl[2] : xsl[0]);
@@ -233,13 +236,13 @@ void launch_fused_l2nn_persist(const void** xsl, const void** csl, const float* 
   const __bf16* c1 = (const __bf16*)(nslice > 1 ? csl[1] : csl[0]);
   const __bf16* c2 = (const __bf16*)(nslice > 2 ? csl[2] : csl[0]);
   if (nslice == 1) {
-    launch_persist<1, 128, 2, 4>(x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, m, n, d, stream);
+    launch_persist<1, 128, 2, 4>(x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d, stream);
   } else if (nslice == 2) {
-    launch_persist<2, 128, 2, 4>(x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, m, n, d, stream);
+    launch_persist<2, 128, 2, 4>(x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d, stream);
   } else if (d <= 128) {
-    launch_persist<3, 128, 2, 4>(x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, m, n, d, stream);
+    launch_persist<3, 128, 2, 4>(x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d, stream);
   } else {
-    launch_persist<3, 64, 1, 8>(x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, m, n, d, stream);
+    launch_persist<3, 64, 1, 8>(x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d, stream);
   }
 }
 

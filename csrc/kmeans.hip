@@ -91,6 +91,74 @@ __global__ void reduce_rows_by_key_sorted_kernel(const float* __restrict__ x,
   }
 }
 
+// Exact-fp32 verification/repair pass for the split-bf16 fused L2-NN
+// (the "verified bf16x2" engine): for every row, recompute the distance to
+// the chosen centroid in exact fp32; rows whose (best, second-best) margin
+// is inside the provable split-emulation error bound rescan ALL centroids
+// in exact fp32 and repair the argmin. Device-side only — no host sync.
+//
+// Error bound (NSLICE=2, per-element bf16 rounding |r_i| <= 2^-9|x_i| twice):
+// |score error| <= 2*(||x1||*||c1|| + ...) <= ~2^-14.5*sqrt(xn*cn); we use
+// E = 2^-13*sqrt(xn*cn_max) + 2^-18*(xn+cn_max) (3x headroom + accumulate).
+__global__ void l2nn_verify_repair_kernel(const float* __restrict__ x,
+                                          const float* __restrict__ c,
+                                          const float* __restrict__ xn,
+                                          float* __restrict__ dmin,
+                                          int* __restrict__ amin,
+                                          const float* __restrict__ dmin2,
+                                          float cn_max, long long m, int n, int d) {
+  const long long waves_per_block = blockDim.x / RAFT_AMD_WAVE;
+  const int lane = threadIdx.x % RAFT_AMD_WAVE;
+  long long row = (long long)blockIdx.x * waves_per_block + threadIdx.x / RAFT_AMD_WAVE;
+  const long long stride = (long long)gridDim.x * waves_per_block;
+  for (; row < m; row += stride) {
+    const float* rp = x + row * d;
+    const float xnr = xn[row];
+    const float margin = dmin2[row] - dmin[row];
+    const float bound = 2.f * (exp2f(-13.f) * sqrtf(fmaxf(xnr * cn_max, 0.f)) +
+                               exp2f(-18.f) * (xnr + cn_max));
+    int a = amin[row];
+    if (margin < bound) {
+      // full exact rescan (rare: near-ties only)
+      float bestv = INFINITY;
+      int besti = 0;
+      for (int j = 0; j < n; j++) {
+        const float* cp = c + (long long)j * d;
+        float acc = 0.f;
+        for (int t = lane; t < d; t += RAFT_AMD_WAVE) {
+          const float diff = rp[t] - cp[t];
+          acc += diff * diff;
+        }
+        acc = wave_reduce_sum(acc);
+        if (acc < bestv) { bestv = acc; besti = j; }
+      }
+      if (lane == 0) {
+        dmin[row] = bestv;
+        amin[row] = besti;
+      }
+    } else {
+      // exact fp32 refinement of the chosen distance
+      const float* cp = c + (long long)a * d;
+      float acc = 0.f;
+      for (int t = lane; t < d; t += RAFT_AMD_WAVE) {
+        const float diff = rp[t] - cp[t];
+        acc += diff * diff;
+      }
+      acc = wave_reduce_sum(acc);
+      if (lane == 0) dmin[row] = acc;
+    }
+  }
+}
+
+void launch_l2nn_verify_repair(const float* x, const float* c, const float* xn,
+                               float* dmin, int* amin, const float* dmin2,
+                               float cn_max, long long m, int n, int d,
+                               hipStream_t stream) {
+  int grid = grid_1d(m * RAFT_AMD_WAVE, 256);
+  hipLaunchKernelGGL(l2nn_verify_repair_kernel, dim3(grid), dim3(256), 0, stream,
+                     x, c, xn, dmin, amin, dmin2, cn_max, m, n, d);
+}
+
 void launch_reduce_rows_by_key_sorted(const float* x, const int* perm,
                                       const int* keys_sorted, float* sums,
                                       long long n_rows, long long d,

@@ -189,7 +189,8 @@ def kmeans_iterate(x: torch.Tensor, centroids: torch.Tensor, n_iters: int,
     (fused L2-NN over all rows) + update (keyed reduction, counts, ONE packed
     allreduce, centroid recompute). Returns (centroids, inertia).
     """
-    from raft_amd.neighbors.fused_l2nn import (_MODE_NSLICE, fused_l2nn_presplit,
+    from raft_amd.neighbors.fused_l2nn import (_MODE_NSLICE, _VERIFY_MODES,
+                                               fused_l2nn_presplit,
                                                split_bf16_slices)
 
     comms = comms or LoopbackComms()
@@ -205,8 +206,9 @@ def kmeans_iterate(x: torch.Tensor, centroids: torch.Tensor, n_iters: int,
     inertia_t = None
     for it in range(n_iters):
         if use_fused:
+            vx = x if fp32_mode in _VERIFY_MODES else None
             dmin, labels = fused_l2nn_presplit(x_slices, xn, centroids,
-                                               int32_labels=True)
+                                               int32_labels=True, verify_x=vx)
         else:
             dmin, labels = fused_l2nn(x, centroids, fp32_mode=fp32_mode,
                                       chunk_rows=chunk_rows)

@@ -23,7 +23,11 @@ from raft_amd._ext import ext_or_none, require_ext
 from raft_amd.utils import on_gpu, row_chunks
 from raft_amd.linalg.gemm import gemm_fp32_emulated
 
-_MODE_NSLICE = {"bf16x2": 2, "bf16x3": 3, "auto": 3, "fused": 3}
+_MODE_NSLICE = {"bf16x2": 2, "bf16x3": 3, "bf16x2v": 2, "auto": 2, "fused": 3}
+#: modes that run the exact-fp32 verification/repair pass (provably exact
+#: argmin: rows inside the split-error margin are rescanned in fp32, and the
+#: chosen distance is recomputed exactly for every row)
+_VERIFY_MODES = {"bf16x2v", "auto"}
 
 
 def split_bf16_slices(t: torch.Tensor, nslice: int):
@@ -50,15 +54,28 @@ def _pad_cols(y: torch.Tensor, yn: torch.Tensor, mult: int = 128):
 
 
 def fused_l2nn_presplit(x_slices, xn: torch.Tensor, y: torch.Tensor,
-                        sqrt: bool = False, int32_labels: bool = False):
-    """Fused kernel entry with precomputed X slices (k-means hot loop)."""
+                        sqrt: bool = False, int32_labels: bool = False,
+                        verify_x: torch.Tensor | None = None):
+    """Fused kernel entry with precomputed X slices (k-means hot loop).
+
+    verify_x: the original fp32 matrix — enables the exact-fp32
+    verification/repair pass (csrc/kmeans.hip l2nn_verify_repair): every row's
+    chosen distance is recomputed in exact fp32, and rows whose
+    (best, second-best) margin falls inside the provable split-emulation error
+    bound rescan all centroids exactly. Result: fp32-exact argmin + fp32
+    distances at split-bf16 MFMA speed.
+    """
     ext = require_ext()
     nslice = len(x_slices)
     yn = (y * y).sum(dim=1)
     yp, ynp, n_true = _pad_cols(y, yn)
     y_slices = split_bf16_slices(yp, nslice)
-    dmin, amin = ext.fused_l2nn_split(list(x_slices), list(y_slices),
-                                      xn.contiguous(), ynp.contiguous())
+    dmin, amin, dmin2 = ext.fused_l2nn_split(list(x_slices), list(y_slices),
+                                             xn.contiguous(), ynp.contiguous())
+    if verify_x is not None:
+        cn_max = float(yn.max().item())
+        ext.l2nn_verify_repair(verify_x.contiguous(), y.contiguous(),
+                               xn.contiguous(), dmin, amin, dmin2, cn_max)
     if sqrt:
         dmin = dmin.clamp_min(0).sqrt()
     return dmin, (amin if int32_labels else amin.to(torch.int64))
@@ -81,8 +98,8 @@ def fused_l2nn(x: torch.Tensor, y: torch.Tensor, sqrt: bool = False,
             yf = y.to(torch.float32)
             yn = (yf * yf).sum(dim=1)
             yp, ynp, _ = _pad_cols(y.contiguous(), yn)
-            dmin, amin = ext.fused_l2nn_split([x.contiguous()], [yp.contiguous()],
-                                              xn.contiguous(), ynp.contiguous())
+            dmin, amin, _ = ext.fused_l2nn_split([x.contiguous()], [yp.contiguous()],
+                                                 xn.contiguous(), ynp.contiguous())
             if sqrt:
                 dmin = dmin.clamp_min(0).sqrt()
             return dmin, amin.to(torch.int64)
@@ -91,7 +108,8 @@ def fused_l2nn(x: torch.Tensor, y: torch.Tensor, sqrt: bool = False,
                 nslice = _MODE_NSLICE[fp32_mode]
                 xs = split_bf16_slices(x, nslice)
                 xn = (x * x).sum(dim=1)
-                return fused_l2nn_presplit(xs, xn, y, sqrt=sqrt)
+                vx = x if fp32_mode in _VERIFY_MODES else None
+                return fused_l2nn_presplit(xs, xn, y, sqrt=sqrt, verify_x=vx)
             return _chunked_gpu(x, y, sqrt, fp32_mode, chunk_rows)
 
     # CPU oracle
