@@ -101,10 +101,15 @@ __global__ void fused_l2nn_kernel(const __bf16* __restrict__ x0,
           const float ov = __shfl_xor(v, off, RAFT_AMD_WAVE);
           const float ov2 = __shfl_xor(v2, off, RAFT_AMD_WAVE);
           const int oi = __shfl_xor(vi, off, RAFT_AMD_WAVE);
-          const float hi = fmaxf(v, ov);
-          v2 = fminf(fminf(v2, ov2), hi);
+          // top-2 merge of two candidate SETS: when both sets share the same
+          // best element (same index — happens after the first exchange),
+          // the loser of the bests is NOT a second-best candidate
+          float new2 = fminf(v2, ov2);
+          if (oi != vi) new2 = fminf(new2, fmaxf(v, ov));
+          v2 = new2;
           if (ov < v || (ov == v && oi < vi)) { v = ov; vi = oi; }
         }
+        // cross-tile merge: disjoint column sets -> plain top-2 merge
         const float hi = fmaxf(best[fr][reg], v);
         const float merged2 = fminf(fminf(best2[fr][reg], v2), hi);
         if (v < best[fr][reg] || (v == best[fr][reg] && vi < bidx[fr][reg])) {
