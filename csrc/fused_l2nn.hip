@@ -156,7 +156,7 @@ __global__ void fused_l2nn_kernel(const __bf16* __restrict__ x0,
         if (row < m) {
           dmin[row] = fmaxf(v + xn[row], 0.f);
           amin[row] = vi;
-          if (dmin2) dmin2[row] = v + xn[row];  // unclamped: margin math
+          if (dmin2) dmin2[row] = s2 + xn[row];  // second-best, unclamped
         }
       }
   }
