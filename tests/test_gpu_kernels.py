@@ -175,13 +175,18 @@ class TestFusedL2NNMfma:
         y[128:] = y[:128] + torch.randn(128, 128, device=dev) * 1e-4
         from raft_amd.neighbors.fused_l2nn import fused_l2nn
         _, av = fused_l2nn(x, y, fp32_mode="bf16x2v")
-        dn, an = fused_l2nn(x, y, fp32_mode="native")
-        # near-ties may legitimately differ between summation orders when the
-        # fp32 distances are EQUAL; require value-equivalence
-        dv_at = torch.cdist(x.double(), y.double())[torch.arange(8192, device=dev), av] ** 2
-        dn_ref = torch.cdist(x.double(), y.double())[torch.arange(8192, device=dev), an] ** 2
-        assert float((dv_at - dn_ref).abs().max()) < 1e-2
-        assert (av == an).float().mean() > 0.99
+        # the meaningful guarantee: the chosen centroid's TRUE distance is
+        # fp32-indistinguishable from the optimum for every row (near-tie
+        # winners may differ between any two fp32 summation orders, including
+        # the native engine's own expanded form)
+        ref = torch.cdist(x.double(), y.double()) ** 2
+        opt = ref.min(dim=1).values
+        chosen = ref[torch.arange(8192, device=dev), av]
+        assert float((chosen - opt).max()) < 1e-2
+        # and on rows with a clear margin the argmin is exactly the optimum
+        margin_ok = (torch.topk(ref, 2, dim=1, largest=False).values.diff(dim=1)
+                     .squeeze(1) > 1.0)
+        assert (av[margin_ok] == ref.argmin(dim=1)[margin_ok]).all()
 
     def test_verified_mode_exact_distances(self, dev, ext):
         from raft_amd.neighbors.fused_l2nn import fused_l2nn
