@@ -127,12 +127,24 @@ def kmeans_fit(x: torch.Tensor, params: KMeansParams,
     else:
         centroids = _init_plusplus(x, k, state, comms, params.fp32_mode)
 
-    prev_shift = None
+    from raft_amd.neighbors.fused_l2nn import (_MODE_NSLICE, _VERIFY_MODES,
+                                               fused_l2nn_presplit,
+                                               split_bf16_slices)
+    use_fused = (x.is_cuda and x.dtype == torch.float32
+                 and params.fp32_mode in _MODE_NSLICE and d % 64 == 0)
+    if use_fused:
+        x_slices = split_bf16_slices(x, _MODE_NSLICE[params.fp32_mode])
+        xn = (x * x).sum(dim=1)
+        vx = x if params.fp32_mode in _VERIFY_MODES else None
+
     inertia = float("inf")
     it = 0
     labels = None
     for it in range(1, params.max_iter + 1):
-        dmin, labels = fused_l2nn(x, centroids, fp32_mode=params.fp32_mode)
+        if use_fused:
+            dmin, labels = fused_l2nn_presplit(x_slices, xn, centroids, verify_x=vx)
+        else:
+            dmin, labels = fused_l2nn(x, centroids, fp32_mode=params.fp32_mode)
         sums = reduce_rows_by_key(x, labels, n_keys=k)
         counts = torch.bincount(labels, minlength=k).to(x.dtype)
         local_inertia = torch.sum(dmin, dtype=torch.float64).to(x.dtype)

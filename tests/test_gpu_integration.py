@@ -1,0 +1,141 @@
+"""GPU integration tests: composed algorithms running end-to-end on device
+(clustering metrics over native pairwise kernels, spectral partition over the
+native SpMV + Lanczos, brute-force knn over MFMA distance + radix select)."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def dev():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from raft_amd._ext import require_ext
+    require_ext()
+    return torch.device("cuda")
+
+
+class TestStatsGpu:
+    def test_silhouette_on_blobs(self, dev):
+        from raft_amd.random import make_blobs, RngState
+        from raft_amd.stats import silhouette_score
+        x, y, _ = make_blobs(20000, 64, n_clusters=8, cluster_std=0.3,
+                             center_box=(-20, 20), state=RngState(seed=1), device=dev)
+        s = silhouette_score(x, y, 8)
+        assert s > 0.8
+
+    def test_moments_and_cov(self, dev):
+        from raft_amd import stats
+        x = torch.randn(50000, 32, device=dev)
+        torch.testing.assert_close(stats.mean(x).double(), x.double().mean(0),
+                                   rtol=1e-4, atol=1e-4)
+        torch.testing.assert_close(stats.cov(x).double(), torch.cov(x.double().t()),
+                                   rtol=1e-3, atol=1e-3)
+
+    def test_clustering_metrics(self, dev):
+        from raft_amd import stats
+        a = torch.randint(0, 5, (10000,), device=dev)
+        assert stats.adjusted_rand_index(a, a) == pytest.approx(1.0)
+        b = torch.randint(0, 5, (10000,), device=dev)
+        assert abs(stats.adjusted_rand_index(a, b)) < 0.05
+
+
+class TestSpectralGpu:
+    def test_partition_two_blocks(self, dev):
+        from raft_amd.sparse import CSR
+        from raft_amd.spectral import partition, analyze_partition
+        n = 400
+        dense = torch.zeros(n, n, device=dev)
+        for lo, hi in ((0, n // 2), (n // 2, n)):
+            blk = torch.rand(hi - lo, hi - lo, device=dev) < 0.2
+            dense[lo:hi, lo:hi] = blk.float()
+        dense = ((dense + dense.t()) > 0).float()
+        dense.fill_diagonal_(0)
+        dense[n // 2 - 1, n // 2] = dense[n // 2, n // 2 - 1] = 0.01
+        g = CSR.from_dense(dense)
+        g.indptr = g.indptr.to(torch.int32)
+        g.indices = g.indices.to(torch.int32)
+        labels, w, v = partition(g, 2, seed=1)
+        l = labels.cpu()
+        first, second = l[: n // 2], l[n // 2:]
+        # each half is (almost entirely) one cluster
+        assert (first == first.mode().values).float().mean() > 0.95
+        assert (second == second.mode().values).float().mean() > 0.95
+        assert first.mode().values != second.mode().values
+
+
+class TestSolverGpu:
+    def test_lap_on_gpu(self, dev):
+        from raft_amd.solver import linear_assignment
+        from scipy.optimize import linear_sum_assignment
+        torch.manual_seed(0)
+        cost = torch.randint(0, 1000, (64, 64), device=dev).float()
+        assign, total = linear_assignment(cost)
+        r, c = linear_sum_assignment(cost.cpu().numpy())
+        assert total == pytest.approx(cost.cpu().numpy()[r, c].sum(), abs=1e-4)
+
+
+class TestRandomGpu:
+    def test_rmat_and_permute(self, dev):
+        from raft_amd.random import rmat, permute, RngState
+        src, dst = rmat(10, 10, 50000, state=RngState(seed=2), device=dev)
+        assert src.max() < 1024 and dst.max() < 1024
+        p = permute(100000, state=RngState(seed=3), device=dev)
+        assert torch.equal(torch.sort(p).values, torch.arange(100000, device=dev))
+
+    def test_sample_and_mvg(self, dev):
+        from raft_amd.random import sample_without_replacement, multi_variable_gaussian, RngState
+        idx = sample_without_replacement(100000, 1000, state=RngState(seed=4), device=dev)
+        assert idx.unique().numel() == 1000
+        mean = torch.tensor([0.0, 1.0], device=dev)
+        cov = torch.tensor([[1.0, 0.3], [0.3, 1.0]], device=dev)
+        s = multi_variable_gaussian(mean, cov, 50000, state=RngState(seed=5))
+        torch.testing.assert_close(torch.cov(s.t()), cov, rtol=0.1, atol=0.05)
+
+
+class TestSparseGpuExtra:
+    def test_spmm_sddmm_symmetrize(self, dev):
+        from raft_amd import sparse as rsp
+        torch.manual_seed(0)
+        dense = torch.randn(300, 200, device=dev)
+        dense[torch.rand_like(dense) > 0.1] = 0
+        csr = rsp.CSR.from_dense(dense)
+        b = torch.randn(200, 16, device=dev)
+        torch.testing.assert_close(rsp.spmm(csr, b).double(), dense.double() @ b.double(),
+                                   rtol=1e-4, atol=1e-4)
+        a2 = torch.randn(300, 8, device=dev)
+        b2 = torch.randn(200, 8, device=dev)
+        out = rsp.sddmm(a2, b2, csr)
+        full = a2.double() @ b2.t().double()
+        coo = rsp.csr_to_coo(csr)
+        torch.testing.assert_close(out.values.double(),
+                                   full[coo.rows.long(), coo.cols.long()],
+                                   rtol=1e-4, atol=1e-4)
+
+    def test_randomized_svds_gpu(self, dev):
+        from raft_amd import sparse as rsp
+        from raft_amd.sparse.solver import randomized_svds
+        torch.manual_seed(1)
+        u0 = torch.rand(400, 5, device=dev)
+        v0 = torch.rand(5, 300, device=dev)
+        dense = u0 @ v0
+        csr = rsp.CSR.from_dense(dense)
+        u, s, v = randomized_svds(csr, k=5, n_iter=6, seed=0)
+        approx = (u * s.unsqueeze(0)) @ v.t()
+        torch.testing.assert_close(approx, dense, rtol=1e-3, atol=1e-3)
+
+
+class TestKnnGpu:
+    def test_knn_bf16_recall(self, dev):
+        from raft_amd.neighbors import knn
+        torch.manual_seed(2)
+        x = torch.randn(50000, 128, device=dev).bfloat16()
+        q = x[:256].clone()
+        d, i = knn(x, q, k=10, index_chunk=20000)
+        assert (i[:, 0] == torch.arange(256, device=dev)).float().mean() > 0.99
+        # recall vs exact fp32 knn
+        ref = torch.cdist(q.float(), x.float()) ** 2
+        ri = torch.topk(ref, 10, dim=1, largest=False).indices
+        hits = (i.unsqueeze(2) == ri.unsqueeze(1)).any(2).float().mean()
+        assert hits > 0.95
