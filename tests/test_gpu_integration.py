@@ -117,13 +117,13 @@ class TestSparseGpuExtra:
         from raft_amd import sparse as rsp
         from raft_amd.sparse.solver import randomized_svds
         torch.manual_seed(1)
-        u0 = torch.rand(400, 5, device=dev)
-        v0 = torch.rand(5, 300, device=dev)
+        u0 = torch.rand(400, 5, device=dev, dtype=torch.float64)
+        v0 = torch.rand(5, 300, device=dev, dtype=torch.float64)
         dense = u0 @ v0
         csr = rsp.CSR.from_dense(dense)
         u, s, v = randomized_svds(csr, k=5, n_iter=6, seed=0)
         approx = (u * s.unsqueeze(0)) @ v.t()
-        torch.testing.assert_close(approx, dense, rtol=1e-3, atol=1e-3)
+        torch.testing.assert_close(approx, dense, rtol=1e-5, atol=1e-5)
 
 
 class TestKnnGpu:
