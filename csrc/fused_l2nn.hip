@@ -31,7 +31,7 @@
 
 namespace raft_amd {
 
-template <int NSLICE>
+template <int NSLICE, bool DB = false>
 __launch_bounds__(256, 2)
 __global__ void fused_l2nn_kernel(const __bf16* __restrict__ x0,
                                   const __bf16* __restrict__ x1,
@@ -47,12 +47,19 @@ __global__ void fused_l2nn_kernel(const __bf16* __restrict__ x0,
   extern __shared__ __bf16 smem[];
   __bf16* xs[NSLICE];
   __bf16* cs[NSLICE];
+  __bf16* xs2[2][NSLICE];
+  __bf16* cs2[2][NSLICE];
   const __bf16* const xg[3] = {x0, x1, x2};
   const __bf16* const cg[3] = {c0, c1, c2};
 #pragma unroll
   for (int s = 0; s < NSLICE; s++) {
     xs[s] = smem + s * 8192;
     cs[s] = smem + (NSLICE + s) * 8192;
+#pragma unroll
+    for (int b = 0; b < 2; b++) {
+      xs2[b][s] = smem + (b * 2 * NSLICE + s) * 4096;
+      cs2[b][s] = smem + (b * 2 * NSLICE + NSLICE + s) * 4096;
+    }
   }
 
   const int lane = threadIdx.x % RAFT_AMD_WAVE;
@@ -76,8 +83,14 @@ __global__ void fused_l2nn_kernel(const __bf16* __restrict__ x0,
 #pragma unroll
       for (int b = 0; b < 4; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-    mfma_tile_kloop<NSLICE>(xg, cg, xs, cs, acc, row0, (long long)nt * 128, d,
-                            m - 1, n - 1, wr, wc, lane);
+    if constexpr (DB) {
+      mfma_tile_kloop_db32<NSLICE>(xg, cg, xs2, cs2, acc, row0,
+                                   (long long)nt * 128, d, m - 1, n - 1, wr, wc,
+                                   lane);
+    } else {
+      mfma_tile_kloop<NSLICE>(xg, cg, xs, cs, acc, row0, (long long)nt * 128, d,
+                              m - 1, n - 1, wr, wc, lane);
+    }
 
     // epilogue: fold this tile's 64 columns-per-wave into the running
     // (best, second-best) pair — second-best feeds the exact-argmin
@@ -168,6 +181,10 @@ void launch_fused_l2nn_split(const void** xsl, const void** csl, const float* xn
                              hipStream_t stream) {
   const int grid = (int)((m + 127) / 128);
   const size_t lds = (size_t)nslice * 2 * 8192 * sizeof(__bf16);
+  static const bool use_db = [] {
+    const char* e = getenv("RAFT_AMD_L2NN_DB");
+    return e && e[0] == '1';
+  }();
   const __bf16* x0 = (const __bf16*)xsl[0];
   const __bf16* x1 = (const __bf16*)(nslice > 1 ? xsl[1] : xsl[0]);
   const __bf16* x2 = (const __bf16*)(nslice > 2 ? xsl[2] : xsl[0]);
@@ -176,12 +193,20 @@ void launch_fused_l2nn_split(const void** xsl, const void** csl, const float* xn
   const __bf16* c2 = (const __bf16*)(nslice > 2 ? csl[2] : csl[0]);
   switch (nslice) {
     case 1:
-      hipLaunchKernelGGL((fused_l2nn_kernel<1>), dim3(grid), dim3(256), lds, stream,
-                         x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d);
+      if (use_db)
+        hipLaunchKernelGGL((fused_l2nn_kernel<1, true>), dim3(grid), dim3(256), lds,
+                           stream, x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d);
+      else
+        hipLaunchKernelGGL((fused_l2nn_kernel<1>), dim3(grid), dim3(256), lds, stream,
+                           x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d);
       break;
     case 2:
-      hipLaunchKernelGGL((fused_l2nn_kernel<2>), dim3(grid), dim3(256), lds, stream,
-                         x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d);
+      if (use_db)
+        hipLaunchKernelGGL((fused_l2nn_kernel<2, true>), dim3(grid), dim3(256), lds,
+                           stream, x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d);
+      else
+        hipLaunchKernelGGL((fused_l2nn_kernel<2>), dim3(grid), dim3(256), lds, stream,
+                           x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d);
       break;
     case 3: {
       static bool attr_set = false;
@@ -189,10 +214,17 @@ void launch_fused_l2nn_split(const void** xsl, const void** csl, const float* xn
         HIP_CHECK(hipFuncSetAttribute((const void*)&fused_l2nn_kernel<3>,
                                       hipFuncAttributeMaxDynamicSharedMemorySize,
                                       96 * 1024));
+        HIP_CHECK(hipFuncSetAttribute((const void*)&fused_l2nn_kernel<3, true>,
+                                      hipFuncAttributeMaxDynamicSharedMemorySize,
+                                      96 * 1024));
         attr_set = true;
       }
-      hipLaunchKernelGGL((fused_l2nn_kernel<3>), dim3(grid), dim3(256), lds, stream,
-                         x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d);
+      if (use_db)
+        hipLaunchKernelGGL((fused_l2nn_kernel<3, true>), dim3(grid), dim3(256), lds,
+                           stream, x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d);
+      else
+        hipLaunchKernelGGL((fused_l2nn_kernel<3>), dim3(grid), dim3(256), lds, stream,
+                           x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d);
       break;
     }
     default:

@@ -103,4 +103,90 @@ __device__ __forceinline__ void mfma_tile_kloop(
   }
 }
 
+// ---------------------------------------------------------------------------
+// BK=32 double-buffered variant: 8 KiB tiles, 2-phase overlap (stage next
+// tile while MFMAing the current one; single vmcnt(0)+barrier per K-step —
+// the guide's minimum 2-phase recipe). LDS rows are 64 B, so the XOR swizzle
+// spreads 4 rows over the 4 in-row 16 B slots (residual 4-way ds_read
+// conflict ~1.6x on the LDS path — the price of fitting 2 blocks/CU).
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ int mfma_swz32(int byte_off) {
+  return byte_off ^ (((byte_off >> 6) & 3) << 4);
+}
+
+// stage a [128][32] bf16 tile (8 KiB) from row-major global
+__device__ __forceinline__ void mfma_stage_tile128_bk32(const __bf16* __restrict__ g,
+                                                        __bf16* lds, long long row0,
+                                                        long long k0, long long ld,
+                                                        long long max_row) {
+  const int t = threadIdx.x;
+  const int w = t / RAFT_AMD_WAVE;
+#pragma unroll
+  for (int j = 0; j < 2; j++) {
+    const int o = j * 4096 + t * 16;
+    const int o_src = mfma_swz32(o);
+    long long r = row0 + (o_src >> 6);
+    if (r > max_row) r = max_row;
+    const long long goff = r * ld + k0 + ((o_src & 63) >> 1);
+    __bf16* lbase = lds + (j * 4096 + w * 1024) / 2;
+    GLOAD_LDS(g + goff, lbase);
+  }
+}
+
+template <int NSLICE>
+__device__ __forceinline__ void mfma_tile_kloop_db32(
+    const __bf16* const (&xg)[3], const __bf16* const (&cg)[3],
+    __bf16* (&xs)[2][NSLICE], __bf16* (&cs)[2][NSLICE],
+    f32x4 (&acc)[4][4], long long row0, long long col0, int d,
+    long long m_max, long long n_max, int wr, int wc, int lane) {
+  const int k_tiles = d / 32;
+  auto stage = [&](int buf, int kt) {
+#pragma unroll
+    for (int s = 0; s < NSLICE; s++) {
+      mfma_stage_tile128_bk32(xg[s], xs[buf][s], row0, (long long)kt * 32, d, m_max);
+      mfma_stage_tile128_bk32(cg[s], cs[buf][s], col0, (long long)kt * 32, d, n_max);
+    }
+  };
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  int cur = 0;
+  for (int kt = 0; kt < k_tiles; kt++) {
+    if (kt + 1 < k_tiles) stage(cur ^ 1, kt + 1);  // overlap next-tile loads
+    bf16x8 a_frag[NSLICE][4], b_frag[NSLICE][4];
+    const int kbyte = (lane >> 4) * 16;  // (lane>>4)*8 bf16
+#pragma unroll
+    for (int fr = 0; fr < 4; fr++) {
+      const int r = wr * 64 + fr * 16 + (lane & 15);
+      const int byte = mfma_swz32(r * 64 + kbyte);
+#pragma unroll
+      for (int s = 0; s < NSLICE; s++)
+        a_frag[s][fr] = *reinterpret_cast<const bf16x8*>((const char*)xs[cur][s] + byte);
+    }
+#pragma unroll
+    for (int fc = 0; fc < 4; fc++) {
+      const int c = wc * 64 + fc * 16 + (lane & 15);
+      const int byte = mfma_swz32(c * 64 + kbyte);
+#pragma unroll
+      for (int s = 0; s < NSLICE; s++)
+        b_frag[s][fc] = *reinterpret_cast<const bf16x8*>((const char*)cs[cur][s] + byte);
+    }
+#pragma unroll
+    for (int fr = 0; fr < 4; fr++)
+#pragma unroll
+      for (int fc = 0; fc < 4; fc++) {
+#pragma unroll
+        for (int p = 0; p < mfma_n_products<NSLICE>(); p++) {
+          acc[fr][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[MFMA_PROD_A[p]][fr], b_frag[MFMA_PROD_B[p]][fc],
+              acc[fr][fc], 0, 0, 0);
+        }
+      }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+}
+
 }  // namespace raft_amd
