@@ -55,6 +55,9 @@ void launch_fused_l2nn_persist(const void**, const void**, const float*, const f
 void launch_pairwise_l2_mfma(const void**, const void**, const float*, const float*,
                              float*, long long, long long, int, long long, int, bool,
                              hipStream_t);
+void launch_pairwise_l2_filter(const void**, const void**, const float*, const float*,
+                               const float*, float*, int*, int*, int, long long,
+                               long long, long long, int, int, hipStream_t);
 // from gemm_rocblas.cpp
 void gemm_bf16_f32_rowmajor(const void*, const void*, float*, long long, long long,
                             long long, float, void*);
@@ -330,6 +333,36 @@ void l2nn_verify_repair(torch::Tensor x, torch::Tensor c, torch::Tensor xn,
                                       (int)x.size(1), cur_stream());
 }
 
+void pairwise_l2_filter(std::vector<torch::Tensor> x_slices,
+                        std::vector<torch::Tensor> y_slices, torch::Tensor xn,
+                        torch::Tensor yn, torch::Tensor thr, torch::Tensor out_d,
+                        torch::Tensor out_i, torch::Tensor cnt, int64_t col_offset) {
+  const int nslice = (int)x_slices.size();
+  TORCH_CHECK(nslice >= 1 && nslice <= 3 && y_slices.size() == x_slices.size());
+  const void* xsl[3];
+  const void* csl[3];
+  for (int s = 0; s < nslice; s++) {
+    TORCH_CHECK(x_slices[s].is_cuda() && x_slices[s].scalar_type() == torch::kBFloat16
+                && x_slices[s].is_contiguous());
+    TORCH_CHECK(y_slices[s].is_cuda() && y_slices[s].scalar_type() == torch::kBFloat16
+                && y_slices[s].is_contiguous());
+    xsl[s] = x_slices[s].data_ptr();
+    csl[s] = y_slices[s].data_ptr();
+  }
+  const long long m = x_slices[0].size(0);
+  const long long n = y_slices[0].size(0);
+  const long long d = x_slices[0].size(1);
+  TORCH_CHECK(d % 64 == 0, "pairwise_l2_filter: d must be a multiple of 64");
+  const int cap = (int)out_d.size(1);
+  TORCH_CHECK(out_d.size(0) == m && out_i.sizes() == out_d.sizes());
+  TORCH_CHECK(cnt.scalar_type() == torch::kInt32 && cnt.numel() == m);
+  raft_amd::launch_pairwise_l2_filter(xsl, csl, xn.data_ptr<float>(),
+                                      yn.data_ptr<float>(), thr.data_ptr<float>(),
+                                      out_d.data_ptr<float>(), out_i.data_ptr<int>(),
+                                      cnt.data_ptr<int>(), cap, col_offset, m, n,
+                                      (int)d, nslice, cur_stream());
+}
+
 torch::Tensor gemm_bf16_f32(torch::Tensor a, torch::Tensor b,
                             c10::optional<torch::Tensor> out, double beta) {
   TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16 && a.is_contiguous());
@@ -399,6 +432,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reduce_rows_by_key_sorted", &reduce_rows_by_key_sorted,
         "keyed row accumulation over a key-sorted permutation");
   m.def("select_k", &select_k, "batched top-k (radix)");
+  m.def("pairwise_l2_filter", &pairwise_l2_filter,
+        "threshold-filtered pairwise L2 candidate emission (fused kNN)");
   m.def("pairwise_l2_mfma", &pairwise_l2_mfma,
         "fused split-bf16 MFMA pairwise L2 tile (single-write epilogue)",
         pybind11::arg("x_slices"), pybind11::arg("y_slices"), pybind11::arg("xn"),

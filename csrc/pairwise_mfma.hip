@@ -77,6 +77,122 @@ __global__ void pairwise_l2_kernel(const __bf16* __restrict__ x0,
   }
 }
 
+// ---------------------------------------------------------------------------
+// filtered top-k emission: same MFMA engine, but instead of writing the tile,
+// emit only candidates with d2 <= thr[row] into per-row bounded buffers
+// (the sample->threshold->filter kNN scheme: the m x n distance tile never
+// touches HBM; expected emissions ~ O(k) per row).
+// ---------------------------------------------------------------------------
+template <int NSLICE>
+__launch_bounds__(256, 2)
+__global__ void pairwise_l2_filter_kernel(const __bf16* __restrict__ x0,
+                                          const __bf16* __restrict__ x1,
+                                          const __bf16* __restrict__ x2,
+                                          const __bf16* __restrict__ c0,
+                                          const __bf16* __restrict__ c1,
+                                          const __bf16* __restrict__ c2,
+                                          const float* __restrict__ xn,
+                                          const float* __restrict__ yn,
+                                          const float* __restrict__ thr,
+                                          float* __restrict__ out_d,
+                                          int* __restrict__ out_i,
+                                          int* __restrict__ cnt, int cap,
+                                          long long col_offset, long long m,
+                                          long long n, int d) {
+  extern __shared__ __bf16 smem[];
+  __bf16* xs[NSLICE];
+  __bf16* cs[NSLICE];
+  const __bf16* const xg[3] = {x0, x1, x2};
+  const __bf16* const cg[3] = {c0, c1, c2};
+#pragma unroll
+  for (int s = 0; s < NSLICE; s++) {
+    xs[s] = smem + s * 8192;
+    cs[s] = smem + (NSLICE + s) * 8192;
+  }
+
+  const int lane = threadIdx.x % RAFT_AMD_WAVE;
+  const int w = threadIdx.x / RAFT_AMD_WAVE;
+  const int wr = w >> 1, wc = w & 1;
+  const long long row0 = (long long)blockIdx.y * 128;
+  const long long col0 = (long long)blockIdx.x * 128;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int a = 0; a < 4; a++)
+#pragma unroll
+    for (int b = 0; b < 4; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  mfma_tile_kloop<NSLICE>(xg, cg, xs, cs, acc, row0, col0, d, m - 1, n - 1,
+                          wr, wc, lane);
+
+#pragma unroll
+  for (int fr = 0; fr < 4; fr++) {
+#pragma unroll
+    for (int reg = 0; reg < 4; reg++) {
+      const long long row = row0 + wr * 64 + fr * 16 + (lane >> 4) * 4 + reg;
+      if (row >= m) continue;
+      const float xv = xn[row];
+      const float t = thr[row];
+#pragma unroll
+      for (int fc = 0; fc < 4; fc++) {
+        const long long col = col0 + wc * 64 + fc * 16 + (lane & 15);
+        if (col < n) {
+          const float d2 = fmaxf(xv + yn[col] - 2.f * acc[fr][fc][reg], 0.f);
+          if (d2 <= t) {
+            const int pos = atomicAdd(&cnt[row], 1);
+            if (pos < cap) {
+              out_d[row * cap + pos] = d2;
+              out_i[row * cap + pos] = (int)(col + col_offset);
+            }
+          }
+        }
+      }
+    }
+  }
+}
+
+void launch_pairwise_l2_filter(const void** xsl, const void** csl, const float* xn,
+                               const float* yn, const float* thr, float* out_d,
+                               int* out_i, int* cnt, int cap, long long col_offset,
+                               long long m, long long n, int d, int nslice,
+                               hipStream_t stream) {
+  dim3 grid((unsigned)((n + 127) / 128), (unsigned)((m + 127) / 128));
+  const size_t lds = (size_t)nslice * 2 * 8192 * sizeof(__bf16);
+  const __bf16* x0 = (const __bf16*)xsl[0];
+  const __bf16* x1 = (const __bf16*)(nslice > 1 ? xsl[1] : xsl[0]);
+  const __bf16* x2 = (const __bf16*)(nslice > 2 ? xsl[2] : xsl[0]);
+  const __bf16* c0 = (const __bf16*)csl[0];
+  const __bf16* c1 = (const __bf16*)(nslice > 1 ? csl[1] : csl[0]);
+  const __bf16* c2 = (const __bf16*)(nslice > 2 ? csl[2] : csl[0]);
+  switch (nslice) {
+    case 1:
+      hipLaunchKernelGGL((pairwise_l2_filter_kernel<1>), grid, dim3(256), lds, stream,
+                         x0, x1, x2, c0, c1, c2, xn, yn, thr, out_d, out_i, cnt, cap,
+                         col_offset, m, n, d);
+      break;
+    case 2:
+      hipLaunchKernelGGL((pairwise_l2_filter_kernel<2>), grid, dim3(256), lds, stream,
+                         x0, x1, x2, c0, c1, c2, xn, yn, thr, out_d, out_i, cnt, cap,
+                         col_offset, m, n, d);
+      break;
+    case 3: {
+      static bool attr_set3 = false;
+      if (!attr_set3) {
+        HIP_CHECK(hipFuncSetAttribute((const void*)&pairwise_l2_filter_kernel<3>,
+                                      hipFuncAttributeMaxDynamicSharedMemorySize,
+                                      96 * 1024));
+        attr_set3 = true;
+      }
+      hipLaunchKernelGGL((pairwise_l2_filter_kernel<3>), grid, dim3(256), lds, stream,
+                         x0, x1, x2, c0, c1, c2, xn, yn, thr, out_d, out_i, cnt, cap,
+                         col_offset, m, n, d);
+      break;
+    }
+    default:
+      throw std::runtime_error("pairwise_l2_filter: nslice must be 1, 2 or 3");
+  }
+}
+
 void launch_pairwise_l2_mfma(const void** xsl, const void** csl, const float* xn,
                              const float* yn, float* out, long long m, long long n,
                              int d, long long ldo, int nslice, bool sqrt_out,

@@ -1,13 +1,26 @@
-"""Brute-force k-nearest-neighbors: pairwise distance tiles + select_k.
+"""Brute-force k-nearest-neighbors.
 
 Reference parity: the historical raft::neighbors brute-force knn (tiled
 distance + k-selection); BASELINE config 5 is this op at k=64 bf16 over 100M
 rows chunked for 288 GB HBM.
+
+MI355X design — the GPU path never materializes the distance matrix:
+  1. SAMPLE: distances to a random ~n/1024 subset of the index give each
+     query row a threshold (the j-th smallest sampled distance, j chosen so
+     the full-index count below it is >= k with high probability).
+  2. FILTER: one MFMA sweep over the whole index emits only candidates with
+     d2 <= threshold into per-row bounded buffers (csrc/pairwise_mfma.hip
+     pairwise_l2_filter_kernel) — expected O(k) emissions per row, so the
+     6.5 TB of distance-tile traffic of the naive tiled path disappears.
+  3. SELECT: native radix select_k over the tiny candidate buffers.
+  4. Rows whose buffer under/overflowed (probabilistically rare) fall back to
+     the exact tiled path; the result is exact for every row.
 """
 from __future__ import annotations
 
 import torch
 
+from raft_amd._ext import require_ext
 from raft_amd.distance import pairwise_distance, DistanceType
 from raft_amd.matrix.select_k import select_k
 from raft_amd.utils import row_chunks
@@ -17,17 +30,84 @@ def knn(x: torch.Tensor, queries: torch.Tensor, k: int,
         metric: DistanceType | str = DistanceType.L2Expanded,
         query_chunk: int = 16384, index_chunk: int = 262144,
         fp32_mode: str = "auto"):
-    """k nearest rows of x for each query row. Returns (dists [q,k], idx [q,k]).
+    """k nearest rows of x for each query row. Returns (dists [q,k], idx [q,k])."""
+    if (queries.is_cuda and metric in (DistanceType.L2Expanded, "sqeuclidean")
+            and x.shape[1] % 64 == 0 and x.shape[0] >= 8 * k
+            and x.dtype in (torch.bfloat16, torch.float32)):
+        return _knn_gpu_filtered(x, queries, k, fp32_mode)
+    return _knn_tiled(x, queries, k, metric, query_chunk, index_chunk, fp32_mode)
 
-    Double-chunked (query rows x index rows) so the distance tile is bounded:
-    tile bytes = query_chunk * index_chunk * 4 — sized for HBM3E residency.
-    Per query chunk, partial top-k results from each index chunk are merged by
-    a select_k over the concatenated candidates (k-way merge, the same scheme
-    the reference uses for multi-block warpsort merges).
-    """
+
+def _slices_of(t: torch.Tensor, fp32_mode: str):
+    from raft_amd.neighbors.fused_l2nn import split_bf16_slices
+    if t.dtype == torch.bfloat16:
+        return [t.contiguous()]
+    nsl = 3 if fp32_mode == "bf16x3" else 2
+    return split_bf16_slices(t, nsl)
+
+
+def _norms(t: torch.Tensor) -> torch.Tensor:
+    return t.float().pow(2).sum(dim=1).contiguous()
+
+
+def _knn_gpu_filtered(x, queries, k, fp32_mode, index_chunk: int = 4_000_000):
+    ext = require_ext()
+    m, d = queries.shape
+    n = x.shape[0]
+    dev = queries.device
+
+    q_slices = _slices_of(queries, fp32_mode)
+    qn = _norms(queries)
+    xn_full = _norms(x)
+
+    # ---- 1. sample -> per-row thresholds -----------------------------------
+    s = max(min(65536, n), n // 1024)
+    gen = torch.Generator(device="cpu").manual_seed(0x5eed)
+    sample_idx = (torch.randint(0, n, (s,), generator=gen)
+                  .to(dev))
+    xs_sample = x[sample_idx].contiguous()
+    sample_slices = _slices_of(xs_sample, fp32_mode)
+    sn = xn_full[sample_idx].contiguous()
+    j = max(2, (4 * k * s) // max(n, 1))
+    thr = torch.empty(m, dtype=torch.float32, device=dev)
+    for s0, s1 in row_chunks(m, 8192):
+        ds = ext.pairwise_l2_mfma([t[s0:s1] for t in q_slices], sample_slices,
+                                  qn[s0:s1].contiguous(), sn)
+        jv, _ = select_k(ds, min(j, s), select_min=True)
+        thr[s0:s1] = jv[:, -1]
+    del ds
+
+    # ---- 2. filtered emission over the full index --------------------------
+    cap = max(16384, 32 * k)
+    cand_d = torch.full((m, cap), float("inf"), dtype=torch.float32, device=dev)
+    cand_i = torch.full((m, cap), -1, dtype=torch.int32, device=dev)
+    cnt = torch.zeros(m, dtype=torch.int32, device=dev)
+    for c0, c1 in row_chunks(n, index_chunk):
+        xc = x[c0:c1]
+        ext.pairwise_l2_filter(q_slices, _slices_of(xc, fp32_mode), qn,
+                               xn_full[c0:c1].contiguous(), thr,
+                               cand_d, cand_i, cnt, c0)
+
+    # ---- 3. select over candidates -----------------------------------------
+    vals, pos = select_k(cand_d, k, select_min=True)
+    idx = torch.gather(cand_i, 1, pos.to(torch.int64)).to(torch.int64)
+
+    # ---- 4. exact fallback for under/overflowed rows ------------------------
+    bad = (cnt < k) | (cnt > cap)
+    n_bad = int(bad.sum().item())
+    if n_bad:
+        rows = bad.nonzero(as_tuple=True)[0]
+        bd, bi = _knn_tiled(x, queries[rows], k, DistanceType.L2Expanded,
+                            8192, index_chunk, fp32_mode)
+        vals[rows] = bd.to(vals.dtype)
+        idx[rows] = bi
+    return vals, idx
+
+
+def _knn_tiled(x, queries, k, metric, query_chunk, index_chunk, fp32_mode):
+    """Exact tiled path (distance tiles + per-tile select + k-way merge)."""
     q = queries.shape[0]
     n = x.shape[0]
-    # distances are fp32 on the GPU bf16 path (MFMA fp32 accumulate)
     out_dtype = torch.float32 if (queries.is_cuda and queries.dtype == torch.bfloat16) \
         else queries.dtype
     out_d = torch.empty((q, k), dtype=out_dtype, device=queries.device)

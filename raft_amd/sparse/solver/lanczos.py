@@ -80,9 +80,14 @@ def lanczos_min_eigenpairs(a, k: int | None = None, config: LanczosConfig | None
         u = u - basis.t() @ (basis @ u)
         return u
 
-    def _extend(start: int):
+    def _extend(start: int, careful: bool = False):
         """Run the three-term recurrence from index `start` to ncv-1, filling
-        t_mat tridiagonally below/right of `start` (lanczos_aux)."""
+        t_mat tridiagonally below/right of `start` (lanczos_aux).
+
+        The loop is host-sync-free: beta degeneracy (invariant subspace) is
+        detected ONCE per cycle by the caller, which re-runs the cycle with
+        careful=True (per-step checks) in that rare case.
+        """
         nonlocal v_next, beta_last
         for i in range(start, ncv):
             u = op(v[i])
@@ -96,7 +101,7 @@ def lanczos_min_eigenpairs(a, k: int | None = None, config: LanczosConfig | None
             u = u - ai * v[i]
             u = _reorth(u, v[: i + 1])
             b = u.norm()
-            if float(b) < 1e-30:
+            if careful and float(b) < 1e-30:
                 # invariant subspace: fresh random orthogonal direction
                 u = uniform((n,), -1.0, 1.0, state=state, device=device, dtype=dtype)
                 u = _reorth(u, v[: i + 1])
@@ -104,12 +109,19 @@ def lanczos_min_eigenpairs(a, k: int | None = None, config: LanczosConfig | None
             if i + 1 < ncv:
                 t_mat[i, i + 1] = b
                 t_mat[i + 1, i] = b
-                v[i + 1] = u / b
+                v[i + 1] = u / b.clamp_min(1e-300)
             else:
                 beta_last = b
-                v_next = u / b
+                v_next = u / b.clamp_min(1e-300)
 
-    _extend(0)
+    def _extend_checked(start: int):
+        """Sync-free extend + one degeneracy check per cycle (rare redo)."""
+        _extend(start, careful=False)
+        betas = torch.diagonal(t_mat, 1)[max(start - 1, 0):]
+        if bool((betas.abs() < 1e-30).any()) or bool(beta_last.abs() < 1e-30):
+            _extend(start, careful=True)
+
+    _extend_checked(0)
     n_iter = 0
     for n_iter in range(1, cfg.max_iterations + 1):
         w, s = torch.linalg.eigh(t_mat)
@@ -126,7 +138,7 @@ def lanczos_min_eigenpairs(a, k: int | None = None, config: LanczosConfig | None
         bk = beta_last * s[ncv - 1, :k]
         t_mat[k, :k] = bk
         t_mat[:k, k] = bk
-        _extend(k)
+        _extend_checked(k)
     w, s = torch.linalg.eigh(t_mat)
     eigvecs = (s[:, :k].t() @ v).t().contiguous()   # [n, k]
     return w[:k].clone(), eigvecs
