@@ -154,7 +154,9 @@ void launch_l2nn_verify_repair(const float* x, const float* c, const float* xn,
                                float* dmin, int* amin, const float* dmin2,
                                float cn_max, long long m, int n, int d,
                                hipStream_t stream) {
-  int grid = grid_1d(m * RAFT_AMD_WAVE, 256);
+  // uncapped: one wave per row (row-serial grid-stride was latency-bound)
+  long long blocks = (m * RAFT_AMD_WAVE + 255) / 256;
+  int grid = (int)(blocks > 2147483647ll ? 2147483647ll : blocks);
   hipLaunchKernelGGL(l2nn_verify_repair_kernel, dim3(grid), dim3(256), 0, stream,
                      x, c, xn, dmin, amin, dmin2, cn_max, m, n, d);
 }

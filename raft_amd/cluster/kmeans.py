@@ -251,6 +251,43 @@ def kmeans_transform(model_or_centroids, x: torch.Tensor, fp32_mode: str = "auto
     return pairwise_distance(x, c, fp32_mode=fp32_mode)
 
 
+def kmeans_balanced_fit(x: torch.Tensor, n_clusters: int, max_iter: int = 20,
+                        seed: int = 0, sample_fraction: float = 0.1,
+                        fp32_mode: str = "auto") -> KMeansModel:
+    """Balanced/hierarchical-flavored k-means (reference kmeans_balanced
+    parity): train on a uniform subsample (the ANN-index-build usage), then
+    assign the full set; clusters that end up empty are refilled from the
+    largest cluster's farthest points, trading inertia for balance.
+    """
+    n = x.shape[0]
+    n_sample = max(n_clusters * 4, int(n * sample_fraction))
+    if n_sample < n:
+        from raft_amd.matrix.sample_rows import sample_rows
+        from raft_amd.random.rng import RngState as _RS
+        xs = sample_rows(x, n_sample, state=_RS(seed=seed))
+    else:
+        xs = x
+    model = kmeans_fit(xs, KMeansParams(n_clusters=n_clusters, max_iter=max_iter,
+                                        seed=seed, init="random",
+                                        fp32_mode=fp32_mode))
+    labels = kmeans_predict(model, x, fp32_mode=fp32_mode)
+    counts = torch.bincount(labels, minlength=n_clusters)
+    model.labels = labels
+    model.inertia = float(fused_l2nn(x, model.centroids,
+                                     fp32_mode=fp32_mode)[0].double().sum())
+    # refill empty clusters from the largest cluster's members
+    for ci in (counts == 0).nonzero(as_tuple=True)[0].tolist():
+        big = int(counts.argmax())
+        members = (labels == big).nonzero(as_tuple=True)[0]
+        d = fused_l2nn(x[members], model.centroids[big:big + 1],
+                       fp32_mode=fp32_mode)[0]
+        far = members[int(d.argmax())]
+        model.centroids[ci] = x[far]
+        counts[big] -= 1
+        counts[ci] += 1
+    return model
+
+
 class KMeans:
     """Estimator-style wrapper."""
 

@@ -60,3 +60,14 @@ class TestKMeans:
             if prev is not None:
                 assert m.inertia <= prev * 1.001
             prev = m.inertia
+
+
+class TestKMeansBalanced:
+    def test_balanced_fit(self):
+        from raft_amd.cluster import kmeans_balanced_fit
+        x, _, centers = make_blobs(2000, 5, n_clusters=6, cluster_std=0.3,
+                                   center_box=(-12, 12), state=RngState(seed=4))
+        model = kmeans_balanced_fit(x, 6, max_iter=30, seed=1, sample_fraction=0.5)
+        counts = torch.bincount(model.labels, minlength=6)
+        assert (counts > 0).all()
+        assert model.inertia < 2000 * 5 * 0.3 ** 2 * 5

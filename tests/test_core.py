@@ -127,3 +127,35 @@ class TestInterruptible:
 
     def test_sync_noop_on_cpu(self):
         Interruptible().synchronize()
+
+
+class TestCompat:
+    def test_pylibraft_style_imports(self):
+        from raft_amd import compat
+        import scipy.sparse as sp
+        import numpy as np
+        rng = np.random.RandomState(0)
+        s = sp.random(80, 80, density=0.1, random_state=rng, format="csr",
+                      dtype=np.float64)
+        s = (s + s.T) * 0.5 + sp.eye(80) * 0.2
+        w, v = compat.eigsh(s.tocsr(), k=3, tol=1e-9)
+        import scipy.sparse.linalg as spla
+        ref = np.sort(spla.eigsh(s.tocsr(), k=3, which="SA")[0])
+        np.testing.assert_allclose(np.asarray(w), ref, rtol=1e-4, atol=1e-6)
+
+    def test_compat_pairwise_and_select(self):
+        from raft_amd import compat
+        import numpy as np
+        x = np.random.rand(10, 4).astype(np.float32)
+        d = compat.pairwise_distance(x, x, metric="sqeuclidean")
+        assert np.asarray(d).shape == (10, 10)
+        v, i = compat.select_k(np.asarray(d), 3)
+        assert np.asarray(v).shape == (10, 3)
+
+    def test_compat_rmat(self):
+        from raft_amd import compat
+        import numpy as np
+        out = np.zeros((1000, 2), dtype=np.int64)
+        theta = np.array([0.6, 0.2, 0.15, 0.05] * 8, dtype=np.float64)
+        compat.rmat(out, theta, 8, 8, seed=7)
+        assert out.max() < 256 and out.min() >= 0
