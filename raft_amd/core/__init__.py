@@ -17,11 +17,16 @@ from .serialize import serialize_mdspan, deserialize_mdspan, save_npy, load_npy
 from .bitset import Bitset
 from .interruptible import Interruptible, synchronize as interruptible_synchronize
 from .logger import get_logger, set_level
+from .trace import annotate, annotated
+from .memory import MemoryStats, TrackingScope, ResourceMonitor
+from .mdbuffer import MDBuffer, MemoryType, memory_type_dispatcher
 
 __all__ = [
     "Resources", "DeviceResources", "Handle", "DeviceResourcesSNMG",
     "DeviceResourcesManager", "get_resources", "device_ndarray",
     "serialize_mdspan", "deserialize_mdspan", "save_npy", "load_npy",
     "Bitset", "Interruptible", "interruptible_synchronize",
-    "get_logger", "set_level",
+    "get_logger", "set_level", "annotate", "annotated",
+    "MemoryStats", "TrackingScope", "ResourceMonitor",
+    "MDBuffer", "MemoryType", "memory_type_dispatcher",
 ]

@@ -14,11 +14,12 @@ from .ops import (
 from .sort import col_wise_sort
 from .sample_rows import sample_rows
 from .norm import l2_norm
+from .print import print_matrix
 
 __all__ = [
     "select_k", "SelectAlgo", "gather", "gather_if", "scatter",
     "argmax", "argmin", "slice_matrix", "get_diagonal", "set_diagonal",
     "upper_triangular", "lower_triangular", "row_reverse", "col_reverse",
     "shift_rows", "eye", "power", "ratio", "reciprocal", "matrix_sqrt",
-    "sign_flip", "threshold", "linewise", "col_wise_sort", "sample_rows", "l2_norm",
+    "sign_flip", "threshold", "linewise", "col_wise_sort", "sample_rows", "l2_norm", "print_matrix",
 ]
