@@ -35,9 +35,14 @@ void launch_csr_spmv(const int*, const int*, const T*, const T*, T*, long long,
 void launch_reduce_rows_by_key(const float*, const int*, float*, float*, long long,
                                long long, long long, int, hipStream_t);
 void launch_reduce_rows_by_key_sorted(const float*, const int*, const int*, float*,
-                                      long long, long long, hipStream_t);
+                                      float*, long long, long long, hipStream_t);
+void launch_split_bf16_norms(const float*, void*, void*, void*, float*, int,
+                             long long, long long, hipStream_t);
+void launch_kmeans_update_centroids(const float*, const float*, float*, long long,
+                                    long long, hipStream_t);
 void launch_l2nn_verify_repair(const float*, const float*, const float*, float*, int*,
-                               const float*, float, long long, int, int, hipStream_t);
+                               const float*, const float*, long long, int, int,
+                               hipStream_t);
 // from select_k.hip
 long long select_k_workspace_bytes(long long batch);
 void launch_select_k(const float*, float*, int*, void*, long long, long long, int,
@@ -211,9 +216,48 @@ torch::Tensor reduce_rows_by_key_sorted(torch::Tensor x, torch::Tensor perm,
   auto sums = torch::zeros({n_keys, x.size(1)}, x.options());
   raft_amd::launch_reduce_rows_by_key_sorted(x.data_ptr<float>(), perm.data_ptr<int>(),
                                              keys_sorted.data_ptr<int>(),
-                                             sums.data_ptr<float>(), x.size(0),
-                                             x.size(1), cur_stream());
+                                             sums.data_ptr<float>(), nullptr,
+                                             x.size(0), x.size(1), cur_stream());
   return sums;
+}
+
+void reduce_rows_by_key_sorted_into(torch::Tensor x, torch::Tensor perm,
+                                    torch::Tensor keys_sorted, torch::Tensor sums,
+                                    torch::Tensor counts) {
+  check_f32_2d(x, "x");
+  TORCH_CHECK(perm.scalar_type() == torch::kInt32 && keys_sorted.scalar_type() == torch::kInt32);
+  TORCH_CHECK(sums.is_contiguous() && counts.is_contiguous());
+  raft_amd::launch_reduce_rows_by_key_sorted(x.data_ptr<float>(), perm.data_ptr<int>(),
+                                             keys_sorted.data_ptr<int>(),
+                                             sums.data_ptr<float>(),
+                                             counts.data_ptr<float>(), x.size(0),
+                                             x.size(1), cur_stream());
+}
+
+void split_bf16_norms(torch::Tensor c, std::vector<torch::Tensor> slices,
+                      torch::Tensor cn) {
+  check_f32_2d(c, "c");
+  const int nslice = (int)slices.size();
+  TORCH_CHECK(nslice >= 1 && nslice <= 3);
+  for (auto& s : slices)
+    TORCH_CHECK(s.scalar_type() == torch::kBFloat16 && s.is_contiguous()
+                && s.sizes() == c.sizes());
+  void* p0 = slices[0].data_ptr();
+  void* p1 = nslice > 1 ? slices[1].data_ptr() : p0;
+  void* p2 = nslice > 2 ? slices[2].data_ptr() : p0;
+  raft_amd::launch_split_bf16_norms(c.data_ptr<float>(), p0, p1, p2,
+                                    cn.data_ptr<float>(), nslice, c.size(0),
+                                    c.size(1), cur_stream());
+}
+
+void kmeans_update_centroids(torch::Tensor sums, torch::Tensor counts,
+                             torch::Tensor centroids) {
+  check_f32_2d(centroids, "centroids");
+  raft_amd::launch_kmeans_update_centroids(sums.data_ptr<float>(),
+                                           counts.data_ptr<float>(),
+                                           centroids.data_ptr<float>(),
+                                           centroids.size(0), centroids.size(1),
+                                           cur_stream());
 }
 
 torch::Tensor reduce_rows_by_key(torch::Tensor x, torch::Tensor keys, int64_t n_keys) {
@@ -323,14 +367,15 @@ torch::Tensor pairwise_l2_mfma(std::vector<torch::Tensor> x_slices,
 
 void l2nn_verify_repair(torch::Tensor x, torch::Tensor c, torch::Tensor xn,
                         torch::Tensor dmin, torch::Tensor amin, torch::Tensor dmin2,
-                        double cn_max) {
+                        torch::Tensor cn_max) {
   check_f32_2d(x, "x");
   check_f32_2d(c, "c");
+  TORCH_CHECK(cn_max.scalar_type() == torch::kFloat32 && cn_max.numel() >= 1);
   raft_amd::launch_l2nn_verify_repair(x.data_ptr<float>(), c.data_ptr<float>(),
                                       xn.data_ptr<float>(), dmin.data_ptr<float>(),
                                       amin.data_ptr<int>(), dmin2.data_ptr<float>(),
-                                      (float)cn_max, x.size(0), (int)c.size(0),
-                                      (int)x.size(1), cur_stream());
+                                      cn_max.data_ptr<float>(), x.size(0),
+                                      (int)c.size(0), (int)x.size(1), cur_stream());
 }
 
 void pairwise_l2_filter(std::vector<torch::Tensor> x_slices,
@@ -431,6 +476,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reduce_rows_by_key", &reduce_rows_by_key, "keyed row accumulation");
   m.def("reduce_rows_by_key_sorted", &reduce_rows_by_key_sorted,
         "keyed row accumulation over a key-sorted permutation");
+  m.def("reduce_rows_by_key_sorted_into", &reduce_rows_by_key_sorted_into,
+        "keyed row accumulation + counts into caller buffers");
+  m.def("split_bf16_norms", &split_bf16_norms,
+        "fused fp32->bf16 slice split + row sq-norms");
+  m.def("kmeans_update_centroids", &kmeans_update_centroids,
+        "centroids = counts>0 ? sums/counts : centroids");
   m.def("select_k", &select_k, "batched top-k (radix)");
   m.def("pairwise_l2_filter", &pairwise_l2_filter,
         "threshold-filtered pairwise L2 candidate emission (fused kNN)");

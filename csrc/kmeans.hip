@@ -53,6 +53,7 @@ __global__ void reduce_rows_by_key_sorted_kernel(const float* __restrict__ x,
                                                  const int* __restrict__ perm,
                                                  const int* __restrict__ keys_sorted,
                                                  float* __restrict__ sums,
+                                                 float* __restrict__ counts,
                                                  long long n_rows, long long d,
                                                  long long chunk) {
   const int lane = threadIdx.x % RAFT_AMD_WAVE;
@@ -67,6 +68,7 @@ __global__ void reduce_rows_by_key_sorted_kernel(const float* __restrict__ x,
   for (int j = 0; j < MAX_DREG; j++) acc[j] = 0.f;
 
   int cur_key = keys_sorted[start];
+  int run_len = 0;
   for (long long i = start; i < stop; i++) {
     const int key = keys_sorted[i];
     if (key != cur_key) {
@@ -76,6 +78,8 @@ __global__ void reduce_rows_by_key_sorted_kernel(const float* __restrict__ x,
         if (col < d) atomicAdd(&sp[col], acc[j]);
         acc[j] = 0.f;
       }
+      if (counts && lane == 0) atomicAdd(&counts[cur_key], (float)run_len);
+      run_len = 0;
       cur_key = key;
     }
     const float* rp = x + (long long)perm[i] * d;
@@ -83,12 +87,80 @@ __global__ void reduce_rows_by_key_sorted_kernel(const float* __restrict__ x,
       const long long col = (long long)j * RAFT_AMD_WAVE + lane;
       if (col < d) acc[j] += rp[col];
     }
+    run_len++;
   }
   float* sp = sums + (long long)cur_key * d;
   for (int j = 0; j < dreg; j++) {
     const long long col = (long long)j * RAFT_AMD_WAVE + lane;
     if (col < d) atomicAdd(&sp[col], acc[j]);
   }
+  if (counts && lane == 0) atomicAdd(&counts[cur_key], (float)run_len);
+}
+
+// fused centroid prep: split fp32 centroids into bf16 slices + row sq-norms
+// in ONE pass (replaces ~6 torch ops per k-means iteration).
+__global__ void split_bf16_norms_kernel(const float* __restrict__ c,
+                                        __bf16* __restrict__ s0,
+                                        __bf16* __restrict__ s1,
+                                        __bf16* __restrict__ s2,
+                                        float* __restrict__ cn, int nslice,
+                                        long long n_rows, long long d) {
+  __shared__ float lds[256 / RAFT_AMD_WAVE];
+  for (long long row = blockIdx.x; row < n_rows; row += gridDim.x) {
+    const float* rp = c + row * d;
+    float acc = 0.f;
+    for (long long j = threadIdx.x; j < d; j += blockDim.x) {
+      const float v = rp[j];
+      acc += v * v;
+      const __bf16 h0 = (__bf16)v;
+      s0[row * d + j] = h0;
+      if (nslice > 1) {
+        const float r1 = v - (float)h0;
+        const __bf16 h1 = (__bf16)r1;
+        s1[row * d + j] = h1;
+        if (nslice > 2) s2[row * d + j] = (__bf16)(r1 - (float)h1);
+      }
+    }
+    acc = wave_reduce_sum(acc);
+    const int wid = threadIdx.x / RAFT_AMD_WAVE, lane = threadIdx.x % RAFT_AMD_WAVE;
+    if (lane == 0) lds[wid] = acc;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      float s = 0.f;
+      for (int w = 0; w < (int)(blockDim.x / RAFT_AMD_WAVE); w++) s += lds[w];
+      cn[row] = s;
+    }
+    __syncthreads();
+  }
+}
+
+void launch_split_bf16_norms(const float* c, void* s0, void* s1, void* s2, float* cn,
+                             int nslice, long long n_rows, long long d,
+                             hipStream_t stream) {
+  int grid = (int)(n_rows < 2048 ? n_rows : 2048);
+  hipLaunchKernelGGL(split_bf16_norms_kernel, dim3(grid), dim3(256), 0, stream, c,
+                     (__bf16*)s0, (__bf16*)s1, (__bf16*)s2, cn, nslice, n_rows, d);
+}
+
+// centroid update: c[k] = counts[k] > 0 ? sums[k]/counts[k] : c[k]
+// (replaces the where/div/clamp torch op chain per iteration)
+__global__ void kmeans_update_centroids_kernel(const float* __restrict__ sums,
+                                               const float* __restrict__ counts,
+                                               float* __restrict__ c, long long k,
+                                               long long d) {
+  const long long total = k * d;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += stride) {
+    const float cnt = counts[t / d];
+    if (cnt > 0.f) c[t] = sums[t] / cnt;
+  }
+}
+
+void launch_kmeans_update_centroids(const float* sums, const float* counts, float* c,
+                                    long long k, long long d, hipStream_t stream) {
+  hipLaunchKernelGGL(kmeans_update_centroids_kernel, dim3(grid_1d(k * d, 256)),
+                     dim3(256), 0, stream, sums, counts, c, k, d);
 }
 
 // Exact-fp32 verification/repair pass for the split-bf16 fused L2-NN
@@ -106,7 +178,9 @@ __global__ void l2nn_verify_repair_kernel(const float* __restrict__ x,
                                           float* __restrict__ dmin,
                                           int* __restrict__ amin,
                                           const float* __restrict__ dmin2,
-                                          float cn_max, long long m, int n, int d) {
+                                          const float* __restrict__ cn_max_p,
+                                          long long m, int n, int d) {
+  const float cn_max = *cn_max_p;
   const long long waves_per_block = blockDim.x / RAFT_AMD_WAVE;
   const int lane = threadIdx.x % RAFT_AMD_WAVE;
   long long row = (long long)blockIdx.x * waves_per_block + threadIdx.x / RAFT_AMD_WAVE;
@@ -152,18 +226,18 @@ __global__ void l2nn_verify_repair_kernel(const float* __restrict__ x,
 
 void launch_l2nn_verify_repair(const float* x, const float* c, const float* xn,
                                float* dmin, int* amin, const float* dmin2,
-                               float cn_max, long long m, int n, int d,
+                               const float* cn_max_dev, long long m, int n, int d,
                                hipStream_t stream) {
   // uncapped: one wave per row (row-serial grid-stride was latency-bound)
   long long blocks = (m * RAFT_AMD_WAVE + 255) / 256;
   int grid = (int)(blocks > 2147483647ll ? 2147483647ll : blocks);
   hipLaunchKernelGGL(l2nn_verify_repair_kernel, dim3(grid), dim3(256), 0, stream,
-                     x, c, xn, dmin, amin, dmin2, cn_max, m, n, d);
+                     x, c, xn, dmin, amin, dmin2, cn_max_dev, m, n, d);
 }
 
 void launch_reduce_rows_by_key_sorted(const float* x, const int* perm,
                                       const int* keys_sorted, float* sums,
-                                      long long n_rows, long long d,
+                                      float* counts, long long n_rows, long long d,
                                       hipStream_t stream) {
   // chunk sized so the grid fills the chip (~2048 blocks * 4 waves)
   const long long n_waves_target = 2048 * 4;
@@ -173,13 +247,13 @@ void launch_reduce_rows_by_key_sorted(const float* x, const int* perm,
   const int grid = (int)((n_waves * RAFT_AMD_WAVE + 255) / 256);
   if (d <= 256) {
     hipLaunchKernelGGL((reduce_rows_by_key_sorted_kernel<4>), dim3(grid), dim3(256),
-                       0, stream, x, perm, keys_sorted, sums, n_rows, d, chunk);
+                       0, stream, x, perm, keys_sorted, sums, counts, n_rows, d, chunk);
   } else if (d <= 1024) {
     hipLaunchKernelGGL((reduce_rows_by_key_sorted_kernel<16>), dim3(grid), dim3(256),
-                       0, stream, x, perm, keys_sorted, sums, n_rows, d, chunk);
+                       0, stream, x, perm, keys_sorted, sums, counts, n_rows, d, chunk);
   } else {
     hipLaunchKernelGGL((reduce_rows_by_key_sorted_kernel<64>), dim3(grid), dim3(256),
-                       0, stream, x, perm, keys_sorted, sums, n_rows, d, chunk);
+                       0, stream, x, perm, keys_sorted, sums, counts, n_rows, d, chunk);
   }
 }
 

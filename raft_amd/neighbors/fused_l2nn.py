@@ -73,7 +73,7 @@ def fused_l2nn_presplit(x_slices, xn: torch.Tensor, y: torch.Tensor,
     dmin, amin, dmin2 = ext.fused_l2nn_split(list(x_slices), list(y_slices),
                                              xn.contiguous(), ynp.contiguous())
     if verify_x is not None:
-        cn_max = float(yn.max().item())
+        cn_max = yn.max().reshape(1)   # device scalar: no host sync
         ext.l2nn_verify_repair(verify_x.contiguous(), y.contiguous(),
                                xn.contiguous(), dmin, amin, dmin2, cn_max)
     if sqrt:
