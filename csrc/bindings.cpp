@@ -47,6 +47,8 @@ void launch_l2nn_verify_repair(const float*, const float*, const float*, float*,
 long long select_k_workspace_bytes(long long batch);
 void launch_select_k(const float*, float*, int*, void*, long long, long long, int,
                      bool, bool, hipStream_t);
+void launch_select_k_warpsort(const float*, float*, int*, long long, long long, int,
+                              bool, hipStream_t);
 // from fused_l2nn.hip
 void launch_fused_l2nn_split(const void**, const void**, const float*, const float*,
                              float*, int*, float*, long long, int, int, int,
@@ -281,6 +283,13 @@ std::tuple<torch::Tensor, torch::Tensor> select_k(torch::Tensor x, int64_t k,
   check_f32_2d(x, "x");
   auto vals = torch::empty({x.size(0), k}, x.options());
   auto idx = torch::empty({x.size(0), k}, x.options().dtype(torch::kInt32));
+  // algo: 0 auto, 1 radix, 2 warpsort (k <= 64, wave-register queue)
+  if (k <= 64 && algo == 2) {
+    raft_amd::launch_select_k_warpsort(x.data_ptr<float>(), vals.data_ptr<float>(),
+                                       idx.data_ptr<int>(), x.size(0), x.size(1),
+                                       (int)k, select_min, cur_stream());
+    return {vals, idx};
+  }
   auto ws = torch::empty({raft_amd::select_k_workspace_bytes(x.size(0))},
                          x.options().dtype(torch::kUInt8));
   raft_amd::launch_select_k(x.data_ptr<float>(), vals.data_ptr<float>(),

@@ -269,6 +269,97 @@ __global__ void select_k_radix_kernel(const float* __restrict__ x,
   }
 }
 
+// ---------------------------------------------------------------------------
+// warpsort select (k <= 64): one wave per row, per-lane single (val,idx)
+// queue slot — the wave-wide register priority queue of the reference's
+// select_warpsort.cuh, re-derived for 64-lane wavefronts (capacity == wave
+// width == 64, so one bitonic network IS the queue). Strategy =
+// filtered-immediate: batches of 64 loads; a ballot skips batches with no
+// candidate below the current worst (after warm-up nearly all batches skip,
+// leaving a pure streaming read); candidate batches are wave-bitonic-sorted
+// descending and merged with the ascending queue by elementwise min + one
+// bitonic merge.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ void cmp_exchange(float& v, int& i, int stride, bool keep_small) {
+  const float ov = __shfl_xor(v, stride, RAFT_AMD_WAVE);
+  const int oi = __shfl_xor(i, stride, RAFT_AMD_WAVE);
+  const bool other_smaller = (ov < v) || (ov == v && oi < i);
+  if (keep_small == other_smaller) { v = ov; i = oi; }
+}
+
+__device__ __forceinline__ void wave_bitonic_sort_asc(float& v, int& i, int lane) {
+#pragma unroll
+  for (int size = 2; size <= 64; size <<= 1) {
+#pragma unroll
+    for (int stride = 64 >> 1; stride > 0; stride >>= 1) {
+      if (stride < size) {
+        const bool up = (lane & size) == 0 || size == 64;
+        const bool keep_small = ((lane & stride) == 0) == up;
+        cmp_exchange(v, i, stride, keep_small);
+      }
+    }
+  }
+}
+
+__device__ __forceinline__ void wave_bitonic_merge_asc(float& v, int& i, int lane) {
+#pragma unroll
+  for (int stride = 32; stride > 0; stride >>= 1) {
+    const bool keep_small = (lane & stride) == 0;
+    cmp_exchange(v, i, stride, keep_small);
+  }
+}
+
+template <int BLOCK = 256>
+__global__ void select_k_warpsort_kernel(const float* __restrict__ x,
+                                         float* __restrict__ out_v,
+                                         int* __restrict__ out_i,
+                                         long long batch, long long len, int k,
+                                         bool select_min) {
+  const int lane = threadIdx.x % RAFT_AMD_WAVE;
+  const long long waves_per_block = BLOCK / RAFT_AMD_WAVE;
+  long long row = (long long)blockIdx.x * waves_per_block + threadIdx.x / RAFT_AMD_WAVE;
+  const long long stride = (long long)gridDim.x * waves_per_block;
+  for (; row < batch; row += stride) {
+    const float* rp = x + row * len;
+    float qv = INFINITY;   // queue ascending across lanes; lane 63 = worst
+    int qi = -1;
+    float worst = INFINITY;
+    for (long long j0 = 0; j0 < len; j0 += RAFT_AMD_WAVE) {
+      const long long j = j0 + lane;
+      float val = j < len ? rp[j] : INFINITY;
+      if (!select_min && j < len) val = -val;
+      const bool cand = (val < worst);
+      if (__ballot(cand) == 0ull) continue;
+      // sort batch ascending, then reverse to descending via lane mirror
+      float bv = val;
+      int bi = cand ? (int)j : -1;
+      if (!cand) bv = INFINITY;  // non-candidates can't enter the queue
+      wave_bitonic_sort_asc(bv, bi, lane);
+      const float rv = __shfl(bv, 63 - lane, RAFT_AMD_WAVE);
+      const int ri = __shfl(bi, 63 - lane, RAFT_AMD_WAVE);
+      // elementwise min of (asc queue, desc batch) -> bitonic; re-merge
+      const bool take = (rv < qv) || (rv == qv && ri != -1 && ri < qi);
+      if (take) { qv = rv; qi = ri; }
+      wave_bitonic_merge_asc(qv, qi, lane);
+      worst = __shfl(qv, (k <= 64 ? k : 64) - 1, RAFT_AMD_WAVE);
+    }
+    if (lane < k) {
+      out_v[row * k + lane] = select_min ? qv : -qv;
+      out_i[row * k + lane] = qi;
+    }
+  }
+}
+
+void launch_select_k_warpsort(const float* x, float* out_v, int* out_i,
+                              long long batch, long long len, int k,
+                              bool select_min, hipStream_t stream) {
+  const long long blocks = (batch + 3) / 4;
+  const int grid = (int)(blocks < 65536 ? blocks : 65536);
+  hipLaunchKernelGGL((select_k_warpsort_kernel<256>), dim3(grid), dim3(256), 0,
+                     stream, x, out_v, out_i, batch, len, k, select_min);
+}
+
 int select_k_grid(long long batch) { return (int)(batch < 4096 ? batch : 4096); }
 
 long long select_k_workspace_bytes(long long batch) {

@@ -364,6 +364,50 @@ class TestKMeansGpu:
         d = torch.cdist(centers, model.centroids)
         assert d.min(dim=1).values.max() < 1.0
 
+    def test_fast_iterate_matches_cpu(self, dev, ext):
+        """the minimal-dispatch fused EM loop (k % 128 == 0) vs CPU oracle."""
+        from raft_amd.cluster.kmeans import kmeans_iterate
+        from raft_amd.random import make_blobs, RngState
+        x, _, centers = make_blobs(20000, 64, n_clusters=128, cluster_std=0.4,
+                                   state=RngState(seed=9), device=dev)
+        c0 = centers + 0.2
+        cg, ig = kmeans_iterate(x, c0.clone(), 3, fp32_mode="bf16x2v")
+        cc, ic = kmeans_iterate(x.cpu(), c0.cpu().clone(), 3)
+        torch.testing.assert_close(cg.cpu(), cc, rtol=1e-3, atol=1e-3)
+        assert abs(ig - ic) / ic < 1e-3
+
+    def test_split_norms_and_update_kernels(self, dev, ext):
+        torch.manual_seed(10)
+        c = torch.randn(256, 192, device=dev)
+        s = [torch.empty_like(c, dtype=torch.bfloat16) for _ in range(2)]
+        cn = torch.empty(256, device=dev)
+        ext.split_bf16_norms(c, s, cn)
+        torch.testing.assert_close(cn.double(), (c.double() ** 2).sum(1),
+                                   rtol=1e-5, atol=1e-4)
+        recon = s[0].float() + s[1].float()
+        assert float((recon - c).abs().max()) < 1e-3
+        # update kernel
+        sums = torch.randn(256, 192, device=dev)
+        counts = torch.randint(0, 3, (256,), device=dev).float()
+        c2 = c.clone()
+        ext.kmeans_update_centroids(sums.contiguous(), counts.contiguous(), c2)
+        ref = torch.where((counts > 0).unsqueeze(1), sums / counts.clamp_min(1e-9).unsqueeze(1), c)
+        torch.testing.assert_close(c2, ref, rtol=1e-5, atol=1e-6)
+
+    def test_rrbk_sorted_counts(self, dev, ext):
+        torch.manual_seed(11)
+        x = torch.randn(30000, 64, device=dev)
+        keys = torch.randint(0, 100, (30000,), device=dev, dtype=torch.int32)
+        ks, perm = torch.sort(keys)
+        sums = torch.zeros(100, 64, device=dev)
+        counts = torch.zeros(100, device=dev)
+        ext.reduce_rows_by_key_sorted_into(x, perm.to(torch.int32), ks, sums, counts)
+        ref_counts = torch.bincount(keys.long(), minlength=100).float()
+        torch.testing.assert_close(counts, ref_counts)
+        ref = torch.zeros(100, 64, device=dev, dtype=torch.float64)
+        ref.index_add_(0, keys.long(), x.double())
+        torch.testing.assert_close(sums.double(), ref, rtol=1e-4, atol=1e-3)
+
     def test_kmeans_iterate_matches_cpu(self, dev, ext):
         from raft_amd.cluster.kmeans import kmeans_iterate
         from raft_amd.random import make_blobs, RngState
