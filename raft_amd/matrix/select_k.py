@@ -76,6 +76,14 @@ def select_k(x: torch.Tensor, k: int, select_min: bool = True,
                 gi = torch.gather(cand_i, 1, fpos.to(torch.int64))
                 return fv, gi
         algo_code = {SelectAlgo.AUTO: 0, SelectAlgo.RADIX: 1, SelectAlgo.WARPSORT: 2}[algo]
+        if algo == SelectAlgo.AUTO and k <= 64 and batch >= 2048 and n >= 100000:
+            # measured on gfx950 (benchmarks 2026-09): the wave-register
+            # warpsort queue is ~6x the radix path on long rows once the
+            # batch fills the chip (it degenerates to a streaming read after
+            # the ballot filter warms up); radix wins on shorter rows / small
+            # batch. This is the re-measured analog of the reference's
+            # learned dispatch tree.
+            algo_code = 2
         vals, idx = ext.select_k(x.contiguous(), k, bool(select_min), algo_code, bool(sorted))
         return vals, idx.to(torch.int64)
 
