@@ -53,8 +53,12 @@ void launch_select_k_warpsort(const float*, float*, int*, long long, long long, 
 void launch_fused_l2nn_split(const void**, const void**, const float*, const float*,
                              float*, int*, float*, long long, int, int, int,
                              hipStream_t);
-// from fused_l2nn_v2.hip (persistent-X variant)
+// from fused_l2nn_v2.hip (persistent-X variant + w8 wide-tile variant)
 bool fused_l2nn_persist_supported(int nslice, int d);
+bool fused_l2nn_w8_supported(int nslice, int n, int d);
+void launch_fused_l2nn_w8(const void**, const void**, const float*, const float*,
+                          float*, int*, float*, long long, int, int, int,
+                          hipStream_t);
 void launch_fused_l2nn_persist(const void**, const void**, const float*, const float*,
                                float*, int*, float*, long long, int, int, int,
                                hipStream_t);
@@ -327,7 +331,16 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> fused_l2nn_split(
     const char* e = getenv("RAFT_AMD_PERSIST_L2NN");
     return e && e[0] == '1';
   }();  // measured slower than v1 at 10M x 256 (occupancy); kept for tuning
-  if (use_persist && raft_amd::fused_l2nn_persist_supported(nslice, (int)d)) {
+  static const bool no_w8 = [] {
+    const char* e = getenv("RAFT_AMD_L2NN_W8");
+    return e && e[0] == '0';
+  }();
+  if (!no_w8 && !use_persist && raft_amd::fused_l2nn_w8_supported(nslice, (int)n, (int)d)) {
+    raft_amd::launch_fused_l2nn_w8(xsl, csl, xn.data_ptr<float>(),
+                                   cn.data_ptr<float>(), dmin.data_ptr<float>(),
+                                   amin.data_ptr<int>(), dmin2.data_ptr<float>(),
+                                   m, (int)n, (int)d, nslice, cur_stream());
+  } else if (use_persist && raft_amd::fused_l2nn_persist_supported(nslice, (int)d)) {
     raft_amd::launch_fused_l2nn_persist(xsl, csl, xn.data_ptr<float>(),
                                         cn.data_ptr<float>(), dmin.data_ptr<float>(),
                                         amin.data_ptr<int>(), dmin2.data_ptr<float>(),

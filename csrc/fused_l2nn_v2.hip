@@ -218,6 +218,221 @@ static void launch_persist(const __bf16* x0, const __bf16* x1, const __bf16* x2,
                      cn, dmin, amin, dmin2, m, n, d);
 }
 
+// ---------------------------------------------------------------------------
+// w8 variant: 512 threads (8 waves, 2x4), output tile 128 rows x 256 cols.
+// Halves the X re-read count AND the barrier count vs the 4-wave 128x128
+// kernel at the same 2-waves/SIMD occupancy (96 KiB LDS for NSLICE=2,
+// 144 KiB for NSLICE=3 -> 1 block/CU of 8 waves). Tracks (best, second-best)
+// like v1 so the verified engine works unchanged.
+// ---------------------------------------------------------------------------
+
+template <int NSLICE>
+__launch_bounds__(512, 2)
+__global__ void fused_l2nn_w8_kernel(
+    const __bf16* __restrict__ x0, const __bf16* __restrict__ x1,
+    const __bf16* __restrict__ x2, const __bf16* __restrict__ c0,
+    const __bf16* __restrict__ c1, const __bf16* __restrict__ c2,
+    const float* __restrict__ xn, const float* __restrict__ cn,
+    float* __restrict__ dmin, int* __restrict__ amin, float* __restrict__ dmin2,
+    long long m, int n, int d) {
+  constexpr int BLOCK = 512;
+  extern __shared__ __bf16 smem[];
+  const __bf16* const xg[3] = {x0, x1, x2};
+  const __bf16* const cg[3] = {c0, c1, c2};
+  __bf16* xs[NSLICE];
+  __bf16* cs[NSLICE];
+#pragma unroll
+  for (int s = 0; s < NSLICE; s++) {
+    xs[s] = smem + s * 8192;                       // [128][64] per slice
+    cs[s] = smem + NSLICE * 8192 + s * 16384;      // [256][64] per slice
+  }
+
+  const int lane = threadIdx.x % RAFT_AMD_WAVE;
+  const int w = threadIdx.x / RAFT_AMD_WAVE;
+  const int wr = w >> 2, wc = w & 3;  // 2x4 wave grid
+  const long long row0 = (long long)blockIdx.x * 128;
+
+  float best[4][4], best2[4][4];
+  int bidx[4][4];
+#pragma unroll
+  for (int a = 0; a < 4; a++)
+#pragma unroll
+    for (int b = 0; b < 4; b++) {
+      best[a][b] = INFINITY; best2[a][b] = INFINITY; bidx[a][b] = 0;
+    }
+
+  const int n_tiles = n / 256;
+  const int k_tiles = d / 64;
+  for (int nt = 0; nt < n_tiles; nt++) {
+    f32x4 acc[4][4];
+#pragma unroll
+    for (int a = 0; a < 4; a++)
+#pragma unroll
+      for (int b = 0; b < 4; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    for (int kt = 0; kt < k_tiles; kt++) {
+#pragma unroll
+      for (int s = 0; s < NSLICE; s++) {
+        stage_rows<128, BLOCK>(xg[s], xs[s], row0, (long long)kt * 64, d, m - 1);
+        stage_rows<256, BLOCK>(cg[s], cs[s], (long long)nt * 256,
+                               (long long)kt * 64, d, n - 1);
+      }
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+
+#pragma unroll
+      for (int kf = 0; kf < 2; kf++) {
+        bf16x8 a_frag[NSLICE][4], b_frag[NSLICE][4];
+        const int kbyte = (kf * 32 + (lane >> 4) * 8) * 2;
+#pragma unroll
+        for (int fr = 0; fr < 4; fr++) {
+          const int r = wr * 64 + fr * 16 + (lane & 15);
+          const int byte = mfma_swz(r * 128 + kbyte);
+#pragma unroll
+          for (int s = 0; s < NSLICE; s++)
+            a_frag[s][fr] = *reinterpret_cast<const bf16x8*>((const char*)xs[s] + byte);
+        }
+#pragma unroll
+        for (int fc = 0; fc < 4; fc++) {
+          const int c = wc * 64 + fc * 16 + (lane & 15);
+          const int byte = mfma_swz(c * 128 + kbyte);
+#pragma unroll
+          for (int s = 0; s < NSLICE; s++)
+            b_frag[s][fc] = *reinterpret_cast<const bf16x8*>((const char*)cs[s] + byte);
+        }
+#pragma unroll
+        for (int fr = 0; fr < 4; fr++)
+#pragma unroll
+          for (int fc = 0; fc < 4; fc++) {
+#pragma unroll
+            for (int p = 0; p < mfma_n_products<NSLICE>(); p++) {
+              acc[fr][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  a_frag[MFMA_PROD_A[p]][fr], b_frag[MFMA_PROD_B[p]][fc],
+                  acc[fr][fc], 0, 0, 0);
+            }
+          }
+      }
+      __syncthreads();
+    }
+
+    const int col_base = nt * 256 + wc * 64;
+#pragma unroll
+    for (int fr = 0; fr < 4; fr++) {
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++) {
+        float v = INFINITY, v2 = INFINITY;
+        int vi = 0;
+#pragma unroll
+        for (int fc = 0; fc < 4; fc++) {
+          const int col = col_base + fc * 16 + (lane & 15);
+          const float s = cn[col] - 2.f * acc[fr][fc][reg];
+          if (s < v) { v2 = v; v = s; vi = col; }
+          else if (s < v2) { v2 = s; }
+        }
+#pragma unroll
+        for (int off = 8; off > 0; off >>= 1) {
+          const float ov = __shfl_xor(v, off, RAFT_AMD_WAVE);
+          const float ov2 = __shfl_xor(v2, off, RAFT_AMD_WAVE);
+          const int oi = __shfl_xor(vi, off, RAFT_AMD_WAVE);
+          float new2 = fminf(v2, ov2);
+          if (oi != vi) new2 = fminf(new2, fmaxf(v, ov));
+          v2 = new2;
+          if (ov < v || (ov == v && oi < vi)) { v = ov; vi = oi; }
+        }
+        const float hi = fmaxf(best[fr][reg], v);
+        const float merged2 = fminf(fminf(best2[fr][reg], v2), hi);
+        if (v < best[fr][reg] || (v == best[fr][reg] && vi < bidx[fr][reg])) {
+          best[fr][reg] = v;
+          bidx[fr][reg] = vi;
+        }
+        best2[fr][reg] = merged2;
+      }
+    }
+  }
+
+  // combine the 4 column-slice waves per row through LDS, then write
+  __syncthreads();
+  float* comb_v = reinterpret_cast<float*>(smem);          // [4][128]
+  int* comb_i = reinterpret_cast<int*>(comb_v + 512);      // [4][128]
+  float* comb_v2 = reinterpret_cast<float*>(comb_i + 512); // [4][128]
+  if ((lane & 15) == 0) {
+#pragma unroll
+    for (int fr = 0; fr < 4; fr++)
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++) {
+        const int rl = wr * 64 + fr * 16 + (lane >> 4) * 4 + reg;  // 0..127
+        comb_v[wc * 128 + rl] = best[fr][reg];
+        comb_i[wc * 128 + rl] = bidx[fr][reg];
+        comb_v2[wc * 128 + rl] = best2[fr][reg];
+      }
+  }
+  __syncthreads();
+  if (wc == 0 && (lane & 15) == 0) {
+#pragma unroll
+    for (int fr = 0; fr < 4; fr++)
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++) {
+        const int rl = wr * 64 + fr * 16 + (lane >> 4) * 4 + reg;
+        float v = comb_v[rl];
+        int vi = comb_i[rl];
+        float s2 = comb_v2[rl];
+        for (int q = 1; q < 4; q++) {
+          const float ov = comb_v[q * 128 + rl];
+          const int oi = comb_i[q * 128 + rl];
+          const float ov2 = comb_v2[q * 128 + rl];
+          s2 = fminf(fminf(s2, ov2), fmaxf(v, ov));
+          if (ov < v || (ov == v && oi < vi)) { v = ov; vi = oi; }
+        }
+        const long long row = row0 + rl;
+        if (row < m) {
+          dmin[row] = fmaxf(v + xn[row], 0.f);
+          amin[row] = vi;
+          if (dmin2) dmin2[row] = s2 + xn[row];
+        }
+      }
+  }
+}
+
+template <int NSLICE>
+static void launch_w8(const __bf16* x0, const __bf16* x1, const __bf16* x2,
+                      const __bf16* c0, const __bf16* c1, const __bf16* c2,
+                      const float* xn, const float* cn, float* dmin, int* amin,
+                      float* dmin2, long long m, int n, int d, hipStream_t stream) {
+  const int grid = (int)((m + 127) / 128);
+  const size_t lds = (size_t)NSLICE * (8192 + 16384) * sizeof(__bf16);
+  static bool attr_set = false;
+  if (!attr_set) {
+    HIP_CHECK(hipFuncSetAttribute((const void*)&fused_l2nn_w8_kernel<NSLICE>,
+                                  hipFuncAttributeMaxDynamicSharedMemorySize,
+                                  160 * 1024));
+    attr_set = true;
+  }
+  hipLaunchKernelGGL((fused_l2nn_w8_kernel<NSLICE>), dim3(grid), dim3(512), lds,
+                     stream, x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2,
+                     m, n, d);
+}
+
+bool fused_l2nn_w8_supported(int nslice, int n, int d) {
+  return n % 256 == 0 && d % 64 == 0 && nslice <= 3;
+}
+
+void launch_fused_l2nn_w8(const void** xsl, const void** csl, const float* xn,
+                          const float* cn, float* dmin, int* amin, float* dmin2,
+                          long long m, int n, int d, int nslice, hipStream_t stream) {
+  const __bf16* x0 = (const __bf16*)xsl[0];
+  const __bf16* x1 = (const __bf16*)(nslice > 1 ? xsl[1] : xsl[0]);
+  const __bf16* x2 = (const __bf16*)(nslice > 2 ? xsl[2] : xsl[0]);
+  const __bf16* c0 = (const __bf16*)csl[0];
+  const __bf16* c1 = (const __bf16*)(nslice > 1 ? csl[1] : csl[0]);
+  const __bf16* c2 = (const __bf16*)(nslice > 2 ? csl[2] : csl[0]);
+  if (nslice == 1)
+    launch_w8<1>(x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d, stream);
+  else if (nslice == 2)
+    launch_w8<2>(x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d, stream);
+  else
+    launch_w8<3>(x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d, stream);
+}
+
 // returns true when a persistent variant covers (nslice, d)
 bool fused_l2nn_persist_supported(int nslice, int d) {
   if (d % 64 != 0) return false;
