@@ -413,7 +413,10 @@ static void launch_w8(const __bf16* x0, const __bf16* x1, const __bf16* x2,
 }
 
 bool fused_l2nn_w8_supported(int nslice, int n, int d) {
-  return n % 256 == 0 && d % 64 == 0 && nslice <= 3;
+  // measured (10M x 256 k=1024): w8 wins only for NSLICE=3 (staging-bound:
+  // halved X re-reads beat the lost cross-block barrier overlap); the 4-wave
+  // 2-block v1 kernel wins for NSLICE<=2 (34.3 vs 38.0 ms/step).
+  return nslice == 3 && n % 256 == 0 && d % 64 == 0;
 }
 
 void launch_fused_l2nn_w8(const void** xsl, const void** csl, const float* xn,
