@@ -114,8 +114,13 @@ def _init_plusplus(x: torch.Tensor, k: int, state: RngState, comms: Comms,
 
 def kmeans_fit(x: torch.Tensor, params: KMeansParams,
                comms: Comms | None = None,
-               init_centroids: torch.Tensor | None = None) -> KMeansModel:
-    """Lloyd EM. `x` is THIS RANK's row shard; pass comms for multi-GPU."""
+               init_centroids: torch.Tensor | None = None,
+               sample_weights: torch.Tensor | None = None) -> KMeansModel:
+    """Lloyd EM. `x` is THIS RANK's row shard; pass comms for multi-GPU.
+
+    sample_weights: optional per-row weights (reference kmeans API parity) —
+    weighted centroid updates and weighted inertia.
+    """
     comms = comms or LoopbackComms()
     state = RngState(seed=params.seed)
     k, (n_local, d) = params.n_clusters, x.shape
@@ -145,9 +150,16 @@ def kmeans_fit(x: torch.Tensor, params: KMeansParams,
             dmin, labels = fused_l2nn_presplit(x_slices, xn, centroids, verify_x=vx)
         else:
             dmin, labels = fused_l2nn(x, centroids, fp32_mode=params.fp32_mode)
-        sums = reduce_rows_by_key(x, labels, n_keys=k)
-        counts = torch.bincount(labels, minlength=k).to(x.dtype)
-        local_inertia = torch.sum(dmin, dtype=torch.float64).to(x.dtype)
+        if sample_weights is None:
+            sums = reduce_rows_by_key(x, labels, n_keys=k)
+            counts = torch.bincount(labels, minlength=k).to(x.dtype)
+            local_inertia = torch.sum(dmin, dtype=torch.float64).to(x.dtype)
+        else:
+            w = sample_weights.to(x.dtype)
+            sums = reduce_rows_by_key(x, labels, n_keys=k, weights=w)
+            counts = torch.zeros(k, dtype=x.dtype, device=x.device)
+            counts.index_add_(0, labels.to(torch.int64), w)
+            local_inertia = torch.sum(dmin * w, dtype=torch.float64).to(x.dtype)
         # ONE packed allreduce: [k, d] sums | [k] counts | [1] inertia
         packed = torch.cat([sums.reshape(-1), counts, local_inertia.reshape(1)])
         if comms.get_size() > 1:

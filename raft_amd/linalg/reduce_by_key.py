@@ -33,6 +33,7 @@ def reduce_rows_by_key(x: torch.Tensor, keys: torch.Tensor, n_keys: int | None =
                                                  perm.to(torch.int32).contiguous(),
                                                  keys_sorted.contiguous(), int(n_keys))
         return ext.reduce_rows_by_key(x.contiguous(), keys.to(torch.int32).contiguous(), int(n_keys))
+    # weighted variant (reference has one): torch index_add over w*x
     out = torch.zeros((n_keys, x.shape[1]), dtype=x.dtype, device=x.device)
     src = x if weights is None else x * weights.unsqueeze(1)
     out.index_add_(0, keys.to(torch.int64), src)

@@ -71,3 +71,21 @@ class TestKMeansBalanced:
         counts = torch.bincount(model.labels, minlength=6)
         assert (counts > 0).all()
         assert model.inertia < 2000 * 5 * 0.3 ** 2 * 5
+
+
+class TestWeightedKMeans:
+    def test_sample_weights_shift_centroids(self):
+        # two tight blobs; heavily weight one point far away in cluster 0
+        x = torch.cat([torch.randn(100, 2) * 0.05,
+                       torch.randn(100, 2) * 0.05 + 10.0])
+        x = torch.cat([x, torch.tensor([[4.0, 0.0]])])
+        w = torch.ones(201)
+        w[200] = 100.0
+        init = torch.tensor([[0.0, 0.0], [10.0, 10.0]])
+        m = kmeans_fit(x, KMeansParams(n_clusters=2, max_iter=10, init="array"),
+                       init_centroids=init, sample_weights=w)
+        # the heavy point drags centroid 0 toward (4, 0)
+        assert m.centroids[0, 0] > 1.0
+        m2 = kmeans_fit(x, KMeansParams(n_clusters=2, max_iter=10, init="array"),
+                        init_centroids=init)
+        assert m2.centroids[0, 0] < m.centroids[0, 0]
