@@ -21,10 +21,16 @@ def test_cpp_smoke(tmp_path):
         if not os.path.exists(o):
             pytest.skip("build/ext objects not present (run build_ext.py)")
     exe = str(tmp_path / "cpp_smoke")
-    cmd = ["hipcc", "--offload-arch=gfx950", "-O2", "-std=c++17",
-           os.path.join(ROOT, "tests", "cpp", "smoke.cpp"), *objs,
-           f"-I{os.path.join(ROOT, 'include')}", "-o", exe]
-    r = subprocess.run(cmd, capture_output=True, timeout=300)
+    obj = str(tmp_path / "smoke.o")
+    # compile then link in two steps (hipcc treats .o inputs as sources when
+    # mixed with .cpp under --offload-arch)
+    r = subprocess.run(["hipcc", "--offload-arch=gfx950", "-O2", "-std=c++17",
+                        "-c", os.path.join(ROOT, "tests", "cpp", "smoke.cpp"),
+                        f"-I{os.path.join(ROOT, 'include')}", "-o", obj],
+                       capture_output=True, timeout=300)
+    assert r.returncode == 0, r.stderr.decode()
+    r = subprocess.run(["hipcc", obj, *objs, "-o", exe],
+                       capture_output=True, timeout=300)
     assert r.returncode == 0, r.stderr.decode()
     r = subprocess.run([exe], capture_output=True, timeout=120)
     assert r.returncode == 0, (r.stdout.decode(), r.stderr.decode())
