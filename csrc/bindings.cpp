@@ -66,6 +66,9 @@ void launch_fused_l2nn_persist(const void**, const void**, const float*, const f
 void launch_pairwise_l2_mfma(const void**, const void**, const float*, const float*,
                              float*, long long, long long, int, long long, int, bool,
                              hipStream_t);
+void launch_pairwise_l2_mfma256(const void**, const void**, const float*, const float*,
+                                float*, long long, long long, int, long long, int, bool,
+                                hipStream_t);
 void launch_pairwise_l2_filter(const void**, const void**, const float*, const float*,
                                const float*, float*, int*, int*, int, long long,
                                long long, long long, int, int, hipStream_t);
@@ -381,9 +384,21 @@ torch::Tensor pairwise_l2_mfma(std::vector<torch::Tensor> x_slices,
   } else {
     o = torch::empty({m, n}, xn.options());
   }
-  raft_amd::launch_pairwise_l2_mfma(xsl, csl, xn.data_ptr<float>(), yn.data_ptr<float>(),
-                                    o.data_ptr<float>(), m, n, (int)d, o.size(1),
-                                    nslice, sqrt_out, cur_stream());
+  static const bool no_256 = [] {
+    const char* e = getenv("RAFT_AMD_PW256");
+    return e && e[0] == '0';
+  }();
+  if (!no_256 && nslice <= 2 && m >= 512 && n >= 512) {
+    // 256x256-tile kernel: 4x fewer workgroups (first-order at small d)
+    raft_amd::launch_pairwise_l2_mfma256(xsl, csl, xn.data_ptr<float>(),
+                                         yn.data_ptr<float>(), o.data_ptr<float>(),
+                                         m, n, (int)d, o.size(1), nslice, sqrt_out,
+                                         cur_stream());
+  } else {
+    raft_amd::launch_pairwise_l2_mfma(xsl, csl, xn.data_ptr<float>(), yn.data_ptr<float>(),
+                                      o.data_ptr<float>(), m, n, (int)d, o.size(1),
+                                      nslice, sqrt_out, cur_stream());
+  }
   return o;
 }
 
