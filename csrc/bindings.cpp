@@ -15,6 +15,7 @@ void launch_reduce_rows(const T*, T*, long long, long long, hipStream_t);
 template <int OP, typename T>
 void launch_reduce_cols(const T*, T*, long long, long long, hipStream_t);
 void launch_row_argmin(const float*, int*, long long, long long, hipStream_t);
+void launch_rows_sqnorm_bf16(const void*, float*, long long, long long, hipStream_t);
 void launch_row_normalize_l2(const float*, float*, long long, long long, float, hipStream_t);
 // from pairwise.hip
 void launch_l2_epilogue(float*, const float*, const float*, long long, long long, hipStream_t);
@@ -126,6 +127,15 @@ torch::Tensor reduce_cols(torch::Tensor x, int64_t op) {
   switch (op) { CASE(0) CASE(1) CASE(2) CASE(3) CASE(4) CASE(5)
     default: TORCH_CHECK(false, "bad op code"); }
 #undef CASE
+  return out;
+}
+
+torch::Tensor rows_sqnorm_bf16(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16 && x.is_contiguous()
+              && x.dim() == 2);
+  auto out = torch::empty({x.size(0)}, x.options().dtype(torch::kFloat32));
+  raft_amd::launch_rows_sqnorm_bf16(x.data_ptr(), out.data_ptr<float>(), x.size(0),
+                                    x.size(1), cur_stream());
   return out;
 }
 
@@ -502,6 +512,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reduce_rows", &reduce_rows, "rowwise reduction (op code)");
   m.def("reduce_cols", &reduce_cols, "columnwise reduction (op code)");
   m.def("row_argmin", &row_argmin, "rowwise argmin");
+  m.def("rows_sqnorm_bf16", &rows_sqnorm_bf16, "bf16 row squared norms -> f32");
   m.def("row_normalize_l2", &row_normalize_l2, "fused L2 row normalize");
   m.def("l2_epilogue_", &l2_epilogue_, "in-place L2 distance epilogue");
   m.def("l2nn_epilogue", &l2nn_epilogue, "fused argmin epilogue over GEMM tile");

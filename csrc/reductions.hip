@@ -259,6 +259,51 @@ void launch_reduce_cols(const T* x, T* out, long long n_rows, long long d,
 INSTANTIATE_OPS(float)
 INSTANTIATE_OPS(double)
 
+// bf16 row squared-norms -> f32 (vectorized bf16x8 loads; avoids the 2x-size
+// fp32 materialization a torch x.float().pow(2).sum(1) chain would need)
+__global__ void rows_sqnorm_bf16_kernel(const __bf16* __restrict__ x,
+                                        float* __restrict__ out,
+                                        long long n_rows, long long d) {
+  typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
+  constexpr int BLOCK = 256;
+  __shared__ float lds[BLOCK / RAFT_AMD_WAVE];
+  for (long long row = blockIdx.x; row < n_rows; row += gridDim.x) {
+    const __bf16* rp = x + row * d;
+    float acc = 0.f;
+    const long long d8 = d / 8;
+    const bf16x8_t* rp8 = reinterpret_cast<const bf16x8_t*>(rp);
+    for (long long j = threadIdx.x; j < d8; j += BLOCK) {
+      const bf16x8_t v = rp8[j];
+#pragma unroll
+      for (int e = 0; e < 8; e++) {
+        const float f = (float)v[e];
+        acc += f * f;
+      }
+    }
+    for (long long j = d8 * 8 + threadIdx.x; j < d; j += BLOCK) {
+      const float f = (float)rp[j];
+      acc += f * f;
+    }
+    acc = wave_reduce_sum(acc);
+    const int wid = threadIdx.x / RAFT_AMD_WAVE, lane = threadIdx.x % RAFT_AMD_WAVE;
+    if (lane == 0) lds[wid] = acc;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      float s = 0.f;
+      for (int w = 0; w < BLOCK / RAFT_AMD_WAVE; w++) s += lds[w];
+      out[row] = s;
+    }
+    __syncthreads();
+  }
+}
+
+void launch_rows_sqnorm_bf16(const void* x, float* out, long long n_rows,
+                             long long d, hipStream_t stream) {
+  long long blocks = n_rows < 65536 ? n_rows : 65536;
+  hipLaunchKernelGGL(rows_sqnorm_bf16_kernel, dim3((unsigned)blocks), dim3(256), 0,
+                     stream, (const __bf16*)x, out, n_rows, d);
+}
+
 void launch_row_argmin(const float* x, int* out, long long n_rows, long long d,
                        hipStream_t stream) {
   int grid = (int)(n_rows < 2048 ? n_rows : 2048);

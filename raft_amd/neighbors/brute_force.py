@@ -47,6 +47,12 @@ def _slices_of(t: torch.Tensor, fp32_mode: str):
 
 
 def _norms(t: torch.Tensor) -> torch.Tensor:
+    if t.is_cuda and t.dtype == torch.bfloat16:
+        # native kernel: a torch float() chain would materialize a 2x-size
+        # fp32 copy (51 GB for the 100M-row index — allocation-bound)
+        return require_ext().rows_sqnorm_bf16(t.contiguous())
+    if t.is_cuda and t.dtype == torch.float32:
+        return require_ext().reduce_rows(t.contiguous(), 1)
     return t.float().pow(2).sum(dim=1).contiguous()
 
 
