@@ -73,6 +73,9 @@ void launch_pairwise_l2_mfma256(const void**, const void**, const float*, const 
 void launch_pairwise_l2_filter(const void**, const void**, const float*, const float*,
                                const float*, float*, int*, int*, int, long long,
                                long long, long long, int, int, hipStream_t);
+void launch_pairwise_l2_filter256(const void**, const void**, const float*, const float*,
+                                  const float*, float*, int*, int*, int, long long,
+                                  long long, long long, int, int, hipStream_t);
 // from gemm_rocblas.cpp
 void gemm_bf16_f32_rowmajor(const void*, const void*, float*, long long, long long,
                             long long, float, void*);
@@ -448,11 +451,19 @@ void pairwise_l2_filter(std::vector<torch::Tensor> x_slices,
   const int cap = (int)out_d.size(1);
   TORCH_CHECK(out_d.size(0) == m && out_i.sizes() == out_d.sizes());
   TORCH_CHECK(cnt.scalar_type() == torch::kInt32 && cnt.numel() == m);
-  raft_amd::launch_pairwise_l2_filter(xsl, csl, xn.data_ptr<float>(),
-                                      yn.data_ptr<float>(), thr.data_ptr<float>(),
-                                      out_d.data_ptr<float>(), out_i.data_ptr<int>(),
-                                      cnt.data_ptr<int>(), cap, col_offset, m, n,
-                                      (int)d, nslice, cur_stream());
+  if (nslice <= 2 && m >= 512 && n >= 512) {
+    raft_amd::launch_pairwise_l2_filter256(xsl, csl, xn.data_ptr<float>(),
+                                           yn.data_ptr<float>(), thr.data_ptr<float>(),
+                                           out_d.data_ptr<float>(), out_i.data_ptr<int>(),
+                                           cnt.data_ptr<int>(), cap, col_offset, m, n,
+                                           (int)d, nslice, cur_stream());
+  } else {
+    raft_amd::launch_pairwise_l2_filter(xsl, csl, xn.data_ptr<float>(),
+                                        yn.data_ptr<float>(), thr.data_ptr<float>(),
+                                        out_d.data_ptr<float>(), out_i.data_ptr<int>(),
+                                        cnt.data_ptr<int>(), cap, col_offset, m, n,
+                                        (int)d, nslice, cur_stream());
+  }
 }
 
 torch::Tensor gemm_bf16_f32(torch::Tensor a, torch::Tensor b,

@@ -406,3 +406,156 @@ void launch_pairwise_l2_mfma256(const void** xsl, const void** csl, const float*
 }
 
 }  // namespace raft_amd
+
+namespace raft_amd {
+
+// 256x256-tile filtered emission (see pairwise_l2_256_kernel geometry +
+// pairwise_l2_filter_kernel semantics)
+template <int NSLICE>
+__launch_bounds__(512, 2)
+__global__ void pairwise_l2_filter256_kernel(const __bf16* __restrict__ x0,
+                                             const __bf16* __restrict__ x1,
+                                             const __bf16* __restrict__ x2,
+                                             const __bf16* __restrict__ c0,
+                                             const __bf16* __restrict__ c1,
+                                             const __bf16* __restrict__ c2,
+                                             const float* __restrict__ xn,
+                                             const float* __restrict__ yn,
+                                             const float* __restrict__ thr,
+                                             float* __restrict__ out_d,
+                                             int* __restrict__ out_i,
+                                             int* __restrict__ cnt, int cap,
+                                             long long col_offset, long long m,
+                                             long long n, int d) {
+  constexpr int BLOCK = 512;
+  extern __shared__ __bf16 smem[];
+  const __bf16* const xg[3] = {x0, x1, x2};
+  const __bf16* const cg[3] = {c0, c1, c2};
+  __bf16* xs[NSLICE];
+  __bf16* cs[NSLICE];
+#pragma unroll
+  for (int s = 0; s < NSLICE; s++) {
+    xs[s] = smem + s * 16384;
+    cs[s] = smem + NSLICE * 16384 + s * 16384;
+  }
+  const int lane = threadIdx.x % RAFT_AMD_WAVE;
+  const int w = threadIdx.x / RAFT_AMD_WAVE;
+  const int wr = w >> 2, wc = w & 3;
+  const long long row0 = (long long)blockIdx.y * 256;
+  const long long col0 = (long long)blockIdx.x * 256;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int a = 0; a < 8; a++)
+#pragma unroll
+    for (int b = 0; b < 4; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int k_tiles = d / 64;
+  for (int kt = 0; kt < k_tiles; kt++) {
+#pragma unroll
+    for (int s = 0; s < NSLICE; s++) {
+      pw_stage_rows<256, BLOCK>(xg[s], xs[s], row0, (long long)kt * 64, d, m - 1);
+      pw_stage_rows<256, BLOCK>(cg[s], cs[s], col0, (long long)kt * 64, d, n - 1);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+#pragma unroll
+    for (int kf = 0; kf < 2; kf++) {
+      bf16x8 a_frag[NSLICE][8], b_frag[NSLICE][4];
+      const int kbyte = (kf * 32 + (lane >> 4) * 8) * 2;
+#pragma unroll
+      for (int fr = 0; fr < 8; fr++) {
+        const int r = wr * 128 + fr * 16 + (lane & 15);
+        const int byte = mfma_swz(r * 128 + kbyte);
+#pragma unroll
+        for (int s = 0; s < NSLICE; s++)
+          a_frag[s][fr] = *reinterpret_cast<const bf16x8*>((const char*)xs[s] + byte);
+      }
+#pragma unroll
+      for (int fc = 0; fc < 4; fc++) {
+        const int c = wc * 64 + fc * 16 + (lane & 15);
+        const int byte = mfma_swz(c * 128 + kbyte);
+#pragma unroll
+        for (int s = 0; s < NSLICE; s++)
+          b_frag[s][fc] = *reinterpret_cast<const bf16x8*>((const char*)cs[s] + byte);
+      }
+#pragma unroll
+      for (int fr = 0; fr < 8; fr++)
+#pragma unroll
+        for (int fc = 0; fc < 4; fc++) {
+#pragma unroll
+          for (int p = 0; p < mfma_n_products<NSLICE>(); p++) {
+            acc[fr][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_frag[MFMA_PROD_A[p]][fr], b_frag[MFMA_PROD_B[p]][fc],
+                acc[fr][fc], 0, 0, 0);
+          }
+        }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int fr = 0; fr < 8; fr++) {
+#pragma unroll
+    for (int reg = 0; reg < 4; reg++) {
+      const long long row = row0 + wr * 128 + fr * 16 + (lane >> 4) * 4 + reg;
+      if (row >= m) continue;
+      const float xv = xn[row];
+      const float t = thr[row];
+#pragma unroll
+      for (int fc = 0; fc < 4; fc++) {
+        const long long col = col0 + wc * 64 + fc * 16 + (lane & 15);
+        if (col < n) {
+          const float d2 = fmaxf(xv + yn[col] - 2.f * acc[fr][fc][reg], 0.f);
+          if (d2 <= t) {
+            const int pos = atomicAdd(&cnt[row], 1);
+            if (pos < cap) {
+              out_d[row * cap + pos] = d2;
+              out_i[row * cap + pos] = (int)(col + col_offset);
+            }
+          }
+        }
+      }
+    }
+  }
+}
+
+void launch_pairwise_l2_filter256(const void** xsl, const void** csl, const float* xn,
+                                  const float* yn, const float* thr, float* out_d,
+                                  int* out_i, int* cnt, int cap, long long col_offset,
+                                  long long m, long long n, int d, int nslice,
+                                  hipStream_t stream) {
+  dim3 grid((unsigned)((n + 255) / 256), (unsigned)((m + 255) / 256));
+  const size_t lds = (size_t)nslice * 2 * 16384 * sizeof(__bf16);
+  const __bf16* x0 = (const __bf16*)xsl[0];
+  const __bf16* x1 = (const __bf16*)(nslice > 1 ? xsl[1] : xsl[0]);
+  const __bf16* x2 = (const __bf16*)(nslice > 2 ? xsl[2] : xsl[0]);
+  const __bf16* c0 = (const __bf16*)csl[0];
+  const __bf16* c1 = (const __bf16*)(nslice > 1 ? csl[1] : csl[0]);
+  const __bf16* c2 = (const __bf16*)(nslice > 2 ? csl[2] : csl[0]);
+  auto set_attr = [](const void* f) {
+    HIP_CHECK(hipFuncSetAttribute(f, hipFuncAttributeMaxDynamicSharedMemorySize,
+                                  160 * 1024));
+  };
+  if (nslice == 1) {
+    static bool a1 = (set_attr((const void*)&pairwise_l2_filter256_kernel<1>), true);
+    (void)a1;
+    hipLaunchKernelGGL((pairwise_l2_filter256_kernel<1>), grid, dim3(512), lds, stream,
+                       x0, x1, x2, c0, c1, c2, xn, yn, thr, out_d, out_i, cnt, cap,
+                       col_offset, m, n, d);
+  } else if (nslice == 2) {
+    static bool a2 = (set_attr((const void*)&pairwise_l2_filter256_kernel<2>), true);
+    (void)a2;
+    hipLaunchKernelGGL((pairwise_l2_filter256_kernel<2>), grid, dim3(512), lds, stream,
+                       x0, x1, x2, c0, c1, c2, xn, yn, thr, out_d, out_i, cnt, cap,
+                       col_offset, m, n, d);
+  } else {
+    static bool a3 = (set_attr((const void*)&pairwise_l2_filter256_kernel<3>), true);
+    (void)a3;
+    hipLaunchKernelGGL((pairwise_l2_filter256_kernel<3>), grid, dim3(512), lds, stream,
+                       x0, x1, x2, c0, c1, c2, xn, yn, thr, out_d, out_i, cnt, cap,
+                       col_offset, m, n, d);
+  }
+}
+
+}  // namespace raft_amd
