@@ -451,7 +451,11 @@ void pairwise_l2_filter(std::vector<torch::Tensor> x_slices,
   const int cap = (int)out_d.size(1);
   TORCH_CHECK(out_d.size(0) == m && out_i.sizes() == out_d.sizes());
   TORCH_CHECK(cnt.scalar_type() == torch::kInt32 && cnt.numel() == m);
-  if (nslice <= 2 && m >= 512 && n >= 512) {
+  static const bool no_f256 = [] {
+    const char* e = getenv("RAFT_AMD_FILTER256");
+    return e && e[0] == '0';
+  }();
+  if (!no_f256 && nslice <= 2 && m >= 512 && n >= 512) {
     raft_amd::launch_pairwise_l2_filter256(xsl, csl, xn.data_ptr<float>(),
                                            yn.data_ptr<float>(), thr.data_ptr<float>(),
                                            out_d.data_ptr<float>(), out_i.data_ptr<int>(),
