@@ -92,45 +92,53 @@ __global__ void fused_l2nn_kernel(const __bf16* __restrict__ x0,
                               m - 1, n - 1, wr, wc, lane);
     }
 
-    // epilogue: fold this tile's 64 columns-per-wave into the running
-    // (best, second-best) pair — second-best feeds the exact-argmin
-    // verification margin (see fused_l2nn.py verify path).
+    // epilogue: fold this tile's columns into each lane's LOCAL running
+    // (best, second-best) — lanes own disjoint column sets, so no cross-lane
+    // work is needed per tile; ONE shuffle reduce happens after the nt loop
+    // (the per-tile 12-shuffle reduce was ~1/3 of the kernel's non-MFMA time).
     const int col_base = nt * 128 + wc * 64;
 #pragma unroll
     for (int fr = 0; fr < 4; fr++) {
 #pragma unroll
       for (int reg = 0; reg < 4; reg++) {
-        float v = INFINITY, v2 = INFINITY;
-        int vi = 0;
 #pragma unroll
         for (int fc = 0; fc < 4; fc++) {
           const int col = col_base + fc * 16 + (lane & 15);
           const float s = cn[col] - 2.f * acc[fr][fc][reg];
-          if (s < v) { v2 = v; v = s; vi = col; }
-          else if (s < v2) { v2 = s; }
+          if (s < best[fr][reg]) {
+            best2[fr][reg] = best[fr][reg];
+            best[fr][reg] = s;
+            bidx[fr][reg] = col;
+          } else if (s < best2[fr][reg]) {
+            best2[fr][reg] = s;
+          }
         }
-#pragma unroll
-        for (int off = 8; off > 0; off >>= 1) {
-          const float ov = __shfl_xor(v, off, RAFT_AMD_WAVE);
-          const float ov2 = __shfl_xor(v2, off, RAFT_AMD_WAVE);
-          const int oi = __shfl_xor(vi, off, RAFT_AMD_WAVE);
-          // top-2 merge of two candidate SETS: when both sets share the same
-          // best element (same index — happens after the first exchange),
-          // the loser of the bests is NOT a second-best candidate
-          float new2 = fminf(v2, ov2);
-          if (oi != vi) new2 = fminf(new2, fmaxf(v, ov));
-          v2 = new2;
-          if (ov < v || (ov == v && oi < vi)) { v = ov; vi = oi; }
-        }
-        // cross-tile merge: disjoint column sets -> plain top-2 merge
-        const float hi = fmaxf(best[fr][reg], v);
-        const float merged2 = fminf(fminf(best2[fr][reg], v2), hi);
-        if (v < best[fr][reg] || (v == best[fr][reg] && vi < bidx[fr][reg])) {
-          best[fr][reg] = v;
-          bidx[fr][reg] = vi;
-        }
-        best2[fr][reg] = merged2;
       }
+    }
+  }
+
+  // ONE cross-lane top-2 reduce per (fr, reg) over the 16 lanes of the
+  // row-group (index-aware merge: shared-best lanes cannot donate their
+  // loser as a second-best candidate)
+#pragma unroll
+  for (int fr = 0; fr < 4; fr++) {
+#pragma unroll
+    for (int reg = 0; reg < 4; reg++) {
+      float v = best[fr][reg], v2 = best2[fr][reg];
+      int vi = bidx[fr][reg];
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1) {
+        const float ov = __shfl_xor(v, off, RAFT_AMD_WAVE);
+        const float ov2 = __shfl_xor(v2, off, RAFT_AMD_WAVE);
+        const int oi = __shfl_xor(vi, off, RAFT_AMD_WAVE);
+        float new2 = fminf(v2, ov2);
+        if (oi != vi) new2 = fminf(new2, fmaxf(v, ov));
+        v2 = new2;
+        if (ov < v || (ov == v && oi < vi)) { v = ov; vi = oi; }
+      }
+      best[fr][reg] = v;
+      best2[fr][reg] = v2;
+      bidx[fr][reg] = vi;
     }
   }
 

@@ -320,33 +320,42 @@ __global__ void fused_l2nn_w8_kernel(
     for (int fr = 0; fr < 4; fr++) {
 #pragma unroll
       for (int reg = 0; reg < 4; reg++) {
-        float v = INFINITY, v2 = INFINITY;
-        int vi = 0;
 #pragma unroll
         for (int fc = 0; fc < 4; fc++) {
           const int col = col_base + fc * 16 + (lane & 15);
           const float s = cn[col] - 2.f * acc[fr][fc][reg];
-          if (s < v) { v2 = v; v = s; vi = col; }
-          else if (s < v2) { v2 = s; }
+          if (s < best[fr][reg]) {
+            best2[fr][reg] = best[fr][reg];
+            best[fr][reg] = s;
+            bidx[fr][reg] = col;
+          } else if (s < best2[fr][reg]) {
+            best2[fr][reg] = s;
+          }
         }
-#pragma unroll
-        for (int off = 8; off > 0; off >>= 1) {
-          const float ov = __shfl_xor(v, off, RAFT_AMD_WAVE);
-          const float ov2 = __shfl_xor(v2, off, RAFT_AMD_WAVE);
-          const int oi = __shfl_xor(vi, off, RAFT_AMD_WAVE);
-          float new2 = fminf(v2, ov2);
-          if (oi != vi) new2 = fminf(new2, fmaxf(v, ov));
-          v2 = new2;
-          if (ov < v || (ov == v && oi < vi)) { v = ov; vi = oi; }
-        }
-        const float hi = fmaxf(best[fr][reg], v);
-        const float merged2 = fminf(fminf(best2[fr][reg], v2), hi);
-        if (v < best[fr][reg] || (v == best[fr][reg] && vi < bidx[fr][reg])) {
-          best[fr][reg] = v;
-          bidx[fr][reg] = vi;
-        }
-        best2[fr][reg] = merged2;
       }
+    }
+  }
+
+  // ONE cross-lane top-2 reduce after the nt loop (lane column sets disjoint)
+#pragma unroll
+  for (int fr = 0; fr < 4; fr++) {
+#pragma unroll
+    for (int reg = 0; reg < 4; reg++) {
+      float v = best[fr][reg], v2 = best2[fr][reg];
+      int vi = bidx[fr][reg];
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1) {
+        const float ov = __shfl_xor(v, off, RAFT_AMD_WAVE);
+        const float ov2 = __shfl_xor(v2, off, RAFT_AMD_WAVE);
+        const int oi = __shfl_xor(vi, off, RAFT_AMD_WAVE);
+        float new2 = fminf(v2, ov2);
+        if (oi != vi) new2 = fminf(new2, fmaxf(v, ov));
+        v2 = new2;
+        if (ov < v || (ov == v && oi < vi)) { v = ov; vi = oi; }
+      }
+      best[fr][reg] = v;
+      best2[fr][reg] = v2;
+      bidx[fr][reg] = vi;
     }
   }
 
