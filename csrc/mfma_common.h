@@ -52,6 +52,9 @@ __device__ constexpr int MFMA_PROD_B[6] = {0, 1, 0, 1, 2, 0};
 
 // Run the K loop for one 128x128 tile pair: stages slices, MFMAs into acc.
 // acc is the per-wave [4][4] fragment grid (wave (wr,wc) of a 2x2 wave grid).
+// Staging source addresses are hoisted: the per-thread (row, kcol) of each
+// gload round is K-invariant, so the 64-bit row*ld multiplies run once per
+// tile pair instead of once per K-step.
 template <int NSLICE>
 __device__ __forceinline__ void mfma_tile_kloop(
     const __bf16* const (&xg)[3], const __bf16* const (&cg)[3],
@@ -59,11 +62,33 @@ __device__ __forceinline__ void mfma_tile_kloop(
     f32x4 (&acc)[4][4], long long row0, long long col0, int d,
     long long m_max, long long n_max, int wr, int wc, int lane) {
   const int k_tiles = d / 64;
+  const int t = threadIdx.x;
+  const int wv = t / RAFT_AMD_WAVE;
+  long long bx[4], bc[4];
+  int ldst[4];
+#pragma unroll
+  for (int j = 0; j < 4; j++) {
+    const int o = j * 4096 + t * 16;
+    const int o_src = mfma_swz(o);
+    const int r = o_src >> 7;
+    const int k = (o_src & 127) >> 1;
+    long long rx = row0 + r;
+    if (rx > m_max) rx = m_max;
+    bx[j] = rx * (long long)d + k;
+    long long rc = col0 + r;
+    if (rc > n_max) rc = n_max;
+    bc[j] = rc * (long long)d + k;
+    ldst[j] = (j * 4096 + wv * 1024) / 2;
+  }
   for (int kt = 0; kt < k_tiles; kt++) {
+    const long long koff = (long long)kt * 64;
 #pragma unroll
     for (int s = 0; s < NSLICE; s++) {
-      mfma_stage_tile128(xg[s], xs[s], row0, (long long)kt * 64, d, m_max);
-      mfma_stage_tile128(cg[s], cs[s], col0, (long long)kt * 64, d, n_max);
+#pragma unroll
+      for (int j = 0; j < 4; j++) {
+        GLOAD_LDS(xg[s] + bx[j] + koff, xs[s] + ldst[j]);
+        GLOAD_LDS(cg[s] + bc[j] + koff, cs[s] + ldst[j]);
+      }
     }
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
