@@ -357,7 +357,9 @@ __global__ void pairwise_l2_256_kernel(const __bf16* __restrict__ x0,
         if (col < n) {
           float v = fmaxf(xv + yn[col] - 2.f * acc[fr][fc][reg], 0.f);
           if (sqrt_out) v = sqrtf(v);
-          out[row * ldo + col] = v;
+          // streaming output (never re-read): non-temporal keeps the L2
+          // clear for the operand tiles
+          __builtin_nontemporal_store(v, &out[row * ldo + col]);
         }
       }
     }
