@@ -4,7 +4,7 @@ Reference parity: raft/matrix/*  — most notably the select_k
 engine (radix + warpsort) that the reference's ANN stack is built on.
 """
 from .select_k import select_k, SelectAlgo
-from .gather import gather, gather_if, scatter
+from .gather import gather, gather_if, scatter, gather_inplace, scatter_inplace
 from .argminmax import argmax, argmin
 from .ops import (
     slice_matrix, get_diagonal, set_diagonal, upper_triangular, lower_triangular,
@@ -18,6 +18,7 @@ from .print import print_matrix
 
 __all__ = [
     "select_k", "SelectAlgo", "gather", "gather_if", "scatter",
+    "gather_inplace", "scatter_inplace",
     "argmax", "argmin", "slice_matrix", "get_diagonal", "set_diagonal",
     "upper_triangular", "lower_triangular", "row_reverse", "col_reverse",
     "shift_rows", "eye", "power", "ratio", "reciprocal", "matrix_sqrt",

@@ -139,3 +139,22 @@ class TestKnnGpu:
         ri = torch.topk(ref, 10, dim=1, largest=False).indices
         hits = (i.unsqueeze(2) == ri.unsqueeze(1)).any(2).float().mean()
         assert hits > 0.95
+
+
+class TestKMeansEndToEnd:
+    def test_full_fit_kmeanspp_quality(self, dev):
+        """end-to-end kmeans_fit (init + EM + convergence) at medium scale."""
+        from raft_amd.cluster import kmeans_fit, KMeansParams
+        from raft_amd.random import make_blobs, RngState
+        x, _, centers = make_blobs(200000, 128, n_clusters=64, cluster_std=0.5,
+                                   center_box=(-25, 25), state=RngState(seed=13),
+                                   device=dev)
+        m = kmeans_fit(x, KMeansParams(n_clusters=64, max_iter=25, seed=2,
+                                       init="kmeans++"))
+        d = torch.cdist(centers, m.centroids)
+        assert float(d.min(dim=1).values.max()) < 2.0
+        # weighted fit API also runs on GPU
+        w = torch.rand(200000, device=dev) + 0.5
+        m2 = kmeans_fit(x, KMeansParams(n_clusters=64, max_iter=5, seed=2,
+                                        init="random"), sample_weights=w)
+        assert torch.isfinite(m2.centroids).all()

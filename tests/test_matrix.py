@@ -88,3 +88,14 @@ class TestSampleRows:
         # every sampled row exists in x
         d = (out.unsqueeze(1) - x.unsqueeze(0)).abs().sum(-1).min(dim=1).values
         assert (d < 1e-6).all()
+
+
+class TestInplace:
+    def test_gather_scatter_inplace(self):
+        from raft_amd.matrix import gather_inplace, scatter_inplace
+        x = torch.arange(12, dtype=torch.float32).reshape(4, 3)
+        perm = torch.tensor([2, 0, 3, 1])
+        g = gather_inplace(x.clone(), perm)
+        assert torch.equal(g, torch.arange(12, dtype=torch.float32).reshape(4, 3)[perm])
+        s = scatter_inplace(x.clone(), perm)
+        assert torch.equal(s[perm], torch.arange(12, dtype=torch.float32).reshape(4, 3))
