@@ -44,6 +44,10 @@ void launch_kmeans_update_centroids(const float*, const float*, float*, long lon
 void launch_l2nn_verify_repair(const float*, const float*, const float*, float*, int*,
                                const float*, const float*, long long, int, int,
                                hipStream_t);
+void launch_kmeans_update_verify(const float*, const int*, const int*, const float*,
+                                 const float*, float*, int*, const float*,
+                                 const float*, float*, float*, long long, long long,
+                                 int, hipStream_t);
 // from select_k.hip
 long long select_k_workspace_bytes(long long batch);
 void launch_select_k(const float*, float*, int*, void*, long long, long long, int,
@@ -270,6 +274,23 @@ void split_bf16_norms(torch::Tensor c, std::vector<torch::Tensor> slices,
   raft_amd::launch_split_bf16_norms(c.data_ptr<float>(), p0, p1, p2,
                                     cn.data_ptr<float>(), nslice, c.size(0),
                                     c.size(1), cur_stream());
+}
+
+void kmeans_update_verify(torch::Tensor x, torch::Tensor perm,
+                          torch::Tensor keys_sorted, torch::Tensor c,
+                          torch::Tensor xn, torch::Tensor dmin, torch::Tensor amin,
+                          torch::Tensor dmin2, torch::Tensor cn_max,
+                          torch::Tensor sums, torch::Tensor counts) {
+  check_f32_2d(x, "x");
+  check_f32_2d(c, "c");
+  TORCH_CHECK(perm.scalar_type() == torch::kInt32 && keys_sorted.scalar_type() == torch::kInt32);
+  TORCH_CHECK(sums.is_contiguous() && counts.is_contiguous());
+  raft_amd::launch_kmeans_update_verify(
+      x.data_ptr<float>(), perm.data_ptr<int>(), keys_sorted.data_ptr<int>(),
+      c.data_ptr<float>(), xn.data_ptr<float>(), dmin.data_ptr<float>(),
+      amin.data_ptr<int>(), dmin2.data_ptr<float>(), cn_max.data_ptr<float>(),
+      sums.data_ptr<float>(), counts.data_ptr<float>(), x.size(0), x.size(1),
+      (int)c.size(0), cur_stream());
 }
 
 void kmeans_update_centroids(torch::Tensor sums, torch::Tensor counts,
@@ -544,6 +565,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "keyed row accumulation + counts into caller buffers");
   m.def("split_bf16_norms", &split_bf16_norms,
         "fused fp32->bf16 slice split + row sq-norms");
+  m.def("kmeans_update_verify", &kmeans_update_verify,
+        "fused centroid-sum accumulation + exact-fp32 verify/refine (one X pass)");
   m.def("kmeans_update_centroids", &kmeans_update_centroids,
         "centroids = counts>0 ? sums/counts : centroids");
   m.def("select_k", &select_k, "batched top-k (radix)");

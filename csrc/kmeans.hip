@@ -97,6 +97,148 @@ __global__ void reduce_rows_by_key_sorted_kernel(const float* __restrict__ x,
   if (counts && lane == 0) atomicAdd(&counts[cur_key], (float)run_len);
 }
 
+// Fused centroid-update + exact-fp32 verify/refine in ONE X pass.
+// Rows arrive key-sorted, so the chosen centroid row is run-constant and
+// lives in registers; each row's x fragment is loaded once and used for BOTH
+// the exact distance refinement and the centroid-sum accumulation. Rows
+// inside the split-error margin rescan all centroids exactly (rare); a
+// rescan that CHANGES the assignment routes that row's contribution to the
+// new cluster via direct atomics (and fixes amin/dmin).
+template <int MAX_DREG>
+__global__ void kmeans_update_verify_kernel(
+    const float* __restrict__ x, const int* __restrict__ perm,
+    const int* __restrict__ keys_sorted, const float* __restrict__ c,
+    const float* __restrict__ xn, float* __restrict__ dmin,
+    int* __restrict__ amin, const float* __restrict__ dmin2,
+    const float* __restrict__ cn_max_p, float* __restrict__ sums,
+    float* __restrict__ counts, long long n_rows, long long d, int n_centroids,
+    long long chunk) {
+  const int lane = threadIdx.x % RAFT_AMD_WAVE;
+  const long long wave_id =
+      ((long long)blockIdx.x * blockDim.x + threadIdx.x) / RAFT_AMD_WAVE;
+  const long long start = wave_id * chunk;
+  if (start >= n_rows) return;
+  const long long stop = min(start + chunk, n_rows);
+  const float cn_max = *cn_max_p;
+
+  float acc[MAX_DREG], creg[MAX_DREG], xv[MAX_DREG];
+#pragma unroll
+  for (int j = 0; j < MAX_DREG; j++) acc[j] = 0.f;
+  int cur_key = -1;
+  int run_len = 0;
+
+  auto flush = [&]() {
+    if (cur_key < 0) return;
+    float* sp = sums + (long long)cur_key * d;
+    #pragma unroll
+    for (int j = 0; j < MAX_DREG; j++) {
+      const long long col = (long long)j * RAFT_AMD_WAVE + lane;
+      if (col < d) atomicAdd(&sp[col], acc[j]);
+      acc[j] = 0.f;
+    }
+    if (lane == 0 && run_len) atomicAdd(&counts[cur_key], (float)run_len);
+    run_len = 0;
+  };
+
+  for (long long i = start; i < stop; i++) {
+    const long long row = perm[i];
+    const int key = keys_sorted[i];
+    if (key != cur_key) {
+      flush();
+      cur_key = key;
+      const float* cp = c + (long long)key * d;
+      #pragma unroll
+    for (int j = 0; j < MAX_DREG; j++) {
+        const long long col = (long long)j * RAFT_AMD_WAVE + lane;
+        creg[j] = col < d ? cp[col] : 0.f;
+      }
+    }
+    const float* rp = x + row * d;
+    #pragma unroll
+    for (int j = 0; j < MAX_DREG; j++) {
+      const long long col = (long long)j * RAFT_AMD_WAVE + lane;
+      xv[j] = col < d ? rp[col] : 0.f;
+    }
+    const float xnr = xn[row];
+    const float margin = dmin2[row] - dmin[row];
+    const float bound = 2.f * (exp2f(-13.f) * sqrtf(fmaxf(xnr * cn_max, 0.f)) +
+                               exp2f(-18.f) * (xnr + cn_max));
+    if (margin < bound) {
+      // exact rescan over all centroids (rare: near-ties only)
+      float bestv = INFINITY;
+      int besti = 0;
+      for (int jc = 0; jc < n_centroids; jc++) {
+        const float* cp = c + (long long)jc * d;
+        float a = 0.f;
+        #pragma unroll
+    for (int j = 0; j < MAX_DREG; j++) {
+          const long long col = (long long)j * RAFT_AMD_WAVE + lane;
+          if (col < d) {
+            const float diff = xv[j] - cp[col];
+            a += diff * diff;
+          }
+        }
+        a = wave_reduce_sum(a);
+        if (a < bestv) { bestv = a; besti = jc; }
+      }
+      if (lane == 0) {
+        dmin[row] = bestv;
+        amin[row] = besti;
+      }
+      if (besti == cur_key) {
+        #pragma unroll
+    for (int j = 0; j < MAX_DREG; j++) acc[j] += xv[j];
+        run_len++;
+      } else {
+        // reassigned: direct atomics into the new cluster
+        float* sp = sums + (long long)besti * d;
+        #pragma unroll
+    for (int j = 0; j < MAX_DREG; j++) {
+          const long long col = (long long)j * RAFT_AMD_WAVE + lane;
+          if (col < d) atomicAdd(&sp[col], xv[j]);
+        }
+        if (lane == 0) atomicAdd(&counts[besti], 1.f);
+      }
+    } else {
+      // exact fp32 refinement of the chosen distance (creg is run-resident)
+      float a = 0.f;
+      #pragma unroll
+    for (int j = 0; j < MAX_DREG; j++) {
+        const float diff = xv[j] - creg[j];
+        a += diff * diff;
+      }
+      a = wave_reduce_sum(a);
+      if (lane == 0) dmin[row] = a;
+      #pragma unroll
+    for (int j = 0; j < MAX_DREG; j++) acc[j] += xv[j];
+      run_len++;
+    }
+  }
+  flush();
+}
+
+void launch_kmeans_update_verify(const float* x, const int* perm,
+                                 const int* keys_sorted, const float* c,
+                                 const float* xn, float* dmin, int* amin,
+                                 const float* dmin2, const float* cn_max_dev,
+                                 float* sums, float* counts, long long n_rows,
+                                 long long d, int n_centroids, hipStream_t stream) {
+  const long long n_waves_target = 2048 * 4;
+  long long chunk = (n_rows + n_waves_target - 1) / n_waves_target;
+  if (chunk < 8) chunk = 8;
+  const long long n_waves = (n_rows + chunk - 1) / chunk;
+  const int grid = (int)((n_waves * RAFT_AMD_WAVE + 255) / 256);
+  if (d <= 256) {
+    hipLaunchKernelGGL((kmeans_update_verify_kernel<4>), dim3(grid), dim3(256), 0,
+                       stream, x, perm, keys_sorted, c, xn, dmin, amin, dmin2,
+                       cn_max_dev, sums, counts, n_rows, d, n_centroids, chunk);
+  } else {
+    hipLaunchKernelGGL((kmeans_update_verify_kernel<16>), dim3(grid), dim3(256), 0,
+                       stream, x, perm, keys_sorted, c, xn, dmin, amin, dmin2,
+                       cn_max_dev, sums, counts, n_rows, d, n_centroids, chunk);
+  }
+}
+
 // fused centroid prep: split fp32 centroids into bf16 slices + row sq-norms
 // in ONE pass (replaces ~6 torch ops per k-means iteration).
 __global__ void split_bf16_norms_kernel(const float* __restrict__ c,

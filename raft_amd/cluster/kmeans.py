@@ -295,15 +295,19 @@ def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify):
         ext.split_bf16_norms(centroids, c_slices, cn)
         dmin, amin, dmin2 = ext.fused_l2nn_split(list(x_slices), c_slices,
                                                  xn, cn)
-        if verify:
-            cn_max = cn.max().reshape(1)
-            ext.l2nn_verify_repair(x, centroids, xn, dmin, amin, dmin2, cn_max)
         keys_sorted, perm = torch.sort(amin)
         packed = torch.zeros(k * d + k + 1, dtype=torch.float32, device=dev)
         sums = packed[: k * d].view(k, d)
         counts = packed[k * d: k * d + k]
-        ext.reduce_rows_by_key_sorted_into(x, perm.to(torch.int32), keys_sorted,
-                                           sums, counts)
+        if verify:
+            # ONE X pass: centroid-sum accumulation + exact-fp32 verify/refine
+            cn_max = cn.max().reshape(1)
+            ext.kmeans_update_verify(x, perm.to(torch.int32), keys_sorted,
+                                     centroids, xn, dmin, amin, dmin2, cn_max,
+                                     sums, counts)
+        else:
+            ext.reduce_rows_by_key_sorted_into(x, perm.to(torch.int32), keys_sorted,
+                                               sums, counts)
         packed[-1] = torch.sum(dmin, dtype=torch.float64).float()
         if comms.get_size() > 1:
             comms.allreduce(packed, op=ReduceOp.SUM)
