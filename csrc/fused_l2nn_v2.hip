@@ -425,7 +425,13 @@ bool fused_l2nn_w8_supported(int nslice, int n, int d) {
   // measured (10M x 256 k=1024): w8 wins only for NSLICE=3 (staging-bound:
   // halved X re-reads beat the lost cross-block barrier overlap); the 4-wave
   // 2-block v1 kernel wins for NSLICE<=2 (34.3 vs 38.0 ms/step).
-  return nslice == 3 && n % 256 == 0 && d % 64 == 0;
+  // RAFT_AMD_L2NN_W8=2 forces w8 for ALL nslice (A/B experiments).
+  static const bool force_all = [] {
+    const char* e = getenv("RAFT_AMD_L2NN_W8");
+    return e && e[0] == '2';
+  }();
+  if (n % 256 != 0 || d % 64 != 0) return false;
+  return force_all || nslice == 3;
 }
 
 void launch_fused_l2nn_w8(const void** xsl, const void** csl, const float* xn,
