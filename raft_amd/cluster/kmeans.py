@@ -34,6 +34,7 @@ class KMeansParams:
     tol: float = 1e-4
     seed: int = 0
     init: str = "kmeans++"      # "kmeans++" | "random" | "array"
+    n_init: int = 1             # restarts with derived seeds; best inertia wins
     oversampling: float = 2.0   # for kmeans|| style init (unused by exact ++)
     fp32_mode: str = "auto"     # GEMM engine for the assignment step
     verbose: bool = False
@@ -120,7 +121,21 @@ def kmeans_fit(x: torch.Tensor, params: KMeansParams,
 
     sample_weights: optional per-row weights (reference kmeans API parity) —
     weighted centroid updates and weighted inertia.
+
+    n_init > 1 (reference kmeans_types n_init): run the whole fit n_init
+    times with derived seeds and keep the lowest-inertia model — the
+    standard guard against kmeans++ local optima. Seeds are derived
+    deterministically so every rank replays the same trials.
     """
+    if params.n_init > 1 and init_centroids is None and params.init != "array":
+        from dataclasses import replace
+        best = None
+        for trial in range(params.n_init):
+            p = replace(params, n_init=1, seed=params.seed + 9973 * trial)
+            m = kmeans_fit(x, p, comms, None, sample_weights)
+            if best is None or m.inertia < best.inertia:
+                best = m
+        return best
     comms = comms or LoopbackComms()
     state = RngState(seed=params.seed)
     k, (n_local, d) = params.n_clusters, x.shape

@@ -50,6 +50,15 @@ class TestKMeans:
         t = km.transform(x)
         assert t.shape == (100, 3)
 
+    def test_n_init_picks_best_restart(self):
+        """n_init (reference kmeans_types n_init): best-of-N restarts."""
+        x, _, _ = make_blobs(800, 8, n_clusters=6, cluster_std=0.4,
+                             state=RngState(seed=7))
+        m1 = kmeans_fit(x, KMeansParams(n_clusters=6, max_iter=15, seed=2))
+        m4 = kmeans_fit(x, KMeansParams(n_clusters=6, max_iter=15, seed=2,
+                                        n_init=4))
+        assert m4.inertia <= m1.inertia + 1e-3
+
     def test_monotone_inertia(self):
         """EM iterations must not increase inertia."""
         x, _, _ = make_blobs(500, 5, n_clusters=4, state=RngState(seed=2))

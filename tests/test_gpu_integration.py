@@ -150,7 +150,7 @@ class TestKMeansEndToEnd:
                                    center_box=(-25, 25), state=RngState(seed=13),
                                    device=dev)
         m = kmeans_fit(x, KMeansParams(n_clusters=64, max_iter=25, seed=2,
-                                       init="kmeans++"))
+                                       init="kmeans++", n_init=3))
         d = torch.cdist(centers, m.centroids)
         # kmeans++ is O(log k)-approx, not exact: allow one unlucky blob
         matched = (d.min(dim=1).values < 2.0).sum()
