@@ -18,10 +18,16 @@ from .bitset import Bitset
 from .interruptible import Interruptible, synchronize as interruptible_synchronize
 from .logger import get_logger, set_level
 from .trace import annotate, annotated
-from .memory import MemoryStats, TrackingScope, ResourceMonitor
+from .memory import MemoryStats, TrackingScope, ResourceMonitor, TemporaryDeviceBuffer
 from .mdbuffer import MDBuffer, MemoryType, memory_type_dispatcher
+from .error import RaftError, LogicError, HipError, expects, fail
+from .kvp import KeyValuePair
+from . import operators
+from . import math
 
 __all__ = [
+    "RaftError", "LogicError", "HipError", "expects", "fail",
+    "KeyValuePair", "operators", "math", "TemporaryDeviceBuffer",
     "Resources", "DeviceResources", "Handle", "DeviceResourcesSNMG",
     "DeviceResourcesManager", "get_resources", "device_ndarray",
     "serialize_mdspan", "deserialize_mdspan", "save_npy", "load_npy",

@@ -107,3 +107,35 @@ class ResourceMonitor:
     def __exit__(self, *exc):
         self.stop()
         return False
+
+
+class TemporaryDeviceBuffer:
+    """Workspace-backed scratch with host fallback
+    (reference: core/temporary_device_buffer.hpp).
+
+    Wraps `data` for device consumption: if `data` is already on the target
+    device it is passed through (zero-copy unless write_back demands
+    isolation); a host tensor is copied in, and copied back on exit when
+    write_back=True.
+    """
+
+    def __init__(self, data: "torch.Tensor", device=None, write_back: bool = False):
+        self._src = data
+        self._write_back = write_back
+        dev = torch.device(device) if device is not None else (
+            torch.device("cuda") if torch.cuda.is_available() else data.device)
+        if data.device == dev:
+            self._buf = data if not write_back else data  # in-place view
+        else:
+            self._buf = data.to(dev)
+
+    def view(self) -> "torch.Tensor":
+        return self._buf
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *exc):
+        if self._write_back and self._buf.data_ptr() != self._src.data_ptr():
+            self._src.copy_(self._buf.to(self._src.device))
+        return False

@@ -200,3 +200,55 @@ class TestTraceMemoryMdbuffer:
         from raft_amd.matrix import print_matrix
         s = print_matrix(torch.eye(3), name="I")
         assert "I" in s and "1" in s
+
+
+class TestOperatorsErrorKvpMath:
+    def test_operators_compose(self):
+        from raft_amd.core import operators as op
+        x = torch.tensor([1.0, -4.0, 9.0])
+        assert torch.equal(op.abs_op(x), x.abs())
+        assert torch.equal(op.sq_op(x), x * x)
+        f = op.compose_op(op.sqrt_op, op.abs_op)   # sqrt(abs(x))
+        torch.testing.assert_close(f(x), x.abs().sqrt())
+        add3 = op.add_const_op(3.0)
+        torch.testing.assert_close(add3(x), x + 3)
+        sqd = op.sqdiff_op
+        torch.testing.assert_close(sqd(x, torch.ones(3)), (x - 1) ** 2)
+        assert op.div_checkzero_op(1.0, 0.0) == 0.0
+        t = op.div_checkzero_op(torch.ones(2), torch.tensor([0.0, 2.0]))
+        assert torch.equal(t, torch.tensor([0.0, 0.5]))
+
+    def test_kvp_argmin_op(self):
+        from raft_amd.core import KeyValuePair
+        from raft_amd.core.operators import argmin_op, argmax_op
+        a, b = KeyValuePair(3, 1.5), KeyValuePair(1, 1.5)
+        assert argmin_op(a, b).key == 1        # tie -> smaller key
+        assert argmax_op(a, KeyValuePair(0, 2.0)).value == 2.0
+        k, v = KeyValuePair("k", 7)
+        assert (k, v) == ("k", 7)
+
+    def test_error_expects(self):
+        from raft_amd.core import expects, fail, LogicError, RaftError
+        expects(True)
+        with pytest.raises(LogicError):
+            expects(False, "nope")
+        with pytest.raises(RaftError):
+            fail("boom")
+
+    def test_math_wrappers(self):
+        from raft_amd.core import math as rmath
+        assert rmath.sigmoid(0.0) == pytest.approx(0.5)
+        assert rmath.sigmoid(-800.0) == pytest.approx(0.0)  # no overflow
+        x = torch.tensor([0.0, 1.0])
+        torch.testing.assert_close(rmath.sigmoid(x), torch.sigmoid(x))
+        assert rmath.max(2, 5) == 5 and rmath.min(2, 5) == 2
+        torch.testing.assert_close(rmath.atan2(torch.ones(2), torch.ones(2)),
+                                   torch.full((2,), 0.7853981633974483))
+
+    def test_temporary_device_buffer(self):
+        from raft_amd.core import TemporaryDeviceBuffer
+        x = torch.arange(4, dtype=torch.float32)
+        with TemporaryDeviceBuffer(x, device="cpu", write_back=True) as buf:
+            v = buf.view()
+            v += 1
+        assert torch.equal(x, torch.tensor([1.0, 2.0, 3.0, 4.0]))
