@@ -31,7 +31,7 @@
 
 namespace raft_amd {
 
-template <int NSLICE, bool DB = false>
+template <int NSLICE, bool DB = false, bool PHASED = false>
 __launch_bounds__(256, 2)
 __global__ void fused_l2nn_kernel(const __bf16* __restrict__ x0,
                                   const __bf16* __restrict__ x1,
@@ -83,7 +83,10 @@ __global__ void fused_l2nn_kernel(const __bf16* __restrict__ x0,
 #pragma unroll
       for (int b = 0; b < 4; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-    if constexpr (DB) {
+    if constexpr (PHASED && NSLICE == 2) {
+      mfma_tile_kloop_p2(xg, cg, xs, cs, acc, row0, (long long)nt * 128, d,
+                         m - 1, n - 1, wr, wc, lane);
+    } else if constexpr (DB) {
       mfma_tile_kloop_db32<NSLICE>(xg, cg, xs2, cs2, acc, row0,
                                    (long long)nt * 128, d, m - 1, n - 1, wr, wc,
                                    lane);
@@ -183,6 +186,14 @@ __global__ void fused_l2nn_kernel(const __bf16* __restrict__ x0,
   }
 }
 
+bool l2nn_phased() {
+  static const bool on = [] {
+    const char* e = getenv("RAFT_AMD_L2NN_PHASED");
+    return e && e[0] == '1';
+  }();
+  return on;
+}
+
 void launch_fused_l2nn_split(const void** xsl, const void** csl, const float* xn,
                              const float* cn, float* dmin, int* amin, float* dmin2,
                              long long m, int n, int d, int nslice,
@@ -212,6 +223,9 @@ void launch_fused_l2nn_split(const void** xsl, const void** csl, const float* xn
       if (use_db)
         hipLaunchKernelGGL((fused_l2nn_kernel<2, true>), dim3(grid), dim3(256), lds,
                            stream, x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d);
+      else if (l2nn_phased())
+        hipLaunchKernelGGL((fused_l2nn_kernel<2, false, true>), dim3(grid), dim3(256),
+                           lds, stream, x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d);
       else
         hipLaunchKernelGGL((fused_l2nn_kernel<2>), dim3(grid), dim3(256), lds, stream,
                            x0, x1, x2, c0, c1, c2, xn, cn, dmin, amin, dmin2, m, n, d);

@@ -1,0 +1,261 @@
+// Fused L2-NN, 2D-grid variant: one block per (row-tile, col-tile-GROUP) with
+// an XCD-contiguous block swizzle + a per-row partials-combine kernel.
+//
+// WHY (measured, profiles/pmc_fused_l2nn_tcc_l2.txt): the v1 kernel loops all
+// col-tiles inside one block, so every block re-stages its 131 KB X tile once
+// per col-tile; the per-XCD working set (64 resident blocks x 131 KB) blows
+// the 4 MiB per-XCD L2 -> 48.6% hit rate, HBM reads == full 8x X re-read.
+// Here the col-tiles of a row tile are TEMPORALLY ADJACENT ON ONE XCD
+// (guide T1: default dispatch round-robins blockIdx across the 8 XCDs, so we
+// remap bijectively so XCD x executes a contiguous range of row-major tile
+// indices): the X tile is read into that XCD's L2 once and the other
+// col-tile groups hit (measured 93.2% hit rate, HBM 49 GB -> 6.7 GB,
+// profiles/pmc_fused_l2nn_2d_tcc.txt). Each block emits per-row
+// (best, second, argmin) PARTIALS for its group; a bandwidth-bound combine
+// kernel merges the n/128/GT partials per row (disjoint column ranges -> a
+// plain top-2 merge) and applies ||x||^2.
+//
+// GT (col-tiles per block) trades L2 working set against epilogue
+// amortization: GT=1 minimizes the per-XCD X footprint (64 resident blocks
+// share 8 row tiles) but pays the reduce epilogue per tile; GT=4 pays it
+// once per 4 tiles but doubles the X footprint. The epilogue itself is an
+// LDS-transpose serial merge (each of 128 threads merges its row's 32 lane
+// candidates) instead of the 12-shuffle butterfly — cheaper, and it
+// subsumes the column-half-wave combine.
+//
+// Reference parity: same fused-L2-NN contract as fused_l2nn.hip (RAFT's
+// fusedL2NN / k-means assignment step).
+
+#include <hip/hip_runtime.h>
+
+#include "mfma_common.h"
+
+namespace raft_amd {
+
+bool l2nn_phased();  // fused_l2nn.hip: RAFT_AMD_L2NN_PHASED
+
+template <int NSLICE, int GT, bool PHASED = false>
+__launch_bounds__(256, 2)
+__global__ void fused_l2nn_2d_kernel(const __bf16* __restrict__ x0,
+                                     const __bf16* __restrict__ x1,
+                                     const __bf16* __restrict__ x2,
+                                     const __bf16* __restrict__ c0,
+                                     const __bf16* __restrict__ c1,
+                                     const __bf16* __restrict__ c2,
+                                     const float* __restrict__ cn,
+                                     float* __restrict__ pd,
+                                     float* __restrict__ pd2,
+                                     int* __restrict__ pi,
+                                     long long m, int n, int d, int n_groups) {
+  extern __shared__ __bf16 smem[];
+  __bf16* xs[NSLICE];
+  __bf16* cs[NSLICE];
+  const __bf16* const xg[3] = {x0, x1, x2};
+  const __bf16* const cg[3] = {c0, c1, c2};
+#pragma unroll
+  for (int s = 0; s < NSLICE; s++) {
+    xs[s] = smem + s * 8192;
+    cs[s] = smem + (NSLICE + s) * 8192;
+  }
+
+  // bijective XCD-contiguous remap (guide T1 / m204 variant): XCD x = bid%8
+  // executes slots bid/8 over a contiguous band of row-major tile indices,
+  // so its resident blocks share X row-tiles through the per-XCD L2.
+  const int nwg = gridDim.x;
+  const int bid = blockIdx.x;
+  const int q = nwg >> 3, r = nwg & 7;
+  const int xcd = bid & 7, slot = bid >> 3;
+  const int t = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + slot;
+  const long long row0 = (long long)(t / n_groups) * 128;
+  const int grp = t % n_groups;
+
+  const int lane = threadIdx.x % RAFT_AMD_WAVE;
+  const int w = threadIdx.x / RAFT_AMD_WAVE;
+  const int wr = w >> 1, wc = w & 1;  // 2x2 wave grid
+
+  float best[4][4], best2[4][4];
+  int bidx[4][4];
+#pragma unroll
+  for (int a = 0; a < 4; a++)
+#pragma unroll
+    for (int b = 0; b < 4; b++) {
+      best[a][b] = INFINITY;
+      best2[a][b] = INFINITY;
+      bidx[a][b] = 0;
+    }
+
+#pragma unroll
+  for (int g = 0; g < GT; g++) {
+    const long long col0 = ((long long)grp * GT + g) * 128;
+    f32x4 acc[4][4];
+#pragma unroll
+    for (int a = 0; a < 4; a++)
+#pragma unroll
+      for (int b = 0; b < 4; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    if constexpr (PHASED && NSLICE == 2) {
+      mfma_tile_kloop_p2(xg, cg, xs, cs, acc, row0, col0, d, m - 1, n - 1, wr,
+                         wc, lane);
+    } else {
+      mfma_tile_kloop<NSLICE>(xg, cg, xs, cs, acc, row0, col0, d, m - 1, n - 1,
+                              wr, wc, lane);
+    }
+
+    // lane-local running top-2 across the group's tiles (disjoint columns)
+    const int col_base = (int)col0 + wc * 64;
+#pragma unroll
+    for (int fr = 0; fr < 4; fr++) {
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++) {
+#pragma unroll
+        for (int fc = 0; fc < 4; fc++) {
+          const int col = col_base + fc * 16 + (lane & 15);
+          const float s = cn[col] - 2.f * acc[fr][fc][reg];
+          if (s < best[fr][reg]) {
+            best2[fr][reg] = best[fr][reg];
+            best[fr][reg] = s;
+            bidx[fr][reg] = col;
+          } else if (s < best2[fr][reg]) {
+            best2[fr][reg] = s;
+          }
+        }
+      }
+    }
+  }
+
+  // LDS-transpose epilogue: each thread parks its 16 (fr,reg) top-2 triples;
+  // then one thread per row serially merges the row's 32 lane-candidates
+  // (both column-half waves at once — subsumes the wc-combine). Stride 33
+  // keeps the 32-candidate rows off a single bank.
+  __syncthreads();
+  float* lv = reinterpret_cast<float*>(smem);   // [128][33]
+  float* lv2 = lv + 128 * 33;                   // [128][33]
+  int* li = reinterpret_cast<int*>(lv2 + 128 * 33);
+  const int cand = wc * 16 + (lane & 15);
+#pragma unroll
+  for (int fr = 0; fr < 4; fr++)
+#pragma unroll
+    for (int reg = 0; reg < 4; reg++) {
+      const int rl = wr * 64 + fr * 16 + ((lane >> 4) & 3) * 4 + reg;
+      lv[rl * 33 + cand] = best[fr][reg];
+      lv2[rl * 33 + cand] = best2[fr][reg];
+      li[rl * 33 + cand] = bidx[fr][reg];
+    }
+  __syncthreads();
+  const int rl = threadIdx.x;
+  if (rl < 128) {
+    const long long row = row0 + rl;
+    if (row < m) {
+      float v = INFINITY, v2 = INFINITY;
+      int vi = 0;
+#pragma unroll 8
+      for (int c = 0; c < 32; c++) {
+        const float b = lv[rl * 33 + c];
+        const float b2 = lv2[rl * 33 + c];
+        const int bi = li[rl * 33 + c];
+        v2 = fminf(fminf(v2, b2), fmaxf(v, b));  // columns all distinct
+        if (b < v || (b == v && bi < vi)) { v = b; vi = bi; }
+      }
+      const long long o = (long long)grp * m + row;
+      pd[o] = v;
+      pd2[o] = v2;
+      pi[o] = vi;
+    }
+  }
+}
+
+// Merge the n_groups partials per row; groups cover disjoint column ranges
+// so the cross-group second-best is the plain top-2 merge of independent
+// lists. Adds ||x||^2 once and clamps the best (matching v1's write).
+__global__ void l2nn_combine_partials_kernel(const float* __restrict__ pd,
+                                             const float* __restrict__ pd2,
+                                             const int* __restrict__ pi,
+                                             const float* __restrict__ xn,
+                                             float* __restrict__ dmin,
+                                             int* __restrict__ amin,
+                                             float* __restrict__ dmin2,
+                                             long long m, int n_groups) {
+  const long long row = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (row >= m) return;
+  float v = INFINITY, v2 = INFINITY;
+  int vi = 0;
+  for (int g = 0; g < n_groups; g++) {
+    const long long o = (long long)g * m + row;
+    const float b = pd[o], b2 = pd2[o];
+    const int bi = pi[o];
+    v2 = fminf(fminf(v2, b2), fmaxf(v, b));
+    if (b < v || (b == v && bi < vi)) { v = b; vi = bi; }
+  }
+  const float x = xn[row];
+  dmin[row] = fmaxf(v + x, 0.f);
+  amin[row] = vi;
+  if (dmin2) dmin2[row] = v2 + x;
+}
+
+static int l2nn_2d_gt() {
+  static const int gt = [] {
+    const char* e = getenv("RAFT_AMD_L2NN_GT");
+    const int v = e ? atoi(e) : 2;
+    return (v == 1 || v == 2 || v == 4) ? v : 2;
+  }();
+  return gt;
+}
+
+bool fused_l2nn_2d_supported(int nslice, long long m, int n, int d) {
+  static const char mode = [] {
+    const char* e = getenv("RAFT_AMD_L2NN_2D");
+    return e ? e[0] : 'a';
+  }();
+  if (mode == '0') return false;
+  if (nslice > 2) return false;  // 3-slice LDS (96 KiB) drops to 1 block/CU
+  if ((n / 128) % l2nn_2d_gt() != 0) return false;
+  if (mode == '1') return true;
+  // auto: pays off when the col-tile loop is long enough that X re-reads
+  // dominate and m is large enough that partial traffic amortizes
+  // (measured: 16.9 ms vs v1's 17.4-18.0 at 10M x 256 k=1024, with HBM
+  // reads 49 GB -> 6.7 GB; see BASELINE.md schedule table).
+  return n >= 512 && m >= 1000000;
+}
+
+void launch_fused_l2nn_2d(const void** xsl, const void** csl, const float* xn,
+                          const float* cn, float* pd, float* pd2, int* pi,
+                          float* dmin, int* amin, float* dmin2,
+                          long long m, int n, int d, int nslice,
+                          hipStream_t stream) {
+  const int gt = l2nn_2d_gt();
+  const int n_row_tiles = (int)((m + 127) / 128);
+  const int n_groups = n / 128 / gt;
+  const int grid = n_row_tiles * n_groups;
+  // kloop needs NSLICE*2*16KiB; the epilogue reuses it ([128][33] f32 x2 + i32)
+  const size_t lds_kloop = (size_t)nslice * 2 * 8192 * sizeof(__bf16);
+  const size_t lds = lds_kloop > 3 * 128 * 33 * 4 ? lds_kloop : 3 * 128 * 33 * 4;
+  const __bf16* x0 = (const __bf16*)xsl[0];
+  const __bf16* x1 = (const __bf16*)(nslice > 1 ? xsl[1] : xsl[0]);
+  const __bf16* x2 = (const __bf16*)(nslice > 2 ? xsl[2] : xsl[0]);
+  const __bf16* c0 = (const __bf16*)csl[0];
+  const __bf16* c1 = (const __bf16*)(nslice > 1 ? csl[1] : csl[0]);
+  const __bf16* c2 = (const __bf16*)(nslice > 2 ? csl[2] : csl[0]);
+#define L2NN2D_LAUNCH(NS, G, P)                                               \
+  hipLaunchKernelGGL((fused_l2nn_2d_kernel<NS, G, P>), dim3(grid), dim3(256), \
+                     lds, stream, x0, x1, x2, c0, c1, c2, cn, pd, pd2, pi, m, \
+                     n, d, n_groups)
+  const bool ph = l2nn_phased();
+  if (nslice == 1) {
+    if (gt == 1) L2NN2D_LAUNCH(1, 1, false);
+    else if (gt == 2) L2NN2D_LAUNCH(1, 2, false);
+    else L2NN2D_LAUNCH(1, 4, false);
+  } else if (nslice == 2) {
+    if (gt == 1) { if (ph) L2NN2D_LAUNCH(2, 1, true); else L2NN2D_LAUNCH(2, 1, false); }
+    else if (gt == 2) { if (ph) L2NN2D_LAUNCH(2, 2, true); else L2NN2D_LAUNCH(2, 2, false); }
+    else { if (ph) L2NN2D_LAUNCH(2, 4, true); else L2NN2D_LAUNCH(2, 4, false); }
+  } else {
+    throw std::runtime_error("fused_l2nn_2d: nslice must be 1 or 2");
+  }
+#undef L2NN2D_LAUNCH
+  const long long cgrid = (m + 255) / 256;
+  hipLaunchKernelGGL(l2nn_combine_partials_kernel, dim3((int)cgrid), dim3(256),
+                     0, stream, pd, pd2, pi, xn, dmin, amin, dmin2, m,
+                     n_groups);
+}
+
+}  // namespace raft_amd

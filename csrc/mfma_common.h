@@ -129,6 +129,104 @@ __device__ __forceinline__ void mfma_tile_kloop(
 }
 
 // ---------------------------------------------------------------------------
+// Phased 2-slice K-loop (guide T3/T4: counted s_waitcnt vmcnt instead of a
+// full drain). Load issue order per K-step: [X0 C0 | C1 | X1] (4 gload
+// rounds each interleaved). The slice-product MFMAs are split into three
+// phases by which tiles they touch:
+//   phase A (vmcnt<=8):  p00 = X0*C0            — first 8 loads landed
+//   phase B (vmcnt<=4):  p01 = X0*C1            — C1 landed
+//   phase C (vmcnt==0):  p10 = X1*C0, p11=X1*C1 — all landed
+// so the tail loads' HBM/L2 latency hides under phase A/B MFMA instead of a
+// serial vmcnt(0) stall. Raw s_barrier via asm — the compiler's
+// __syncthreads() would conservatively drain vmcnt(0) and defeat the
+// counting. Cross-wave safety: every thread waits its OWN vmcnt before the
+// barrier, so after it the corresponding loads of ALL threads have landed.
+// ---------------------------------------------------------------------------
+
+#define RAFT_AMD_WAIT_BARRIER(N)                                     \
+  asm volatile("s_waitcnt vmcnt(" #N ")\n\ts_barrier" ::: "memory")
+
+__device__ __forceinline__ void mfma_p2_frags(__bf16* ts, int wrc, int lane,
+                                              bf16x8 (&frag)[2][4]) {
+#pragma unroll
+  for (int kf = 0; kf < 2; kf++)
+#pragma unroll
+    for (int f = 0; f < 4; f++) {
+      const int r = wrc * 64 + f * 16 + (lane & 15);
+      const int byte = mfma_swz(r * 128 + (kf * 32 + (lane >> 4) * 8) * 2);
+      frag[kf][f] = *reinterpret_cast<const bf16x8*>((const char*)ts + byte);
+    }
+}
+
+__device__ __forceinline__ void mfma_p2_prod(const bf16x8 (&a)[2][4],
+                                             const bf16x8 (&b)[2][4],
+                                             f32x4 (&acc)[4][4]) {
+#pragma unroll
+  for (int kf = 0; kf < 2; kf++)
+#pragma unroll
+    for (int fr = 0; fr < 4; fr++)
+#pragma unroll
+      for (int fc = 0; fc < 4; fc++)
+        acc[fr][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a[kf][fr], b[kf][fc], acc[fr][fc], 0, 0, 0);
+}
+
+__device__ __forceinline__ void mfma_tile_kloop_p2(
+    const __bf16* const (&xg)[3], const __bf16* const (&cg)[3],
+    __bf16* (&xs)[2], __bf16* (&cs)[2],
+    f32x4 (&acc)[4][4], long long row0, long long col0, int d,
+    long long m_max, long long n_max, int wr, int wc, int lane) {
+  const int k_tiles = d / 64;
+  const int t = threadIdx.x;
+  const int wv = t / RAFT_AMD_WAVE;
+  long long bx[4], bc[4];
+  int ldst[4];
+#pragma unroll
+  for (int j = 0; j < 4; j++) {
+    const int o = j * 4096 + t * 16;
+    const int o_src = mfma_swz(o);
+    const int r = o_src >> 7;
+    const int k = (o_src & 127) >> 1;
+    long long rx = row0 + r;
+    if (rx > m_max) rx = m_max;
+    bx[j] = rx * (long long)d + k;
+    long long rc = col0 + r;
+    if (rc > n_max) rc = n_max;
+    bc[j] = rc * (long long)d + k;
+    ldst[j] = (j * 4096 + wv * 1024) / 2;
+  }
+  for (int kt = 0; kt < k_tiles; kt++) {
+    const long long koff = (long long)kt * 64;
+#pragma unroll
+    for (int j = 0; j < 4; j++) {  // first 8: X0 + C0
+      GLOAD_LDS(xg[0] + bx[j] + koff, xs[0] + ldst[j]);
+      GLOAD_LDS(cg[0] + bc[j] + koff, cs[0] + ldst[j]);
+    }
+#pragma unroll
+    for (int j = 0; j < 4; j++)    // next 4: C1
+      GLOAD_LDS(cg[1] + bc[j] + koff, cs[1] + ldst[j]);
+#pragma unroll
+    for (int j = 0; j < 4; j++)    // last 4: X1
+      GLOAD_LDS(xg[1] + bx[j] + koff, xs[1] + ldst[j]);
+
+    bf16x8 a0[2][4], a1[2][4], b0[2][4], b1[2][4];
+    RAFT_AMD_WAIT_BARRIER(8);      // X0,C0 landed (phase A)
+    mfma_p2_frags(xs[0], wr, lane, a0);
+    mfma_p2_frags(cs[0], wc, lane, b0);
+    mfma_p2_prod(a0, b0, acc);     // p00
+    RAFT_AMD_WAIT_BARRIER(4);      // C1 landed (phase B)
+    mfma_p2_frags(cs[1], wc, lane, b1);
+    mfma_p2_prod(a0, b1, acc);     // p01
+    RAFT_AMD_WAIT_BARRIER(0);      // X1 landed (phase C)
+    mfma_p2_frags(xs[1], wr, lane, a1);
+    mfma_p2_prod(a1, b0, acc);     // p10
+    mfma_p2_prod(a1, b1, acc);     // p11
+    // all ds_reads must retire before the next K-step's stores
+    asm volatile("s_waitcnt lgkmcnt(0)\n\ts_barrier" ::: "memory");
+  }
+}
+
+// ---------------------------------------------------------------------------
 // BK=32 double-buffered variant: 8 KiB tiles, 2-phase overlap (stage next
 // tile while MFMAing the current one; single vmcnt(0)+barrier per K-step —
 // the guide's minimum 2-phase recipe). LDS rows are 64 B, so the XOR swizzle
