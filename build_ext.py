@@ -74,7 +74,7 @@ def build(verbose: bool = True) -> Path:
     link = (["hipcc", "-shared", "-fPIC", "-o", str(out_so)] + [str(o) for o in objs]
             + [f"-L{d}" for d in libdirs]
             + ["-ltorch", "-ltorch_cpu", "-ltorch_hip", "-lc10", "-lc10_hip",
-               "-ltorch_python", "-lrocblas", "-lamdhip64"]
+               "-ltorch_python", "-lrocblas", "-lhipblaslt", "-lamdhip64"]
             + [f"-Wl,-rpath,{d}" for d in libdirs])
     if verbose:
         print("[build_ext] link", out_so.name, flush=True)

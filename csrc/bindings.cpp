@@ -90,6 +90,10 @@ void gemm_bf16_f32_rowmajor(const void*, const void*, float*, long long, long lo
                             long long, float, void*);
 void gemm_f32_rowmajor(const float*, const float*, float*, long long, long long,
                        long long, float, void*);
+void gemm_bf16_f32_rowmajor_lt(const void*, const void*, float*, long long, long long,
+                               long long, float, void*);
+void gemm_bf16_f32_nt_rowmajor_lt(const void*, const void*, float*, long long, long long,
+                                  long long, float, void*);
 void gemm_bf16_f32_nt_rowmajor(const void*, const void*, float*, long long, long long,
                                long long, float, void*);
 }  // namespace raft_amd
@@ -508,6 +512,17 @@ void pairwise_l2_filter(std::vector<torch::Tensor> x_slices,
   }
 }
 
+// backend toggle (reference parity: cublas vs cublasLt wrapper pair) —
+// RAFT_AMD_GEMM_BACKEND=hipblaslt routes the bf16->f32 GEMMs through the
+// hipBLASLt heuristic path (csrc/gemm_hipblaslt.cpp)
+static bool use_hipblaslt() {
+  static const bool on = [] {
+    const char* e = getenv("RAFT_AMD_GEMM_BACKEND");
+    return e && std::string(e) == "hipblaslt";
+  }();
+  return on;
+}
+
 torch::Tensor gemm_bf16_f32(torch::Tensor a, torch::Tensor b,
                             c10::optional<torch::Tensor> out, double beta) {
   TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16 && a.is_contiguous());
@@ -523,9 +538,14 @@ torch::Tensor gemm_bf16_f32(torch::Tensor a, torch::Tensor b,
                      a.options().dtype(torch::kFloat32));
     bt = 0.0f;
   }
-  raft_amd::gemm_bf16_f32_rowmajor(a.data_ptr(), b.data_ptr(), c.data_ptr<float>(),
-                                   a.size(0), b.size(1), a.size(1), bt,
-                                   (void*)cur_stream());
+  if (use_hipblaslt())
+    raft_amd::gemm_bf16_f32_rowmajor_lt(a.data_ptr(), b.data_ptr(), c.data_ptr<float>(),
+                                        a.size(0), b.size(1), a.size(1), bt,
+                                        (void*)cur_stream());
+  else
+    raft_amd::gemm_bf16_f32_rowmajor(a.data_ptr(), b.data_ptr(), c.data_ptr<float>(),
+                                     a.size(0), b.size(1), a.size(1), bt,
+                                     (void*)cur_stream());
   return c;
 }
 
@@ -543,9 +563,14 @@ torch::Tensor gemm_bf16_f32_nt(torch::Tensor a, torch::Tensor b,
     c = torch::empty({a.size(0), b.size(0)}, a.options().dtype(torch::kFloat32));
     bt = 0.0f;
   }
-  raft_amd::gemm_bf16_f32_nt_rowmajor(a.data_ptr(), b.data_ptr(), c.data_ptr<float>(),
-                                      a.size(0), b.size(0), a.size(1), bt,
-                                      (void*)cur_stream());
+  if (use_hipblaslt())
+    raft_amd::gemm_bf16_f32_nt_rowmajor_lt(a.data_ptr(), b.data_ptr(), c.data_ptr<float>(),
+                                           a.size(0), b.size(0), a.size(1), bt,
+                                           (void*)cur_stream());
+  else
+    raft_amd::gemm_bf16_f32_nt_rowmajor(a.data_ptr(), b.data_ptr(), c.data_ptr<float>(),
+                                        a.size(0), b.size(0), a.size(1), bt,
+                                        (void*)cur_stream());
   return c;
 }
 
