@@ -164,6 +164,141 @@ __global__ void fused_l2nn_2d_kernel(const __bf16* __restrict__ x0,
   }
 }
 
+// BK=32 / 4-blocks-per-CU variant: 8 KiB LDS tiles (32 KiB total for
+// 2 slices) double the resident blocks per CU, so 4 independent load-drains
+// interleave per SIMD instead of 2. Epilogue is the v1-style shuffle reduce
+// (once per GT tiles) + tiny LDS half-combine — the [128][33] transpose
+// epilogue would need 50 KiB and cap occupancy back at 2.
+template <int NSLICE, int GT>
+__launch_bounds__(256, 4)
+__global__ void fused_l2nn_2d_bk32_kernel(const __bf16* __restrict__ x0,
+                                          const __bf16* __restrict__ x1,
+                                          const __bf16* __restrict__ x2,
+                                          const __bf16* __restrict__ c0,
+                                          const __bf16* __restrict__ c1,
+                                          const __bf16* __restrict__ c2,
+                                          const float* __restrict__ cn,
+                                          float* __restrict__ pd,
+                                          float* __restrict__ pd2,
+                                          int* __restrict__ pi,
+                                          long long m, int n, int d,
+                                          int n_groups) {
+  extern __shared__ __bf16 smem[];
+  __bf16* xs[NSLICE];
+  __bf16* cs[NSLICE];
+  const __bf16* const xg[3] = {x0, x1, x2};
+  const __bf16* const cg[3] = {c0, c1, c2};
+#pragma unroll
+  for (int s = 0; s < NSLICE; s++) {
+    xs[s] = smem + s * 4096;
+    cs[s] = smem + (NSLICE + s) * 4096;
+  }
+  const int nwg = gridDim.x;
+  const int bid = blockIdx.x;
+  const int q = nwg >> 3, r = nwg & 7;
+  const int xcd = bid & 7, slot = bid >> 3;
+  const int t = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + slot;
+  const long long row0 = (long long)(t / n_groups) * 128;
+  const int grp = t % n_groups;
+
+  const int lane = threadIdx.x % RAFT_AMD_WAVE;
+  const int w = threadIdx.x / RAFT_AMD_WAVE;
+  const int wr = w >> 1, wc = w & 1;
+
+  float best[4][4], best2[4][4];
+  int bidx[4][4];
+#pragma unroll
+  for (int a = 0; a < 4; a++)
+#pragma unroll
+    for (int b = 0; b < 4; b++) {
+      best[a][b] = INFINITY;
+      best2[a][b] = INFINITY;
+      bidx[a][b] = 0;
+    }
+
+#pragma unroll
+  for (int g = 0; g < GT; g++) {
+    const long long col0 = ((long long)grp * GT + g) * 128;
+    f32x4 acc[4][4];
+#pragma unroll
+    for (int a = 0; a < 4; a++)
+#pragma unroll
+      for (int b = 0; b < 4; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+    mfma_tile_kloop_s32<NSLICE>(xg, cg, xs, cs, acc, row0, col0, d, m - 1,
+                                n - 1, wr, wc, lane);
+    const int col_base = (int)col0 + wc * 64;
+#pragma unroll
+    for (int fr = 0; fr < 4; fr++)
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++)
+#pragma unroll
+        for (int fc = 0; fc < 4; fc++) {
+          const int col = col_base + fc * 16 + (lane & 15);
+          const float s = cn[col] - 2.f * acc[fr][fc][reg];
+          if (s < best[fr][reg]) {
+            best2[fr][reg] = best[fr][reg];
+            best[fr][reg] = s;
+            bidx[fr][reg] = col;
+          } else if (s < best2[fr][reg]) {
+            best2[fr][reg] = s;
+          }
+        }
+  }
+
+  // ONE cross-lane top-2 reduce (index-aware), then LDS half-combine
+#pragma unroll
+  for (int fr = 0; fr < 4; fr++)
+#pragma unroll
+    for (int reg = 0; reg < 4; reg++) {
+      float v = best[fr][reg], v2 = best2[fr][reg];
+      int vi = bidx[fr][reg];
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1) {
+        const float ov = __shfl_xor(v, off, RAFT_AMD_WAVE);
+        const float ov2 = __shfl_xor(v2, off, RAFT_AMD_WAVE);
+        const int oi = __shfl_xor(vi, off, RAFT_AMD_WAVE);
+        float new2 = fminf(v2, ov2);
+        if (oi != vi) new2 = fminf(new2, fmaxf(v, ov));
+        v2 = new2;
+        if (ov < v || (ov == v && oi < vi)) { v = ov; vi = oi; }
+      }
+      best[fr][reg] = v;
+      best2[fr][reg] = v2;
+      bidx[fr][reg] = vi;
+    }
+  __syncthreads();
+  float* comb_v = reinterpret_cast<float*>(smem);
+  int* comb_i = reinterpret_cast<int*>(comb_v + 256);
+  float* comb_v2 = reinterpret_cast<float*>(comb_i + 256);
+  if ((lane & 15) == 0) {
+#pragma unroll
+    for (int fr = 0; fr < 4; fr++)
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++) {
+        const int rl = wr * 64 + fr * 16 + (lane >> 4) * 4 + reg;
+        comb_v[wc * 128 + rl] = best[fr][reg];
+        comb_i[wc * 128 + rl] = bidx[fr][reg];
+        comb_v2[wc * 128 + rl] = best2[fr][reg];
+      }
+  }
+  __syncthreads();
+  const int rl = threadIdx.x;
+  if (rl < 128) {
+    const long long row = row0 + rl;
+    if (row < m) {
+      const float v0 = comb_v[rl], v1 = comb_v[128 + rl];
+      const int i0 = comb_i[rl], i1 = comb_i[128 + rl];
+      const float s2 = fminf(fminf(comb_v2[rl], comb_v2[128 + rl]),
+                             fmaxf(v0, v1));
+      const bool take1 = (v1 < v0) || (v1 == v0 && i1 < i0);
+      const long long o = (long long)grp * m + row;
+      pd[o] = take1 ? v1 : v0;
+      pi[o] = take1 ? i1 : i0;
+      pd2[o] = s2;
+    }
+  }
+}
+
 // Merge the n_groups partials per row; groups cover disjoint column ranges
 // so the cross-group second-best is the plain top-2 merge of independent
 // lists. Adds ||x||^2 once and clamps the best (matching v1's write).
@@ -239,6 +374,32 @@ void launch_fused_l2nn_2d(const void** xsl, const void** csl, const float* xn,
   hipLaunchKernelGGL((fused_l2nn_2d_kernel<NS, G, P>), dim3(grid), dim3(256), \
                      lds, stream, x0, x1, x2, c0, c1, c2, cn, pd, pd2, pi, m, \
                      n, d, n_groups)
+  static const bool bk32 = [] {
+    const char* e = getenv("RAFT_AMD_L2NN_BK32");
+    return e && e[0] == '1';
+  }();
+  if (bk32 && (nslice == 1 || nslice == 2) && d % 32 == 0) {
+    const size_t lds32 = (size_t)nslice * 2 * 4096 * sizeof(__bf16);
+#define L2NN2D_BK32(NS, G)                                                     \
+  hipLaunchKernelGGL((fused_l2nn_2d_bk32_kernel<NS, G>), dim3(grid),           \
+                     dim3(256), lds32, stream, x0, x1, x2, c0, c1, c2, cn, pd, \
+                     pd2, pi, m, n, d, n_groups)
+    if (nslice == 1) {
+      if (gt == 1) L2NN2D_BK32(1, 1);
+      else if (gt == 2) L2NN2D_BK32(1, 2);
+      else L2NN2D_BK32(1, 4);
+    } else {
+      if (gt == 1) L2NN2D_BK32(2, 1);
+      else if (gt == 2) L2NN2D_BK32(2, 2);
+      else L2NN2D_BK32(2, 4);
+    }
+#undef L2NN2D_BK32
+    const long long cgrid32 = (m + 255) / 256;
+    hipLaunchKernelGGL(l2nn_combine_partials_kernel, dim3((int)cgrid32),
+                       dim3(256), 0, stream, pd, pd2, pi, xn, dmin, amin,
+                       dmin2, m, n_groups);
+    return;
+  }
   const bool ph = l2nn_phased();
   if (nslice == 1) {
     if (gt == 1) L2NN2D_LAUNCH(1, 1, false);

@@ -257,6 +257,78 @@ __device__ __forceinline__ void mfma_stage_tile128_bk32(const __bf16* __restrict
   }
 }
 
+// Single-buffered BK=32 K-loop: 8 KiB tiles -> 32 KiB LDS total for 2 slices,
+// enabling 4 blocks/CU (vs BK=64's 2) so more independent load-drains
+// interleave on each SIMD. Pays the 64 B-row residual LDS read conflict.
+template <int NSLICE>
+__device__ __forceinline__ void mfma_tile_kloop_s32(
+    const __bf16* const (&xg)[3], const __bf16* const (&cg)[3],
+    __bf16* (&xs)[NSLICE], __bf16* (&cs)[NSLICE],
+    f32x4 (&acc)[4][4], long long row0, long long col0, int d,
+    long long m_max, long long n_max, int wr, int wc, int lane) {
+  const int k_tiles = d / 32;
+  const int t = threadIdx.x;
+  const int wv = t / RAFT_AMD_WAVE;
+  long long bx[2], bc[2];
+  int ldst[2];
+#pragma unroll
+  for (int j = 0; j < 2; j++) {
+    const int o = j * 4096 + t * 16;
+    const int o_src = mfma_swz32(o);
+    const int r = o_src >> 6;
+    const int k = (o_src & 63) >> 1;
+    long long rx = row0 + r;
+    if (rx > m_max) rx = m_max;
+    bx[j] = rx * (long long)d + k;
+    long long rc = col0 + r;
+    if (rc > n_max) rc = n_max;
+    bc[j] = rc * (long long)d + k;
+    ldst[j] = (j * 4096 + wv * 1024) / 2;
+  }
+  for (int kt = 0; kt < k_tiles; kt++) {
+    const long long koff = (long long)kt * 32;
+#pragma unroll
+    for (int s = 0; s < NSLICE; s++)
+#pragma unroll
+      for (int j = 0; j < 2; j++) {
+        GLOAD_LDS(xg[s] + bx[j] + koff, xs[s] + ldst[j]);
+        GLOAD_LDS(cg[s] + bc[j] + koff, cs[s] + ldst[j]);
+      }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    bf16x8 a_frag[NSLICE][4], b_frag[NSLICE][4];
+    const int kbyte = (lane >> 4) * 16;
+#pragma unroll
+    for (int fr = 0; fr < 4; fr++) {
+      const int r = wr * 64 + fr * 16 + (lane & 15);
+      const int byte = mfma_swz32(r * 64 + kbyte);
+#pragma unroll
+      for (int s = 0; s < NSLICE; s++)
+        a_frag[s][fr] = *reinterpret_cast<const bf16x8*>((const char*)xs[s] + byte);
+    }
+#pragma unroll
+    for (int fc = 0; fc < 4; fc++) {
+      const int c = wc * 64 + fc * 16 + (lane & 15);
+      const int byte = mfma_swz32(c * 64 + kbyte);
+#pragma unroll
+      for (int s = 0; s < NSLICE; s++)
+        b_frag[s][fc] = *reinterpret_cast<const bf16x8*>((const char*)cs[s] + byte);
+    }
+#pragma unroll
+    for (int fr = 0; fr < 4; fr++)
+#pragma unroll
+      for (int fc = 0; fc < 4; fc++) {
+#pragma unroll
+        for (int p = 0; p < mfma_n_products<NSLICE>(); p++) {
+          acc[fr][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[MFMA_PROD_A[p]][fr], b_frag[MFMA_PROD_B[p]][fc],
+              acc[fr][fc], 0, 0, 0);
+        }
+      }
+    __syncthreads();
+  }
+}
+
 template <int NSLICE>
 __device__ __forceinline__ void mfma_tile_kloop_db32(
     const __bf16* const (&xg)[3], const __bf16* const (&cg)[3],
