@@ -29,13 +29,12 @@ def _torch_flags():
     return inc, lib, abi
 
 
-def build(verbose: bool = True) -> Path:
-    inc, libdirs, abi = _torch_flags()
+def compile_flags(inc=None, abi=None):
+    """The per-TU hipcc flag set (shared with scripts/analyze_build_log.py)."""
+    if inc is None or abi is None:
+        inc, _, abi = _torch_flags()
     py_inc = sysconfig.get_paths()["include"]
-    BUILD.mkdir(parents=True, exist_ok=True)
-
-    sources = sorted(CSRC.glob("*.hip")) + sorted(CSRC.glob("*.cpp"))
-    common = [
+    return [
         "-O3", "-std=c++17", "-fPIC", f"--offload-arch={ARCH}",
         "-D__HIP_PLATFORM_AMD__=1", "-DUSE_ROCM=1", "-DHIPBLAS_V2",
         "-DCUDA_HAS_FP16=1", "-D__HIP_NO_HALF_OPERATORS__=1",
@@ -45,13 +44,22 @@ def build(verbose: bool = True) -> Path:
         "-Wno-unused-result", "-Wno-ignored-attributes",
     ] + [f"-I{p}" for p in inc + [py_inc, str(CSRC)]]
 
+
+def build(verbose: bool = True) -> Path:
+    inc, libdirs, abi = _torch_flags()
+    common = compile_flags(inc, abi)
+    BUILD.mkdir(parents=True, exist_ok=True)
+
+    sources = sorted(CSRC.glob("*.hip")) + sorted(CSRC.glob("*.cpp"))
+    headers_mtime = max(h.stat().st_mtime for h in CSRC.glob("*.h"))
+
     objs = []
     procs = []
     for src in sources:
         obj = BUILD / (src.stem + ".o")
         objs.append(obj)
         if obj.exists() and obj.stat().st_mtime > max(src.stat().st_mtime,
-                                                     (CSRC / "common.h").stat().st_mtime):
+                                                      headers_mtime):
             continue
         cmd = ["hipcc", "-c", "-x", "hip", str(src), "-o", str(obj)] + common
         if verbose:
