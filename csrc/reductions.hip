@@ -135,21 +135,80 @@ __global__ void reduce_rows_block_kernel(const T* __restrict__ x, T* __restrict_
 // column reduce: consecutive lanes own consecutive columns (coalesced);
 // each block tiles BLOCK columns and walks all rows.
 // --------------------------------------------------------------------------
+// 2D grid: blockIdx.x tiles columns, blockIdx.y tiles rows — a skinny
+// matrix (d << 256) still fills the chip. Each block reduces its row chunk
+// (Kahan for sums) then combines across row tiles with device atomics
+// (CAS loop for min/max since HIP has no float atomicMin/Max) — the same
+// partial+atomic shape as the reference stridedSummationKernel.
 template <int OP, typename T, int BLOCK = 256>
 __global__ void reduce_cols_kernel(const T* __restrict__ x, T* __restrict__ out,
-                                   long long n_rows, long long d) {
-  const long long col0 = (long long)blockIdx.x * BLOCK;
-  const long long col = col0 + threadIdx.x;
+                                   long long n_rows, long long d,
+                                   long long rows_per_tile) {
+  const long long col = (long long)blockIdx.x * BLOCK + threadIdx.x;
   if (col >= d) return;
+  const long long r0 = (long long)blockIdx.y * rows_per_tile;
+  const long long r1 = min(n_rows, r0 + rows_per_tile);
   if constexpr (is_sum<OP>()) {
-    KahanAcc<T> acc;
-    for (long long r = 0; r < n_rows; r++) acc.add(main_op<OP>(x[r * d + col]));
-    out[col] = acc.get();
+    // 4 independent Kahan accumulators: the compensated add is a serial
+    // dependency chain, so a single accumulator caps each thread at
+    // ~1 element / chain-latency — 4-way ILP restores bandwidth
+    KahanAcc<T> a0, a1, a2, a3;
+    long long r = r0;
+    for (; r + 3 < r1; r += 4) {
+      a0.add(main_op<OP>(x[r * d + col]));
+      a1.add(main_op<OP>(x[(r + 1) * d + col]));
+      a2.add(main_op<OP>(x[(r + 2) * d + col]));
+      a3.add(main_op<OP>(x[(r + 3) * d + col]));
+    }
+    for (; r < r1; r++) a0.add(main_op<OP>(x[r * d + col]));
+    a0.add(a1.get());
+    a0.add(a2.get());
+    a0.add(a3.get());
+    if (gridDim.y == 1) {
+      out[col] = a0.get();
+    } else {
+      atomicAdd(&out[col], a0.get());
+    }
   } else {
-    T v = red_init<OP, T>();
-    for (long long r = 0; r < n_rows; r++) v = red_op<OP>(v, main_op<OP>(x[r * d + col]));
-    out[col] = v;
+    T v0 = red_init<OP, T>(), v1 = v0, v2 = v0, v3 = v0;
+    long long r = r0;
+    for (; r + 3 < r1; r += 4) {
+      v0 = red_op<OP>(v0, main_op<OP>(x[r * d + col]));
+      v1 = red_op<OP>(v1, main_op<OP>(x[(r + 1) * d + col]));
+      v2 = red_op<OP>(v2, main_op<OP>(x[(r + 2) * d + col]));
+      v3 = red_op<OP>(v3, main_op<OP>(x[(r + 3) * d + col]));
+    }
+    for (; r < r1; r++) v0 = red_op<OP>(v0, main_op<OP>(x[r * d + col]));
+    T v = red_op<OP>(red_op<OP>(v0, v1), red_op<OP>(v2, v3));
+    if (gridDim.y == 1) {
+      out[col] = v;
+    } else if constexpr (sizeof(T) == 4) {
+      int* addr = reinterpret_cast<int*>(&out[col]);
+      int cur = __float_as_int(*reinterpret_cast<volatile float*>(addr));
+      while (true) {
+        const float merged = (float)red_op<OP>((T)__int_as_float(cur), v);
+        const int old = atomicCAS(addr, cur, __float_as_int(merged));
+        if (old == cur) break;
+        cur = old;
+      }
+    } else {
+      unsigned long long* addr = reinterpret_cast<unsigned long long*>(&out[col]);
+      unsigned long long cur = *reinterpret_cast<volatile unsigned long long*>(addr);
+      while (true) {
+        double merged = (double)red_op<OP>((T)__longlong_as_double(cur), v);
+        const unsigned long long old =
+            atomicCAS(addr, cur, __double_as_longlong(merged));
+        if (old == cur) break;
+        cur = old;
+      }
+    }
   }
+}
+
+template <typename T>
+__global__ void fill_kernel(T* __restrict__ out, long long n, T v) {
+  const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) out[i] = v;
 }
 
 // --------------------------------------------------------------------------
@@ -238,8 +297,24 @@ void launch_reduce_rows(const T* x, T* out, long long n_rows, long long d,
 template <int OP, typename T>
 void launch_reduce_cols(const T* x, T* out, long long n_rows, long long d,
                         hipStream_t stream) {
-  int grid = (int)((d + 255) / 256);
-  hipLaunchKernelGGL((reduce_cols_kernel<OP, T>), dim3(grid), dim3(256), 0, stream, x, out, n_rows, d);
+  const int gx = (int)((d + 255) / 256);
+  // enough row tiles to fill 256 CUs x 2 blocks even for skinny d, but keep
+  // each chunk >= 1024 rows so atomic traffic stays negligible
+  long long gy = (512 + gx - 1) / gx;
+  gy = min(gy, (n_rows + 1023) / 1024);
+  if (gy < 1) gy = 1;
+  const long long rows_per_tile = (n_rows + gy - 1) / gy;
+  if (gy > 1) {
+    if constexpr (OP <= 2) {
+      hipMemsetAsync(out, 0, d * sizeof(T), stream);
+    } else {
+      const T init = (OP == 3 || OP == 5) ? (T)-INFINITY : (T)INFINITY;
+      hipLaunchKernelGGL((fill_kernel<T>), dim3((int)((d + 255) / 256)),
+                         dim3(256), 0, stream, out, d, init);
+    }
+  }
+  hipLaunchKernelGGL((reduce_cols_kernel<OP, T>), dim3(gx, (int)gy), dim3(256),
+                     0, stream, x, out, n_rows, d, rows_per_tile);
 }
 
 #define INSTANTIATE_OPS(T)                                                              \

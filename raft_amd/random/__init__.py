@@ -12,12 +12,13 @@ from .rng import (
 from .make_blobs import make_blobs
 from .make_regression import make_regression
 from .rmat import rmat
-from .permute import permute
+from .permute import permute, permute_rows
 from .mvg import multi_variable_gaussian
 
 __all__ = [
     "RngState", "uniform", "uniform_int", "normal", "lognormal", "logistic",
     "exponential", "rayleigh", "laplace", "gumbel", "bernoulli",
     "sample_with_replacement", "sample_without_replacement",
-    "make_blobs", "make_regression", "rmat", "permute", "multi_variable_gaussian",
+    "make_blobs", "make_regression", "rmat", "permute", "permute_rows",
+    "multi_variable_gaussian",
 ]

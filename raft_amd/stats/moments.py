@@ -21,8 +21,10 @@ def mean(x: torch.Tensor, sample: bool = False) -> torch.Tensor:
     return sum_cols(x) / x.shape[0]
 
 
-def vars_(x: torch.Tensor, sample: bool = True) -> torch.Tensor:
-    mu = mean(x)
+def vars_(x: torch.Tensor, sample: bool = True,
+          mu: torch.Tensor | None = None) -> torch.Tensor:
+    if mu is None:
+        mu = mean(x)
     n = x.shape[0]
     ss = strided_reduction(x - mu.unsqueeze(0), main_op="sq", reduce_op="sum")
     return ss / (n - 1 if sample else n)
@@ -33,10 +35,10 @@ def stddev(x: torch.Tensor, sample: bool = True) -> torch.Tensor:
 
 
 def meanvar(x: torch.Tensor, sample: bool = True):
-    """Single-pass mean+variance (detail/meanvar.cuh mean_var<T> aggregates)."""
-    n = x.shape[0]
+    """Mean+variance in one sweep pair (detail/meanvar.cuh): the mean is
+    reused for the centered sq-sum instead of being recomputed."""
     mu = mean(x)
-    var = vars_(x, sample=sample)
+    var = vars_(x, sample=sample, mu=mu)
     return mu, var
 
 

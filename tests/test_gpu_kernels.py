@@ -55,6 +55,17 @@ class TestReductions:
         ref = {0: xr.sum(0), 1: (xr * xr).sum(0)}[op]
         torch.testing.assert_close(out.double(), ref, rtol=1e-5, atol=1e-4)
 
+    @pytest.mark.parametrize("op", [0, 3, 4])
+    def test_reduce_cols_tall_skinny_atomic_path(self, dev, ext, op):
+        """d << 256 forces the multi-row-tile grid (atomic combine across
+        tiles incl. the CAS min/max path)."""
+        torch.manual_seed(3)
+        x = torch.randn(1_000_000, 8, device=dev)
+        out = ext.reduce_cols(x, op)
+        xr = x.double()
+        ref = {0: xr.sum(0), 3: xr.max(0).values, 4: xr.min(0).values}[op]
+        torch.testing.assert_close(out.double(), ref, rtol=1e-5, atol=1e-3)
+
     def test_row_argmin(self, dev, ext):
         torch.manual_seed(3)
         x = torch.randn(5000, 777, device=dev)
@@ -199,6 +210,25 @@ class TestFusedL2NNMfma:
         # distances are exact-fp32 recomputed: error = fp32 rounding only
         rel = ((dv.double() - rd).abs() / rd.clamp_min(1e-3)).max()
         assert float(rel) < 1e-4, float(rel)
+
+    def test_2d_xcd_engine_at_scale(self, dev, ext):
+        """m >= 1M, n >= 512 routes to the 2D XCD-swizzled tile-pair engine
+        (fused_l2nn_2d_kernel + partials combine) by default — verify it on
+        sampled rows against an exact fp64 reference, including dmin2."""
+        torch.manual_seed(7)
+        m, n, d = 1_000_100, 512, 128   # odd m exercises the edge row tile
+        from raft_amd.neighbors.fused_l2nn import fused_l2nn
+        x = torch.randn(m, d, device=dev)
+        y = torch.randn(n, d, device=dev)
+        dmin, amin = fused_l2nn(x, y, fp32_mode="bf16x2v")
+        idx = torch.randperm(m, device=dev)[:2000]
+        idx = torch.cat([idx, torch.arange(m - 130, m, device=dev)])  # tail rows
+        ref = torch.cdist(x[idx].double(), y.double()) ** 2
+        rd, ra = ref.min(dim=1)
+        chosen = ref[torch.arange(idx.numel(), device=dev), amin[idx]]
+        assert float((chosen - rd).max()) < 1e-2          # argmin fp32-exact
+        rel = ((dmin[idx].double() - rd).abs() / rd.clamp_min(1e-3)).max()
+        assert float(rel) < 1e-4                          # exact distances
 
     def test_blob_data_agreement_is_exact(self, dev, ext):
         """On clustered data (the bench workload) assignments must match
