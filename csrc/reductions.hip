@@ -211,6 +211,65 @@ __global__ void fill_kernel(T* __restrict__ out, long long n, T v) {
   if (i < n) out[i] = v;
 }
 
+// skinny-d column reduce (d < 32): a flat col-per-thread map would idle
+// 64-d lanes of every wave, so lanes cover (row_sub, col) instead —
+// RPT=64/ceil_pow2(d) rows per wave step, full coalescing — and the row_sub
+// partials fold with log2(RPT) shuffles before the cross-tile atomic.
+template <int OP, typename T, int BLOCK = 256>
+__global__ void reduce_cols_skinny_kernel(const T* __restrict__ x,
+                                          T* __restrict__ out, long long n_rows,
+                                          long long d, long long rows_per_tile) {
+  int dp2 = 1;
+  while (dp2 < d) dp2 <<= 1;          // <= 32
+  const int rpt = RAFT_AMD_WAVE / dp2; // rows per wave step
+  const int lane = threadIdx.x % RAFT_AMD_WAVE;
+  const int wv = threadIdx.x / RAFT_AMD_WAVE;
+  const int col = lane % dp2;
+  const int rsub = lane / dp2;
+  const long long r0 = (long long)blockIdx.x * rows_per_tile;
+  const long long r1 = min(n_rows, r0 + rows_per_tile);
+  const int waves = BLOCK / RAFT_AMD_WAVE;
+  T v = red_init<OP, T>();
+  KahanAcc<T> acc;
+  if (col < d) {
+    for (long long r = r0 + (long long)wv * rpt + rsub; r < r1;
+         r += (long long)waves * rpt) {
+      const T e = main_op<OP>(x[r * d + col]);
+      if constexpr (is_sum<OP>()) acc.add(e);
+      else v = red_op<OP>(v, e);
+    }
+  }
+  // fold row_subs: shuffle down by dp2 strides within the wave
+  if constexpr (is_sum<OP>()) v = acc.get();
+  for (int off = RAFT_AMD_WAVE >> 1; off >= dp2; off >>= 1)
+    v = red_op<OP>(v, __shfl_xor(v, off, RAFT_AMD_WAVE));
+  // one lane per (wave, col) merges across waves + tiles via atomics
+  if (rsub == 0 && col < d) {
+    if constexpr (is_sum<OP>()) {
+      atomicAdd(&out[col], v);
+    } else if constexpr (sizeof(T) == 4) {
+      int* addr = reinterpret_cast<int*>(&out[col]);
+      int cur = __float_as_int(*reinterpret_cast<volatile float*>(addr));
+      while (true) {
+        const float merged = (float)red_op<OP>((T)__int_as_float(cur), v);
+        const int old = atomicCAS(addr, cur, __float_as_int(merged));
+        if (old == cur) break;
+        cur = old;
+      }
+    } else {
+      unsigned long long* addr = reinterpret_cast<unsigned long long*>(&out[col]);
+      unsigned long long cur = *reinterpret_cast<volatile unsigned long long*>(addr);
+      while (true) {
+        double merged = (double)red_op<OP>((T)__longlong_as_double(cur), v);
+        const unsigned long long old =
+            atomicCAS(addr, cur, __double_as_longlong(merged));
+        if (old == cur) break;
+        cur = old;
+      }
+    }
+  }
+}
+
 // --------------------------------------------------------------------------
 // row argmin: one wave per row (thin) or block per row (wide)
 // --------------------------------------------------------------------------
@@ -297,6 +356,20 @@ void launch_reduce_rows(const T* x, T* out, long long n_rows, long long d,
 template <int OP, typename T>
 void launch_reduce_cols(const T* x, T* out, long long n_rows, long long d,
                         hipStream_t stream) {
+  if (d < 32 && n_rows >= 4096) {
+    long long gy = min((long long)1024, (n_rows + 4095) / 4096);
+    const long long rows_per_tile = (n_rows + gy - 1) / gy;
+    if constexpr (OP <= 2) {
+      hipMemsetAsync(out, 0, d * sizeof(T), stream);
+    } else {
+      const T init = (OP == 3 || OP == 5) ? (T)-INFINITY : (T)INFINITY;
+      hipLaunchKernelGGL((fill_kernel<T>), dim3(1), dim3(256), 0, stream, out,
+                         d, init);
+    }
+    hipLaunchKernelGGL((reduce_cols_skinny_kernel<OP, T>), dim3((int)gy),
+                       dim3(256), 0, stream, x, out, n_rows, d, rows_per_tile);
+    return;
+  }
   const int gx = (int)((d + 255) / 256);
   // enough row tiles to fill 256 CUs x 2 blocks even for skinny d, but keep
   // each chunk >= 1024 rows so atomic traffic stays negligible

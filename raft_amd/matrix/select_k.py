@@ -42,10 +42,21 @@ def select_k(x: torch.Tensor, k: int, select_min: bool = True,
     assert 0 < k <= n, f"k={k} out of range for row length {n}"
 
     if (on_gpu(x) and x.dtype == torch.float32 and algo != SelectAlgo.TORCH
+            and k <= 64 and n <= 4096
+            and algo in (SelectAlgo.AUTO, SelectAlgo.WARPSORT)):
+        # short rows, small k: the wave-register warpsort queue beats both
+        # rocPRIM topk and the radix path (measured [20000 x 500] k=32:
+        # warpsort 0.108 ms vs torch.topk 0.27 ms vs radix 0.43 ms)
+        ext = require_ext()
+        vals, idx = ext.select_k(x.contiguous(), k, bool(select_min), 2,
+                                 bool(sorted))
+        return vals, idx.to(torch.int64)
+
+    if (on_gpu(x) and x.dtype == torch.float32 and algo != SelectAlgo.TORCH
             and k <= 2048 and n > 4096):
-        # n <= 4096 rows are cheapest through the vendor segmented sort
-        # (rocPRIM topk) — the same shape-dispatch idea as the reference's
-        # learned tree, re-measured on gfx950 (see benchmarks)
+        # remaining n <= 4096 shapes (k > 64) are cheapest through the vendor
+        # segmented sort (rocPRIM topk) — the same shape-dispatch idea as the
+        # reference's learned tree, re-measured on gfx950 (see benchmarks)
         ext = require_ext()
         # two-level split: a small batch over a huge row leaves the chip idle
         # (one workgroup per row); split rows into S segments, select per
