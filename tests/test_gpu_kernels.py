@@ -308,7 +308,8 @@ class TestRngGpu:
 class TestSelectKGpu:
     @pytest.mark.parametrize("batch,n,k", [(32, 1000, 10), (4, 100000, 64),
                                            (128, 512, 256), (2, 1000000, 100),
-                                           (16, 2048, 1024)])
+                                           (16, 2048, 1024), (20000, 500, 32),
+                                           (128, 500, 64), (64, 65, 64)])
     def test_vs_topk(self, dev, ext, batch, n, k):
         from raft_amd.matrix import select_k
         torch.manual_seed(0)
@@ -324,6 +325,16 @@ class TestSelectKGpu:
         vals, idx = select_k(x, 32, select_min=False)
         ref_v, _ = torch.topk(x, 32, dim=1, largest=True)
         torch.testing.assert_close(vals, ref_v)
+
+    def test_duplicates_short_row_warpsort(self, dev, ext):
+        """short-row path (native warpsort): duplicate minima straddling k."""
+        from raft_amd.matrix import select_k
+        x = torch.zeros(5, 300, device=dev)
+        x[:, :40] = -1.0
+        vals, idx = select_k(x, 64, select_min=True)
+        assert (vals[:, :40] == -1).all() and (vals[:, 40:] == 0).all()
+        for r in range(5):
+            assert idx[r].unique().numel() == 64
 
     def test_duplicates_at_kth(self, dev, ext):
         from raft_amd.matrix import select_k
