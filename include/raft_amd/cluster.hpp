@@ -1,0 +1,34 @@
+// k-means device primitives (csrc/kmeans.hip): keyed row reductions, the
+// fused one-X-pass update+verify kernel, and centroid helpers.
+#pragma once
+
+#include "core.hpp"
+
+namespace raft_amd {
+
+// naive atomic keyed reduction (small n) — workspace holds `replicas` copies
+void launch_reduce_rows_by_key(const float* x, const int* keys, float* workspace,
+                               float* out, long long n_rows, long long d,
+                               long long n_keys, int replicas, hipStream_t s);
+// sort-based: perm/keys_sorted from an argsort of keys; atomics only at run
+// boundaries; fused per-key counts
+void launch_reduce_rows_by_key_sorted(const float* x, const int* perm,
+                                      const int* keys_sorted, float* sums,
+                                      float* counts, long long n_rows, long long d,
+                                      hipStream_t s);
+// fused one-X-pass centroid-sum accumulation + exact-fp32 verify/refine of
+// rows whose split-error margin is inconclusive (the bf16x2v engine's
+// update step)
+void launch_kmeans_update_verify(const float* x, const int* perm,
+                                 const int* keys_sorted, const float* c,
+                                 const float* xn, float* dmin, int* amin,
+                                 const float* dmin2, const float* cn_max_dev,
+                                 float* sums, float* counts, long long n_rows,
+                                 long long d, int n_centroids, hipStream_t s);
+// split fp32 -> nslice bf16 slices + squared row norms in one pass
+void launch_split_bf16_norms(const float* c, void* s0, void* s1, void* s2, float* cn,
+                             int nslice, long long n_rows, long long d, hipStream_t s);
+void launch_kmeans_update_centroids(const float* sums, const float* counts, float* c,
+                                    long long k, long long d, hipStream_t s);
+
+}  // namespace raft_amd

@@ -8,11 +8,11 @@ import sys
 import pytest
 import torch
 
-pytestmark = pytest.mark.gpu
 
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
+@pytest.mark.gpu
 @pytest.mark.skipif(not torch.cuda.is_available(), reason="no GPU")
 def test_cpp_smoke(tmp_path):
     objs = [os.path.join(ROOT, "build", "ext", f"{n}.o")
@@ -35,3 +35,51 @@ def test_cpp_smoke(tmp_path):
     r = subprocess.run([exe], capture_output=True, timeout=120)
     assert r.returncode == 0, (r.stdout.decode(), r.stderr.decode())
     assert b"cpp smoke OK" in r.stdout
+
+
+def test_umbrella_header_links(tmp_path):
+    """All per-domain headers (include/raft_amd/*.hpp) must compile together
+    and their declarations must match the kernel objects' symbols — catches
+    header/implementation signature drift. CPU-only: compile + link."""
+    import glob
+    objs = sorted(glob.glob(os.path.join(ROOT, "build", "ext", "*.o")))
+    objs = [o for o in objs if not o.endswith("bindings.o")]
+    if len(objs) < 5:
+        pytest.skip("build/ext objects not present (run build_ext.py)")
+    src = tmp_path / "consumer.cpp"
+    src.write_text(
+        "#include <raft_amd/raft_amd.hpp>\n"
+        "int main() {\n"
+        "  void* fns[] = {\n"
+        "    (void*)&raft_amd::launch_fused_l2nn_split,\n"
+        "    (void*)&raft_amd::launch_fused_l2nn_2d,\n"
+        "    (void*)&raft_amd::fused_l2nn_2d_supported,\n"
+        "    (void*)&raft_amd::launch_l2nn_verify_repair,\n"
+        "    (void*)&raft_amd::launch_select_k,\n"
+        "    (void*)&raft_amd::launch_select_k_warpsort,\n"
+        "    (void*)&raft_amd::launch_reduce_rows<0, float>,\n"
+        "    (void*)&raft_amd::launch_reduce_cols<3, double>,\n"
+        "    (void*)&raft_amd::launch_rows_sqnorm_bf16,\n"
+        "    (void*)&raft_amd::launch_pairwise_l2_mfma,\n"
+        "    (void*)&raft_amd::launch_pairwise_l2_filter,\n"
+        "    (void*)&raft_amd::launch_kmeans_update_verify,\n"
+        "    (void*)&raft_amd::launch_split_bf16_norms,\n"
+        "    (void*)&raft_amd::launch_reduce_rows_by_key_sorted,\n"
+        "    (void*)&raft_amd::launch_csr_spmv<float>,\n"
+        "    (void*)&raft_amd::launch_rng_uniform,\n"
+        "    (void*)&raft_amd::launch_make_blobs,\n"
+        "    (void*)&raft_amd::gemm_bf16_f32_rowmajor,\n"
+        "    (void*)&raft_amd::gemm_bf16_f32_rowmajor_lt,\n"
+        "  };\n"
+        "  for (void* f : fns) if (!f) return 1;\n"
+        "  return 0;\n"
+        "}\n")
+    obj = str(tmp_path / "consumer.o")
+    r = subprocess.run(["hipcc", "--offload-arch=gfx950", "-O1", "-std=c++17",
+                        "-c", str(src), f"-I{os.path.join(ROOT, 'include')}",
+                        "-o", obj], capture_output=True, timeout=300)
+    assert r.returncode == 0, r.stderr.decode()
+    r = subprocess.run(["hipcc", obj, *objs, "-L/opt/rocm/lib", "-lrocblas",
+                        "-lhipblaslt", "-o", str(tmp_path / "consumer")],
+                       capture_output=True, timeout=300)
+    assert r.returncode == 0, r.stderr.decode()
