@@ -16,30 +16,34 @@ import torch
 
 
 def linear_assignment(cost: torch.Tensor, eps_scale: float = 0.15,
-                      max_rounds: int | None = None):
+                      max_rounds: int | None = None, tol: float | None = None):
     """Minimize sum cost[i, assign[i]] over permutations.
 
     Returns (row_assignment [n] int64, total_cost float).
+
+    Auction theory: the final assignment is within n*eps_final of optimal.
+    The classic "eps < 1/n => exact" bound holds for INTEGER costs only;
+    for float costs we eps-scale down to eps_final = tol/n where tol is the
+    requested ABSOLUTE optimality gap (default span * 1e-9).
     """
     assert cost.dim() == 2 and cost.shape[0] == cost.shape[1], "square cost matrix"
     n = cost.shape[0]
     c = -cost.double()                   # auction maximizes value
     span = float((c.max() - c.min()).item()) or 1.0
-    eps = span / 2.0
-    eps_min = 1.0 / (n + 1) * span * 1e-9 + 1e-12
-    # final phase epsilon guarantees optimality margin for the scaled problem
-    final_eps = max(span * 1e-12, 1e-12)
+    if tol is None:
+        tol = span * 1e-9 + 1e-12
+    eps_final = max(tol / n, 1e-14)
 
     price = torch.zeros(n, dtype=torch.float64, device=cost.device)
     owner = torch.full((n,), -1, dtype=torch.int64, device=cost.device)      # col -> row
     assign = torch.full((n,), -1, dtype=torch.int64, device=cost.device)     # row -> col
 
     phases = []
-    e = eps
-    while e > 1.0 / (n + 1):
+    e = span / 2.0
+    while e > eps_final:
         phases.append(e)
         e *= eps_scale
-    phases.append(1.0 / (n + 1))
+    phases.append(eps_final)
 
     for e in phases:
         owner.fill_(-1)
