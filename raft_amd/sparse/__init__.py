@@ -13,7 +13,8 @@ from .linalg import (
     symmetrize_coo, knn_graph_symmetrize, csr_transpose, csr_row_norm,
     csr_degree, csr_add,
 )
-from .op import coo_sort, filter_zeros, dedupe_coo, slice_csr_rows, csr_row_op
+from .op import (coo_sort, filter_zeros, dedupe_coo, slice_csr_rows, csr_row_op,
+                 csr_diagonal, csr_set_diagonal)
 from .select_k import csr_select_k
 from .preprocessing import tfidf_transform, bm25_transform
 from . import solver

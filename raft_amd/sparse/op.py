@@ -49,3 +49,29 @@ def csr_row_op(a: CSR, fn) -> CSR:
     seg = torch.repeat_interleave(torch.arange(a.n_rows, device=a.device), lengths)
     new_vals = fn(a.values, seg)
     return CSR(a.indptr, a.indices, new_vals, a.n_rows, a.n_cols)
+
+
+def csr_diagonal(a: CSR) -> torch.Tensor:
+    """Extract the main diagonal (reference: sparse/matrix/diagonal.cuh).
+    Missing diagonal entries read as 0."""
+    out = torch.zeros(min(a.n_rows, a.n_cols), dtype=a.values.dtype,
+                      device=a.values.device)
+    lengths = a.indptr[1:] - a.indptr[:-1]
+    rows = torch.repeat_interleave(
+        torch.arange(a.n_rows, device=a.values.device), lengths.long())
+    on_diag = a.indices.long() == rows
+    out[rows[on_diag]] = a.values[on_diag]
+    return out
+
+
+def csr_set_diagonal(a: CSR, vec: torch.Tensor) -> CSR:
+    """Set existing diagonal entries to vec[i] (reference diagonal.cuh
+    set_diagonal: only stored positions are written — the sparsity pattern
+    is unchanged)."""
+    lengths = a.indptr[1:] - a.indptr[:-1]
+    rows = torch.repeat_interleave(
+        torch.arange(a.n_rows, device=a.values.device), lengths.long())
+    on_diag = a.indices.long() == rows
+    values = a.values.clone()
+    values[on_diag] = vec.to(values.dtype)[rows[on_diag]]
+    return CSR(a.indptr, a.indices, values, a.n_rows, a.n_cols)

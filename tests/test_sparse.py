@@ -198,3 +198,18 @@ class TestRandomizedSvds:
         u, sv, v = randomized_svds(csr, k=4, n_iter=6, seed=0)
         approx = (u * sv.unsqueeze(0)) @ v.t()
         np.testing.assert_allclose(approx.numpy(), dense, atol=1e-6)
+
+
+class TestCsrDiagonal:
+    def test_extract_and_set(self):
+        from raft_amd.sparse import CSR, csr_diagonal, csr_set_diagonal
+        d = torch.tensor([[1.0, 2.0, 0.0],
+                          [0.0, 0.0, 3.0],
+                          [4.0, 0.0, 5.0]])
+        a = CSR.from_dense(d)
+        torch.testing.assert_close(csr_diagonal(a), torch.tensor([1.0, 0.0, 5.0]))
+        b = csr_set_diagonal(a, torch.tensor([9.0, 8.0, 7.0]))
+        bd = b.to_torch_sparse().to_dense()
+        assert bd[0, 0] == 9.0 and bd[2, 2] == 7.0
+        assert bd[1, 1] == 0.0  # not stored -> pattern unchanged
+        assert bd[0, 1] == 2.0
