@@ -28,8 +28,9 @@ FORBIDDEN_ANYWHERE = ("cuda_runtime", "cublas", "cusparse", "cusolver",
 
 def main() -> int:
     bad = 0
-    for f in sorted(CSRC.glob("*")):
-        if f.suffix not in (".h", ".hip", ".cpp"):
+    files = sorted(CSRC.glob("*")) + sorted((ROOT / "include" / "raft_amd").glob("*.hpp"))
+    for f in files:
+        if f.suffix not in (".h", ".hpp", ".hip", ".cpp"):
             continue
         text = f.read_text()
         rel = f.relative_to(ROOT)
@@ -38,7 +39,7 @@ def main() -> int:
             bad += 1
         for m in re.finditer(r'#include\s+"([^"]+)"', text):
             inc = m.group(1)
-            if not (CSRC / inc).exists():
+            if not (f.parent / inc).exists():
                 print(f"{rel}: quoted include not found: {inc}")
                 bad += 1
         for m in re.finditer(r"#include\s+<([^>]+)>", text):
