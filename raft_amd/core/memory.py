@@ -124,10 +124,9 @@ class TemporaryDeviceBuffer:
         self._write_back = write_back
         dev = torch.device(device) if device is not None else (
             torch.device("cuda") if torch.cuda.is_available() else data.device)
-        if data.device == dev:
-            self._buf = data if not write_back else data  # in-place view
-        else:
-            self._buf = data.to(dev)
+        # same device: pass through zero-copy (writes hit `data` directly);
+        # otherwise copy in (and back on exit when write_back)
+        self._buf = data if data.device == dev else data.to(dev)
 
     def view(self) -> "torch.Tensor":
         return self._buf
