@@ -42,6 +42,25 @@ __device__ __forceinline__ void mfma_stage_tile128(const __bf16* __restrict__ g,
   }
 }
 
+// XCD-contiguous 8x8 super-tiled (row-tile, col-tile) decode for 2D tile
+// grids (guide T1 + super-tiling): the bijective remap gives each XCD a
+// contiguous slot range, and consecutive 64-slot windows decode to an
+// 8x8 super-tile — the window's 8 X panels + 8 C panels (512 KiB at
+// 128-tiles) are read into that XCD's 4 MiB L2 once and reused 8x each.
+// Launch with grid = rg*cg*64 (rg=ceil(R/8), cg=ceil(C/8)); callers must
+// early-return when rt/ct land past R/C (grid inflation on ragged edges).
+__device__ __forceinline__ void xcd_supertile_decode(int rg, long long* rt,
+                                                     long long* ct) {
+  const int nwg = gridDim.x;
+  const int bid = blockIdx.x;
+  const int q = nwg >> 3, r = nwg & 7;
+  const int xcd = bid & 7, slot = bid >> 3;
+  const int t = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + slot;
+  const int g = t >> 6, w = t & 63;
+  *rt = (long long)(g % rg) * 8 + (w & 7);
+  *ct = (long long)(g / rg) * 8 + (w >> 3);
+}
+
 template <int NSLICE>
 __device__ __forceinline__ constexpr int mfma_n_products() {
   return NSLICE == 1 ? 1 : (NSLICE == 2 ? 3 : 6);

@@ -28,7 +28,7 @@ __global__ void pairwise_l2_kernel(const __bf16* __restrict__ x0,
                                    const float* __restrict__ yn,
                                    float* __restrict__ out,
                                    long long m, long long n, int d,
-                                   long long ldo, int sqrt_out) {
+                                   long long ldo, int sqrt_out, int rg) {
   extern __shared__ __bf16 smem[];
   __bf16* xs[NSLICE];
   __bf16* cs[NSLICE];
@@ -43,8 +43,11 @@ __global__ void pairwise_l2_kernel(const __bf16* __restrict__ x0,
   const int lane = threadIdx.x % RAFT_AMD_WAVE;
   const int w = threadIdx.x / RAFT_AMD_WAVE;
   const int wr = w >> 1, wc = w & 1;
-  const long long row0 = (long long)blockIdx.y * 128;
-  const long long col0 = (long long)blockIdx.x * 128;
+  long long rt_, ct_;
+  xcd_supertile_decode(rg, &rt_, &ct_);
+  if (rt_ * 128 >= m || ct_ * 128 >= n) return;  // super-tile ragged edge
+  const long long row0 = rt_ * 128;
+  const long long col0 = ct_ * 128;
 
   f32x4 acc[4][4];
 #pragma unroll
@@ -98,7 +101,7 @@ __global__ void pairwise_l2_filter_kernel(const __bf16* __restrict__ x0,
                                           int* __restrict__ out_i,
                                           int* __restrict__ cnt, int cap,
                                           long long col_offset, long long m,
-                                          long long n, int d) {
+                                          long long n, int d, int rg) {
   extern __shared__ __bf16 smem[];
   __bf16* xs[NSLICE];
   __bf16* cs[NSLICE];
@@ -113,8 +116,11 @@ __global__ void pairwise_l2_filter_kernel(const __bf16* __restrict__ x0,
   const int lane = threadIdx.x % RAFT_AMD_WAVE;
   const int w = threadIdx.x / RAFT_AMD_WAVE;
   const int wr = w >> 1, wc = w & 1;
-  const long long row0 = (long long)blockIdx.y * 128;
-  const long long col0 = (long long)blockIdx.x * 128;
+  long long rt_, ct_;
+  xcd_supertile_decode(rg, &rt_, &ct_);
+  if (rt_ * 128 >= m || ct_ * 128 >= n) return;  // super-tile ragged edge
+  const long long row0 = rt_ * 128;
+  const long long col0 = ct_ * 128;
 
   f32x4 acc[4][4];
 #pragma unroll
@@ -156,7 +162,9 @@ void launch_pairwise_l2_filter(const void** xsl, const void** csl, const float* 
                                int* out_i, int* cnt, int cap, long long col_offset,
                                long long m, long long n, int d, int nslice,
                                hipStream_t stream) {
-  dim3 grid((unsigned)((n + 127) / 128), (unsigned)((m + 127) / 128));
+  const int rg = (int)((m + 1023) / 1024);       // ceil(R/8), R=ceil(m/128)
+  const int cg = (int)((n + 1023) / 1024);
+  dim3 grid((unsigned)((long long)rg * cg * 64));
   const size_t lds = (size_t)nslice * 2 * 8192 * sizeof(__bf16);
   const __bf16* x0 = (const __bf16*)xsl[0];
   const __bf16* x1 = (const __bf16*)(nslice > 1 ? xsl[1] : xsl[0]);
@@ -168,12 +176,12 @@ void launch_pairwise_l2_filter(const void** xsl, const void** csl, const float* 
     case 1:
       hipLaunchKernelGGL((pairwise_l2_filter_kernel<1>), grid, dim3(256), lds, stream,
                          x0, x1, x2, c0, c1, c2, xn, yn, thr, out_d, out_i, cnt, cap,
-                         col_offset, m, n, d);
+                         col_offset, m, n, d, rg);
       break;
     case 2:
       hipLaunchKernelGGL((pairwise_l2_filter_kernel<2>), grid, dim3(256), lds, stream,
                          x0, x1, x2, c0, c1, c2, xn, yn, thr, out_d, out_i, cnt, cap,
-                         col_offset, m, n, d);
+                         col_offset, m, n, d, rg);
       break;
     case 3: {
       static bool attr_set3 = false;
@@ -185,7 +193,7 @@ void launch_pairwise_l2_filter(const void** xsl, const void** csl, const float* 
       }
       hipLaunchKernelGGL((pairwise_l2_filter_kernel<3>), grid, dim3(256), lds, stream,
                          x0, x1, x2, c0, c1, c2, xn, yn, thr, out_d, out_i, cnt, cap,
-                         col_offset, m, n, d);
+                         col_offset, m, n, d, rg);
       break;
     }
     default:
@@ -197,7 +205,9 @@ void launch_pairwise_l2_mfma(const void** xsl, const void** csl, const float* xn
                              const float* yn, float* out, long long m, long long n,
                              int d, long long ldo, int nslice, bool sqrt_out,
                              hipStream_t stream) {
-  dim3 grid((unsigned)((n + 127) / 128), (unsigned)((m + 127) / 128));
+  const int rg = (int)((m + 1023) / 1024);       // ceil(R/8), R=ceil(m/128)
+  const int cg = (int)((n + 1023) / 1024);
+  dim3 grid((unsigned)((long long)rg * cg * 64));
   const size_t lds = (size_t)nslice * 2 * 8192 * sizeof(__bf16);
   const __bf16* x0 = (const __bf16*)xsl[0];
   const __bf16* x1 = (const __bf16*)(nslice > 1 ? xsl[1] : xsl[0]);
@@ -208,11 +218,11 @@ void launch_pairwise_l2_mfma(const void** xsl, const void** csl, const float* xn
   switch (nslice) {
     case 1:
       hipLaunchKernelGGL((pairwise_l2_kernel<1>), grid, dim3(256), lds, stream,
-                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out);
+                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out, rg);
       break;
     case 2:
       hipLaunchKernelGGL((pairwise_l2_kernel<2>), grid, dim3(256), lds, stream,
-                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out);
+                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out, rg);
       break;
     case 3: {
       static bool attr_set = false;
@@ -223,7 +233,7 @@ void launch_pairwise_l2_mfma(const void** xsl, const void** csl, const float* xn
         attr_set = true;
       }
       hipLaunchKernelGGL((pairwise_l2_kernel<3>), grid, dim3(256), lds, stream,
-                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out);
+                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out, rg);
       break;
     }
     default:
@@ -274,7 +284,7 @@ __global__ void pairwise_l2_256_kernel(const __bf16* __restrict__ x0,
                                        const float* __restrict__ yn,
                                        float* __restrict__ out,
                                        long long m, long long n, int d,
-                                       long long ldo, int sqrt_out) {
+                                       long long ldo, int sqrt_out, int rg) {
   constexpr int BLOCK = 512;
   extern __shared__ __bf16 smem[];
   const __bf16* const xg[3] = {x0, x1, x2};
@@ -290,8 +300,11 @@ __global__ void pairwise_l2_256_kernel(const __bf16* __restrict__ x0,
   const int lane = threadIdx.x % RAFT_AMD_WAVE;
   const int w = threadIdx.x / RAFT_AMD_WAVE;
   const int wr = w >> 2, wc = w & 3;  // 2x4: wave tile 128 rows x 64 cols
-  const long long row0 = (long long)blockIdx.y * 256;
-  const long long col0 = (long long)blockIdx.x * 256;
+  long long rt_, ct_;
+  xcd_supertile_decode(rg, &rt_, &ct_);
+  if (rt_ * 256 >= m || ct_ * 256 >= n) return;  // super-tile ragged edge
+  const long long row0 = rt_ * 256;
+  const long long col0 = ct_ * 256;
 
   f32x4 acc[8][4];
 #pragma unroll
@@ -370,7 +383,9 @@ void launch_pairwise_l2_mfma256(const void** xsl, const void** csl, const float*
                                 const float* yn, float* out, long long m, long long n,
                                 int d, long long ldo, int nslice, bool sqrt_out,
                                 hipStream_t stream) {
-  dim3 grid((unsigned)((n + 255) / 256), (unsigned)((m + 255) / 256));
+  const int rg = (int)((m + 2047) / 2048);       // ceil(R/8), R=ceil(m/256)
+  const int cg = (int)((n + 2047) / 2048);
+  dim3 grid((unsigned)((long long)rg * cg * 64));
   const size_t lds = (size_t)nslice * 2 * 16384 * sizeof(__bf16);
   const __bf16* x0 = (const __bf16*)xsl[0];
   const __bf16* x1 = (const __bf16*)(nslice > 1 ? xsl[1] : xsl[0]);
@@ -387,21 +402,21 @@ void launch_pairwise_l2_mfma256(const void** xsl, const void** csl, const float*
       static bool a1 = (set_attr((const void*)&pairwise_l2_256_kernel<1>), true);
       (void)a1;
       hipLaunchKernelGGL((pairwise_l2_256_kernel<1>), grid, dim3(512), lds, stream,
-                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out);
+                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out, rg);
       break;
     }
     case 2: {
       static bool a2 = (set_attr((const void*)&pairwise_l2_256_kernel<2>), true);
       (void)a2;
       hipLaunchKernelGGL((pairwise_l2_256_kernel<2>), grid, dim3(512), lds, stream,
-                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out);
+                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out, rg);
       break;
     }
     default: {
       static bool a3 = (set_attr((const void*)&pairwise_l2_256_kernel<3>), true);
       (void)a3;
       hipLaunchKernelGGL((pairwise_l2_256_kernel<3>), grid, dim3(512), lds, stream,
-                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out);
+                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out, rg);
       break;
     }
   }
@@ -428,7 +443,7 @@ __global__ void pairwise_l2_filter256_kernel(const __bf16* __restrict__ x0,
                                              int* __restrict__ out_i,
                                              int* __restrict__ cnt, int cap,
                                              long long col_offset, long long m,
-                                             long long n, int d) {
+                                             long long n, int d, int rg) {
   constexpr int BLOCK = 512;
   extern __shared__ __bf16 smem[];
   const __bf16* const xg[3] = {x0, x1, x2};
@@ -443,8 +458,11 @@ __global__ void pairwise_l2_filter256_kernel(const __bf16* __restrict__ x0,
   const int lane = threadIdx.x % RAFT_AMD_WAVE;
   const int w = threadIdx.x / RAFT_AMD_WAVE;
   const int wr = w >> 2, wc = w & 3;
-  const long long row0 = (long long)blockIdx.y * 256;
-  const long long col0 = (long long)blockIdx.x * 256;
+  long long rt_, ct_;
+  xcd_supertile_decode(rg, &rt_, &ct_);
+  if (rt_ * 256 >= m || ct_ * 256 >= n) return;  // super-tile ragged edge
+  const long long row0 = rt_ * 256;
+  const long long col0 = ct_ * 256;
 
   f32x4 acc[8][4];
 #pragma unroll
@@ -527,7 +545,9 @@ void launch_pairwise_l2_filter256(const void** xsl, const void** csl, const floa
                                   int* out_i, int* cnt, int cap, long long col_offset,
                                   long long m, long long n, int d, int nslice,
                                   hipStream_t stream) {
-  dim3 grid((unsigned)((n + 255) / 256), (unsigned)((m + 255) / 256));
+  const int rg = (int)((m + 2047) / 2048);       // ceil(R/8), R=ceil(m/256)
+  const int cg = (int)((n + 2047) / 2048);
+  dim3 grid((unsigned)((long long)rg * cg * 64));
   const size_t lds = (size_t)nslice * 2 * 16384 * sizeof(__bf16);
   const __bf16* x0 = (const __bf16*)xsl[0];
   const __bf16* x1 = (const __bf16*)(nslice > 1 ? xsl[1] : xsl[0]);
@@ -544,19 +564,19 @@ void launch_pairwise_l2_filter256(const void** xsl, const void** csl, const floa
     (void)a1;
     hipLaunchKernelGGL((pairwise_l2_filter256_kernel<1>), grid, dim3(512), lds, stream,
                        x0, x1, x2, c0, c1, c2, xn, yn, thr, out_d, out_i, cnt, cap,
-                       col_offset, m, n, d);
+                       col_offset, m, n, d, rg);
   } else if (nslice == 2) {
     static bool a2 = (set_attr((const void*)&pairwise_l2_filter256_kernel<2>), true);
     (void)a2;
     hipLaunchKernelGGL((pairwise_l2_filter256_kernel<2>), grid, dim3(512), lds, stream,
                        x0, x1, x2, c0, c1, c2, xn, yn, thr, out_d, out_i, cnt, cap,
-                       col_offset, m, n, d);
+                       col_offset, m, n, d, rg);
   } else {
     static bool a3 = (set_attr((const void*)&pairwise_l2_filter256_kernel<3>), true);
     (void)a3;
     hipLaunchKernelGGL((pairwise_l2_filter256_kernel<3>), grid, dim3(512), lds, stream,
                        x0, x1, x2, c0, c1, c2, xn, yn, thr, out_d, out_i, cnt, cap,
-                       col_offset, m, n, d);
+                       col_offset, m, n, d, rg);
   }
 }
 
