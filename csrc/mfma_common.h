@@ -148,6 +148,81 @@ __device__ __forceinline__ void mfma_tile_kloop(
 }
 
 // ---------------------------------------------------------------------------
+// 1-slice paired K-loop: stage TWO k-tiles (64 KiB LDS) under ONE
+// vmcnt(0)+barrier drain. MEASURED LOSER on the kNN filter sweep
+// (9873 -> 7409 q/s, d=128): halving the drain count does not pay for the
+// doubled per-drain wait (16 outstanding gloads) — kept only as a recorded
+// experiment, not routed from any kernel.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ void mfma_tile_kloop_1s_pair(
+    const __bf16* __restrict__ xg, const __bf16* __restrict__ cg,
+    __bf16* (&xs)[2], __bf16* (&cs)[2],
+    f32x4 (&acc)[4][4], long long row0, long long col0, int d,
+    long long m_max, long long n_max, int wr, int wc, int lane) {
+  const int k_tiles = d / 64;
+  const int t = threadIdx.x;
+  const int wv = t / RAFT_AMD_WAVE;
+  long long bx[4], bc[4];
+  int ldst[4];
+#pragma unroll
+  for (int j = 0; j < 4; j++) {
+    const int o = j * 4096 + t * 16;
+    const int o_src = mfma_swz(o);
+    const int r = o_src >> 7;
+    const int k = (o_src & 127) >> 1;
+    long long rx = row0 + r;
+    if (rx > m_max) rx = m_max;
+    bx[j] = rx * (long long)d + k;
+    long long rc = col0 + r;
+    if (rc > n_max) rc = n_max;
+    bc[j] = rc * (long long)d + k;
+    ldst[j] = (j * 4096 + wv * 1024) / 2;
+  }
+  for (int kt = 0; kt < k_tiles; kt += 2) {
+    const int pair = (kt + 1 < k_tiles) ? 2 : 1;
+#pragma unroll
+    for (int h = 0; h < 2; h++) {
+      if (h >= pair) break;
+      const long long koff = (long long)(kt + h) * 64;
+#pragma unroll
+      for (int j = 0; j < 4; j++) {
+        GLOAD_LDS(xg + bx[j] + koff, xs[h] + ldst[j]);
+        GLOAD_LDS(cg + bc[j] + koff, cs[h] + ldst[j]);
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+#pragma unroll
+    for (int h = 0; h < 2; h++) {
+      if (h >= pair) break;
+#pragma unroll
+      for (int kf = 0; kf < 2; kf++) {
+        bf16x8 a_frag[4], b_frag[4];
+#pragma unroll
+        for (int fr = 0; fr < 4; fr++) {
+          const int r = wr * 64 + fr * 16 + (lane & 15);
+          const int byte = mfma_swz(r * 128 + (kf * 32 + (lane >> 4) * 8) * 2);
+          a_frag[fr] = *reinterpret_cast<const bf16x8*>((const char*)xs[h] + byte);
+        }
+#pragma unroll
+        for (int fc = 0; fc < 4; fc++) {
+          const int c = wc * 64 + fc * 16 + (lane & 15);
+          const int byte = mfma_swz(c * 128 + (kf * 32 + (lane >> 4) * 8) * 2);
+          b_frag[fc] = *reinterpret_cast<const bf16x8*>((const char*)cs[h] + byte);
+        }
+#pragma unroll
+        for (int fr = 0; fr < 4; fr++)
+#pragma unroll
+          for (int fc = 0; fc < 4; fc++)
+            acc[fr][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_frag[fr], b_frag[fc], acc[fr][fc], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Phased 2-slice K-loop (guide T3/T4: counted s_waitcnt vmcnt instead of a
 // full drain). Load issue order per K-step: [X0 C0 | C1 | X1] (4 gload
 // rounds each interleaved). The slice-product MFMAs are split into three
