@@ -18,8 +18,11 @@ from .types import CSR, COO
 from .convert import csr_to_coo, coo_to_csr, sorted_coo_to_csr
 
 
-def spmv(a: CSR, x: torch.Tensor, out: torch.Tensor | None = None) -> torch.Tensor:
-    """y = A @ x for CSR A [m,n], dense x [n]."""
+def spmv(a, x: torch.Tensor, out: torch.Tensor | None = None) -> torch.Tensor:
+    """y = A @ x for CSR or COO A [m,n], dense x [n] (reference SpMV accepts
+    both container types)."""
+    if isinstance(a, COO):
+        a = coo_to_csr(a)
     assert x.dim() == 1 and x.numel() == a.n_cols
     if a.values.is_cuda and a.values.dtype in (torch.float32, torch.float64):
         ext = require_ext()
@@ -38,8 +41,10 @@ def spmv(a: CSR, x: torch.Tensor, out: torch.Tensor | None = None) -> torch.Tens
     return y
 
 
-def spmm(a: CSR, b: torch.Tensor) -> torch.Tensor:
-    """C = A @ B for dense B [n, k] (rocSPARSE via torch.sparse)."""
+def spmm(a, b: torch.Tensor) -> torch.Tensor:
+    """C = A @ B for dense B [n, k] (rocSPARSE via torch.sparse); CSR or COO."""
+    if isinstance(a, COO):
+        a = coo_to_csr(a)
     return a.to_torch_sparse() @ b
 
 

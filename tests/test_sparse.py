@@ -213,3 +213,18 @@ class TestCsrDiagonal:
         assert bd[0, 0] == 9.0 and bd[2, 2] == 7.0
         assert bd[1, 1] == 0.0  # not stored -> pattern unchanged
         assert bd[0, 1] == 2.0
+
+
+class TestCooSpmv:
+    def test_spmv_spmm_accept_coo(self):
+        from raft_amd.sparse import COO, spmv, spmm
+        d = torch.tensor([[1.0, 0.0, 2.0], [0.0, 3.0, 0.0], [4.0, 0.0, 0.0]])
+        nz = d.nonzero(as_tuple=False)
+        # deliberately unsorted triplets
+        perm = torch.tensor([2, 0, 3, 1])
+        coo = COO(nz[perm, 0].to(torch.int32), nz[perm, 1].to(torch.int32),
+                  d[nz[perm, 0], nz[perm, 1]], 3, 3)
+        x = torch.tensor([1.0, 2.0, 3.0])
+        torch.testing.assert_close(spmv(coo, x), d @ x)
+        b = torch.randn(3, 4)
+        torch.testing.assert_close(spmm(coo, b), d @ b)
