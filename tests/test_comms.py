@@ -79,6 +79,14 @@ def _collective_worker(rank, world):
     c.device_sendrecv(send, dst, recv, src)
     assert torch.equal(recv, torch.full((3,), float(src)))
 
+    # multicast sendrecv: each rank sends to every peer, receives from all
+    msend = torch.full((2,), float(rank) + 10.0)
+    peers = [r for r in range(world) if r != rank]
+    mrecvs = [torch.empty(2) for _ in peers]
+    c.device_multicast_sendrecv(msend, peers, mrecvs, peers)
+    for r, peer in zip(mrecvs, peers):
+        assert torch.equal(r, torch.full((2,), float(peer) + 10.0))
+
     # comm_split: even/odd colors
     sub = c.comm_split(color=rank % 2, key=rank)
     t = torch.ones(1)

@@ -263,3 +263,27 @@ class TestPreprocessingLabelsGpu:
         assert merged[:4].unique().numel() == 1
         mono = make_monotonic(torch.tensor([5, 9, 5], device=dev))
         assert mono.tolist() == [0, 1, 0]
+
+
+class TestBalancedTrustGpu:
+    def test_kmeans_balanced_gpu(self, dev):
+        from raft_amd.cluster import kmeans_balanced_fit
+        from raft_amd.random import make_blobs, RngState
+        x, _, _ = make_blobs(100000, 64, n_clusters=16, cluster_std=0.4,
+                             state=RngState(seed=6), device=dev)
+        model = kmeans_balanced_fit(x, 16, max_iter=10, seed=0)
+        counts = torch.bincount(model.labels, minlength=16).float()
+        # balanced: no cluster more than 3x the mean size
+        assert float(counts.max()) < 3.0 * float(counts.mean())
+        assert torch.isfinite(model.centroids).all()
+
+    def test_trustworthiness_gpu(self, dev):
+        from raft_amd.stats import trustworthiness_score as trustworthiness
+        torch.manual_seed(7)
+        x = torch.randn(2000, 32, device=dev)
+        # identity embedding is perfectly trustworthy
+        t = trustworthiness(x, x.clone(), n_neighbors=8)
+        assert t == pytest.approx(1.0, abs=1e-6)
+        # random embedding is not
+        t2 = trustworthiness(x, torch.randn(2000, 2, device=dev), n_neighbors=8)
+        assert t2 < 0.8
