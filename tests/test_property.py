@@ -102,8 +102,9 @@ class TestLapProps:
         cost = torch.rand(n, n) * 10
         assign, my_cost = linear_assignment(cost)
         assert assign.long().unique().numel() == n  # a permutation
-        ri, ci = linear_sum_assignment(cost.numpy())
-        opt = float(cost.numpy()[ri, ci].sum())
+        cd = cost.double().numpy()       # compare in fp64 (fp32 ties differ)
+        ri, ci = linear_sum_assignment(cd)
+        opt = float(cd[ri, ci].sum())
         # eps scales down to tol/n (absolute), so the gap is ~1e-8-tiny
         assert my_cost <= opt + 1e-6
 
