@@ -33,9 +33,9 @@ class KMeansParams:
     max_iter: int = 20
     tol: float = 1e-4
     seed: int = 0
-    init: str = "kmeans++"      # "kmeans++" | "random" | "array"
+    init: str = "kmeans++"      # "kmeans++" | "scalable" (kmeans||) | "random" | "array"
     n_init: int = 1             # restarts with derived seeds; best inertia wins
-    oversampling: float = 2.0   # for kmeans|| style init (unused by exact ++)
+    oversampling: float = 2.0   # kmeans||: l = oversampling * k samples/round
     fp32_mode: str = "auto"     # GEMM engine for the assignment step
     verbose: bool = False
 
@@ -133,6 +133,67 @@ def _init_plusplus(x: torch.Tensor, k: int, state: RngState, comms: Comms,
     return torch.stack(centers, dim=0)
 
 
+def _init_scalable(x: torch.Tensor, k: int, state: RngState, comms: Comms,
+                   fp32_mode: str, oversampling: float = 2.0,
+                   n_rounds: int = 5) -> torch.Tensor:
+    """kmeans|| (scalable k-means++), distributed-aware.
+
+    Reference parity: the reference's initScalableKMeansPlusPlus — instead of
+    k sequential D^2 draws (k synchronization points), do ~5 rounds that each
+    sample l = oversampling*k points INDEPENDENTLY with probability
+    min(1, l*d2/total), then weight the ~l*rounds candidates by how many
+    points they own and reduce them to k centers with (weighted) k-means++ +
+    a few Lloyd steps on the tiny candidate set.
+    """
+    n_local, d = x.shape
+    world = comms.get_size()
+    l = max(int(oversampling * k), 2)
+    # first center (rank 0's draw wins)
+    u = uniform((1,), state=state, device=x.device)
+    j0 = int(u.item() * n_local) % n_local
+    c0 = x[j0:j0 + 1].clone()
+    if world > 1:
+        c0 = comms.bcast(c0, root=0)
+    cands = [c0]
+    mind2 = fused_l2nn(x, c0, fp32_mode=fp32_mode)[0]
+    for _ in range(n_rounds):
+        total = mind2.sum().reshape(1)
+        if world > 1:
+            comms.allreduce(total, op=ReduceOp.SUM)
+        tot = float(total.item())
+        if tot <= 0:
+            break
+        p = (mind2 * (l / tot)).clamp_(max=1.0)
+        draw = uniform((n_local,), state=state, device=x.device) < p
+        new_c = x[draw]
+        if world > 1:
+            counts = torch.tensor([new_c.shape[0]], device=x.device)
+            all_counts = comms.allgather(counts).reshape(-1)
+            new_c = comms.allgatherv(new_c.reshape(new_c.shape[0], d),
+                                     [int(c) for c in all_counts])
+        if new_c.shape[0] == 0:
+            continue
+        cands.append(new_c)
+        nd2 = fused_l2nn(x, new_c, fp32_mode=fp32_mode)[0]
+        mind2 = torch.minimum(mind2, nd2)
+    cand = torch.cat(cands, dim=0)          # identical on every rank
+    # weight candidates by ownership over the (global) data
+    _, owner = fused_l2nn(x, cand, fp32_mode=fp32_mode)
+    w = torch.bincount(owner, minlength=cand.shape[0]).to(x.dtype)
+    if world > 1:
+        comms.allreduce(w, op=ReduceOp.SUM)
+    if cand.shape[0] <= k:
+        # degenerate: too few candidates — pad with random rows
+        extra = _init_random(x, k - cand.shape[0], state, comms)
+        return torch.cat([cand, extra], dim=0)[:k]
+    # cluster the candidates: weighted k-means++ seeding + weighted Lloyd
+    sub = kmeans_fit(cand, KMeansParams(n_clusters=k, max_iter=10,
+                                        seed=state.seed ^ 0x5bd1e995,
+                                        init="kmeans++", fp32_mode=fp32_mode),
+                     sample_weights=w)
+    return sub.centroids
+
+
 def kmeans_fit(x: torch.Tensor, params: KMeansParams,
                comms: Comms | None = None,
                init_centroids: torch.Tensor | None = None,
@@ -164,6 +225,9 @@ def kmeans_fit(x: torch.Tensor, params: KMeansParams,
         centroids = init_centroids.to(x.device, x.dtype).clone()
     elif params.init == "random":
         centroids = _init_random(x, k, state, comms)
+    elif params.init in ("scalable", "kmeans||"):
+        centroids = _init_scalable(x, k, state, comms, params.fp32_mode,
+                                   oversampling=params.oversampling)
     else:
         centroids = _init_plusplus(x, k, state, comms, params.fp32_mode)
 

@@ -98,3 +98,20 @@ class TestWeightedKMeans:
         m2 = kmeans_fit(x, KMeansParams(n_clusters=2, max_iter=10, init="array"),
                         init_centroids=init)
         assert m2.centroids[0, 0] < m.centroids[0, 0]
+
+
+class TestKMeansScalable:
+    def test_scalable_init_recovers_blobs(self):
+        x, _, centers = make_blobs(5000, 8, n_clusters=10, cluster_std=0.3,
+                                   center_box=(-15, 15), state=RngState(seed=4))
+        m = kmeans_fit(x, KMeansParams(n_clusters=10, max_iter=30, seed=1,
+                                       init="scalable", oversampling=2.0))
+        d = torch.cdist(centers, m.centroids)
+        assert d.min(dim=1).values.max() < 1.0
+        assert m.inertia < 5000 * 8 * 0.3 ** 2 * 3
+
+    def test_scalable_handles_tiny_k(self):
+        x, _, _ = make_blobs(500, 4, n_clusters=2, state=RngState(seed=5))
+        m = kmeans_fit(x, KMeansParams(n_clusters=2, max_iter=10, seed=0,
+                                       init="kmeans||"))
+        assert m.centroids.shape == (2, 4)

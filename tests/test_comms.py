@@ -128,3 +128,26 @@ def test_distributed_kmeans_gloo_world2():
     from tests.conftest import spawn_gloo
 
     spawn_gloo(_kmeans_worker, world_size=2)
+
+
+def _scalable_init_worker(rank, world):
+    from raft_amd.comms import TorchDistComms
+    from raft_amd.cluster import kmeans_fit, KMeansParams
+    from raft_amd.random import make_blobs, RngState
+
+    x, _, centers = make_blobs(800, 6, n_clusters=5, cluster_std=0.3,
+                               center_box=(-12, 12), state=RngState(seed=21))
+    shard = x[rank * 400:(rank + 1) * 400]
+    comms = TorchDistComms()
+    model = kmeans_fit(shard, KMeansParams(n_clusters=5, max_iter=25, seed=2,
+                                           init="scalable"), comms=comms)
+    d = torch.cdist(centers, model.centroids)
+    assert d.min(dim=1).values.max() < 1.0, d.min(dim=1).values
+    g = comms.allgather(model.centroids)
+    assert torch.allclose(g[0], g[world - 1], atol=1e-6)
+
+
+def test_distributed_scalable_init_gloo_world2():
+    from tests.conftest import spawn_gloo
+
+    spawn_gloo(_scalable_init_worker, world_size=2)
