@@ -287,3 +287,17 @@ class TestBalancedTrustGpu:
         # random embedding is not
         t2 = trustworthiness(x, torch.randn(2000, 2, device=dev), n_neighbors=8)
         assert t2 < 0.8
+
+
+class TestScalableInitGpu:
+    def test_kmeans_scalable_gpu(self, dev):
+        from raft_amd.cluster import kmeans_fit, KMeansParams
+        from raft_amd.random import make_blobs, RngState
+        x, _, centers = make_blobs(200000, 64, n_clusters=32, cluster_std=0.4,
+                                   center_box=(-20, 20), state=RngState(seed=9),
+                                   device=dev)
+        m = kmeans_fit(x, KMeansParams(n_clusters=32, max_iter=20, seed=3,
+                                       init="scalable"))
+        d = torch.cdist(centers, m.centroids)
+        assert int((d.min(dim=1).values < 2.0).sum()) >= 31
+        assert m.inertia < 3 * 200000 * 64 * 0.4 ** 2
