@@ -299,7 +299,6 @@ def kmeans_fit(x: torch.Tensor, params: KMeansParams,
             print(f"[kmeans] iter {it} inertia {inertia:.4e} shift {shift:.3e}")
         if shift <= params.tol * params.tol:
             break
-        prev_shift = shift
     return KMeansModel(centroids=centroids, inertia=inertia, n_iter=it, labels=labels)
 
 

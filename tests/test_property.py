@@ -4,7 +4,6 @@ The unit suites pin exact oracles; these check structural invariants over
 randomized shapes/values — the reference's randomized-input test style
  extended with shrinking.
 """
-import numpy as np
 import pytest
 import torch
 from hypothesis import given, settings, strategies as st
