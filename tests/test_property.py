@@ -133,3 +133,64 @@ class TestMatrixProps:
         assert torch.equal(matrix.col_reverse(matrix.col_reverse(x)), x)
         sh = matrix.shift_rows(x, k % (r + 1), fill_value=0.0)
         assert sh.shape == x.shape
+
+
+class TestStatsProps:
+    @given(n=st.integers(2, 400), d=st.integers(1, 10), seed=st.integers(0, 999))
+    def test_meanvar_matches_torch(self, n, d, seed):
+        from raft_amd.stats import meanvar
+        torch.manual_seed(seed)
+        x = torch.randn(n, d) * 3 + 1
+        mu, var = meanvar(x)
+        torch.testing.assert_close(mu.double(), x.double().mean(0), atol=1e-5, rtol=1e-5)
+        torch.testing.assert_close(var.double(), x.double().var(0), atol=1e-4, rtol=1e-4)
+
+    @given(n=st.integers(1, 2000), bins=st.integers(1, 64), seed=st.integers(0, 999))
+    def test_histogram_total_and_support(self, n, bins, seed):
+        from raft_amd.stats import histogram
+        torch.manual_seed(seed)
+        x = torch.randn(n)
+        h = histogram(x, n_bins=bins)
+        assert int(h.sum()) == n
+        assert (h >= 0).all()
+
+    @given(n=st.integers(1, 300), k=st.integers(1, 8), seed=st.integers(0, 99))
+    def test_contingency_marginals(self, n, k, seed):
+        from raft_amd.stats import contingency_matrix
+        torch.manual_seed(seed)
+        a = torch.randint(0, k, (n,))
+        b = torch.randint(0, k, (n,))
+        c = contingency_matrix(a, b)
+        assert int(c.sum()) == n
+        # rows span the label RANGE a.min()..a.max() (reference's
+        # label-range reduction), so marginals are the shifted bincount
+        torch.testing.assert_close(c.sum(dim=1),
+                                   torch.bincount(a - a.min(),
+                                                  minlength=c.shape[0]))
+
+
+class TestBitsetProps:
+    @given(n=st.integers(1, 500), seed=st.integers(0, 999))
+    def test_set_test_flip_count(self, n, seed):
+        from raft_amd.core import Bitset
+        torch.manual_seed(seed)
+        bs = Bitset(n, default=False)  # reference default is all-set
+        idx = torch.randperm(n)[: max(1, n // 3)]
+        bs.set(idx)
+        assert bs.count() == idx.numel()
+        assert bool(bs.test(idx).all())
+        bs.flip()
+        assert bs.count() == n - idx.numel()
+
+
+class TestNormalizeProps:
+    @given(n=st.integers(1, 100), d=st.integers(1, 50), seed=st.integers(0, 999))
+    def test_row_normalize_unit_norm(self, n, d, seed):
+        from raft_amd.linalg import normalize
+        torch.manual_seed(seed)
+        x = torch.randn(n, d) * 10
+        y = normalize(x)
+        norms = y.norm(dim=1)
+        nz = x.norm(dim=1) > 1e-6
+        torch.testing.assert_close(norms[nz], torch.ones(int(nz.sum())),
+                                   atol=1e-4, rtol=1e-4)
