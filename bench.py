@@ -10,10 +10,12 @@ the BASELINE.json config. Strong scaling: the 10M rows are sharded across
 ranks; metric = Lloyd iterations/sec of the WHOLE job (one iteration = full
 assignment of all 10M rows + centroid update + allreduce).
 
-fp32 on CDNA4 has no MFMA; the default engine is split-bf16 fp32 emulation
-(bf16x3: fp32-class accuracy, validated in tests/test_linalg.py and below via
-an inertia cross-check) on the 2.5 PF bf16 matrix cores. --fp32-mode native
-uses rocBLAS SGEMM (157 TF vector ALU) for comparison.
+fp32 on CDNA4 has no MFMA; the default engine is bf16x2v — split-bf16
+emulation on the 2.5 PF bf16 matrix cores with the in-kernel second-best
+margin + exact-fp32 rescan that makes the argmin provably fp32-exact
+(tests/test_gpu_kernels.py). --fp32-mode native uses rocBLAS SGEMM (157 TF
+vector ALU) for comparison; --check measures assignment agreement vs the
+native engine over ALL local rows.
 """
 from __future__ import annotations
 
@@ -144,8 +146,8 @@ def main():
 
     if args.check and use_gpu:
         # assignment agreement between emulated and native fp32 engines
-        d_emul, a_emul = fused_l2nn(x[:65536], c, fp32_mode=args.fp32_mode)
-        d_nat, a_nat = fused_l2nn(x[:65536], c, fp32_mode="native")
+        d_emul, a_emul = fused_l2nn(x, c, fp32_mode=args.fp32_mode)
+        d_nat, a_nat = fused_l2nn(x, c, fp32_mode="native")
         agree = float((a_emul == a_nat).float().mean().item())
         result["config"]["assign_agreement_vs_native_fp32"] = agree
 
