@@ -166,6 +166,17 @@ class TestFusedL2NNMfma:
         assert (amin == ra).float().mean() > 0.999
         assert (amin < 200).all()  # padded columns never win
 
+    def test_unaligned_d_fallback(self, dev, ext):
+        """d % 64 != 0 cannot use the MFMA tile staging -> the chunked
+        expanded-GEMM fallback must produce the same argmin."""
+        torch.manual_seed(8)
+        x = torch.randn(3000, 100, device=dev)
+        y = torch.randn(300, 100, device=dev)
+        from raft_amd.neighbors.fused_l2nn import fused_l2nn
+        dmin, amin = fused_l2nn(x, y, fp32_mode="bf16x3")
+        ref = torch.cdist(x.double(), y.double()) ** 2
+        assert (amin == ref.argmin(dim=1)).float().mean() > 0.999
+
     def test_bf16_input_path(self, dev, ext):
         torch.manual_seed(2)
         x = torch.randn(2048, 128, device=dev).bfloat16()
