@@ -4,11 +4,15 @@ The unit suites pin exact oracles; these check structural invariants over
 randomized shapes/values — the reference's randomized-input test style
  extended with shrinking.
 """
+import os
+
 import pytest
 import torch
 from hypothesis import given, settings, strategies as st
 
-settings.register_profile("ci", deadline=None, max_examples=25)
+# HYP_EXAMPLES=200 for an extended local fuzz; 25 keeps CI fast
+settings.register_profile("ci", deadline=None,
+                          max_examples=int(os.environ.get("HYP_EXAMPLES", "25")))
 settings.load_profile("ci")
 
 
