@@ -115,6 +115,34 @@ class Resources:
         if self.device.type == "cuda":
             torch.cuda.synchronize(self.device)
 
+    # -- workspace memory resource (reference: resource/workspace_resource,
+    # set_workspace_resource / set_workspace_to_pool_resource). On MI355X the
+    # pool IS torch's caching allocator over the 288 GB HBM3E; these
+    # accessors expose the reference's workspace surface over it. -----------
+    def get_workspace(self, shape, dtype=torch.uint8) -> torch.Tensor:
+        """Workspace-backed scratch allocation (pool-recycled, uninitialized)."""
+        return torch.empty(shape, dtype=dtype, device=self.device)
+
+    def set_workspace_limit(self, nbytes: int) -> None:
+        """Cap the device pool (reference limiting-adaptor analog)."""
+        if self.device.type == "cuda":
+            total = torch.cuda.get_device_properties(self.device).total_memory
+            torch.cuda.set_per_process_memory_fraction(
+                min(1.0, nbytes / total), self.device)
+
+    def workspace_stats(self):
+        """(allocated, reserved) bytes of the pool backing workspaces."""
+        if self.device.type != "cuda":
+            return (0, 0)
+        s = torch.cuda.memory_stats(self.device)
+        return (s.get("allocated_bytes.all.current", 0),
+                s.get("reserved_bytes.all.current", 0))
+
+    def empty_workspace_pool(self) -> None:
+        """Release cached pool blocks back to the driver (pool flush)."""
+        if self.device.type == "cuda":
+            torch.cuda.empty_cache()
+
     # -- comms (set by raft_amd.comms) --------------------------------------
     def set_comms(self, comms) -> None:
         self._comms = comms

@@ -252,3 +252,13 @@ class TestOperatorsErrorKvpMath:
             v = buf.view()
             v += 1
         assert torch.equal(x, torch.tensor([1.0, 2.0, 3.0, 4.0]))
+
+
+class TestWorkspaceResource:
+    def test_workspace_accessors(self):
+        from raft_amd.core import Resources
+        r = Resources(torch.device("cpu"))
+        w = r.get_workspace((16, 4), dtype=torch.float32)
+        assert w.shape == (16, 4) and w.dtype == torch.float32
+        assert r.workspace_stats() == (0, 0)   # cpu: no device pool
+        r.empty_workspace_pool()               # no-op, must not raise
