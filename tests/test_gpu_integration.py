@@ -301,3 +301,25 @@ class TestScalableInitGpu:
         d = torch.cdist(centers, m.centroids)
         assert int((d.min(dim=1).values < 2.0).sum()) >= 31
         assert m.inertia < 3 * 200000 * 64 * 0.4 ** 2
+
+
+class TestKMeansWideDim:
+    def test_fast_iterate_d512(self, dev):
+        """d=512 routes update_verify to the MAX_DREG=16 variant; the fast
+        EM loop must match the native-fp32 engine's assignments per step."""
+        from raft_amd.cluster import kmeans_iterate
+        from raft_amd.neighbors.fused_l2nn import fused_l2nn
+        from raft_amd.random import make_blobs, RngState
+        x, _, _ = make_blobs(200000, 512, n_clusters=128, cluster_std=1.0,
+                             center_box=(-8, 8), state=RngState(seed=17),
+                             device=dev)
+        torch.manual_seed(0)
+        c0 = x[torch.randperm(200000, device=dev)[:128]].clone()
+        cv, inertia_v = kmeans_iterate(x, c0.clone(), 2, fp32_mode="bf16x2v")
+        # per-step exactness: assignments for the SAME centroids match native
+        _, av = fused_l2nn(x, c0, fp32_mode="bf16x2v")
+        _, an = fused_l2nn(x, c0, fp32_mode="native")
+        assert float((av == an).float().mean()) == 1.0
+        # the fast loop's result is a valid EM trajectory: inertia decreases
+        _, inertia_v2 = kmeans_iterate(x, cv, 1, fp32_mode="bf16x2v")
+        assert inertia_v2 <= inertia_v * 1.0001
