@@ -316,10 +316,17 @@ class TestKMeansWideDim:
         torch.manual_seed(0)
         c0 = x[torch.randperm(200000, device=dev)[:128]].clone()
         cv, inertia_v = kmeans_iterate(x, c0.clone(), 2, fp32_mode="bf16x2v")
-        # per-step exactness: assignments for the SAME centroids match native
+        # per-step exactness: every chosen centroid's TRUE distance is
+        # fp32-indistinguishable from the optimum (near-ties may resolve
+        # differently than the native engine's expanded-form rounding)
         _, av = fused_l2nn(x, c0, fp32_mode="bf16x2v")
+        idx = torch.randperm(200000, device=dev)[:4000]
+        ref = torch.cdist(x[idx].double(), c0.double()) ** 2
+        opt = ref.min(dim=1).values
+        chosen = ref[torch.arange(4000, device=dev), av[idx]]
+        assert float((chosen - opt).max()) < 5e-2
         _, an = fused_l2nn(x, c0, fp32_mode="native")
-        assert float((av == an).float().mean()) == 1.0
+        assert float((av == an).float().mean()) > 0.9999
         # the fast loop's result is a valid EM trajectory: inertia decreases
         _, inertia_v2 = kmeans_iterate(x, cv, 1, fp32_mode="bf16x2v")
         assert inertia_v2 <= inertia_v * 1.0001
