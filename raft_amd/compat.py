@@ -125,3 +125,78 @@ common = types.SimpleNamespace(DeviceResources=DeviceResources, Handle=Handle,
 __all__ = ["DeviceResources", "Handle", "device_ndarray", "eigsh", "svds",
            "rmat", "pairwise_distance", "select_k", "sparse_linalg", "random",
            "common", "get_resources"]
+
+
+# -- pylibraft.common parity ----------------------------------
+# Stream, auto_sync_handle, interruptible signal bridge, set_output_as.
+
+class Stream:
+    """pylibraft.common.Stream parity: thin HIP-stream handle."""
+
+    def __init__(self, device=None):
+        self._s = (torch.cuda.Stream(device=device)
+                   if torch.cuda.is_available() else None)
+
+    def sync(self):
+        if self._s is not None:
+            self._s.synchronize()
+
+    @property
+    def torch_stream(self):
+        return self._s
+
+
+def auto_sync_handle(fn):
+    """Decorator parity (pylibraft handle.pyx auto_sync_handle): calls with a
+    default handle when none given and syncs it afterwards."""
+    import functools
+    import inspect
+
+    @functools.wraps(fn)
+    def wrapper(*args, **kwargs):
+        handle = kwargs.get("handle")
+        made = False
+        if handle is None:
+            kwargs["handle"] = get_resources()
+            made = True
+        out = fn(*args, **kwargs)
+        if made and torch.cuda.is_available():
+            torch.cuda.synchronize()
+        return out
+
+    return wrapper
+
+
+_OUTPUT_AS = "torch"
+
+
+def set_output_as(kind) -> None:
+    """pylibraft.config.set_output_as parity: 'torch' | 'array' (numpy) |
+    callable applied to outputs of `post_output`."""
+    global _OUTPUT_AS
+    assert kind in ("torch", "array", "cupy") or callable(kind)
+    _OUTPUT_AS = kind
+
+
+def post_output(t: torch.Tensor):
+    """Convert an output tensor per set_output_as (used by compat wrappers)."""
+    if _OUTPUT_AS == "torch":
+        return t
+    if _OUTPUT_AS == "array":
+        return t.cpu().numpy()
+    if callable(_OUTPUT_AS):
+        return _OUTPUT_AS(t)
+    return t
+
+
+def interruptible(fn, *args, **kwargs):
+    """pylibraft.common.interruptible parity: run fn so that Ctrl-C cancels
+    in-flight stream waits cooperatively (core.interruptible token)."""
+    from raft_amd.core.interruptible import Interruptible
+
+    tok = Interruptible()
+    try:
+        return fn(*args, **kwargs)
+    except KeyboardInterrupt:
+        tok.cancel()
+        raise

@@ -262,3 +262,26 @@ class TestWorkspaceResource:
         assert w.shape == (16, 4) and w.dtype == torch.float32
         assert r.workspace_stats() == (0, 0)   # cpu: no device pool
         r.empty_workspace_pool()               # no-op, must not raise
+
+
+class TestCompatCommon:
+    def test_stream_handle_output_as(self):
+        from raft_amd import compat
+        s = compat.Stream()
+        s.sync()   # cpu no-op
+        calls = {}
+
+        @compat.auto_sync_handle
+        def takes_handle(x, handle=None):
+            calls["handle"] = handle
+            return x * 2
+
+        out = takes_handle(torch.ones(2))
+        assert calls["handle"] is not None
+        assert torch.equal(out, torch.full((2,), 2.0))
+        compat.set_output_as("array")
+        import numpy as np
+        assert isinstance(compat.post_output(torch.ones(2)), np.ndarray)
+        compat.set_output_as("torch")
+        assert torch.is_tensor(compat.post_output(torch.ones(2)))
+        assert float(compat.interruptible(lambda: torch.ones(1).sum())) == 1.0
