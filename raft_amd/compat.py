@@ -124,7 +124,8 @@ common = types.SimpleNamespace(DeviceResources=DeviceResources, Handle=Handle,
 
 __all__ = ["DeviceResources", "Handle", "device_ndarray", "eigsh", "svds",
            "rmat", "pairwise_distance", "select_k", "sparse_linalg", "random",
-           "common", "get_resources"]
+           "common", "get_resources", "Stream", "auto_sync_handle",
+           "set_output_as", "post_output", "interruptible"]
 
 
 # -- pylibraft.common parity ----------------------------------
