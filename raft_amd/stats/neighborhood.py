@@ -33,7 +33,9 @@ def trustworthiness_score(x: torch.Tensor, x_embedded: torch.Tensor,
     from raft_amd.distance import pairwise_distance
     d_orig = pairwise_distance(x, x, metric=DistanceType.L2Expanded)
     ranks = d_orig.argsort(dim=1).argsort(dim=1)  # rank of each point per row
-    r = ranks.gather(1, emb_nn) - 1               # self occupies rank 0
+    # self occupies rank 0, so the full-list rank of a non-self point IS its
+    # 1-based rank among non-self points — the standard formula's r(i, j)
+    r = ranks.gather(1, emb_nn)
     penalty = (r - k).clamp_min(0).double().sum()
     norm = n * k * (2.0 * n - 3.0 * k - 1.0)
     return float(1.0 - 2.0 / norm * penalty)

@@ -144,3 +144,55 @@ class TestNeighborhood:
         x = torch.randn(60, 5)
         s = stats.trustworthiness_score(x, x.clone(), n_neighbors=5)
         assert s == pytest.approx(1.0, abs=1e-6)
+
+
+class TestSklearnCrossValidation:
+    """Second independent oracle: scikit-learn (the CPU torch path already
+    pins against hand-derived formulas; sklearn guards formula drift)."""
+
+    def test_clustering_metrics_vs_sklearn(self):
+        import sklearn.metrics as skm
+        from raft_amd import stats
+        torch.manual_seed(0)
+        a = torch.randint(0, 6, (3000,))
+        b = (a + (torch.rand(3000) < 0.3).long() * torch.randint(0, 6, (3000,))) % 6
+        an, bn = a.numpy(), b.numpy()
+        assert stats.adjusted_rand_index(a, b) == pytest.approx(
+            skm.adjusted_rand_score(an, bn), abs=1e-6)
+        assert stats.rand_index(a, b) == pytest.approx(
+            skm.rand_score(an, bn), abs=1e-6)
+        assert stats.mutual_info_score(a, b) == pytest.approx(
+            skm.mutual_info_score(an, bn), abs=1e-6)
+        assert stats.homogeneity_score(a, b) == pytest.approx(
+            skm.homogeneity_score(an, bn), abs=1e-6)
+        assert stats.completeness_score(a, b) == pytest.approx(
+            skm.completeness_score(an, bn), abs=1e-6)
+        assert stats.v_measure(a, b) == pytest.approx(
+            skm.v_measure_score(an, bn), abs=1e-6)
+
+    def test_silhouette_r2_vs_sklearn(self):
+        import sklearn.metrics as skm
+        from raft_amd import stats
+        from raft_amd.random import make_blobs, RngState
+        x, y, _ = make_blobs(1500, 8, n_clusters=4, cluster_std=1.0,
+                             state=RngState(seed=3))
+        assert stats.silhouette_score(x, y, 4) == pytest.approx(
+            skm.silhouette_score(x.numpy(), y.numpy()), abs=1e-4)
+        yt = torch.randn(500)
+        yp = yt + 0.3 * torch.randn(500)
+        assert stats.r2_score(yt, yp) == pytest.approx(
+            skm.r2_score(yt.numpy(), yp.numpy()), abs=1e-5)
+
+    def test_accuracy_trustworthiness_vs_sklearn(self):
+        import sklearn.metrics as skm
+        from sklearn.manifold import trustworthiness as sk_trust
+        from raft_amd import stats
+        torch.manual_seed(1)
+        yt = torch.randint(0, 3, (1000,))
+        yp = torch.where(torch.rand(1000) < 0.8, yt, (yt + 1) % 3)
+        assert stats.accuracy_score(yt, yp) == pytest.approx(
+            skm.accuracy_score(yt.numpy(), yp.numpy()), abs=1e-6)
+        x = torch.randn(400, 16)
+        emb = torch.randn(400, 2)
+        assert stats.trustworthiness_score(x, emb, n_neighbors=7) == pytest.approx(
+            sk_trust(x.numpy(), emb.numpy(), n_neighbors=7), abs=1e-4)
