@@ -198,3 +198,32 @@ class TestRsvdPca:
         assert m.components.shape == (3, 8)
         u, s, vh = torch.linalg.svd(x, full_matrices=False)
         torch.testing.assert_close(m.singular_values, s[:3], rtol=1e-8, atol=1e-8)
+
+
+class TestDecompCrossValidation:
+    def test_pca_matches_sklearn(self):
+        from sklearn.decomposition import PCA
+        from raft_amd.linalg import pca_fit, pca_transform
+        torch.manual_seed(0)
+        x = torch.randn(500, 12) @ torch.randn(12, 12)
+        m = pca_fit(x, 4)
+        sk = PCA(n_components=4).fit(x.numpy())
+        # eigenvalue spectra agree; components up to sign
+        torch.testing.assert_close(m.explained_variance.double(),
+                                   torch.from_numpy(sk.explained_variance_).double(),
+                                   rtol=1e-4, atol=1e-4)
+        z = pca_transform(m, x).numpy()
+        zs = sk.transform(x.numpy())
+        import numpy as np
+        for c in range(4):
+            assert min(np.abs(z[:, c] - zs[:, c]).max(),
+                       np.abs(z[:, c] + zs[:, c]).max()) < 1e-3
+
+    def test_rsvd_matches_scipy_spectrum(self):
+        import numpy as np
+        from raft_amd.linalg import rsvd
+        torch.manual_seed(1)
+        a = torch.randn(300, 6) @ torch.randn(6, 200) + 0.01 * torch.randn(300, 200)
+        _, s, _ = rsvd(a, k=6, n_iter=4, seed=0)
+        ref = np.linalg.svd(a.numpy(), compute_uv=False)[:6]
+        np.testing.assert_allclose(s.numpy(), ref, rtol=1e-3)

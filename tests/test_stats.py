@@ -196,3 +196,14 @@ class TestSklearnCrossValidation:
         emb = torch.randn(400, 2)
         assert stats.trustworthiness_score(x, emb, n_neighbors=7) == pytest.approx(
             sk_trust(x.numpy(), emb.numpy(), n_neighbors=7), abs=1e-4)
+
+    def test_regression_metrics_vs_sklearn(self):
+        import sklearn.metrics as skm
+        from raft_amd.stats import regression_metrics
+        torch.manual_seed(2)
+        yt = torch.randn(999)  # odd n: torch.median = lower middle, sklearn
+        yp = yt + torch.randn(999) * 0.5   # interpolates — odd length matches
+        mae, mse, medae = regression_metrics(yt, yp)
+        assert mae == pytest.approx(skm.mean_absolute_error(yt, yp), abs=1e-6)
+        assert mse == pytest.approx(skm.mean_squared_error(yt, yp), abs=1e-6)
+        assert medae == pytest.approx(skm.median_absolute_error(yt, yp), abs=1e-6)
