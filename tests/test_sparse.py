@@ -228,3 +228,24 @@ class TestCooSpmv:
         torch.testing.assert_close(spmv(coo, x), d @ x)
         b = torch.randn(3, 4)
         torch.testing.assert_close(spmm(coo, b), d @ b)
+
+
+class TestLaplacianVsScipy:
+    def test_laplacian_matches_scipy(self):
+        import numpy as np
+        import scipy.sparse as sp
+        from scipy.sparse.csgraph import laplacian as sp_lap
+        from raft_amd.sparse import CSR, laplacian, laplacian_normalized
+        torch.manual_seed(0)
+        dense = (torch.rand(40, 40) < 0.2).float()
+        dense = ((dense + dense.T) > 0).float()
+        dense.fill_diagonal_(0)
+        a = CSR.from_dense(dense)
+        ref = sp_lap(sp.csr_matrix(dense.numpy())).toarray()
+        torch.testing.assert_close(laplacian(a).to_torch_sparse().to_dense(),
+                                   torch.from_numpy(ref).float(),
+                                   atol=1e-5, rtol=1e-5)
+        refn = sp_lap(sp.csr_matrix(dense.numpy()), normed=True).toarray()
+        torch.testing.assert_close(
+            laplacian_normalized(a).to_torch_sparse().to_dense(),
+            torch.from_numpy(refn).float(), atol=1e-5, rtol=1e-5)

@@ -207,3 +207,13 @@ class TestSklearnCrossValidation:
         assert mae == pytest.approx(skm.mean_absolute_error(yt, yp), abs=1e-6)
         assert mse == pytest.approx(skm.mean_squared_error(yt, yp), abs=1e-6)
         assert medae == pytest.approx(skm.median_absolute_error(yt, yp), abs=1e-6)
+
+    def test_entropy_vs_scipy(self):
+        from scipy.stats import entropy as sp_entropy
+        from raft_amd.stats import entropy
+        import numpy as np
+        torch.manual_seed(3)
+        labels = torch.randint(0, 7, (2000,))
+        counts = torch.bincount(labels).numpy()
+        assert entropy(labels) == pytest.approx(
+            float(sp_entropy(counts / counts.sum())), abs=1e-6)
