@@ -151,3 +151,30 @@ def test_distributed_scalable_init_gloo_world2():
     from tests.conftest import spawn_gloo
 
     spawn_gloo(_scalable_init_worker, world_size=2)
+
+
+class TestLoopbackFidelity:
+    """LoopbackComms must be shape/dtype/value-faithful for every op the
+    algorithms use — it stands in for RCCL in most unit tests."""
+
+    def test_all_ops_shapes_dtypes(self):
+        from raft_amd.comms import LoopbackComms, ReduceOp
+        c = LoopbackComms()
+        assert c.get_size() == 1 and c.get_rank() == 0
+        for dtype in (torch.float32, torch.float64, torch.int64):
+            for shape in ((3,), (2, 5), (1,)):
+                t = (torch.randn(shape) * 10).to(dtype)
+                orig = t.clone()
+                c.allreduce(t, op=ReduceOp.SUM)
+                assert torch.equal(t, orig)          # world=1: identity
+                g = c.allgather(orig)
+                assert g.shape == (1,) + tuple(shape)
+                assert torch.equal(g[0], orig)
+                b = c.bcast(orig.clone(), root=0)
+                assert torch.equal(b, orig)
+        v = torch.arange(6, dtype=torch.float32).reshape(3, 2)
+        av = c.allgatherv(v, [3])
+        assert torch.equal(av, v)
+        c.barrier()
+        sub = c.comm_split(color=0, key=0)
+        assert sub.get_size() == 1
