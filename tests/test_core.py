@@ -285,3 +285,13 @@ class TestCompatCommon:
         compat.set_output_as("torch")
         assert torch.is_tensor(compat.post_output(torch.ones(2)))
         assert float(compat.interruptible(lambda: torch.ones(1).sum())) == 1.0
+
+
+class TestUtils:
+    def test_helpers(self):
+        from raft_amd.utils import ceil_div, next_pow2, row_chunks, as_2d
+        assert ceil_div(10, 3) == 4 and ceil_div(9, 3) == 3
+        assert next_pow2(1) == 1 and next_pow2(5) == 8 and next_pow2(64) == 64
+        chunks = list(row_chunks(10, 4))
+        assert chunks == [(0, 4), (4, 8), (8, 10)]
+        assert as_2d(torch.arange(6)).shape[0] == 1 or as_2d(torch.arange(6)).dim() == 2
