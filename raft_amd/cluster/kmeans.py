@@ -97,6 +97,20 @@ def _init_plusplus(x: torch.Tensor, k: int, state: RngState, comms: Comms,
             sums = local_sum.reshape(1)
         total = float(sums.sum().item())
         cdf = torch.cumsum(mind2, dim=0)
+        if world == 1:
+            # vectorized trial sampling: ONE uniform draw + ONE searchsorted
+            # for all L candidates (the per-trial .item() loop costs ~2
+            # host syncs x k x L on GPU — ~12k syncs at k=1024)
+            us = uniform((n_trials,), state=state, device=x.device).double() * total
+            js = torch.searchsorted(cdf, us.to(cdf.dtype)).clamp_(max=n_local - 1)
+            cmat = x[js]                                       # [L, d]
+            d2 = (xsq.unsqueeze(1) + (cmat * cmat).sum(dim=1)
+                  - 2.0 * (x @ cmat.T)).clamp_min_(0).double()
+            pot = torch.minimum(mind2.unsqueeze(1), d2).sum(dim=0)
+            best = int(pot.argmin().item())
+            centers.append(cmat[best])
+            mind2 = torch.minimum(mind2, d2[:, best])
+            continue
         cands = []
         for t in range(n_trials):
             u = float(uniform((1,), state=state, device=x.device).item()) * total
