@@ -480,10 +480,13 @@ class KMeans:
     """Estimator-style wrapper."""
 
     def __init__(self, n_clusters: int = 8, max_iter: int = 20, tol: float = 1e-4,
-                 seed: int = 0, init: str = "kmeans++", fp32_mode: str = "auto",
+                 seed: int = 0, init: str = "kmeans++", n_init: int = 1,
+                 oversampling: float = 2.0, fp32_mode: str = "auto",
                  verbose: bool = False):
         self.params = KMeansParams(n_clusters=n_clusters, max_iter=max_iter, tol=tol,
-                                   seed=seed, init=init, fp32_mode=fp32_mode, verbose=verbose)
+                                   seed=seed, init=init, n_init=n_init,
+                                   oversampling=oversampling, fp32_mode=fp32_mode,
+                                   verbose=verbose)
         self.model: KMeansModel | None = None
 
     def fit(self, x: torch.Tensor, comms: Comms | None = None) -> "KMeans":
@@ -503,3 +506,22 @@ class KMeans:
 
     def transform(self, x: torch.Tensor) -> torch.Tensor:
         return kmeans_transform(self.model, x, fp32_mode=self.params.fp32_mode)
+
+    def fit_predict(self, x: torch.Tensor, comms: Comms | None = None) -> torch.Tensor:
+        return self.fit(x, comms=comms).predict(x)
+
+    def fit_transform(self, x: torch.Tensor, comms: Comms | None = None) -> torch.Tensor:
+        return self.fit(x, comms=comms).transform(x)
+
+    def score(self, x: torch.Tensor) -> float:
+        """Negative inertia of x under the fitted centroids (sklearn convention)."""
+        dmin, _ = fused_l2nn(x, self.model.centroids, fp32_mode=self.params.fp32_mode)
+        return -float(torch.sum(dmin, dtype=torch.float64))
+
+    @property
+    def labels_(self):
+        return self.model.labels
+
+    @property
+    def n_iter_(self):
+        return self.model.n_iter

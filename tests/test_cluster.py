@@ -115,3 +115,17 @@ class TestKMeansScalable:
         m = kmeans_fit(x, KMeansParams(n_clusters=2, max_iter=10, seed=0,
                                        init="kmeans||"))
         assert m.centroids.shape == (2, 4)
+
+
+class TestEstimatorSurface:
+    def test_fit_predict_transform_score(self):
+        x, y, _ = make_blobs(600, 6, n_clusters=4, cluster_std=0.3,
+                             state=RngState(seed=8))
+        km = KMeans(n_clusters=4, max_iter=20, seed=1, n_init=2)
+        labels = km.fit_predict(x)
+        assert labels.shape == (600,)
+        t = km.fit_transform(x)
+        assert t.shape == (600, 4)
+        s = km.score(x)
+        assert s <= 0 and s == pytest.approx(-km.inertia_, rel=1e-3)
+        assert km.labels_ is not None and km.n_iter_ >= 1
