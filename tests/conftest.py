@@ -49,6 +49,12 @@ def spawn_gloo(fn, world_size=2, args=()):
         procs.append(p)
     for p in procs:
         p.join(180)
+    # terminate any worker that outlived the join timeout before asserting,
+    # so a hang can't leak processes past the test session
+    hung = [p for p in procs if p.exitcode is None]
+    for p in hung:
+        p.terminate()
+        p.join(10)
     for p in procs:
         assert p.exitcode == 0, f"worker exited with {p.exitcode}"
 
