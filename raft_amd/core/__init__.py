@@ -19,7 +19,7 @@ from .interruptible import Interruptible, synchronize as interruptible_synchroni
 from .logger import get_logger, set_level
 from .trace import annotate, annotated
 from .memory import MemoryStats, TrackingScope, ResourceMonitor, TemporaryDeviceBuffer
-from .mdbuffer import MDBuffer, MemoryType, memory_type_dispatcher
+from .mdbuffer import MDBuffer, MemoryType, memory_type_dispatcher, copy_mdspan
 from .error import RaftError, LogicError, HipError, expects, fail
 from .kvp import KeyValuePair
 from . import operators
@@ -34,5 +34,5 @@ __all__ = [
     "Bitset", "Interruptible", "interruptible_synchronize",
     "get_logger", "set_level", "annotate", "annotated",
     "MemoryStats", "TrackingScope", "ResourceMonitor",
-    "MDBuffer", "MemoryType", "memory_type_dispatcher",
+    "MDBuffer", "MemoryType", "memory_type_dispatcher", "copy_mdspan",
 ]

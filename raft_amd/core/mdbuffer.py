@@ -60,3 +60,17 @@ def memory_type_dispatcher(buf: "MDBuffer | torch.Tensor", device_fn, host_fn):
     """Route by location (memory_type_dispatcher.cuh parity)."""
     t = buf.view() if isinstance(buf, MDBuffer) else buf
     return device_fn(t) if t.is_cuda else host_fn(t)
+
+
+def copy_mdspan(dst: torch.Tensor, src: torch.Tensor) -> torch.Tensor:
+    """Layout/location/dtype-converting copy (reference: core/detail/copy.hpp
+    mdspan_copyable dispatch — cudaMemcpy for same-layout, tiled conversion
+    kernel otherwise). Torch's copy_ engine performs the same dispatch on
+    ROCm: a contiguous same-dtype pair becomes one hipMemcpy; transposed or
+    dtype-converting pairs run a vectorized conversion kernel; cross-device
+    pairs stage through DMA. Shapes must match; dst's layout/device/dtype
+    win."""
+    if dst.shape != src.shape:
+        raise ValueError(f"shape mismatch {tuple(dst.shape)} vs {tuple(src.shape)}")
+    dst.copy_(src)
+    return dst

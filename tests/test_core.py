@@ -295,3 +295,16 @@ class TestUtils:
         chunks = list(row_chunks(10, 4))
         assert chunks == [(0, 4), (4, 8), (8, 10)]
         assert as_2d(torch.arange(6)).shape[0] == 1 or as_2d(torch.arange(6)).dim() == 2
+
+
+class TestCopyMdspan:
+    def test_layout_dtype_conversion(self):
+        from raft_amd.core import copy_mdspan
+        src = torch.arange(12, dtype=torch.float64).reshape(3, 4)
+        # layout conversion: dst is a transposed (non-contiguous) view target
+        dst = torch.empty(4, 3, dtype=torch.float32).t()
+        copy_mdspan(dst, src)
+        torch.testing.assert_close(dst, src.float())
+        assert not dst.is_contiguous()
+        with pytest.raises(ValueError):
+            copy_mdspan(torch.empty(2, 2), src)
