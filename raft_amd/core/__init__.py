@@ -22,6 +22,7 @@ from .memory import MemoryStats, TrackingScope, ResourceMonitor, TemporaryDevice
 from .mdbuffer import MDBuffer, MemoryType, memory_type_dispatcher, copy_mdspan
 from .error import RaftError, LogicError, HipError, expects, fail
 from .kvp import KeyValuePair
+from .span import host_span, device_span, subspan
 from . import operators
 from . import math
 
@@ -35,4 +36,5 @@ __all__ = [
     "get_logger", "set_level", "annotate", "annotated",
     "MemoryStats", "TrackingScope", "ResourceMonitor",
     "MDBuffer", "MemoryType", "memory_type_dispatcher", "copy_mdspan",
+    "host_span", "device_span", "subspan",
 ]

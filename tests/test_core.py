@@ -308,3 +308,19 @@ class TestCopyMdspan:
         assert not dst.is_contiguous()
         with pytest.raises(ValueError):
             copy_mdspan(torch.empty(2, 2), src)
+
+
+class TestSpan:
+    def test_spans(self):
+        from raft_amd.core import host_span, subspan, device_span
+        x = torch.arange(12, dtype=torch.float32).reshape(3, 4)
+        s = host_span(x)
+        assert s.data_ptr() == x.data_ptr() and s.numel() == 12
+        sub = subspan(s, 4, 4)
+        assert torch.equal(sub, torch.tensor([4.0, 5, 6, 7]))
+        sub[0] = 99.0                      # non-owning: writes through
+        assert x[1, 0] == 99.0
+        with pytest.raises(IndexError):
+            subspan(s, 10, 5)
+        with pytest.raises(TypeError):
+            device_span(x)                 # host tensor -> must raise on CPU
