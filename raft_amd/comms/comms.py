@@ -152,6 +152,9 @@ class LoopbackComms(Comms):
     def device_sendrecv(self, send, dst, recv, src):
         raise RuntimeError("p2p on a 1-rank loopback communicator")
 
+    def device_multicast_sendrecv(self, send, dsts, recvs, srcs):
+        raise RuntimeError("p2p on a 1-rank loopback communicator")
+
 
 class TorchDistComms(Comms):
     """torch.distributed-backed communicator (RCCL on GPU, gloo on CPU)."""
