@@ -82,6 +82,18 @@ def _knn_gpu_filtered(x, queries, k, fp32_mode, index_chunk: int = 4_000_000):
         jv, _ = select_k(ds, min(j, s), select_min=True)
         thr[s0:s1] = jv[:, -1]
     del ds
+    # inflate thresholds by 2x the provable split-emulation error bound so a
+    # true neighbor can never be filtered out by split rounding (one bound
+    # covers the sampled threshold reading low, one the candidate reading
+    # high). bf16 input (1 slice) has no split error. Same bound family as
+    # the verified L2NN engine (csrc/kmeans.hip l2nn verify).
+    nslice = len(q_slices)
+    if nslice >= 2:
+        xm = float(xn_full.max())
+        lead = 2.0 ** -13 if nslice == 2 else 2.0 ** -21
+        tail = 2.0 ** -18 if nslice == 2 else 2.0 ** -24
+        thr = thr + 4.0 * (lead * torch.sqrt(qn.clamp_min(0) * xm)
+                           + tail * (qn + xm))
 
     # ---- 2. filtered emission over the full index --------------------------
     cap = max(16384, 32 * k)
