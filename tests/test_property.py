@@ -198,3 +198,42 @@ class TestNormalizeProps:
         nz = x.norm(dim=1) > 1e-6
         torch.testing.assert_close(norms[nz], torch.ones(int(nz.sum())),
                                    atol=1e-4, rtol=1e-4)
+
+
+class TestSerializeProps:
+    @given(r=st.integers(0, 20), c=st.integers(1, 20),
+           dt=st.sampled_from(["float32", "float64", "int32", "int64", "uint8"]),
+           seed=st.integers(0, 999))
+    def test_npy_roundtrip(self, r, c, dt, seed):
+        import io
+        from raft_amd.core import serialize_mdspan, deserialize_mdspan
+        torch.manual_seed(seed)
+        dtype = getattr(torch, dt)
+        if dtype.is_floating_point:
+            x = torch.randn(r, c).to(dtype)
+        else:
+            x = torch.randint(0, 100, (r, c), dtype=dtype)
+        buf = io.BytesIO()
+        serialize_mdspan(buf, x)
+        buf.seek(0)
+        y = deserialize_mdspan(buf)
+        assert y.dtype == x.dtype and torch.equal(y, x)
+
+    @given(r=st.integers(1, 16), c=st.integers(1, 16), seed=st.integers(0, 99))
+    def test_npy_numpy_interchange(self, r, c, seed):
+        """Our .npy bytes load in NumPy and vice versa (format parity)."""
+        import io
+        import numpy as np
+        from raft_amd.core import serialize_mdspan, deserialize_mdspan
+        torch.manual_seed(seed)
+        x = torch.randn(r, c)
+        buf = io.BytesIO()
+        serialize_mdspan(buf, x)
+        buf.seek(0)
+        arr = np.load(buf)
+        assert np.array_equal(arr, x.numpy())
+        buf2 = io.BytesIO()
+        np.save(buf2, arr)
+        buf2.seek(0)
+        y = deserialize_mdspan(buf2)
+        assert torch.equal(y, x)
