@@ -110,3 +110,26 @@ class TestMvg:
         emp_cov = torch.cov(s.t())
         torch.testing.assert_close(emp_mean, mean, rtol=0.05, atol=0.05)
         torch.testing.assert_close(emp_cov, cov, rtol=0.08, atol=0.08)
+
+
+class TestDistributionShapes:
+    """Kolmogorov-Smirnov vs scipy's exact CDFs (beyond the moment checks):
+    catches shape errors moments can't (e.g. a wrong tail transform)."""
+
+    @pytest.mark.parametrize("name,fn_kwargs,scipy_dist", [
+        ("uniform", {}, ("uniform", ())),
+        ("normal", {}, ("norm", ())),
+        ("exponential", {"lambda_": 1.0}, ("expon", ())),
+        ("rayleigh", {"sigma": 1.0}, ("rayleigh", ())),
+        ("laplace", {"mu": 0.0, "scale": 1.0}, ("laplace", ())),
+        ("gumbel", {"mu": 0.0, "beta": 1.0}, ("gumbel_r", ())),
+        ("logistic", {"mu": 0.0, "scale": 1.0}, ("logistic", ())),
+        ("lognormal", {"mu": 0.0, "sigma": 1.0}, ("lognorm", (1.0,))),
+    ])
+    def test_ks_vs_scipy(self, name, fn_kwargs, scipy_dist):
+        import scipy.stats as ss
+        fn = getattr(rnd, name)
+        x = fn((100000,), state=RngState(seed=11), **fn_kwargs).double().numpy()
+        dist_name, args = scipy_dist
+        stat, pvalue = ss.kstest(x, dist_name, args=args)
+        assert pvalue > 1e-4, f"{name}: KS stat {stat:.4f}, p {pvalue:.2e}"
