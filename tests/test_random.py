@@ -94,6 +94,17 @@ class TestRmat:
         # skewed quadrant probabilities put most mass at low ids
         assert (src < 128).float().mean() > 0.7
 
+    def test_bit_recursion_matches_theta(self):
+        """Each of the r_scale recursion levels independently picks the
+        row-half with P(top) = a + b: the per-LEVEL frequency must match,
+        not just the top-level split."""
+        a, b, c = 0.6, 0.15, 0.15
+        src, dst = rnd.rmat(10, 10, 100000, a=a, b=b, c=c, state=RngState(seed=3))
+        for level in range(10):
+            bit = (src >> (9 - level)) & 1
+            p_top = float((bit == 0).float().mean())
+            assert abs(p_top - (a + b)) < 0.02, (level, p_top)
+
 
 class TestPermute:
     def test_is_permutation(self):
