@@ -144,3 +144,14 @@ class TestDistributionShapes:
         dist_name, args = scipy_dist
         stat, pvalue = ss.kstest(x, dist_name, args=args)
         assert pvalue > 1e-4, f"{name}: KS stat {stat:.4f}, p {pvalue:.2e}"
+
+
+class TestMvgDecomposers:
+    @pytest.mark.parametrize("method", ["chol", "eig", "qr"])
+    def test_all_decomposers_recover_cov(self, method):
+        mean = torch.tensor([1.0, -2.0, 0.5])
+        cov = torch.tensor([[2.0, 0.5, 0.1], [0.5, 1.0, 0.2], [0.1, 0.2, 1.5]])
+        s = rnd.multi_variable_gaussian(mean, cov, 200_000, method=method,
+                                        state=RngState(seed=6))
+        torch.testing.assert_close(s.mean(dim=0), mean, atol=0.03, rtol=0)
+        torch.testing.assert_close(torch.cov(s.t()), cov, atol=0.05, rtol=0.05)
