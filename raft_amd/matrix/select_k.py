@@ -35,6 +35,12 @@ def select_k(x: torch.Tensor, k: int, select_min: bool = True,
     """Per-row k smallest (or largest) values of a [batch, len] matrix.
 
     Returns (values [batch,k], indices [batch,k] int64).
+
+    NaN semantics: for select_min=True both native engines order NaN after
+    every finite value (never selected while >= k finite values exist).
+    For select_min=False the engines differ (radix orders NaN above +inf,
+    warpsort's comparison filter drops it) — don't rely on NaN placement
+    in max-selection.
     """
     assert x.dim() == 2
     batch, n = x.shape

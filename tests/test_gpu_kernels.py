@@ -498,3 +498,24 @@ class TestGemmGpu:
         err_native = ((a @ b).double() - ref).abs().max()
         err3 = (gemm_fp32_emulated(a, b, "bf16x3").double() - ref).abs().max()
         assert float(err3) < float(err_native) * 8, (float(err3), float(err_native))
+
+
+class TestSelectKNaN:
+    """NaN semantics (documented in matrix/select_k.py): for MIN selection
+    both engines order NaN after every finite value (ordered-uint encoding /
+    comparison filtering), so NaNs are never selected while enough finite
+    values exist. MAX selection is engine-dependent (radix treats NaN as
+    above +inf, warpsort's comparisons drop it) — not asserted."""
+
+    @pytest.mark.parametrize("algo_name", ["radix", "warpsort"])
+    def test_nan_never_selected_min(self, dev, ext, algo_name):
+        from raft_amd.matrix import select_k, SelectAlgo
+        torch.manual_seed(0)
+        x = torch.randn(8, 6000 if algo_name == "radix" else 500, device=dev)
+        x[:, :7] = float("nan")
+        algo = SelectAlgo.RADIX if algo_name == "radix" else SelectAlgo.WARPSORT
+        vals, idx = select_k(x, 16, select_min=True, algo=algo)
+        assert not torch.isnan(vals).any()
+        finite = torch.nan_to_num(x, nan=float("inf"))
+        ref = torch.topk(finite, 16, dim=1, largest=False).values
+        torch.testing.assert_close(vals, ref)
