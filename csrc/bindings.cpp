@@ -1,6 +1,9 @@
 // Python bindings for the raft_amd native extension (raft_amd._C).
 // Torch tensors in, torch tensors out; every launch goes onto the current
 // PyTorch HIP stream so the ops compose with torch's own kernels.
+// Reference parity: the raft_runtime instantiation layer + pylibraft's
+// Cython bridges (cpp/src/*, python/pylibraft/**/*.pyx) collapsed into one
+// torch-extension TU.
 
 #include <torch/extension.h>
 

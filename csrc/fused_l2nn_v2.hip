@@ -1,4 +1,6 @@
 // Fused L2-NN, persistent-X variant (the k-means assignment hot kernel).
+// Reference parity: the same fusedL2NN contract as fused_l2nn.hip — these
+// are alternative CDNA4 schedules (A/B results in BASELINE.md).
 //
 // v1 (fused_l2nn.hip) re-stages the X tile for every 128-centroid tile:
 // X HBM traffic = n/128 full passes (measured ~60% of kernel time at

@@ -1,4 +1,6 @@
 // Shared machinery for split-bf16 MFMA contraction kernels (gfx950).
+// Reference parity: the contractions engine role (raft/linalg/contractions.cuh
+// Contractions_NT, KernelPolicy) re-derived for 64-wide wavefronts and MFMA.
 // See fused_l2nn.hip header comment for the design rationale (tile geometry,
 // XOR swizzle + global_load_lds both-sides rule, slice-product emulation).
 #pragma once
