@@ -324,3 +324,24 @@ class TestSpan:
             subspan(s, 10, 5)
         with pytest.raises(TypeError):
             device_span(x)                 # host tensor -> must raise on CPU
+
+
+class TestResourcesStreamPool:
+    def test_pool_round_robin_indexing(self):
+        from raft_amd.core import Resources
+        r = Resources(torch.device("cpu"), stream_pool_size=0)
+        # empty pool: stream_from_pool falls back to the main stream (None on CPU)
+        assert r.stream_from_pool(3) is r.stream
+        # type-indexed registry: custom resources are lazy and cached
+        calls = []
+        r.add_resource_factory("my_res", lambda: calls.append(1) or {"x": 1})
+        a = r.get_resource("my_res")
+        b = r.get_resource("my_res")
+        assert a is b and calls == [1]
+
+    def test_clone_shares_factories(self):
+        from raft_amd.core import Resources
+        r = Resources(torch.device("cpu"))
+        r.add_resource_factory("shared", lambda: object())
+        c = r.clone()
+        assert c.get_resource("shared") is not None
