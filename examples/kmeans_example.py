@@ -18,12 +18,15 @@ from raft_amd.random import make_blobs, RngState
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--rows", type=int, default=100_000)
+    ap.add_argument("--rows", type=int, default=None,
+                    help="default: 1M on GPU, 20k on CPU (oracle path is slow)")
     ap.add_argument("--dim", type=int, default=64)
     ap.add_argument("--k", type=int, default=64)
     args = ap.parse_args()
 
     dev = "cuda" if torch.cuda.is_available() else "cpu"
+    if args.rows is None:
+        args.rows = 1_000_000 if dev == "cuda" else 20_000
     x, y_true, centers = make_blobs(args.rows, args.dim, n_clusters=args.k,
                                     cluster_std=0.5, state=RngState(seed=0),
                                     device=dev)
