@@ -128,6 +128,30 @@ class TestInterruptible:
     def test_sync_noop_on_cpu(self):
         Interruptible().synchronize()
 
+    def test_cross_thread_cancel(self):
+        """reference semantics: cancel() from ANOTHER thread interrupts the
+        spinning waiter (interruptible.hpp:64 token-per-thread design)."""
+        import threading
+        import time as _t
+        tok = Interruptible()
+        hit = {}
+
+        def waiter():
+            try:
+                for _ in range(20000):
+                    tok.check()
+                    _t.sleep(0.001)
+                hit["r"] = "timeout"
+            except InterruptedException:
+                hit["r"] = "interrupted"
+
+        th = threading.Thread(target=waiter)
+        th.start()
+        _t.sleep(0.05)
+        tok.cancel()
+        th.join(30)
+        assert hit.get("r") == "interrupted"
+
 
 class TestCompat:
     def test_pylibraft_style_imports(self):
