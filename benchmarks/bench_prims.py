@@ -95,6 +95,17 @@ def main():
     t = timeit(lambda: ramd.permute_rows(x[:, :64]), it)
     rec("permute rows", "1Mx64", t, f"{gb((1<<20)*64*8, t):.0f} GB/s")
 
+    # --- more reductions / norms ---
+    import raft_amd._C as C
+    t = timeit(lambda: C.row_argmin(x), it)
+    rec("row_argmin", "1Mx256", t, f"{gb(x.numel()*4, t):.0f} GB/s")
+    xb = x.bfloat16().contiguous()
+    t = timeit(lambda: C.rows_sqnorm_bf16(xb), it)
+    rec("rows_sqnorm_bf16", "1Mx256", t, f"{gb(x.numel()*2, t):.0f} GB/s")
+    vr = torch.randn(1 << 20, device=dev)
+    t = timeit(lambda: linalg.matrix_vector_op(x, vr, op="add", along_rows=False), it)
+    rec("matrix_vector_op(cols)", "1Mx256", t, f"{gb(x.numel()*8, t):.0f} GB/s")
+
     # --- matrix ops ---
     t = timeit(lambda: matrix.gather(x, torch.randint(0, 1 << 20, (1 << 19,),
                                                       device=dev)), it)
