@@ -430,11 +430,13 @@ def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify):
 
 
 def kmeans_predict(model_or_centroids, x: torch.Tensor, fp32_mode: str = "auto") -> torch.Tensor:
+    """Assign each row of x to its nearest centroid (fused L2-NN argmin)."""
     c = getattr(model_or_centroids, "centroids", model_or_centroids)
     return fused_l2nn(x, c, fp32_mode=fp32_mode)[1]
 
 
 def kmeans_transform(model_or_centroids, x: torch.Tensor, fp32_mode: str = "auto") -> torch.Tensor:
+    """Distance of each row of x to every centroid ([n, k] matrix)."""
     c = getattr(model_or_centroids, "centroids", model_or_centroids)
     return pairwise_distance(x, c, fp32_mode=fp32_mode)
 

@@ -80,6 +80,7 @@ class Comms:
         raise NotImplementedError
 
     def gather(self, t: torch.Tensor, root: int = 0) -> Optional[torch.Tensor]:
+        """Gather equal-size `t` to `root` (others return None)."""
         raise NotImplementedError
 
     def gatherv(self, t: torch.Tensor, counts: Sequence[int], root: int = 0) -> Optional[torch.Tensor]:
@@ -234,6 +235,7 @@ class TorchDistComms(Comms):
                 w.wait()
 
     def gather(self, t, root=0):
+        """Gather equal-size `t` to `root` (ncclRecv loop analog)."""
         world, rank = self.get_size(), self.get_rank()
         if rank == root:
             outs = [torch.empty_like(t) for _ in range(world)]

@@ -28,6 +28,7 @@ def log(x):
 
 
 def sqrt(x):
+    """Scalar/tensor-safe sqrt (reference math.hpp)."""
     return _t(torch.sqrt, _pymath.sqrt, x)
 
 

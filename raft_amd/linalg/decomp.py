@@ -60,6 +60,7 @@ def qr(a: torch.Tensor, mode: str = "reduced"):
 
 
 def cholesky(a: torch.Tensor, upper: bool = False) -> torch.Tensor:
+    """Cholesky factor (rocSOLVER potrf analog via torch.linalg)."""
     l = torch.linalg.cholesky(a)
     return l.t() if upper else l
 

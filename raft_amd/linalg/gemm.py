@@ -41,6 +41,7 @@ def gemm(a: torch.Tensor, b: torch.Tensor, alpha: float = 1.0, beta: float = 0.0
 
 def gemv(a: torch.Tensor, x: torch.Tensor, alpha: float = 1.0, beta: float = 0.0,
          y: torch.Tensor | None = None, trans: bool = False) -> torch.Tensor:
+    """y = alpha*A@x + beta*y (rocBLAS gemv analog)."""
     m = a.t() if trans else a
     out = torch.mv(m, x) * alpha
     if y is not None and beta != 0.0:
@@ -49,6 +50,7 @@ def gemv(a: torch.Tensor, x: torch.Tensor, alpha: float = 1.0, beta: float = 0.0
 
 
 def dot(x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
+    """BLAS-1 dot product (rocBLAS via torch)."""
     return torch.dot(x.reshape(-1), y.reshape(-1))
 
 

@@ -32,42 +32,52 @@ def map_offset(fn: Callable, n: int, *tensors: torch.Tensor, device=None,
 
 
 def unary_op(fn, x):
+    """Apply fn(x) elementwise (reference unary_op)."""
     return fn(x)
 
 
 def binary_op(fn, x, y):
+    """Apply fn(x, y) elementwise (reference binary_op)."""
     return fn(x, y)
 
 
 def ternary_op(fn, x, y, z):
+    """Apply fn(x, y, z) elementwise (reference ternary_op)."""
     return fn(x, y, z)
 
 
 def add(x, y):
+    """Elementwise a + b (map family)."""
     return x + y
 
 
 def subtract(x, y):
+    """Elementwise a - b (map family)."""
     return x - y
 
 
 def multiply(x, y):
+    """Elementwise a * b (map family)."""
     return x * y
 
 
 def divide(x, y):
+    """Elementwise a / b (map family)."""
     return x / y
 
 
 def power(x, y):
+    """Elementwise power (map family)."""
     return torch.pow(x, y)
 
 
 def sqrt(x):
+    """Elementwise sqrt (map family)."""
     return torch.sqrt(x)
 
 
 def eltwise(fn, *tensors):
+    """Variadic elementwise apply (reference eltwise)."""
     return fn(*tensors)
 
 

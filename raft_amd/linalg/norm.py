@@ -27,10 +27,12 @@ class NormType(Enum):
 
 
 def row_norm(x: torch.Tensor, norm_type: NormType = NormType.L2, sqrt: bool = True) -> torch.Tensor:
+    """Per-row norms (native fused kernel on GPU)."""
     return _norm_along(x, norm_type, sqrt, along_rows=True)
 
 
 def col_norm(x: torch.Tensor, norm_type: NormType = NormType.L2, sqrt: bool = True) -> torch.Tensor:
+    """Per-column norms (reduce + final op)."""
     return _norm_along(x, norm_type, sqrt, along_rows=False)
 
 
@@ -50,6 +52,7 @@ def _norm_along(x, norm_type, sqrt, along_rows: bool):
 
 def norm(x: torch.Tensor, norm_type: NormType = NormType.L2, along_rows: bool = True,
          sqrt: bool = True) -> torch.Tensor:
+    """Row or column norms with L0/L1/L2/Linf types (reference norm.cuh)."""
     return _norm_along(x, norm_type, sqrt, along_rows)
 
 

@@ -23,6 +23,7 @@ class PCAModel:
 
 
 def pca_fit(x: torch.Tensor, n_components: int, algo: str = "eig") -> PCAModel:
+    """Fit PCA (cov->eig or svd path; explained variance retained)."""
     n, d = x.shape
     mean = x.mean(dim=0)
     xc = x - mean
@@ -50,10 +51,12 @@ def pca_fit(x: torch.Tensor, n_components: int, algo: str = "eig") -> PCAModel:
 
 
 def pca_transform(model: PCAModel, x: torch.Tensor) -> torch.Tensor:
+    """Project rows onto the fitted principal components."""
     return (x - model.mean) @ model.components.t()
 
 
 def pca_inverse_transform(model: PCAModel, z: torch.Tensor) -> torch.Tensor:
+    """Map scores back to the original feature space."""
     return z @ model.components + model.mean
 
 
@@ -71,4 +74,5 @@ def tsvd_fit(x: torch.Tensor, n_components: int) -> TSVDModel:
 
 
 def tsvd_transform(model: TSVDModel, x: torch.Tensor) -> torch.Tensor:
+    """Project rows onto the truncated-SVD components."""
     return x @ model.components.t()

@@ -14,6 +14,7 @@ from raft_amd.utils import on_gpu
 
 
 def argmin(x: torch.Tensor) -> torch.Tensor:
+    """Per-row argmin (reference matrix::argmin)."""
     assert x.dim() == 2
     if on_gpu(x) and x.dtype == torch.float32:
         ext = require_ext()
@@ -22,6 +23,7 @@ def argmin(x: torch.Tensor) -> torch.Tensor:
 
 
 def argmax(x: torch.Tensor) -> torch.Tensor:
+    """Per-row argmax (reference matrix::argmax)."""
     assert x.dim() == 2
     if on_gpu(x) and x.dtype == torch.float32:
         ext = require_ext()

@@ -11,6 +11,7 @@ import torch
 
 
 def gather(x: torch.Tensor, indices: torch.Tensor, out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Gather rows x[indices[i]] (reference matrix::gather)."""
     res = torch.index_select(x, 0, indices.to(torch.int64))
     if out is not None:
         out.copy_(res)

@@ -10,14 +10,17 @@ import torch
 
 
 def slice_matrix(x: torch.Tensor, r0: int, c0: int, r1: int, c1: int) -> torch.Tensor:
+    """Rectangular sub-matrix copy (reference slice)."""
     return x[r0:r1, c0:c1].contiguous()
 
 
 def get_diagonal(x: torch.Tensor) -> torch.Tensor:
+    """Copy of the main diagonal (reference diagonal copy)."""
     return torch.diagonal(x).contiguous()
 
 
 def set_diagonal(x: torch.Tensor, vec: torch.Tensor) -> torch.Tensor:
+    """Write vec onto the main diagonal in place."""
     n = min(x.shape)
     idx = torch.arange(n, device=x.device)
     x[idx, idx] = vec[:n].to(x.dtype)
@@ -25,18 +28,22 @@ def set_diagonal(x: torch.Tensor, vec: torch.Tensor) -> torch.Tensor:
 
 
 def upper_triangular(x: torch.Tensor) -> torch.Tensor:
+    """Upper-triangular copy (reference upper-triangular copy)."""
     return torch.triu(x)
 
 
 def lower_triangular(x: torch.Tensor) -> torch.Tensor:
+    """Lower-triangular copy (reference tril analog)."""
     return torch.tril(x)
 
 
 def row_reverse(x: torch.Tensor) -> torch.Tensor:
+    """Reverse row order (reference rowReverse)."""
     return torch.flip(x, dims=[0])
 
 
 def col_reverse(x: torch.Tensor) -> torch.Tensor:
+    """Reverse column order (reference colReverse)."""
     return torch.flip(x, dims=[1])
 
 
@@ -52,10 +59,12 @@ def shift_rows(x: torch.Tensor, k: int, fill_value: float = 0.0) -> torch.Tensor
 
 
 def eye(n: int, m: int | None = None, device=None, dtype=torch.float32) -> torch.Tensor:
+    """Identity matrix (reference eye initializer)."""
     return torch.eye(n, m if m is not None else n, device=device, dtype=dtype)
 
 
 def power(x: torch.Tensor, p: float) -> torch.Tensor:
+    """Elementwise power (map family)."""
     return torch.pow(x, p)
 
 
@@ -71,6 +80,7 @@ def reciprocal(x: torch.Tensor, scalar: float = 1.0, thres: float = 0.0) -> torc
 
 
 def sqrt(x: torch.Tensor) -> torch.Tensor:
+    """Elementwise sqrt (map family)."""
     return torch.sqrt(x)
 
 
@@ -83,9 +93,11 @@ def sign_flip(x: torch.Tensor) -> torch.Tensor:
 
 
 def threshold(x: torch.Tensor, thres: float) -> torch.Tensor:
+    """Zero out entries below thres (reference threshold op)."""
     return torch.where(x < thres, torch.zeros_like(x), x)
 
 
 def linewise(x: torch.Tensor, vec: torch.Tensor, fn, along_rows: bool = True) -> torch.Tensor:
+    """Broadcast a vector op along rows/columns (reference linewise_op)."""
     from raft_amd.linalg.matrix_vector import matrix_vector_op
     return matrix_vector_op(x, vec, op=fn, along_rows=along_rows)

@@ -145,4 +145,5 @@ def _gemm_xyt(x, y, fp32_mode):
 
 
 def fused_l2nn_argmin(x: torch.Tensor, y: torch.Tensor, **kw) -> torch.Tensor:
+    """Argmin-only convenience over fused_l2nn (reference fusedL2NNArgmin)."""
     return fused_l2nn(x, y, **kw)[1]

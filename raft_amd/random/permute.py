@@ -9,6 +9,7 @@ from .rng import RngState, _draw_u32, require_ext
 
 
 def permute(n: int, state: RngState | None = None, device=None) -> torch.Tensor:
+    """Random permutation of 0..n-1 (counter-based RNG)."""
     state = state or RngState(seed=0)
     device = torch.device(device) if device is not None else torch.device("cpu")
     if device.type == "cuda":
@@ -21,5 +22,6 @@ def permute(n: int, state: RngState | None = None, device=None) -> torch.Tensor:
 
 
 def permute_rows(x: torch.Tensor, state: RngState | None = None) -> torch.Tensor:
+    """Randomly permute the rows of x (reference permute)."""
     perm = permute(x.shape[0], state=state, device=x.device)
     return x[perm]

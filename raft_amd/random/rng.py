@@ -98,6 +98,7 @@ def _norm_shape(shape) -> tuple:
 
 def uniform(shape, low: float = 0.0, high: float = 1.0, state: RngState | None = None,
             device=None, dtype=torch.float32) -> torch.Tensor:
+    """Uniform[low, high) samples (counter-based PCG32; CPU/GPU bitwise-identical)."""
     state = state or RngState()
     u = _gpu_or_cpu_uniform01(_norm_shape(shape), state, device, dtype)
     return u * (high - low) + low
@@ -105,6 +106,7 @@ def uniform(shape, low: float = 0.0, high: float = 1.0, state: RngState | None =
 
 def uniform_int(shape, low: int, high: int, state: RngState | None = None, device=None,
                 dtype=torch.int64) -> torch.Tensor:
+    """Uniform integer samples in [low, high)."""
     state = state or RngState()
     n = 1
     shape = _norm_shape(shape)
@@ -144,35 +146,42 @@ def normal(shape, mu: float = 0.0, sigma: float = 1.0, state: RngState | None = 
 
 
 def lognormal(shape, mu=0.0, sigma=1.0, state=None, device=None, dtype=torch.float32):
+    """LogNormal(mu, sigma) samples (exp of Box-Muller normal)."""
     return torch.exp(normal(shape, mu, sigma, state, device, torch.float64)).to(dtype)
 
 
 def logistic(shape, mu=0.0, scale=1.0, state=None, device=None, dtype=torch.float32):
+    """Logistic(mu, scale) samples via inverse CDF."""
     u = _u01(shape, state, device)
     return (mu - scale * torch.log(1.0 / u - 1.0)).to(dtype)
 
 
 def exponential(shape, lambda_: float = 1.0, state=None, device=None, dtype=torch.float32):
+    """Exponential(lambda) samples via inverse CDF."""
     u = _u01(shape, state, device)
     return (-torch.log(1.0 - u) / lambda_).to(dtype)
 
 
 def rayleigh(shape, sigma: float = 1.0, state=None, device=None, dtype=torch.float32):
+    """Rayleigh(sigma) samples via inverse CDF."""
     u = _u01(shape, state, device)
     return (sigma * torch.sqrt(-2.0 * torch.log(1.0 - u))).to(dtype)
 
 
 def laplace(shape, mu: float = 0.0, scale: float = 1.0, state=None, device=None, dtype=torch.float32):
+    """Laplace(mu, scale) samples via inverse CDF."""
     u = _u01(shape, state, device) - 0.5
     return (mu - scale * torch.sign(u) * torch.log(1.0 - 2.0 * u.abs())).to(dtype)
 
 
 def gumbel(shape, mu: float = 0.0, beta: float = 1.0, state=None, device=None, dtype=torch.float32):
+    """Gumbel(mu, beta) samples via inverse CDF."""
     u = _u01(shape, state, device)
     return (mu - beta * torch.log(-torch.log(u))).to(dtype)
 
 
 def bernoulli(shape, p: float = 0.5, state=None, device=None, dtype=torch.bool):
+    """Bernoulli(p) samples (reference distribution set)."""
     u = _u01(shape, state, device)
     return (u < p).to(dtype)
 

@@ -9,6 +9,7 @@ from .types import CSR, COO
 
 
 def coo_to_csr(coo: COO) -> CSR:
+    """COO -> CSR conversion (sorts by row; reference convert/csr)."""
     order = torch.argsort(coo.rows * coo.n_cols + coo.cols)
     rows = coo.rows[order]
     cols = coo.cols[order]
@@ -28,6 +29,7 @@ def sorted_coo_to_csr(coo: COO) -> CSR:
 
 
 def csr_to_coo(csr: CSR) -> COO:
+    """CSR -> COO expansion (reference convert)."""
     lengths = (csr.indptr[1:] - csr.indptr[:-1]).to(torch.int64)
     rows = torch.repeat_interleave(
         torch.arange(csr.n_rows, device=csr.device, dtype=torch.int64), lengths)
@@ -35,10 +37,12 @@ def csr_to_coo(csr: CSR) -> COO:
 
 
 def csr_to_dense(csr: CSR) -> torch.Tensor:
+    """CSR -> dense matrix (reference csr2dense)."""
     return csr.to_torch_sparse().to_dense()
 
 
 def dense_to_csr(d: torch.Tensor) -> CSR:
+    """Dense -> CSR of the nonzero entries (reference dense2csr)."""
     return CSR.from_dense(d)
 
 

@@ -64,10 +64,12 @@ def masked_matmul(a: torch.Tensor, b: torch.Tensor, bitmask: Bitset) -> CSR:
 
 
 def csr_degree(a: CSR) -> torch.Tensor:
+    """Per-row nonzero counts (reference degree kernels)."""
     return (a.indptr[1:] - a.indptr[:-1]).to(torch.int64)
 
 
 def csr_row_norm(a: CSR, norm_type: str = "l2") -> torch.Tensor:
+    """Per-row L1/L2/Linf norms of CSR values (reference sparse norm)."""
     seg = torch.repeat_interleave(torch.arange(a.n_rows, device=a.device),
                                   (a.indptr[1:] - a.indptr[:-1]).to(torch.int64))
     out = torch.zeros(a.n_rows, dtype=a.values.dtype, device=a.device)
@@ -91,6 +93,7 @@ def csr_add(a: CSR, b: CSR) -> CSR:
 
 
 def csr_transpose(a: CSR) -> CSR:
+    """CSR transpose (rocSPARSE csr2csc analog)."""
     coo = csr_to_coo(a)
     return coo_to_csr(COO(coo.cols, coo.rows, coo.values, a.n_cols, a.n_rows))
 

@@ -8,6 +8,7 @@ from .types import CSR, COO
 
 
 def coo_sort(coo: COO) -> COO:
+    """Sort COO triplets by (row, col) (reference coo sort)."""
     order = torch.argsort(coo.rows.to(torch.int64) * coo.n_cols + coo.cols.to(torch.int64))
     return COO(coo.rows[order], coo.cols[order], coo.values[order],
                coo.n_rows, coo.n_cols)
@@ -36,6 +37,7 @@ def dedupe_coo(coo: COO, op: str = "max") -> COO:
 
 
 def slice_csr_rows(a: CSR, start: int, stop: int) -> CSR:
+    """Row-range slice of a CSR matrix (reference slice csr rows)."""
     lo = int(a.indptr[start].item())
     hi = int(a.indptr[stop].item())
     indptr = a.indptr[start:stop + 1] - lo

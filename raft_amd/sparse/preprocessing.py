@@ -30,12 +30,14 @@ def _idf(a: CSR) -> torch.Tensor:
 
 
 def tfidf_transform(a: CSR) -> CSR:
+    """TF-IDF weighting of a term-document CSR (reference preprocessing)."""
     idf = _idf(a)
     vals = a.values.double() * idf[a.indices.to(torch.int64)]
     return CSR(a.indptr, a.indices, vals.to(a.values.dtype), a.n_rows, a.n_cols)
 
 
 def bm25_transform(a: CSR, k1: float = 1.6, b: float = 0.75) -> CSR:
+    """BM25 weighting of a term-document CSR (reference preprocessing)."""
     dl, seg = _doc_lengths(a)
     avgdl = dl.mean().clamp_min(1e-12)
     idf = _idf(a)

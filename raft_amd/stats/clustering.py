@@ -21,6 +21,7 @@ def _comb2(x: torch.Tensor) -> torch.Tensor:
 
 
 def adjusted_rand_index(a: torch.Tensor, b: torch.Tensor) -> float:
+    """Adjusted Rand index between two labelings (reference ARI)."""
     c = contingency_matrix(a, b).double()
     n = c.sum()
     sum_comb = _comb2(c).sum()
@@ -46,6 +47,7 @@ def rand_index(a: torch.Tensor, b: torch.Tensor) -> float:
 
 
 def mutual_info_score(a: torch.Tensor, b: torch.Tensor) -> float:
+    """Mutual information between two labelings (nats)."""
     c = contingency_matrix(a, b).double()
     n = c.sum()
     p = c / n
@@ -58,6 +60,7 @@ def mutual_info_score(a: torch.Tensor, b: torch.Tensor) -> float:
 
 
 def entropy(labels: torch.Tensor, n_classes: int | None = None) -> float:
+    """Shannon entropy of a labeling (nats; reference stats::entropy)."""
     l = labels.to(torch.int64)
     counts = torch.bincount(l - int(l.min()), minlength=n_classes or 0).double()
     p = counts[counts > 0] / counts.sum()
@@ -65,6 +68,7 @@ def entropy(labels: torch.Tensor, n_classes: int | None = None) -> float:
 
 
 def homogeneity_score(truth: torch.Tensor, pred: torch.Tensor) -> float:
+    """Homogeneity of pred w.r.t. truth (reference metric)."""
     h_c = entropy(truth)
     if h_c == 0.0:
         return 1.0
@@ -73,10 +77,12 @@ def homogeneity_score(truth: torch.Tensor, pred: torch.Tensor) -> float:
 
 
 def completeness_score(truth: torch.Tensor, pred: torch.Tensor) -> float:
+    """Completeness of pred w.r.t. truth (reference metric)."""
     return homogeneity_score(pred, truth)
 
 
 def v_measure(truth: torch.Tensor, pred: torch.Tensor, beta: float = 1.0) -> float:
+    """Harmonic mean of homogeneity and completeness (reference v_measure)."""
     h = homogeneity_score(truth, pred)
     c = completeness_score(truth, pred)
     if h + c == 0.0:
@@ -85,6 +91,7 @@ def v_measure(truth: torch.Tensor, pred: torch.Tensor, beta: float = 1.0) -> flo
 
 
 def kl_divergence(p: torch.Tensor, q: torch.Tensor) -> float:
+    """KL divergence between two distributions (reference metric)."""
     pd, qd = p.double(), q.double()
     mask = pd > 0
     return float(torch.where(mask, pd * (pd / qd.clamp_min(1e-300)).log(),

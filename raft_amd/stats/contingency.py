@@ -8,6 +8,7 @@ import torch
 def contingency_matrix(labels_a: torch.Tensor, labels_b: torch.Tensor,
                        n_classes_a: int | None = None,
                        n_classes_b: int | None = None) -> torch.Tensor:
+    """Label contingency table over the label RANGES (reference)."""
     a = labels_a.to(torch.int64)
     b = labels_b.to(torch.int64)
     amin, bmin = int(a.min()), int(b.min())

@@ -7,6 +7,7 @@ from .moments import mean
 
 
 def cov(x: torch.Tensor, sample: bool = True, centered: bool = False) -> torch.Tensor:
+    """Covariance matrix: mean-center + GEMM (reference stats::cov)."""
     xc = x if centered else x - mean(x).unsqueeze(0)
     n = x.shape[0]
     return (xc.t() @ xc) / (n - 1 if sample else n)

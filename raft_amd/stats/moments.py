@@ -18,11 +18,13 @@ def sum_cols(x: torch.Tensor) -> torch.Tensor:
 
 
 def mean(x: torch.Tensor, sample: bool = False) -> torch.Tensor:
+    """Per-column means (reference stats::mean)."""
     return sum_cols(x) / x.shape[0]
 
 
 def vars_(x: torch.Tensor, sample: bool = True,
           mu: torch.Tensor | None = None) -> torch.Tensor:
+    """Per-column variances (reference stats::vars)."""
     if mu is None:
         mu = mean(x)
     n = x.shape[0]
@@ -31,6 +33,7 @@ def vars_(x: torch.Tensor, sample: bool = True,
 
 
 def stddev(x: torch.Tensor, sample: bool = True) -> torch.Tensor:
+    """Per-column standard deviations (reference stats::stddev)."""
     return vars_(x, sample=sample).sqrt()
 
 
@@ -62,8 +65,10 @@ def weighted_mean(x: torch.Tensor, weights: torch.Tensor, along_rows: bool = Tru
 
 
 def mean_center(x: torch.Tensor) -> torch.Tensor:
+    """Subtract per-column means (reference mean_center)."""
     return x - mean(x).unsqueeze(0)
 
 
 def mean_add(x: torch.Tensor, mu: torch.Tensor) -> torch.Tensor:
+    """Add the column means back (inverse of mean_center)."""
     return x + mu.unsqueeze(0)

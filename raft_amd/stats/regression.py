@@ -9,6 +9,7 @@ import torch
 
 
 def r2_score(y_true: torch.Tensor, y_pred: torch.Tensor) -> float:
+    """Coefficient of determination (reference r2_score)."""
     yt, yp = y_true.double(), y_pred.double()
     ss_res = ((yt - yp) ** 2).sum()
     ss_tot = ((yt - yt.mean()) ** 2).sum()
@@ -26,6 +27,7 @@ def regression_metrics(y_true: torch.Tensor, y_pred: torch.Tensor):
 
 def information_criterion(log_likelihood: float, n_params: int, n_samples: int,
                           kind: str = "aic") -> float:
+    """AIC/AICc/BIC from log-likelihood (reference information_criterion)."""
     ll, k, n = float(log_likelihood), int(n_params), int(n_samples)
     if kind == "aic":
         return -2.0 * ll + 2.0 * k
