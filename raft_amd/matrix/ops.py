@@ -64,7 +64,7 @@ def eye(n: int, m: int | None = None, device=None, dtype=torch.float32) -> torch
 
 
 def power(x: torch.Tensor, p: float) -> torch.Tensor:
-    """Elementwise power (map family)."""
+    """Elementwise power (reference matrix power op)."""
     return torch.pow(x, p)
 
 
@@ -80,7 +80,7 @@ def reciprocal(x: torch.Tensor, scalar: float = 1.0, thres: float = 0.0) -> torc
 
 
 def sqrt(x: torch.Tensor) -> torch.Tensor:
-    """Elementwise sqrt (map family)."""
+    """Elementwise sqrt (exported as matrix_sqrt; reference matrix sqrt op)."""
     return torch.sqrt(x)
 
 
