@@ -191,3 +191,11 @@ def test_interface_fully_overridden():
     for impl in (LoopbackComms, TorchDistComms):
         missing = [n for n in abstract if getattr(impl, n) is getattr(Comms, n)]
         assert not missing, f"{impl.__name__} missing {missing}"
+
+
+def test_collectives_gloo_world3():
+    """Odd world size: ragged allgatherv/gatherv chunks and the even/odd
+    comm_split take different branches than world=2."""
+    from tests.conftest import spawn_gloo
+
+    spawn_gloo(_collective_worker, world_size=3)
