@@ -369,3 +369,26 @@ class TestResourcesStreamPool:
         r.add_resource_factory("shared", lambda: object())
         c = r.clone()
         assert c.get_resource("shared") is not None
+
+
+class TestLoggerEnvSinks:
+    def test_file_sink_and_level_env(self, tmp_path):
+        """RAFT_AMD_DEBUG file sink + RAFT_AMD_LOG_LEVEL (reference RAFT_DEBUG
+        env + compile-time level, logger.hpp:25-49) — in a subprocess so the
+        module-level logger singleton initializes from the env."""
+        import subprocess
+        import sys as _sys
+        f = tmp_path / "raft.log"
+        code = (
+            "from raft_amd.core import get_logger\n"
+            "log = get_logger()\n"
+            "log.info('should-appear')\n"
+            "log.debug('should-not-appear')\n"
+        )
+        r = subprocess.run([_sys.executable, "-c", code], env={
+            **__import__("os").environ,
+            "RAFT_AMD_DEBUG": str(f), "RAFT_AMD_LOG_LEVEL": "info"},
+            capture_output=True, timeout=120)
+        assert r.returncode == 0, r.stderr.decode()
+        text = f.read_text()
+        assert "should-appear" in text and "should-not-appear" not in text
