@@ -392,3 +392,19 @@ class TestLoggerEnvSinks:
         assert r.returncode == 0, r.stderr.decode()
         text = f.read_text()
         assert "should-appear" in text and "should-not-appear" not in text
+
+
+class TestManagerRoundRobin:
+    def test_pool_bounded_and_cycling(self):
+        mgr = DeviceResourcesManager(pool_size=3)
+        seen = [mgr.get_resources(torch.device("cpu")) for _ in range(9)]
+        distinct = {id(r) for r in seen}
+        assert len(distinct) == 3            # pool capped at pool_size
+        # round-robin: i and i+3 land on the same Resources
+        for i in range(6):
+            assert seen[i] is seen[i + 3]
+
+    def test_singleton_instance(self):
+        a = DeviceResourcesManager.instance()
+        b = DeviceResourcesManager.instance()
+        assert a is b
