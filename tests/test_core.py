@@ -408,3 +408,18 @@ class TestManagerRoundRobin:
         a = DeviceResourcesManager.instance()
         b = DeviceResourcesManager.instance()
         assert a is b
+
+
+class TestDeviceNdarraySurface:
+    def test_constructors_and_props(self):
+        import numpy as np
+        a = device_ndarray.zeros((3, 4), dtype=np.float32)
+        assert a.shape == (3, 4) and a.c_contiguous
+        b = device_ndarray.empty((2, 2), dtype=np.float64)
+        assert b.torch.dtype == torch.float64
+        c = device_ndarray(np.arange(6, dtype=np.int32).reshape(2, 3))
+        assert torch.equal(c.torch, torch.arange(6, dtype=torch.int32).reshape(2, 3))
+        # numpy round trip preserves values and dtype
+        back = np.asarray(c.copy_to_host() if hasattr(c, "copy_to_host")
+                          else c.torch.cpu().numpy())
+        assert back.dtype == np.int32 and back.sum() == 15
