@@ -23,9 +23,19 @@
 namespace raft_amd {
 
 __device__ __forceinline__ uint32_t f32_to_ord(float f, bool select_min) {
+  // Canonicalize NaN (any payload, either sign) to the maximum ordinal so it
+  // orders AFTER every finite value and +/-inf in BOTH selection directions.
+  // Without this a negative-sign NaN maps below -inf and is selected first.
+  if (f != f) return 0xFFFFFFFFu;
   uint32_t u = __float_as_uint(f);
   u = (u & 0x80000000u) ? ~u : (u | 0x80000000u);  // monotone: asc float -> asc u
   return select_min ? u : ~u;
+}
+
+__device__ __forceinline__ float ord_to_f32(uint32_t u, bool select_min) {
+  if (!select_min) u = ~u;
+  u = (u & 0x80000000u) ? (u & 0x7FFFFFFFu) : ~u;
+  return __uint_as_float(u);
 }
 
 constexpr int SELECT_K_MAX = 2048;
@@ -41,8 +51,10 @@ __device__ __forceinline__ void select_k_finish(const float* __restrict__ rp,
   if (do_sort) {
     int kp = 1;
     while (kp < k) kp <<= 1;
+    // padding must tie-break AFTER real entries (canonical NaN shares the
+    // max ordinal), so the index sentinel is INT_MAX, not -1
     for (int j = threadIdx.x + k; j < kp; j += BLOCK) {
-      if (j < SELECT_K_MAX) { pair_u[j] = 0xFFFFFFFFu; pair_i[j] = -1; }
+      if (j < SELECT_K_MAX) { pair_u[j] = 0xFFFFFFFFu; pair_i[j] = INT_MAX; }
     }
     __syncthreads();
     for (int size = 2; size <= kp; size <<= 1) {
@@ -281,14 +293,14 @@ __global__ void select_k_radix_kernel(const float* __restrict__ x,
 // bitonic merge.
 // ---------------------------------------------------------------------------
 
-__device__ __forceinline__ void cmp_exchange(float& v, int& i, int stride, bool keep_small) {
-  const float ov = __shfl_xor(v, stride, RAFT_AMD_WAVE);
+__device__ __forceinline__ void cmp_exchange(uint32_t& v, int& i, int stride, bool keep_small) {
+  const uint32_t ov = __shfl_xor(v, stride, RAFT_AMD_WAVE);
   const int oi = __shfl_xor(i, stride, RAFT_AMD_WAVE);
   const bool other_smaller = (ov < v) || (ov == v && oi < i);
   if (keep_small == other_smaller) { v = ov; i = oi; }
 }
 
-__device__ __forceinline__ void wave_bitonic_sort_asc(float& v, int& i, int lane) {
+__device__ __forceinline__ void wave_bitonic_sort_asc(uint32_t& v, int& i, int lane) {
 #pragma unroll
   for (int size = 2; size <= 64; size <<= 1) {
 #pragma unroll
@@ -302,7 +314,7 @@ __device__ __forceinline__ void wave_bitonic_sort_asc(float& v, int& i, int lane
   }
 }
 
-__device__ __forceinline__ void wave_bitonic_merge_asc(float& v, int& i, int lane) {
+__device__ __forceinline__ void wave_bitonic_merge_asc(uint32_t& v, int& i, int lane) {
 #pragma unroll
   for (int stride = 32; stride > 0; stride >>= 1) {
     const bool keep_small = (lane & stride) == 0;
@@ -310,6 +322,10 @@ __device__ __forceinline__ void wave_bitonic_merge_asc(float& v, int& i, int lan
   }
 }
 
+// The queue runs in the monotone ORDINAL domain (f32_to_ord), so NaN values
+// participate with their real indices (canonical max ordinal = ordered last)
+// and every output slot carries a valid in-range index — the float-domain
+// version dropped NaN/inf ties and could leave -1 sentinels in the output.
 template <int BLOCK = 256>
 __global__ void select_k_warpsort_kernel(const float* __restrict__ x,
                                          float* __restrict__ out_v,
@@ -322,31 +338,35 @@ __global__ void select_k_warpsort_kernel(const float* __restrict__ x,
   const long long stride = (long long)gridDim.x * waves_per_block;
   for (; row < batch; row += stride) {
     const float* rp = x + row * len;
-    float qv = INFINITY;   // queue ascending across lanes; lane 63 = worst
-    int qi = -1;
-    float worst = INFINITY;
+    uint32_t qv = 0xFFFFFFFFu;  // queue ascending across lanes; lane 63 = worst
+    int qi = INT_MAX;           // index tie-break prefers real (smaller) indices
+    uint32_t worst = 0xFFFFFFFFu;
     for (long long j0 = 0; j0 < len; j0 += RAFT_AMD_WAVE) {
       const long long j = j0 + lane;
-      float val = j < len ? rp[j] : INFINITY;
-      if (!select_min && j < len) val = -val;
-      const bool cand = (val < worst);
+      const uint32_t val =
+          j < len ? f32_to_ord(rp[j], select_min) : 0xFFFFFFFFu;
+      // admit strictly-better candidates; while the worst slot is still at the
+      // max ordinal (unfilled or NaN) also admit max-ordinal elements so their
+      // real indices displace the INT_MAX sentinels (index tie-break wins)
+      const bool cand =
+          (val < worst) || (val == worst && worst == 0xFFFFFFFFu && j < len);
       if (__ballot(cand) == 0ull) continue;
       // sort batch ascending, then reverse to descending via lane mirror
-      float bv = val;
-      int bi = cand ? (int)j : -1;
-      if (!cand) bv = INFINITY;  // non-candidates can't enter the queue
+      uint32_t bv = cand ? val : 0xFFFFFFFFu;
+      int bi = cand ? (int)j : INT_MAX;
       wave_bitonic_sort_asc(bv, bi, lane);
-      const float rv = __shfl(bv, 63 - lane, RAFT_AMD_WAVE);
+      const uint32_t rv = __shfl(bv, 63 - lane, RAFT_AMD_WAVE);
       const int ri = __shfl(bi, 63 - lane, RAFT_AMD_WAVE);
       // elementwise min of (asc queue, desc batch) -> bitonic; re-merge
-      const bool take = (rv < qv) || (rv == qv && ri != -1 && ri < qi);
+      const bool take = (rv < qv) || (rv == qv && ri < qi);
       if (take) { qv = rv; qi = ri; }
       wave_bitonic_merge_asc(qv, qi, lane);
       worst = __shfl(qv, (k <= 64 ? k : 64) - 1, RAFT_AMD_WAVE);
     }
     if (lane < k) {
-      out_v[row * k + lane] = select_min ? qv : -qv;
-      out_i[row * k + lane] = qi;
+      out_v[row * k + lane] = qi == INT_MAX ? ord_to_f32(qv, select_min)
+                                            : rp[qi];
+      out_i[row * k + lane] = qi == INT_MAX ? 0 : qi;
     }
   }
 }

@@ -62,23 +62,28 @@ def _init_random(x: torch.Tensor, k: int, state: RngState, comms: Comms) -> torc
 
 
 def _init_plusplus(x: torch.Tensor, k: int, state: RngState, comms: Comms,
-                   fp32_mode: str) -> torch.Tensor:
-    """Greedy k-means++ (D^2 sampling with local trials), distributed-aware.
+                   fp32_mode: str,
+                   sample_weights: torch.Tensor | None = None) -> torch.Tensor:
+    """Greedy k-means++ (weighted D^2 sampling with local trials), distributed-aware.
 
     Reference parity: the reference's kmeansPlusPlus (and sklearn) draw
     2+log(k) candidates per step by D^2 and keep the one that minimizes the
     resulting potential — pure sequential D^2 misses a blob with measurable
-    probability at large k; greedy selection drives that to ~0.
+    probability at large k; greedy selection drives that to ~0. With
+    sample_weights (the kmeans|| candidate-reduction path), both the D^2
+    sampling CDF and the candidate potentials are weighted by w, matching the
+    reference's initScalableKMeansPlusPlus ownership weighting.
 
     Each step: every rank holds min-sq-distances to chosen centers for its
-    shard; ranks compute local D^2 sums, candidates are sampled rank/row
-    proportionally and broadcast; candidate potentials are computed from one
-    [n_local, L] GEMM-shaped distance block and allreduced.
+    shard; ranks compute local (weighted) D^2 sums, candidates are sampled
+    rank/row proportionally and broadcast; candidate potentials are computed
+    from one [n_local, L] GEMM-shaped distance block and allreduced.
     """
     import math
     n_local, d = x.shape
     world = comms.get_size()
     rank = comms.get_rank()
+    w = None if sample_weights is None else sample_weights.double().clamp_min(0)
     n_trials = 2 + int(math.log(max(2, k)))
     # first center: global row 0 owner = rank 0 (deterministic from seed)
     u = uniform((1,), state=state, device=x.device)
@@ -90,13 +95,14 @@ def _init_plusplus(x: torch.Tensor, k: int, state: RngState, comms: Comms,
     mind2 = fused_l2nn(x, c0, fp32_mode=fp32_mode)[0].double()
     xsq = (x * x).sum(dim=1)
     for _ in range(1, k):
-        local_sum = mind2.sum()
+        wd2 = mind2 if w is None else mind2 * w
+        local_sum = wd2.sum()
         if world > 1:
             sums = comms.allgather(local_sum.reshape(1)).reshape(-1)
         else:
             sums = local_sum.reshape(1)
         total = float(sums.sum().item())
-        cdf = torch.cumsum(mind2, dim=0)
+        cdf = torch.cumsum(wd2, dim=0)
         if world == 1:
             # vectorized trial sampling: ONE uniform draw + ONE searchsorted
             # for all L candidates (the per-trial .item() loop costs ~2
@@ -106,7 +112,8 @@ def _init_plusplus(x: torch.Tensor, k: int, state: RngState, comms: Comms,
             cmat = x[js]                                       # [L, d]
             d2 = (xsq.unsqueeze(1) + (cmat * cmat).sum(dim=1)
                   - 2.0 * (x @ cmat.T)).clamp_min_(0).double()
-            pot = torch.minimum(mind2.unsqueeze(1), d2).sum(dim=0)
+            pot = torch.minimum(mind2.unsqueeze(1), d2)
+            pot = (pot if w is None else pot * w.unsqueeze(1)).sum(dim=0)
             best = int(pot.argmin().item())
             centers.append(cmat[best])
             mind2 = torch.minimum(mind2, d2[:, best])
@@ -138,7 +145,8 @@ def _init_plusplus(x: torch.Tensor, k: int, state: RngState, comms: Comms,
         # ||x||^2 + ||c||^2 - 2 x.c — selection only needs ~1e-7 relative
         d2 = (xsq.unsqueeze(1) + (cmat * cmat).sum(dim=1)
               - 2.0 * (x @ cmat.T)).clamp_min_(0).double()
-        pot = torch.minimum(mind2.unsqueeze(1), d2).sum(dim=0)  # [L]
+        pot = torch.minimum(mind2.unsqueeze(1), d2)             # [n_local, L]
+        pot = (pot if w is None else pot * w.unsqueeze(1)).sum(dim=0)  # [L]
         if world > 1:
             comms.allreduce(pot, op=ReduceOp.SUM)
         best = int(pot.argmin().item())
@@ -243,7 +251,8 @@ def kmeans_fit(x: torch.Tensor, params: KMeansParams,
         centroids = _init_scalable(x, k, state, comms, params.fp32_mode,
                                    oversampling=params.oversampling)
     else:
-        centroids = _init_plusplus(x, k, state, comms, params.fp32_mode)
+        centroids = _init_plusplus(x, k, state, comms, params.fp32_mode,
+                                   sample_weights=sample_weights)
 
     from raft_amd.neighbors.fused_l2nn import (_MODE_NSLICE, _VERIFY_MODES,
                                                fused_l2nn_presplit,
@@ -263,16 +272,21 @@ def kmeans_fit(x: torch.Tensor, params: KMeansParams,
             dmin, labels = fused_l2nn_presplit(x_slices, xn, centroids, verify_x=vx)
         else:
             dmin, labels = fused_l2nn(x, centroids, fp32_mode=params.fp32_mode)
+        # accumulate sums/counts/inertia in fp32 regardless of x.dtype —
+        # bf16 cannot represent counts > 256 exactly, so half-precision
+        # accumulation silently corrupts centroids of large clusters
+        acc = torch.float32 if x.dtype in (torch.bfloat16, torch.float16) \
+            else x.dtype
         if sample_weights is None:
-            sums = reduce_rows_by_key(x, labels, n_keys=k)
-            counts = torch.bincount(labels, minlength=k).to(x.dtype)
-            local_inertia = torch.sum(dmin, dtype=torch.float64).to(x.dtype)
+            sums = reduce_rows_by_key(x, labels, n_keys=k, out_dtype=acc)
+            counts = torch.bincount(labels, minlength=k).to(acc)
+            local_inertia = torch.sum(dmin, dtype=torch.float64).to(acc)
         else:
-            w = sample_weights.to(x.dtype)
-            sums = reduce_rows_by_key(x, labels, n_keys=k, weights=w)
-            counts = torch.zeros(k, dtype=x.dtype, device=x.device)
+            w = sample_weights.to(acc)
+            sums = reduce_rows_by_key(x, labels, n_keys=k, weights=w, out_dtype=acc)
+            counts = torch.zeros(k, dtype=acc, device=x.device)
             counts.index_add_(0, labels.to(torch.int64), w)
-            local_inertia = torch.sum(dmin * w, dtype=torch.float64).to(x.dtype)
+            local_inertia = torch.sum(dmin.to(acc) * w, dtype=torch.float64).to(acc)
         # ONE packed allreduce: [k, d] sums | [k] counts | [1] inertia
         packed = torch.cat([sums.reshape(-1), counts, local_inertia.reshape(1)])
         if comms.get_size() > 1:
@@ -282,7 +296,8 @@ def kmeans_fit(x: torch.Tensor, params: KMeansParams,
         inertia = float(packed[-1].item())
         nonzero = counts > 0
         new_centroids = centroids.clone()
-        new_centroids[nonzero] = sums[nonzero] / counts[nonzero].unsqueeze(1)
+        new_centroids[nonzero] = (sums[nonzero]
+                                  / counts[nonzero].unsqueeze(1)).to(x.dtype)
         # empty clusters: relocate to the globally farthest point (reference
         # relocates empties; here: owner rank = argmax of local max-dmin)
         empties = (~nonzero).nonzero(as_tuple=True)[0]
@@ -372,9 +387,12 @@ def kmeans_iterate(x: torch.Tensor, centroids: torch.Tensor, n_iters: int,
         else:
             dmin, labels = fused_l2nn(x, centroids, fp32_mode=fp32_mode,
                                       chunk_rows=chunk_rows)
-        sums = reduce_rows_by_key(x, labels, n_keys=k)
-        counts = torch.bincount(labels, minlength=k).to(x.dtype)
-        local_inertia = torch.sum(dmin, dtype=torch.float64).to(x.dtype)
+        # fp32 accumulation regardless of x.dtype (bf16 counts >256 are inexact)
+        acc = torch.float32 if x.dtype in (torch.bfloat16, torch.float16) \
+            else x.dtype
+        sums = reduce_rows_by_key(x, labels, n_keys=k, out_dtype=acc)
+        counts = torch.bincount(labels, minlength=k).to(acc)
+        local_inertia = torch.sum(dmin, dtype=torch.float64).to(acc)
         packed = torch.cat([sums.reshape(-1), counts, local_inertia.reshape(1)])
         if comms.get_size() > 1:
             comms.allreduce(packed, op=ReduceOp.SUM)
@@ -383,7 +401,8 @@ def kmeans_iterate(x: torch.Tensor, centroids: torch.Tensor, n_iters: int,
         inertia_t = packed[-1]
         nonzero = counts > 0
         centroids = torch.where(nonzero.unsqueeze(1),
-                                sums / counts.clamp_min(1).unsqueeze(1), centroids)
+                                (sums / counts.clamp_min(1).unsqueeze(1)).to(x.dtype),
+                                centroids)
     # single host sync at the end (a per-iter .item() serializes the pipeline)
     inertia = float(inertia_t.item()) if inertia_t is not None else float("inf")
     return centroids, inertia

@@ -17,12 +17,22 @@ from raft_amd.utils import on_gpu
 
 
 def reduce_rows_by_key(x: torch.Tensor, keys: torch.Tensor, n_keys: int | None = None,
-                       weights: torch.Tensor | None = None) -> torch.Tensor:
-    """sums[k, :] = sum over rows i with keys[i]==k of (w_i *) x[i, :]."""
+                       weights: torch.Tensor | None = None,
+                       out_dtype: torch.dtype | None = None) -> torch.Tensor:
+    """sums[k, :] = sum over rows i with keys[i]==k of (w_i *) x[i, :].
+
+    out_dtype: accumulation/output dtype. Defaults to fp32 for half-precision
+    inputs (bf16 cannot represent sums of >256 unit-scale terms exactly) and
+    x.dtype otherwise.
+    """
     assert x.dim() == 2 and keys.dim() == 1 and keys.shape[0] == x.shape[0]
     if n_keys is None:
         n_keys = int(keys.max().item()) + 1 if keys.numel() else 0
-    if on_gpu(x, keys) and x.dtype == torch.float32 and weights is None:
+    if out_dtype is None:
+        out_dtype = torch.float32 if x.dtype in (torch.bfloat16, torch.float16) \
+            else x.dtype
+    if on_gpu(x, keys) and x.dtype == torch.float32 and weights is None \
+            and out_dtype == torch.float32:
         ext = require_ext()
         if x.shape[0] >= 65536:
             # sort-based: atomics only at run boundaries (the naive atomic
@@ -34,9 +44,9 @@ def reduce_rows_by_key(x: torch.Tensor, keys: torch.Tensor, n_keys: int | None =
                                                  keys_sorted.contiguous(), int(n_keys))
         return ext.reduce_rows_by_key(x.contiguous(), keys.to(torch.int32).contiguous(), int(n_keys))
     # weighted variant (reference has one): torch index_add over w*x
-    out = torch.zeros((n_keys, x.shape[1]), dtype=x.dtype, device=x.device)
+    out = torch.zeros((n_keys, x.shape[1]), dtype=out_dtype, device=x.device)
     src = x if weights is None else x * weights.unsqueeze(1)
-    out.index_add_(0, keys.to(torch.int64), src)
+    out.index_add_(0, keys.to(torch.int64), src.to(out_dtype))
     return out
 
 

@@ -36,11 +36,10 @@ def select_k(x: torch.Tensor, k: int, select_min: bool = True,
 
     Returns (values [batch,k], indices [batch,k] int64).
 
-    NaN semantics: for select_min=True both native engines order NaN after
-    every finite value (never selected while >= k finite values exist).
-    For select_min=False the engines differ (radix orders NaN above +inf,
-    warpsort's comparison filter drops it) — don't rely on NaN placement
-    in max-selection.
+    NaN semantics: both native engines canonicalize NaN (any sign/payload) to
+    the maximum ordinal, so NaN orders after every finite value and +/-inf in
+    BOTH selection directions; NaNs are only selected when a row has fewer
+    than k non-NaN values, and then they carry their real in-range indices.
     """
     assert x.dim() == 2
     batch, n = x.shape
@@ -91,6 +90,10 @@ def select_k(x: torch.Tensor, k: int, select_min: bool = True,
                 fv, fpos = ext.select_k(cand_v.contiguous(), k, bool(select_min),
                                         0, bool(sorted))
                 gi = torch.gather(cand_i, 1, fpos.to(torch.int64))
+                # padded +/-inf slots carry fabricated indices base+pos >= n;
+                # they are only selected when a row has < k real candidates —
+                # clamp so callers never see an out-of-range index
+                gi.clamp_(max=n - 1)
                 return fv, gi
         algo_code = {SelectAlgo.AUTO: 0, SelectAlgo.RADIX: 1, SelectAlgo.WARPSORT: 2}[algo]
         if algo == SelectAlgo.AUTO and k <= 64 and batch >= 2048 and n >= 100000:

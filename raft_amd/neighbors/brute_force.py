@@ -13,8 +13,15 @@ MI355X design — the GPU path never materializes the distance matrix:
      pairwise_l2_filter_kernel) — expected O(k) emissions per row, so the
      6.5 TB of distance-tile traffic of the naive tiled path disappears.
   3. SELECT: native radix select_k over the tiny candidate buffers.
-  4. Rows whose buffer under/overflowed (probabilistically rare) fall back to
-     the exact tiled path; the result is exact for every row.
+  4. fp32 inputs: the selected candidates are RE-RANKED by exact-fp32
+     distances (gather + einsum over ~2k candidates/row) and a per-row margin
+     check proves the exact top-k is inside the re-ranked set; rows failing
+     the margin, or whose buffer under/overflowed (probabilistically rare),
+     fall back to the exact tiled path. The returned neighbors are therefore
+     exact w.r.t. fp32 expanded distances for every row (the same guarantee
+     and mechanism as the verified kmeans L2-NN engine, csrc/kmeans.hip).
+     fp32_mode="native" skips the MFMA filter entirely and runs the tiled
+     fp32 rocBLAS path.
 """
 from __future__ import annotations
 
@@ -33,7 +40,8 @@ def knn(x: torch.Tensor, queries: torch.Tensor, k: int,
     """k nearest rows of x for each query row. Returns (dists [q,k], idx [q,k])."""
     if (queries.is_cuda and metric in (DistanceType.L2Expanded, "sqeuclidean")
             and x.shape[1] % 64 == 0 and x.shape[0] >= 8 * k
-            and x.dtype in (torch.bfloat16, torch.float32)):
+            and x.dtype in (torch.bfloat16, torch.float32)
+            and not (x.dtype == torch.float32 and fp32_mode == "native")):
         return _knn_gpu_filtered(x, queries, k, fp32_mode)
     return _knn_tiled(x, queries, k, metric, query_chunk, index_chunk, fp32_mode)
 
@@ -107,11 +115,37 @@ def _knn_gpu_filtered(x, queries, k, fp32_mode, index_chunk: int = 4_000_000):
                                cand_d, cand_i, cnt, c0)
 
     # ---- 3. select over candidates -----------------------------------------
-    vals, pos = select_k(cand_d, k, select_min=True)
-    idx = torch.gather(cand_i, 1, pos.to(torch.int64)).to(torch.int64)
-
-    # ---- 4. exact fallback for under/overflowed rows ------------------------
     bad = (cnt < k) | (cnt > cap)
+    if nslice >= 2:
+        # fp32 input: candidate distances are split-bf16 EMULATED — re-rank
+        # the top 2k candidates by EXACT fp32 distances, then prove per row
+        # that no candidate outside the re-ranked set can be a true neighbor:
+        # exact_d >= emul_d - eps for every candidate, so if the exact k-th
+        # selected <= (emulated (2k)-th) - eps, the exact top-k is inside the
+        # 2k set. Rows failing the margin join the exact-fallback set.
+        k2 = min(2 * k, cap)
+        evals, pos2 = select_k(cand_d, k2, select_min=True)
+        idx2 = torch.gather(cand_i, 1, pos2.to(torch.int64)).to(torch.int64)
+        safe_idx2 = idx2.clamp_min(0)           # -1 slots carry +inf distances
+        xg = x[safe_idx2.reshape(-1)].reshape(m, k2, d)
+        dots = torch.einsum("md,mkd->mk", queries.float(), xg.float())
+        exact = (qn.unsqueeze(1) + xn_full[safe_idx2] - 2.0 * dots).clamp_min_(0)
+        exact = torch.where(idx2 < 0, torch.full_like(exact, float("inf")), exact)
+        vals, rpos = torch.sort(exact, dim=1)
+        vals = vals[:, :k]
+        idx = torch.gather(idx2, 1, rpos[:, :k])
+        eps = lead * torch.sqrt(qn.clamp_min(0) * xm) + tail * (qn + xm)
+        # rows with cnt <= k2 re-ranked their ENTIRE candidate set (complete
+        # by threshold inflation) — exact with no margin needed; rows with
+        # more candidates need the emulated-(k2)th margin to clear eps
+        need_margin = cnt.to(torch.int64) > k2
+        margin_fail = need_margin & ~(vals[:, -1] <= evals[:, -1] - eps)
+        bad = bad | margin_fail
+    else:
+        vals, pos = select_k(cand_d, k, select_min=True)
+        idx = torch.gather(cand_i, 1, pos.to(torch.int64)).to(torch.int64)
+
+    # ---- 4. exact fallback for margin-failed / under/overflowed rows --------
     n_bad = int(bad.sum().item())
     if n_bad:
         rows = bad.nonzero(as_tuple=True)[0]

@@ -129,3 +129,43 @@ class TestEstimatorSurface:
         s = km.score(x)
         assert s <= 0 and s == pytest.approx(-km.inertia_, rel=1e-3)
         assert km.labels_ is not None and km.n_iter_ >= 1
+
+
+class TestHalfPrecisionAccumulation:
+    def test_bf16_counts_and_centroids_exact(self):
+        # bf16 accumulation made counts >256 inexact. One cluster
+        # with 2000 members at a constant offset must produce the exact mean.
+        n, d = 2000, 8
+        x = torch.full((n, d), 1.0, dtype=torch.bfloat16)
+        x[:, 0] = 3.0
+        init = torch.zeros((1, d), dtype=torch.bfloat16)
+        m = kmeans_fit(x, KMeansParams(n_clusters=1, max_iter=1, init="array"),
+                       init_centroids=init)
+        # mean of 2000 identical bf16 rows == the row itself, exactly
+        assert torch.equal(m.centroids[0].float(), x[0].float())
+
+    def test_bf16_iterate_matches_fp32(self):
+        from raft_amd.cluster.kmeans import kmeans_iterate
+        x32, _, centers = make_blobs(4000, 8, n_clusters=4, cluster_std=0.2,
+                                     state=RngState(seed=11))
+        c0 = centers.clone() + 0.05
+        c32, i32 = kmeans_iterate(x32, c0.clone(), 2)
+        c16, i16 = kmeans_iterate(x32.bfloat16(), c0.bfloat16(), 2)
+        # quantization-level agreement only (inputs rounded to bf16), but the
+        # old bf16 count accumulation was off by >10% on 1000-member clusters
+        assert torch.allclose(c16.float(), c32, rtol=0.02, atol=0.05)
+        assert abs(i16 - i32) / i32 < 0.05
+
+
+class TestWeightedSeeding:
+    def test_weighted_kmeanspp_prefers_heavy_points(self):
+        # two tight blobs; blob A carries 100x the weight. With k=1 the single
+        # center must land on A's side: the weighted potential demands it.
+        torch.manual_seed(0)
+        a = torch.randn(50, 2) * 0.05 + torch.tensor([5.0, 0.0])
+        b = torch.randn(50, 2) * 0.05 + torch.tensor([-5.0, 0.0])
+        x = torch.cat([a, b])
+        w = torch.cat([torch.full((50,), 100.0), torch.full((50,), 1.0)])
+        m = kmeans_fit(x, KMeansParams(n_clusters=1, max_iter=5, seed=3,
+                                       init="kmeans++"), sample_weights=w)
+        assert m.centroids[0, 0] > 4.0  # weighted mean ~ 4.9, unweighted ~ 0

@@ -501,21 +501,43 @@ class TestGemmGpu:
 
 
 class TestSelectKNaN:
-    """NaN semantics (documented in matrix/select_k.py): for MIN selection
-    both engines order NaN after every finite value (ordered-uint encoding /
-    comparison filtering), so NaNs are never selected while enough finite
-    values exist. MAX selection is engine-dependent (radix treats NaN as
-    above +inf, warpsort's comparisons drop it) — not asserted."""
+    """NaN semantics (documented in matrix/select_k.py): NaN of ANY sign or
+    payload canonicalizes to the max ordinal — ordered after every finite
+    value and +/-inf in BOTH selection directions; NaN-padded outputs carry
+    real in-range indices (no -1 sentinels)."""
 
     @pytest.mark.parametrize("algo_name", ["radix", "warpsort"])
-    def test_nan_never_selected_min(self, dev, ext, algo_name):
+    @pytest.mark.parametrize("select_min", [True, False])
+    def test_nan_never_selected(self, dev, ext, algo_name, select_min):
         from raft_amd.matrix import select_k, SelectAlgo
         torch.manual_seed(0)
         x = torch.randn(8, 6000 if algo_name == "radix" else 500, device=dev)
-        x[:, :7] = float("nan")
+        x[:, :4] = float("nan")
+        # negative-sign NaN: maps below -inf unless canonicalized (ADVICE r1)
+        neg_nan = torch.tensor(float("nan"), device=dev).view(torch.int32) \
+            | torch.tensor(-0x80000000, device=dev, dtype=torch.int32)
+        x[:, 4:7] = neg_nan.view(torch.float32)
+        algo = SelectAlgo.RADIX if algo_name == "radix" else SelectAlgo.WARPSORT
+        vals, idx = select_k(x, 16, select_min=select_min, algo=algo)
+        assert not torch.isnan(vals).any()
+        fill = float("inf") if select_min else float("-inf")
+        finite = torch.nan_to_num(x, nan=fill, posinf=float("inf"),
+                                  neginf=float("-inf"))
+        ref = torch.topk(finite, 16, dim=1, largest=not select_min).values
+        torch.testing.assert_close(vals, ref)
+        assert (idx >= 0).all() and (idx < x.shape[1]).all()
+
+    @pytest.mark.parametrize("algo_name", ["radix", "warpsort"])
+    def test_nan_heavy_rows_valid_indices(self, dev, ext, algo_name):
+        # rows with FEWER than k finite values: NaNs fill the tail slots and
+        # every index must still be in-range (the old warpsort emitted -1)
+        from raft_amd.matrix import select_k, SelectAlgo
+        torch.manual_seed(1)
+        n = 6000 if algo_name == "radix" else 500
+        x = torch.full((4, n), float("nan"), device=dev)
+        x[:, :5] = torch.randn(4, 5, device=dev)
         algo = SelectAlgo.RADIX if algo_name == "radix" else SelectAlgo.WARPSORT
         vals, idx = select_k(x, 16, select_min=True, algo=algo)
-        assert not torch.isnan(vals).any()
-        finite = torch.nan_to_num(x, nan=float("inf"))
-        ref = torch.topk(finite, 16, dim=1, largest=False).values
-        torch.testing.assert_close(vals, ref)
+        assert (idx >= 0).all() and (idx < n).all()
+        torch.testing.assert_close(vals[:, :5], torch.sort(x[:, :5], dim=1).values)
+        assert torch.isnan(vals[:, 5:]).all()
