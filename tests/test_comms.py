@@ -20,81 +20,13 @@ class TestLoopback:
 
 
 def _collective_worker(rank, world):
-    from raft_amd.comms import TorchDistComms, ReduceOp
+    from raft_amd.comms import TorchDistComms
+    from tests.comms_suite import collective_suite
 
     c = TorchDistComms()
     assert c.get_size() == world
     assert c.get_rank() == rank
-
-    # allreduce (test_collective_allreduce parity)
-    t = torch.full((4,), float(rank + 1))
-    c.allreduce(t)
-    expected = sum(r + 1 for r in range(world))
-    assert torch.equal(t, torch.full((4,), float(expected)))
-
-    # bcast
-    t = torch.full((3,), float(rank))
-    c.bcast(t, root=0)
-    assert torch.equal(t, torch.zeros(3))
-
-    # reduce
-    t = torch.full((2,), 1.0)
-    c.reduce(t, root=0, op=ReduceOp.SUM)
-    if rank == 0:
-        assert torch.equal(t, torch.full((2,), float(world)))
-
-    # allgather
-    g = c.allgather(torch.full((2,), float(rank)))
-    for r in range(world):
-        assert torch.equal(g[r], torch.full((2,), float(r)))
-
-    # allgatherv (ragged)
-    counts = [r + 1 for r in range(world)]
-    mine = torch.full((rank + 1,), float(rank))
-    cat = c.allgatherv(mine, counts)
-    assert cat.numel() == sum(counts)
-    off = 0
-    for r in range(world):
-        assert torch.equal(cat[off:off + r + 1], torch.full((r + 1,), float(r)))
-        off += r + 1
-
-    # gather / gatherv
-    got = c.gather(torch.full((2,), float(rank)), root=0)
-    if rank == 0:
-        assert got.shape[0] == world
-    gv = c.gatherv(mine, counts, root=0)
-    if rank == 0:
-        assert gv.numel() == sum(counts)
-
-    # reducescatter
-    t = torch.arange(float(world * 2))
-    out = c.reducescatter(t.clone())
-    assert torch.equal(out, t[rank * 2:(rank + 1) * 2] * world)
-
-    # p2p sendrecv ring
-    send = torch.full((3,), float(rank))
-    recv = torch.empty(3)
-    dst = (rank + 1) % world
-    src = (rank - 1) % world
-    c.device_sendrecv(send, dst, recv, src)
-    assert torch.equal(recv, torch.full((3,), float(src)))
-
-    # multicast sendrecv: each rank sends to every peer, receives from all
-    msend = torch.full((2,), float(rank) + 10.0)
-    peers = [r for r in range(world) if r != rank]
-    mrecvs = [torch.empty(2) for _ in peers]
-    c.device_multicast_sendrecv(msend, peers, mrecvs, peers)
-    for r, peer in zip(mrecvs, peers):
-        assert torch.equal(r, torch.full((2,), float(peer) + 10.0))
-
-    # comm_split: even/odd colors
-    sub = c.comm_split(color=rank % 2, key=rank)
-    t = torch.ones(1)
-    sub.allreduce(t)
-    n_same_color = len([r for r in range(world) if r % 2 == rank % 2])
-    assert torch.equal(t, torch.full((1,), float(n_same_color)))
-
-    c.barrier()
+    collective_suite(c, "cpu")
 
 
 def test_collectives_gloo_world2():
