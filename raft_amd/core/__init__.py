@@ -19,6 +19,16 @@ from .interruptible import Interruptible, synchronize as interruptible_synchroni
 from .logger import get_logger, set_level
 from .trace import annotate, annotated
 from .memory import MemoryStats, TrackingScope, ResourceMonitor, TemporaryDeviceBuffer
+from .mr import (
+    DeviceMemoryResource,
+    TorchMemoryResource,
+    PoolMemoryResource,
+    LimitingAdaptor,
+    TrackingAdaptor,
+    MemoryLimitExceeded,
+    WorkspaceBuffer,
+    AllocationStats,
+)
 from .mdbuffer import MDBuffer, MemoryType, memory_type_dispatcher, copy_mdspan
 from .error import RaftError, LogicError, HipError, expects, fail
 from .kvp import KeyValuePair
@@ -35,6 +45,9 @@ __all__ = [
     "Bitset", "Interruptible", "interruptible_synchronize",
     "get_logger", "set_level", "annotate", "annotated",
     "MemoryStats", "TrackingScope", "ResourceMonitor",
+    "DeviceMemoryResource", "TorchMemoryResource", "PoolMemoryResource",
+    "LimitingAdaptor", "TrackingAdaptor", "MemoryLimitExceeded",
+    "WorkspaceBuffer", "AllocationStats",
     "MDBuffer", "MemoryType", "memory_type_dispatcher", "copy_mdspan",
     "host_span", "device_span", "subspan",
 ]

@@ -115,20 +115,54 @@ class Resources:
         if self.device.type == "cuda":
             torch.cuda.synchronize(self.device)
 
-    # -- workspace memory resource (reference: resource/workspace_resource,
-    # set_workspace_resource / set_workspace_to_pool_resource). On MI355X the
-    # pool IS torch's caching allocator over the 288 GB HBM3E; these
-    # accessors expose the reference's workspace surface over it. -----------
-    def get_workspace(self, shape, dtype=torch.uint8) -> torch.Tensor:
-        """Workspace-backed scratch allocation (pool-recycled, uninitialized)."""
-        return torch.empty(shape, dtype=dtype, device=self.device)
+    # -- workspace memory resource (reference: resource/workspace_resource /
+    # large_workspace_resource, resource_types.hpp:37-40; mr adaptors). The
+    # default upstream is torch's caching allocator over the 288 GB HBM3E;
+    # set_workspace_resource injects a PoolMemoryResource / LimitingAdaptor /
+    # TrackingAdaptor chain (core/mr.py) when bounded or audited workspaces
+    # are required. Chunked algorithms size tiles from workspace_budget(). ---
+    def get_workspace_resource(self):
+        if not self.has_resource_factory("workspace_mr"):
+            from raft_amd.core.mr import TorchMemoryResource
+            self.add_resource_factory(
+                "workspace_mr", lambda: TorchMemoryResource(self.device))
+        return self.get_resource("workspace_mr")
+
+    def set_workspace_resource(self, mr) -> None:
+        """Inject a DeviceMemoryResource (pool/limiting/tracking chain)."""
+        self.add_resource_factory("workspace_mr", lambda: mr)
+
+    def get_large_workspace_resource(self):
+        """LARGE_WORKSPACE_RESOURCE slot: unbounded scratch for rare
+        huge temporaries (falls back to the workspace resource)."""
+        if self.has_resource_factory("large_workspace_mr"):
+            return self.get_resource("large_workspace_mr")
+        return self.get_workspace_resource()
+
+    def set_large_workspace_resource(self, mr) -> None:
+        self.add_resource_factory("large_workspace_mr", lambda: mr)
+
+    def get_workspace(self, shape, dtype=torch.uint8):
+        """Workspace-backed scratch (WorkspaceBuffer; use as context manager).
+        Raises MemoryLimitExceeded when a cap is configured and exceeded."""
+        return self.get_workspace_resource().allocate_tensor(shape, dtype)
 
     def set_workspace_limit(self, nbytes: int) -> None:
-        """Cap the device pool (reference limiting-adaptor analog)."""
-        if self.device.type == "cuda":
-            total = torch.cuda.get_device_properties(self.device).total_memory
-            torch.cuda.set_per_process_memory_fraction(
-                min(1.0, nbytes / total), self.device)
+        """Cap workspace allocations (limiting-adaptor over the current MR)."""
+        from raft_amd.core.mr import LimitingAdaptor
+        base = self.get_workspace_resource()
+        self.set_workspace_resource(LimitingAdaptor(base, nbytes))
+
+    def workspace_budget(self) -> int:
+        """Bytes an algorithm may use for scratch: the configured cap minus
+        outstanding, else a free-HBM estimate. Chunked algorithms (kNN,
+        pairwise) derive tile sizes from this instead of hardcoded rows."""
+        mr = self.get_workspace_resource()
+        avail = mr.available_bytes()
+        if avail is not None:
+            return avail
+        from raft_amd.core.mr import default_workspace_budget
+        return default_workspace_budget(self.device)
 
     def workspace_stats(self):
         """(allocated, reserved) bytes of the pool backing workspaces."""

@@ -35,15 +35,42 @@ from raft_amd.utils import row_chunks
 
 def knn(x: torch.Tensor, queries: torch.Tensor, k: int,
         metric: DistanceType | str = DistanceType.L2Expanded,
-        query_chunk: int = 16384, index_chunk: int = 262144,
-        fp32_mode: str = "auto"):
-    """k nearest rows of x for each query row. Returns (dists [q,k], idx [q,k])."""
+        query_chunk: int | None = None, index_chunk: int | None = None,
+        fp32_mode: str = "auto", res=None):
+    """k nearest rows of x for each query row. Returns (dists [q,k], idx [q,k]).
+
+    query_chunk/index_chunk default to sizes derived from the Resources
+    workspace budget (reference: workspace-resource-driven batching) — set a
+    workspace limit via res.set_workspace_limit(nbytes) to bound scratch.
+    """
+    from raft_amd.core.resources import get_resources
+    res = get_resources(res if res is not None else queries.device)
+    if query_chunk is None or index_chunk is None:
+        qc, ic = _tiles_from_budget(res.workspace_budget(), queries.shape[0],
+                                    x.shape[0], x.shape[1], k)
+        query_chunk = query_chunk or qc
+        index_chunk = index_chunk or ic
     if (queries.is_cuda and metric in (DistanceType.L2Expanded, "sqeuclidean")
             and x.shape[1] % 64 == 0 and x.shape[0] >= 8 * k
             and x.dtype in (torch.bfloat16, torch.float32)
             and not (x.dtype == torch.float32 and fp32_mode == "native")):
-        return _knn_gpu_filtered(x, queries, k, fp32_mode)
+        return _knn_gpu_filtered(x, queries, k, fp32_mode, res=res)
     return _knn_tiled(x, queries, k, metric, query_chunk, index_chunk, fp32_mode)
+
+
+def _tiles_from_budget(budget_bytes: int, q: int, n: int, d: int, k: int):
+    """Size the (query_chunk x index_chunk) distance tile from the workspace
+    budget: the fp32 tile is the dominant scratch of the tiled path; keep it
+    under half the budget, with floors that keep the chip busy."""
+    q_chunk = max(min(q, 16384), 1)
+    max_tile = max(budget_bytes // 2, 1 << 22)
+    ic_floor = min(max(n, 1), max(4 * k, 1024))
+    per_row = 4 * q_chunk + 8  # tile column + select candidates
+    index_chunk = max(min(min(max(n, 1), 262144), max_tile // per_row), ic_floor)
+    if index_chunk == ic_floor:
+        # the floor won: shrink the query chunk instead so the tile still fits
+        q_chunk = max(min(q_chunk, max_tile // (4 * ic_floor + 8)), 64)
+    return int(q_chunk), int(index_chunk)
 
 
 def _slices_of(t: torch.Tensor, fp32_mode: str):
@@ -64,7 +91,31 @@ def _norms(t: torch.Tensor) -> torch.Tensor:
     return t.float().pow(2).sum(dim=1).contiguous()
 
 
-def _knn_gpu_filtered(x, queries, k, fp32_mode, index_chunk: int = 4_000_000):
+def _knn_gpu_filtered(x, queries, k, fp32_mode, index_chunk: int = 4_000_000,
+                      res=None):
+    """Query-blocked driver: splits queries so the candidate buffers
+    (cap x 8 B/row) + the exact-rerank gather fit the workspace budget, then
+    runs the sample->filter->select pipeline per block."""
+    from raft_amd.core.resources import get_resources
+    res = get_resources(res if res is not None else queries.device)
+    m = queries.shape[0]
+    cap = max(16384, 32 * k)
+    per_row = cap * 8 + min(2 * k, cap) * x.shape[1] * 4 + 64
+    budget = max(res.workspace_budget(), 1 << 22)
+    q_block = max(256, min(m, int(budget // (2 * per_row)) or 1))
+    if q_block >= m:
+        return _knn_gpu_filtered_block(x, queries, k, fp32_mode, index_chunk, res)
+    out_d = torch.empty((m, k), dtype=torch.float32, device=queries.device)
+    out_i = torch.empty((m, k), dtype=torch.int64, device=queries.device)
+    for s0, s1 in row_chunks(m, q_block):
+        dv, iv = _knn_gpu_filtered_block(x, queries[s0:s1], k, fp32_mode,
+                                         index_chunk, res)
+        out_d[s0:s1] = dv
+        out_i[s0:s1] = iv
+    return out_d, out_i
+
+
+def _knn_gpu_filtered_block(x, queries, k, fp32_mode, index_chunk, res):
     ext = require_ext()
     m, d = queries.shape
     n = x.shape[0]
@@ -104,9 +155,14 @@ def _knn_gpu_filtered(x, queries, k, fp32_mode, index_chunk: int = 4_000_000):
                            + tail * (qn + xm))
 
     # ---- 2. filtered emission over the full index --------------------------
+    # candidate buffers come from the workspace resource (tracked/capped)
     cap = max(16384, 32 * k)
-    cand_d = torch.full((m, cap), float("inf"), dtype=torch.float32, device=dev)
-    cand_i = torch.full((m, cap), -1, dtype=torch.int32, device=dev)
+    ws_d = res.get_workspace((m, cap), torch.float32)
+    ws_i = res.get_workspace((m, cap), torch.int32)
+    cand_d = ws_d.view((m, cap), torch.float32)
+    cand_d.fill_(float("inf"))
+    cand_i = ws_i.view((m, cap), torch.int32)
+    cand_i.fill_(-1)
     cnt = torch.zeros(m, dtype=torch.int32, device=dev)
     for c0, c1 in row_chunks(n, index_chunk):
         xc = x[c0:c1]
@@ -144,6 +200,9 @@ def _knn_gpu_filtered(x, queries, k, fp32_mode, index_chunk: int = 4_000_000):
     else:
         vals, pos = select_k(cand_d, k, select_min=True)
         idx = torch.gather(cand_i, 1, pos.to(torch.int64)).to(torch.int64)
+
+    ws_d.free()
+    ws_i.free()
 
     # ---- 4. exact fallback for margin-failed / under/overflowed rows --------
     n_bad = int(bad.sum().item())

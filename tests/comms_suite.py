@@ -68,21 +68,23 @@ def collective_suite(c, device):
     out = c.reducescatter(t.clone())
     assert torch.equal(out, t[rank * 2:(rank + 1) * 2] * world)
 
-    # p2p sendrecv ring
-    send = torch.full((3,), float(rank), device=dev)
-    recv = torch.empty(3, device=dev)
-    dst = (rank + 1) % world
-    src = (rank - 1) % world
-    c.device_sendrecv(send, dst, recv, src)
-    assert torch.equal(recv, torch.full((3,), float(src), device=dev))
+    if world > 1:
+        # p2p sendrecv ring (self-send at world 1 is undefined for NCCL/RCCL)
+        send = torch.full((3,), float(rank), device=dev)
+        recv = torch.empty(3, device=dev)
+        dst = (rank + 1) % world
+        src = (rank - 1) % world
+        c.device_sendrecv(send, dst, recv, src)
+        assert torch.equal(recv, torch.full((3,), float(src), device=dev))
 
-    # multicast sendrecv: each rank sends to every peer, receives from all
-    msend = torch.full((2,), float(rank) + 10.0, device=dev)
-    peers = [r for r in range(world) if r != rank]
-    mrecvs = [torch.empty(2, device=dev) for _ in peers]
-    c.device_multicast_sendrecv(msend, peers, mrecvs, peers)
-    for r, peer in zip(mrecvs, peers):
-        assert torch.equal(r, torch.full((2,), float(peer) + 10.0, device=dev))
+        # multicast sendrecv: each rank sends to every peer, receives from all
+        msend = torch.full((2,), float(rank) + 10.0, device=dev)
+        peers = [r for r in range(world) if r != rank]
+        mrecvs = [torch.empty(2, device=dev) for _ in peers]
+        c.device_multicast_sendrecv(msend, peers, mrecvs, peers)
+        for r, peer in zip(mrecvs, peers):
+            assert torch.equal(r, torch.full((2,), float(peer) + 10.0,
+                                             device=dev))
 
     # comm_split: even/odd colors
     sub = c.comm_split(color=rank % 2, key=rank)

@@ -41,6 +41,17 @@ def main():
     torch.cuda.synchronize()
     print(f"[rank {rank}] RCCL collective suite OK", flush=True)
 
+    # flagship-shaped packed allreduce ([k*d + k + 1] fp32 — the kmeans
+    # per-iteration collective) through RCCL explicitly, so even a world-1
+    # run launches the real ncclAllReduce on a production-sized buffer
+    k_, d_ = 1024, 256
+    g = torch.Generator(device="cpu").manual_seed(1234)   # identical per rank
+    packed = torch.randn(k_ * d_ + k_ + 1, generator=g).to(dev)
+    ref = packed.clone()
+    c.allreduce(packed)
+    torch.cuda.synchronize()
+    assert torch.allclose(packed, ref * world, rtol=1e-5, atol=1e-5)
+
     # ---- distributed k-means on device: the bench.py --gpus N inner loop ----
     from raft_amd.cluster.kmeans import kmeans_iterate
     from raft_amd.comms import LoopbackComms

@@ -282,8 +282,9 @@ class TestWorkspaceResource:
     def test_workspace_accessors(self):
         from raft_amd.core import Resources
         r = Resources(torch.device("cpu"))
-        w = r.get_workspace((16, 4), dtype=torch.float32)
-        assert w.shape == (16, 4) and w.dtype == torch.float32
+        with r.get_workspace((16, 4), dtype=torch.float32) as w:
+            t = w.view((16, 4), torch.float32)
+            assert t.shape == (16, 4) and t.dtype == torch.float32
         assert r.workspace_stats() == (0, 0)   # cpu: no device pool
         r.empty_workspace_pool()               # no-op, must not raise
 

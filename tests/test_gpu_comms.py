@@ -39,29 +39,67 @@ def _run_torchrun(script_args, nproc=2, timeout=420):
 
 
 @pytest.mark.gpu
-def test_rccl_collectives_and_kmeans_2rank():
-    """Full comms_t suite + distributed kmeans over RCCL, 2 ranks, device
-    tensors. Both ranks map to GPU 0 on a single-GPU box."""
-    r = _run_torchrun(["tests/gpu_comms_worker.py"])
+def test_rccl_collectives_world1():
+    """Real RCCL communicator + every device collective at world size 1
+    (ncclAllReduce/Broadcast/AllGather/ReduceScatter kernels actually launch
+    on the device) plus the flagship-shaped packed allreduce and the device
+    kmeans loop. Runs on ANY GPU box — RCCL refuses two ranks on one device
+    ('Duplicate GPU detected', measured 2026-09-12), so single-box coverage
+    is world-1 RCCL here + the multi-rank suite below on multi-GPU nodes."""
+    r = _run_torchrun(["tests/gpu_comms_worker.py"], nproc=1)
     out = r.stdout + r.stderr
     assert r.returncode == 0, out[-4000:]
-    assert out.count("GPU_COMMS_WORKER_OK") == 2, out[-4000:]
+    assert out.count("GPU_COMMS_WORKER_OK") == 1, out[-4000:]
 
 
 @pytest.mark.gpu
-def test_bench_2rank_end_to_end():
-    """bench.py --gpus 2 exactly as the driver launches it (small problem so
+def test_rccl_collectives_and_kmeans_multirank():
+    """Full comms_t suite + distributed kmeans over RCCL with one rank per
+    GPU. Needs >= 2 devices (RCCL forbids rank sharing a device)."""
+    n = torch.cuda.device_count()
+    if n < 2:
+        pytest.skip("needs >= 2 GPUs: RCCL rejects two ranks on one device")
+    r = _run_torchrun(["tests/gpu_comms_worker.py"], nproc=min(n, 8))
+    out = r.stdout + r.stderr
+    assert r.returncode == 0, out[-4000:]
+    assert out.count("GPU_COMMS_WORKER_OK") == min(n, 8), out[-4000:]
+
+
+@pytest.mark.gpu
+def test_bench_multirank_end_to_end():
+    """bench.py --gpus N exactly as the driver launches it (small problem so
     it finishes in seconds); validates the whole distributed bench contract
-    on RCCL before the round-end 8-GPU run."""
+    on RCCL before the round-end 8-GPU run. Needs >= 2 devices."""
     import json
 
-    r = _run_torchrun(["bench.py", "--gpus", "2", "--steps", "3",
+    n = torch.cuda.device_count()
+    if n < 2:
+        pytest.skip("needs >= 2 GPUs: RCCL rejects two ranks on one device")
+    nproc = min(n, 8)
+    r = _run_torchrun(["bench.py", "--gpus", str(nproc), "--steps", "3",
                        "--warmup", "1", "--rows", "200000", "--dim", "256",
-                       "--k", "256"])
+                       "--k", "256", "--no-pairwise"], nproc=nproc)
     out = r.stdout + r.stderr
     assert r.returncode == 0, out[-4000:]
     line = [l for l in r.stdout.splitlines() if l.strip().startswith("{")]
     assert line, out[-4000:]
     j = json.loads(line[-1])
-    assert j["n_gpus"] == 2 and j["steps"] == 3
+    assert j["n_gpus"] == nproc and j["steps"] == 3
     assert j["value"] > 0 and j["config"]["inertia"] > 0
+
+
+@pytest.mark.gpu
+def test_bench_world1_rccl_contract():
+    """bench.py under torchrun with WORLD_SIZE=1: validates the driver's
+    launch contract (env rendezvous, JSON line) on a single GPU."""
+    import json
+
+    r = _run_torchrun(["bench.py", "--gpus", "1", "--steps", "2",
+                       "--warmup", "1", "--rows", "100000", "--dim", "64",
+                       "--k", "128", "--no-pairwise"], nproc=1)
+    out = r.stdout + r.stderr
+    assert r.returncode == 0, out[-4000:]
+    line = [l for l in r.stdout.splitlines() if l.strip().startswith("{")]
+    assert line, out[-4000:]
+    j = json.loads(line[-1])
+    assert j["n_gpus"] == 1 and j["value"] > 0
