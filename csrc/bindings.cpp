@@ -66,6 +66,11 @@ bool fused_l2nn_2d_supported(int nslice, long long m, int n, int d);
 void launch_fused_l2nn_2d(const void**, const void**, const float*, const float*,
                           float*, float*, int*, float*, int*, float*,
                           long long, int, int, int, hipStream_t);
+// from fused_l2nn_256.hip (256^2 counted-vmcnt double-buffered engine)
+bool fused_l2nn_256_supported(int nslice, long long m, int n, int d);
+void launch_fused_l2nn_256(const void**, const void**, const float*, const float*,
+                           float*, float*, int*, float*, int*, float*,
+                           long long, int, int, int, hipStream_t);
 // from fused_l2nn_v2.hip (persistent-X variant + w8 wide-tile variant)
 bool fused_l2nn_persist_supported(int nslice, int d);
 bool fused_l2nn_w8_supported(int nslice, int n, int d);
@@ -384,7 +389,18 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> fused_l2nn_split(
     const char* e = getenv("RAFT_AMD_L2NN_W8");
     return e && e[0] == '0';
   }();
-  if (!use_persist && raft_amd::fused_l2nn_2d_supported(nslice, m, (int)n, (int)d)) {
+  if (!use_persist && raft_amd::fused_l2nn_256_supported(nslice, m, (int)n, (int)d)) {
+    const long long n_col_tiles = n / 256;
+    auto pd = torch::empty({n_col_tiles * m}, xn.options());
+    auto pd2 = torch::empty({n_col_tiles * m}, xn.options());
+    auto pi = torch::empty({n_col_tiles * m}, xn.options().dtype(torch::kInt32));
+    raft_amd::launch_fused_l2nn_256(xsl, csl, xn.data_ptr<float>(),
+                                    cn.data_ptr<float>(), pd.data_ptr<float>(),
+                                    pd2.data_ptr<float>(), pi.data_ptr<int>(),
+                                    dmin.data_ptr<float>(), amin.data_ptr<int>(),
+                                    dmin2.data_ptr<float>(), m, (int)n, (int)d,
+                                    nslice, cur_stream());
+  } else if (!use_persist && raft_amd::fused_l2nn_2d_supported(nslice, m, (int)n, (int)d)) {
     const long long n_col_tiles = n / 128;
     auto pd = torch::empty({n_col_tiles * m}, xn.options());
     auto pd2 = torch::empty({n_col_tiles * m}, xn.options());

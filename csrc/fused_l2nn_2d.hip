@@ -34,7 +34,7 @@ namespace raft_amd {
 
 bool l2nn_phased();  // fused_l2nn.hip: RAFT_AMD_L2NN_PHASED
 
-template <int NSLICE, int GT, bool PHASED = false>
+template <int NSLICE, int GT, bool PHASED = false, bool ADIRECT = false>
 __launch_bounds__(256, 2)
 __global__ void fused_l2nn_2d_kernel(const __bf16* __restrict__ x0,
                                      const __bf16* __restrict__ x1,
@@ -54,8 +54,13 @@ __global__ void fused_l2nn_2d_kernel(const __bf16* __restrict__ x0,
   const __bf16* const cg[3] = {c0, c1, c2};
 #pragma unroll
   for (int s = 0; s < NSLICE; s++) {
-    xs[s] = smem + s * 8192;
-    cs[s] = smem + (NSLICE + s) * 8192;
+    if constexpr (ADIRECT) {
+      cs[s] = smem + s * 8192;   // C-only LDS; X reads are direct
+      xs[s] = smem;
+    } else {
+      xs[s] = smem + s * 8192;
+      cs[s] = smem + (NSLICE + s) * 8192;
+    }
   }
 
   // bijective XCD-contiguous remap (guide T1 / m204 variant): XCD x = bid%8
@@ -93,7 +98,10 @@ __global__ void fused_l2nn_2d_kernel(const __bf16* __restrict__ x0,
 #pragma unroll
       for (int b = 0; b < 4; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-    if constexpr (PHASED && NSLICE == 2) {
+    if constexpr (ADIRECT) {
+      mfma_tile_kloop_ad<NSLICE>(xg, cg, cs, acc, row0, col0, d, m - 1, n - 1,
+                                 wr, wc, lane);
+    } else if constexpr (PHASED && NSLICE == 2) {
       mfma_tile_kloop_p2(xg, cg, xs, cs, acc, row0, col0, d, m - 1, n - 1, wr,
                          wc, lane);
     } else {
@@ -327,6 +335,16 @@ __global__ void l2nn_combine_partials_kernel(const float* __restrict__ pd,
   if (dmin2) dmin2[row] = v2 + x;
 }
 
+// shared combine launch (also used by the 256^2 engine in fused_l2nn_256.hip)
+void launch_l2nn_combine(const float* pd, const float* pd2, const int* pi,
+                         const float* xn, float* dmin, int* amin, float* dmin2,
+                         long long m, int n_groups, hipStream_t stream) {
+  const long long cgrid = (m + 255) / 256;
+  hipLaunchKernelGGL(l2nn_combine_partials_kernel, dim3((int)cgrid), dim3(256),
+                     0, stream, pd, pd2, pi, xn, dmin, amin, dmin2, m,
+                     n_groups);
+}
+
 static int l2nn_2d_gt() {
   static const int gt = [] {
     const char* e = getenv("RAFT_AMD_L2NN_GT");
@@ -361,8 +379,13 @@ void launch_fused_l2nn_2d(const void** xsl, const void** csl, const float* xn,
   const int n_row_tiles = (int)((m + 127) / 128);
   const int n_groups = n / 128 / gt;
   const int grid = n_row_tiles * n_groups;
-  // kloop needs NSLICE*2*16KiB; the epilogue reuses it ([128][33] f32 x2 + i32)
-  const size_t lds_kloop = (size_t)nslice * 2 * 8192 * sizeof(__bf16);
+  static const bool adirect = [] {
+    const char* e = getenv("RAFT_AMD_L2NN_AD");
+    return e && e[0] == '1';
+  }();
+  // kloop needs NSLICE*2*16KiB (C-only when A-direct); the epilogue reuses
+  // it ([128][33] f32 x2 + i32)
+  const size_t lds_kloop = (size_t)nslice * (adirect ? 1 : 2) * 8192 * sizeof(__bf16);
   const size_t lds = lds_kloop > 3 * 128 * 33 * 4 ? lds_kloop : 3 * 128 * 33 * 4;
   const __bf16* x0 = (const __bf16*)xsl[0];
   const __bf16* x1 = (const __bf16*)(nslice > 1 ? xsl[1] : xsl[0]);
@@ -371,9 +394,16 @@ void launch_fused_l2nn_2d(const void** xsl, const void** csl, const float* xn,
   const __bf16* c1 = (const __bf16*)(nslice > 1 ? csl[1] : csl[0]);
   const __bf16* c2 = (const __bf16*)(nslice > 2 ? csl[2] : csl[0]);
 #define L2NN2D_LAUNCH(NS, G, P)                                               \
-  hipLaunchKernelGGL((fused_l2nn_2d_kernel<NS, G, P>), dim3(grid), dim3(256), \
-                     lds, stream, x0, x1, x2, c0, c1, c2, cn, pd, pd2, pi, m, \
-                     n, d, n_groups)
+  do {                                                                         \
+    if (adirect)                                                               \
+      hipLaunchKernelGGL((fused_l2nn_2d_kernel<NS, G, false, true>),           \
+                         dim3(grid), dim3(256), lds, stream, x0, x1, x2, c0,   \
+                         c1, c2, cn, pd, pd2, pi, m, n, d, n_groups);          \
+    else                                                                       \
+      hipLaunchKernelGGL((fused_l2nn_2d_kernel<NS, G, P>), dim3(grid),         \
+                         dim3(256), lds, stream, x0, x1, x2, c0, c1, c2, cn,   \
+                         pd, pd2, pi, m, n, d, n_groups);                      \
+  } while (0)
   static const bool bk32 = [] {
     const char* e = getenv("RAFT_AMD_L2NN_BK32");
     return e && e[0] == '1';

@@ -150,6 +150,93 @@ __device__ __forceinline__ void mfma_tile_kloop(
 }
 
 // ---------------------------------------------------------------------------
+// A-direct K-loop: X fragments load straight global->register (no LDS leg).
+// Within a block each X element is read by exactly ONE wave exactly once, so
+// LDS-staging X is a pure write+read round-trip — the 2-block/CU 128^2
+// engine is LDS-write bound (128 KiB staged per CU per K-tile vs 768 MFMA
+// ~= 1024 vs 768 cycles). Dropping the X leg halves LDS writes, halves the
+// per-K-step drain (8 gloads instead of 16) and shrinks LDS to C-only
+// (NSLICE*16 KiB). X reads hit the XCD-local L2 (row-tile remap keeps the
+// panel hot). C stays LDS-staged: its fragments are read by all 4 waves.
+// ---------------------------------------------------------------------------
+template <int NSLICE>
+__device__ __forceinline__ void mfma_tile_kloop_ad(
+    const __bf16* const (&xg)[3], const __bf16* const (&cg)[3],
+    __bf16* (&cs)[NSLICE],
+    f32x4 (&acc)[4][4], long long row0, long long col0, int d,
+    long long m_max, long long n_max, int wr, int wc, int lane) {
+  const int k_tiles = d / 64;
+  const int t = threadIdx.x;
+  const int wv = t / RAFT_AMD_WAVE;
+  long long bc[4];
+  int ldst[4];
+#pragma unroll
+  for (int j = 0; j < 4; j++) {
+    const int o = j * 4096 + t * 16;
+    const int o_src = mfma_swz(o);
+    const int r = o_src >> 7;
+    const int k = (o_src & 127) >> 1;
+    long long rc = col0 + r;
+    if (rc > n_max) rc = n_max;
+    bc[j] = rc * (long long)d + k;
+    ldst[j] = (j * 4096 + wv * 1024) / 2;
+  }
+  // hoisted per-fragment X row base offsets (global elements)
+  long long ax[4];
+#pragma unroll
+  for (int fr = 0; fr < 4; fr++) {
+    long long r = row0 + wr * 64 + fr * 16 + (lane & 15);
+    if (r > m_max) r = m_max;
+    ax[fr] = r * (long long)d + (lane >> 4) * 8;
+  }
+  for (int kt = 0; kt < k_tiles; kt++) {
+    const long long koff = (long long)kt * 64;
+#pragma unroll
+    for (int s = 0; s < NSLICE; s++)
+#pragma unroll
+      for (int j = 0; j < 4; j++)
+        GLOAD_LDS(cg[s] + bc[j] + koff, cs[s] + ldst[j]);
+    // X fragments direct from global, issued before the drain so their
+    // latency overlaps the C-staging wait
+    bf16x8 a_frag[NSLICE][2][4];
+#pragma unroll
+    for (int s = 0; s < NSLICE; s++)
+#pragma unroll
+      for (int kf = 0; kf < 2; kf++)
+#pragma unroll
+        for (int fr = 0; fr < 4; fr++)
+          a_frag[s][kf][fr] = *reinterpret_cast<const bf16x8*>(
+              xg[s] + ax[fr] + koff + kf * 32);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+#pragma unroll
+    for (int kf = 0; kf < 2; kf++) {
+      bf16x8 b_frag[NSLICE][4];
+#pragma unroll
+      for (int fc = 0; fc < 4; fc++) {
+        const int c = wc * 64 + fc * 16 + (lane & 15);
+        const int byte = mfma_swz(c * 128 + (kf * 32 + (lane >> 4) * 8) * 2);
+#pragma unroll
+        for (int s = 0; s < NSLICE; s++)
+          b_frag[s][fc] = *reinterpret_cast<const bf16x8*>((const char*)cs[s] + byte);
+      }
+#pragma unroll
+      for (int fr = 0; fr < 4; fr++)
+#pragma unroll
+        for (int fc = 0; fc < 4; fc++) {
+#pragma unroll
+          for (int p = 0; p < mfma_n_products<NSLICE>(); p++) {
+            acc[fr][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_frag[MFMA_PROD_A[p]][kf][fr], b_frag[MFMA_PROD_B[p]][fc],
+                acc[fr][fc], 0, 0, 0);
+          }
+        }
+    }
+    __syncthreads();
+  }
+}
+
+// ---------------------------------------------------------------------------
 // 1-slice paired K-loop: stage TWO k-tiles (64 KiB LDS) under ONE
 // vmcnt(0)+barrier drain. MEASURED LOSER on the kNN filter sweep
 // (9873 -> 7409 q/s, d=128): halving the drain count does not pay for the
