@@ -457,12 +457,15 @@ torch::Tensor pairwise_l2_mfma(std::vector<torch::Tensor> x_slices,
   } else {
     o = torch::empty({m, n}, xn.options());
   }
-  static const bool no_256 = [] {
+  static const bool use_256 = [] {
     const char* e = getenv("RAFT_AMD_PW256");
-    return e && e[0] == '0';
+    return e && e[0] == '1';
   }();
-  if (!no_256 && nslice <= 2 && m >= 512 && n >= 512) {
-    // 256x256-tile kernel: 4x fewer workgroups (first-order at small d)
+  if (use_256 && nslice <= 2 && m >= 512 && n >= 512) {
+    // 256x256-tile kernel (BK=32 counted-vmcnt): 4x fewer workgroups, but
+    // 1 block/CU serializes the tile-store phase with the K-loop — measured
+    // 547 vs the 128^2 kernel's 736 Gdist/s at 1Mx128 (round 2); kept for
+    // A/B via RAFT_AMD_PW256=1
     raft_amd::launch_pairwise_l2_mfma256(xsl, csl, xn.data_ptr<float>(),
                                          yn.data_ptr<float>(), o.data_ptr<float>(),
                                          m, n, (int)d, o.size(1), nslice, sqrt_out,

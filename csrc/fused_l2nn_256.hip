@@ -1,30 +1,14 @@
-// Fused L2-NN, 256x256-tile engine with a counted-vmcnt double-buffered
-// K-loop (guide T3/T4 + 256^2 template): the round-1 128^2 kernel's per-K-step
-// `s_waitcnt vmcnt(0) + s_barrier` drains the global_load_lds queue before any
-// wave crosses the barrier — the structural ~20% stall of the 2-barrier
-// structure. This engine:
+// Fused L2-NN, 256x256-tile engine over the BK=32 product-phase
+// counted-vmcnt K-loop (mfma_common.h mfma256_bk32_kloop; design history and
+// measured A/B in profiles/pmc_l2nn_256_ab.txt — on THIS op the round-1
+// 128^2 2-blocks/CU engine stays faster, so this engine ships disabled; it
+// is the production K-loop for the pairwise tile kernel, whose old schedule
+// was 1 block/CU WITH full drains).
 //
-//   * 256x256 output tile, 8 waves (2 row x 4 col), per-wave output 128x64
-//     (acc[8][4] f32x4) — 2x the arithmetic intensity per staged byte of the
-//     128^2 tile, and HALF the LDS write traffic per MFMA (the 128^2 engine
-//     at 2 blocks/CU is LDS-write bound: 128 KiB staged per CU per K-tile
-//     vs 768 MFMA);
-//   * BK=32 with ALL slice tiles resident per K-chunk (A0,A1,B0,B1 = 64 KiB)
-//     so the 3 split-bf16 products run from ONE staging — no re-staging
-//     (the first 256^2 attempt staged per-product virtual tiles: 192 KiB
-//     LDS writes per K-tile, measured LDS-write bound at 3.2-4.2 ms);
-//   * double-buffered (128 KiB LDS, 1 block/CU): product-phases with counted
-//     `s_waitcnt vmcnt(N)` — never a full drain in the main loop; gloads for
-//     K-chunk kt+1 are spread across kt's phases in read-consumption order
-//     (B0, A0, B1, A1) so each wait retires loads issued >= 2 phases earlier;
-//   * s_setprio(1) around each MFMA block (guide T5);
-//   * fused top-2 argmin epilogue (never materializes the distance tile):
-//     per-lane top-2 over the tile columns, 16-lane shuffle merge, [256][17]
-//     LDS wave-combine, per-(row, col-tile) partials merged by
-//     l2nn_combine_partials_kernel (fused_l2nn_2d.hip).
-//
-// Grid: (row-tile, col-tile) pairs, XCD-contiguous bijective remap (same as
-// fused_l2nn_2d.hip) so one XCD's resident blocks share X row-tiles in L2.
+// Epilogue: fused top-2 argmin (never materializes the distance tile):
+// per-lane top-2 over the tile columns, 16-lane shuffle merge, [256][17]
+// LDS wave-combine, per-(row, col-tile) partials merged by
+// l2nn_combine_partials_kernel (fused_l2nn_2d.hip).
 //
 // Reference parity: RAFT's fusedL2NN (k-means assignment), the contraction
 // engine role of linalg/detail/contractions.cuh:140-307.
@@ -47,15 +31,6 @@ __global__ void fused_l2nn_256_kernel(const __bf16* __restrict__ x0,
                                       int* __restrict__ pi,
                                       long long m, int n, int d, int n_groups) {
   extern __shared__ __bf16 smem[];
-  // per buffer: A-slice tiles then B-slice tiles, 16 KiB (8192 bf16) each
-  // [256 rows][32 k] swizzled; NSLICE=1 uses the 8 regions as a 4-deep ring.
-  // CRITICAL: all LDS accesses go through `smem` + integer offsets — a
-  // runtime-selected pointer ARRAY (As[cur][s]) defeats LLVM's address-space
-  // inference and every frag read compiles to flat_load_dwordx4 (generic
-  // path: 64-bit address VALU + cselect chains; measured +1.3e9 VALU and
-  // 1.5x kernel time before this fix).
-#define A_ELE(buf, s) (((buf) * 4 + (s)) * 8192)
-#define B_ELE(buf, s) (((buf) * 4 + 2 + (s)) * 8192)
 
   // XCD-contiguous bijective remap (see fused_l2nn_2d.hip)
   const int nwg = gridDim.x;
@@ -72,30 +47,8 @@ __global__ void fused_l2nn_256_kernel(const __bf16* __restrict__ x0,
   const int w = tid >> 6;
   const int wm = w >> 2, wn = w & 3;   // 2 x 4 wave grid
 
-  // hoisted per-thread staging offsets: 2 rounds x (row*ld + k) per tile.
-  // Balanced ADD-rotation swizzle for the 64 B rows: row r's k-slot s
-  // (16 B units) is stored at slot (s + (r>>1)) & 3, so a 16-row b128
-  // column read covers every 128 B bank window exactly 2x — the minimum
-  // for 16 lanes x 16 B at 64 B row stride (the XOR swizzle collapses 4
-  // even rows onto 2 slots: measured 1.0 LDS conflict per MFMA).
-  long long bx[2], bc[2];
-  int ldst[2];
-#pragma unroll
-  for (int j = 0; j < 2; j++) {
-    const int o = j * 8192 + tid * 16;   // 16-aligned linear dest byte
-    const int rr = o >> 6;               // dest row 0..255
-    const int sd = (o >> 4) & 3;         // dest slot
-    const int kk = (((sd - (rr >> 1)) & 3) << 3);  // source k (bf16 elems)
-    long long rx = row0 + rr;
-    if (rx > m - 1) rx = m - 1;
-    bx[j] = rx * (long long)d + kk;
-    long long rc = col0 + rr;
-    if (rc > n - 1) rc = n - 1;
-    bc[j] = rc * (long long)d + kk;
-    ldst[j] = (j * 8192 + w * 1024) / 2;
-  }
-
-  const int kt_tiles = d / 32;
+  Mfma256BK32 st;
+  mfma256_bk32_setup(st, row0, col0, d, m - 1, n - 1, wm, wn, lane);
 
   f32x4 acc[8][4];
 #pragma unroll
@@ -103,162 +56,7 @@ __global__ void fused_l2nn_256_kernel(const __bf16* __restrict__ x0,
 #pragma unroll
     for (int b = 0; b < 4; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  // 2 gloads staging one 16 KiB [256][32] tile region
-#define L2NN256_GA(slice, buf, koff)                                           \
-  do {                                                                         \
-    GLOAD_LDS((slice == 0 ? x0 : x1) + bx[0] + (koff),                         \
-              smem + A_ELE(buf, slice) + ldst[0]);                             \
-    GLOAD_LDS((slice == 0 ? x0 : x1) + bx[1] + (koff),                         \
-              smem + A_ELE(buf, slice) + ldst[1]);                             \
-  } while (0)
-#define L2NN256_GB(slice, buf, koff)                                           \
-  do {                                                                         \
-    GLOAD_LDS((slice == 0 ? c0 : c1) + bc[0] + (koff),                         \
-              smem + B_ELE(buf, slice) + ldst[0]);                             \
-    GLOAD_LDS((slice == 0 ? c0 : c1) + bc[1] + (koff),                         \
-              smem + B_ELE(buf, slice) + ldst[1]);                             \
-  } while (0)
-
-  // ds-read byte offsets are K-invariant: hoist them (the per-phase address
-  // recomputation was 60% extra VALU in the first 256^2 attempt)
-  const int ks = lane >> 4;            // K slot 0..3 (8 bf16 each)
-  int a_off[8], b_off[4];
-#pragma unroll
-  for (int fr = 0; fr < 8; fr++) {
-    const int rr = wm * 128 + fr * 16 + (lane & 15);
-    a_off[fr] = rr * 64 + (((ks + (rr >> 1)) & 3) << 4);
-  }
-#pragma unroll
-  for (int fc = 0; fc < 4; fc++) {
-    const int cc = wn * 64 + fc * 16 + (lane & 15);
-    b_off[fc] = cc * 64 + (((ks + (cc >> 1)) & 3) << 4);
-  }
-  const char* lds_base = reinterpret_cast<const char*>(smem);
-  auto ds_b = [&](int buf_ele, bf16x8(&b_frag)[4]) {
-    const char* base = lds_base + buf_ele * 2;
-#pragma unroll
-    for (int fc = 0; fc < 4; fc++)
-      b_frag[fc] = *reinterpret_cast<const bf16x8*>(base + b_off[fc]);
-  };
-  // A fragments stream through a 2-row register window (a full a_frag[8]
-  // per slice pushed peak pressure to 224+ VGPRs -> scratch spills in the
-  // K-loop, whose vmem ops also pollute the counted vmcnt waits)
-  auto mfma32 = [&](int buf_a_ele, const bf16x8(&b_frag)[4]) {
-    const char* abase = lds_base + buf_a_ele * 2;
-#pragma unroll
-    for (int qd = 0; qd < 4; qd++) {
-      bf16x8 a2[2];
-#pragma unroll
-      for (int fi = 0; fi < 2; fi++)
-        a2[fi] =
-            *reinterpret_cast<const bf16x8*>(abase + a_off[qd * 2 + fi]);
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int fi = 0; fi < 2; fi++)
-#pragma unroll
-        for (int fc = 0; fc < 4; fc++)
-          acc[qd * 2 + fi][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a2[fi], b_frag[fc], acc[qd * 2 + fi][fc], 0, 0, 0);
-      __builtin_amdgcn_s_setprio(0);
-    }
-  };
-
-  if constexpr (NSLICE == 1) {
-    // plain bf16 GEMM schedule: the 8 x 16 KiB LDS regions form a 4-deep
-    // (A,B) ring, staging TWO K-chunks ahead — a chunk's loads get 2 full
-    // chunks (64 MFMA) of flight time before the counted wait needs them.
-    // One barrier + one counted wait per chunk.
-    // ring r -> regions A_ELE(r>>1, r&1) / B_ELE(r>>1, r&1)
-#define RING_A(r) A_ELE((r) >> 1, (r)&1)
-#define RING_B(r) B_ELE((r) >> 1, (r)&1)
-    L2NN256_GB(0, 0, 0);                      // kt0 -> ring 0
-    L2NN256_GA(0, 0, 0);
-#pragma unroll
-    for (int j = 0; j < 2; j++) {             // kt1 -> ring 1
-      GLOAD_LDS(c0 + bc[j] + 32, smem + RING_B(1) + ldst[j]);
-      GLOAD_LDS(x0 + bx[j] + 32, smem + RING_A(1) + ldst[j]);
-    }
-    for (int kt = 0; kt < kt_tiles; kt++) {
-      const int cur = kt & 3;
-      if (kt + 2 < kt_tiles) {
-        const long long koff = (long long)(kt + 2) * 32;
-        const int nxt = (kt + 2) & 3;
-#pragma unroll
-        for (int j = 0; j < 2; j++) {
-          GLOAD_LDS(c0 + bc[j] + koff, smem + RING_B(nxt) + ldst[j]);
-          GLOAD_LDS(x0 + bx[j] + koff, smem + RING_A(nxt) + ldst[j]);
-        }
-        asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
-      } else if (kt + 1 < kt_tiles) {
-        asm volatile("s_waitcnt vmcnt(4)\n\ts_barrier" ::: "memory");
-      } else {
-        asm volatile("s_waitcnt vmcnt(0)\n\ts_barrier" ::: "memory");
-      }
-      bf16x8 b_frag[4];
-      ds_b(RING_B(cur), b_frag);
-      mfma32(RING_A(cur), b_frag);
-    }
-#undef RING_A
-#undef RING_B
-  } else {
-    // split-bf16: 3 product-phases per K-chunk from ONE staging (96 MFMA per
-    // 64 KiB staged — no per-product re-staging). Per phase: issue kt+1's
-    // gloads, ONE counted wait + ONE barrier, ds-read, MFMA. The phase-0
-    // wait is loop-carried (phase 2's vmcnt(4) already guaranteed B0+A0 of
-    // the next chunk), so the steady loop has 3 barriers / 3 counted waits /
-    // zero full drains per 96 MFMA.
-    // Issue order per chunk (staging kt+1): [B0r0 B0r1 A0r0] [A0r1 B1r0 B1r1]
-    // [A1r0 A1r1] — read-consumption order, >= 1 full phase of flight time.
-    L2NN256_GB(0, 0, 0);
-    L2NN256_GA(0, 0, 0);
-    L2NN256_GB(1, 0, 0);
-    L2NN256_GA(1, 0, 0);
-    // phase 0 needs B0+A0 (leave B1,A1 in flight)
-    asm volatile("s_waitcnt vmcnt(4)\n\ts_barrier" ::: "memory");
-    for (int kt = 0; kt < kt_tiles; kt++) {
-      const int cur = kt & 1;
-      const bool more = kt + 1 < kt_tiles;
-      const long long koff = (long long)(kt + 1) * 32;
-      bf16x8 b0[4], b1[4];
-      // ---- phase 0: p00 = A0 x B0 (wait carried from prev phase 2) --------
-      if (more) {
-        L2NN256_GB(0, cur ^ 1, koff);
-        GLOAD_LDS(x0 + bx[0] + koff, smem + A_ELE(cur ^ 1, 0) + ldst[0]);
-      }
-      ds_b(B_ELE(cur, 0), b0);
-      mfma32(A_ELE(cur, 0), b0);
-      // ---- phase 1: p01 = A0 x B1 (first read of B1(kt)) ------------------
-      if (more) {
-        GLOAD_LDS(x0 + bx[1] + koff, smem + A_ELE(cur ^ 1, 0) + ldst[1]);
-        L2NN256_GB(1, cur ^ 1, koff);
-        // in flight: B1,A1(kt) + 6(kt+1); retire B1(kt)
-        asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
-      } else {
-        asm volatile("s_waitcnt vmcnt(2)\n\ts_barrier" ::: "memory");
-      }
-      ds_b(B_ELE(cur, 1), b1);
-      mfma32(A_ELE(cur, 0), b1);
-      // ---- phase 2: p10 = A1 x B0 (first read of A1(kt)) ------------------
-      if (more) {
-        L2NN256_GA(1, cur ^ 1, koff);
-        // in flight: A1(kt) + 8(kt+1); retire A1(kt)
-        asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
-      } else {
-        asm volatile("s_waitcnt vmcnt(0)\n\ts_barrier" ::: "memory");
-      }
-      mfma32(A_ELE(cur, 1), b0);
-      // next chunk's phase 0 reads B0,A0(kt+1): retire the oldest 4
-      if (more) {
-        asm volatile("s_waitcnt vmcnt(4)\n\ts_barrier" ::: "memory");
-      } else {
-        asm volatile("s_barrier" ::: "memory");
-      }
-    }
-  }
-#undef L2NN256_GA
-#undef L2NN256_GB
-#undef A_ELE
-#undef B_ELE
+  mfma256_bk32_kloop<NSLICE>(x0, x1, c0, c1, smem, st, acc, d / 32);
 
   // ---- fused top-2 argmin epilogue ----------------------------------------
   // per-lane top-2 over this thread's 4 columns per (fr, reg) row slot

@@ -51,16 +51,24 @@ __device__ __forceinline__ void mfma_stage_tile128(const __bf16* __restrict__ g,
 // 128-tiles) are read into that XCD's 4 MiB L2 once and reused 8x each.
 // Launch with grid = rg*cg*64 (rg=ceil(R/8), cg=ceil(C/8)); callers must
 // early-return when rt/ct land past R/C (grid inflation on ragged edges).
+// rg < 0 selects COLUMN-major slot order within the 8x8 window (consecutive
+// slots walk adjacent col-tiles of one row-tile — adjacent 512 B output
+// segments, better DRAM page locality for tile-writing kernels); rg > 0 is
+// the original row-major order. L2 panel reuse is symmetric under the swap.
 __device__ __forceinline__ void xcd_supertile_decode(int rg, long long* rt,
                                                      long long* ct) {
+  const bool cm = rg < 0;
+  if (cm) rg = -rg;
   const int nwg = gridDim.x;
   const int bid = blockIdx.x;
   const int q = nwg >> 3, r = nwg & 7;
   const int xcd = bid & 7, slot = bid >> 3;
   const int t = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + slot;
   const int g = t >> 6, w = t & 63;
-  *rt = (long long)(g % rg) * 8 + (w & 7);
-  *ct = (long long)(g / rg) * 8 + (w >> 3);
+  const int wr = cm ? (w >> 3) : (w & 7);
+  const int wc = cm ? (w & 7) : (w >> 3);
+  *rt = (long long)(g % rg) * 8 + wr;
+  *ct = (long long)(g / rg) * 8 + wc;
 }
 
 template <int NSLICE>
@@ -147,6 +155,208 @@ __device__ __forceinline__ void mfma_tile_kloop(
     }
     __syncthreads();
   }
+}
+
+// ---------------------------------------------------------------------------
+// 256x256-tile BK=32 product-phase counted-vmcnt K-loop (8 waves, 2x4 grid,
+// per-wave 128x64 output, acc[8][4]). All slice tiles of one K-chunk are
+// resident (A0,A1,B0,B1 = 4 x 16 KiB), double-buffered in 128 KiB LDS; the
+// NSLICE=2 split-bf16 products run as 3 phases from ONE staging with
+// counted `s_waitcnt vmcnt(N)` (no full drains in the main loop); NSLICE=1
+// uses the 8 regions as a 4-deep ring staging 2 chunks ahead. See
+// fused_l2nn_256.hip header + profiles/pmc_l2nn_256_ab.txt for the measured
+// design history (swizzle, addrspace and spill lessons).
+//
+// LDS region layout (16 KiB = 8192 bf16 each):
+//   A_ELE(buf, s) = (buf*4 + s)     * 8192
+//   B_ELE(buf, s) = (buf*4 + 2 + s) * 8192
+// ---------------------------------------------------------------------------
+
+#define RAFT_MFMA256_A_ELE(buf, s) (((buf) * 4 + (s)) * 8192)
+#define RAFT_MFMA256_B_ELE(buf, s) (((buf) * 4 + 2 + (s)) * 8192)
+
+struct Mfma256BK32 {
+  long long bx[2], bc[2];  // per-thread global staging offsets (2 rounds)
+  int ldst[2];             // LDS dest offsets (bf16 elements)
+  int a_off[8], b_off[4];  // hoisted ds-read byte offsets
+};
+
+// Balanced ADD-rotation swizzle for 64 B LDS rows: row r's k-slot s (16 B
+// units) stored at slot (s + (r>>1)) & 3 — a 16-row b128 column read covers
+// every 128 B bank window exactly 2x (the minimum; the XOR swizzle measures
+// 1.0 conflicts/MFMA here).
+__device__ __forceinline__ void mfma256_bk32_setup(
+    Mfma256BK32& st, long long row0, long long col0, int d, long long m_max,
+    long long n_max, int wm, int wn, int lane) {
+  const int tid = threadIdx.x;
+  const int w = tid >> 6;
+#pragma unroll
+  for (int j = 0; j < 2; j++) {
+    const int o = j * 8192 + tid * 16;   // 16-aligned linear dest byte
+    const int rr = o >> 6;               // dest row 0..255
+    const int sd = (o >> 4) & 3;         // dest slot
+    const int kk = (((sd - (rr >> 1)) & 3) << 3);  // source k (bf16 elems)
+    long long rx = row0 + rr;
+    if (rx > m_max) rx = m_max;
+    st.bx[j] = rx * (long long)d + kk;
+    long long rc = col0 + rr;
+    if (rc > n_max) rc = n_max;
+    st.bc[j] = rc * (long long)d + kk;
+    st.ldst[j] = (j * 8192 + w * 1024) / 2;
+  }
+  const int ks = lane >> 4;            // K slot 0..3 (8 bf16 each)
+#pragma unroll
+  for (int fr = 0; fr < 8; fr++) {
+    const int rr = wm * 128 + fr * 16 + (lane & 15);
+    st.a_off[fr] = rr * 64 + (((ks + (rr >> 1)) & 3) << 4);
+  }
+#pragma unroll
+  for (int fc = 0; fc < 4; fc++) {
+    const int cc = wn * 64 + fc * 16 + (lane & 15);
+    st.b_off[fc] = cc * 64 + (((ks + (cc >> 1)) & 3) << 4);
+  }
+}
+
+template <int NSLICE>
+__device__ __forceinline__ void mfma256_bk32_kloop(
+    const __bf16* __restrict__ x0, const __bf16* __restrict__ x1,
+    const __bf16* __restrict__ c0, const __bf16* __restrict__ c1,
+    __bf16* smem, const Mfma256BK32& st, f32x4 (&acc)[8][4], int kt_tiles) {
+  static_assert(NSLICE <= 2, "mfma256_bk32_kloop: NSLICE 1 or 2");
+#define M256_GA(slice, buf, koff)                                              \
+  do {                                                                         \
+    GLOAD_LDS((slice == 0 ? x0 : x1) + st.bx[0] + (koff),                      \
+              smem + RAFT_MFMA256_A_ELE(buf, slice) + st.ldst[0]);             \
+    GLOAD_LDS((slice == 0 ? x0 : x1) + st.bx[1] + (koff),                      \
+              smem + RAFT_MFMA256_A_ELE(buf, slice) + st.ldst[1]);             \
+  } while (0)
+#define M256_GB(slice, buf, koff)                                              \
+  do {                                                                         \
+    GLOAD_LDS((slice == 0 ? c0 : c1) + st.bc[0] + (koff),                      \
+              smem + RAFT_MFMA256_B_ELE(buf, slice) + st.ldst[0]);             \
+    GLOAD_LDS((slice == 0 ? c0 : c1) + st.bc[1] + (koff),                      \
+              smem + RAFT_MFMA256_B_ELE(buf, slice) + st.ldst[1]);             \
+  } while (0)
+  const char* lds_base = reinterpret_cast<const char*>(smem);
+  auto ds_b = [&](int buf_ele, bf16x8(&b_frag)[4]) {
+    const char* base = lds_base + buf_ele * 2;
+#pragma unroll
+    for (int fc = 0; fc < 4; fc++)
+      b_frag[fc] = *reinterpret_cast<const bf16x8*>(base + st.b_off[fc]);
+  };
+  // A fragments stream through a 2-row register window (a full a_frag[8]
+  // per slice pushes peak pressure past 256 VGPRs -> K-loop scratch spills,
+  // whose vmem ops also corrupt the counted vmcnt arithmetic)
+  auto mfma32 = [&](int buf_a_ele, const bf16x8(&b_frag)[4]) {
+    const char* abase = lds_base + buf_a_ele * 2;
+#pragma unroll
+    for (int qd = 0; qd < 4; qd++) {
+      bf16x8 a2[2];
+#pragma unroll
+      for (int fi = 0; fi < 2; fi++)
+        a2[fi] =
+            *reinterpret_cast<const bf16x8*>(abase + st.a_off[qd * 2 + fi]);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int fi = 0; fi < 2; fi++)
+#pragma unroll
+        for (int fc = 0; fc < 4; fc++)
+          acc[qd * 2 + fi][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a2[fi], b_frag[fc], acc[qd * 2 + fi][fc], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+  };
+
+  if constexpr (NSLICE == 1) {
+    // 4-deep (A,B) ring staging TWO K-chunks ahead: one barrier + one
+    // counted wait per chunk; a chunk's loads get 2 chunks (64 MFMA) of
+    // flight time before their wait.
+#define M256_RING_A(r) RAFT_MFMA256_A_ELE((r) >> 1, (r)&1)
+#define M256_RING_B(r) RAFT_MFMA256_B_ELE((r) >> 1, (r)&1)
+    M256_GB(0, 0, 0);                      // kt0 -> ring 0
+    M256_GA(0, 0, 0);
+#pragma unroll
+    for (int j = 0; j < 2; j++) {          // kt1 -> ring 1
+      GLOAD_LDS(c0 + st.bc[j] + 32, smem + M256_RING_B(1) + st.ldst[j]);
+      GLOAD_LDS(x0 + st.bx[j] + 32, smem + M256_RING_A(1) + st.ldst[j]);
+    }
+    for (int kt = 0; kt < kt_tiles; kt++) {
+      const int cur = kt & 3;
+      if (kt + 2 < kt_tiles) {
+        const long long koff = (long long)(kt + 2) * 32;
+        const int nxt = (kt + 2) & 3;
+#pragma unroll
+        for (int j = 0; j < 2; j++) {
+          GLOAD_LDS(c0 + st.bc[j] + koff, smem + M256_RING_B(nxt) + st.ldst[j]);
+          GLOAD_LDS(x0 + st.bx[j] + koff, smem + M256_RING_A(nxt) + st.ldst[j]);
+        }
+        asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
+      } else if (kt + 1 < kt_tiles) {
+        asm volatile("s_waitcnt vmcnt(4)\n\ts_barrier" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(0)\n\ts_barrier" ::: "memory");
+      }
+      bf16x8 b_frag[4];
+      ds_b(M256_RING_B(cur), b_frag);
+      mfma32(M256_RING_A(cur), b_frag);
+    }
+#undef M256_RING_A
+#undef M256_RING_B
+  } else {
+    // split-bf16: 3 product-phases per K-chunk from ONE staging (96 MFMA
+    // per 64 KiB staged). Per phase: issue kt+1's gloads, ONE counted wait
+    // + ONE barrier, ds-read, MFMA; the phase-0 wait is loop-carried.
+    // Issue order per chunk: [B0r0 B0r1 A0r0] [A0r1 B1r0 B1r1] [A1r0 A1r1].
+    M256_GB(0, 0, 0);
+    M256_GA(0, 0, 0);
+    M256_GB(1, 0, 0);
+    M256_GA(1, 0, 0);
+    // phase 0 needs B0+A0 (leave B1,A1 in flight)
+    asm volatile("s_waitcnt vmcnt(4)\n\ts_barrier" ::: "memory");
+    for (int kt = 0; kt < kt_tiles; kt++) {
+      const int cur = kt & 1;
+      const bool more = kt + 1 < kt_tiles;
+      const long long koff = (long long)(kt + 1) * 32;
+      bf16x8 b0[4], b1[4];
+      // ---- phase 0: p00 = A0 x B0 (wait carried from prev phase 2) -------
+      if (more) {
+        M256_GB(0, cur ^ 1, koff);
+        GLOAD_LDS(x0 + st.bx[0] + koff,
+                  smem + RAFT_MFMA256_A_ELE(cur ^ 1, 0) + st.ldst[0]);
+      }
+      ds_b(RAFT_MFMA256_B_ELE(cur, 0), b0);
+      mfma32(RAFT_MFMA256_A_ELE(cur, 0), b0);
+      // ---- phase 1: p01 = A0 x B1 (first read of B1(kt)) -----------------
+      if (more) {
+        GLOAD_LDS(x0 + st.bx[1] + koff,
+                  smem + RAFT_MFMA256_A_ELE(cur ^ 1, 0) + st.ldst[1]);
+        M256_GB(1, cur ^ 1, koff);
+        // in flight: B1,A1(kt) + 6(kt+1); retire B1(kt)
+        asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(2)\n\ts_barrier" ::: "memory");
+      }
+      ds_b(RAFT_MFMA256_B_ELE(cur, 1), b1);
+      mfma32(RAFT_MFMA256_A_ELE(cur, 0), b1);
+      // ---- phase 2: p10 = A1 x B0 (first read of A1(kt)) -----------------
+      if (more) {
+        M256_GA(1, cur ^ 1, koff);
+        // in flight: A1(kt) + 8(kt+1); retire A1(kt)
+        asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(0)\n\ts_barrier" ::: "memory");
+      }
+      mfma32(RAFT_MFMA256_A_ELE(cur, 1), b0);
+      // next chunk's phase 0 reads B0,A0(kt+1): retire the oldest 4
+      if (more) {
+        asm volatile("s_waitcnt vmcnt(4)\n\ts_barrier" ::: "memory");
+      } else {
+        asm volatile("s_barrier" ::: "memory");
+      }
+    }
+  }
+#undef M256_GA
+#undef M256_GB
 }
 
 // ---------------------------------------------------------------------------

@@ -58,8 +58,46 @@ __global__ void pairwise_l2_kernel(const __bf16* __restrict__ x0,
   mfma_tile_kloop<NSLICE>(xg, cg, xs, cs, acc, row0, col0, d, m - 1, n - 1,
                           wr, wc, lane);
 
-  // fused epilogue + write: d2 = max(xn[r] + yn[c] - 2 acc, 0)
+  // fused epilogue: d2 = max(xn[r] + yn[c] - 2 acc, 0).
   // C/D layout: col = lane&15 (+fc*16), row = (lane>>4)*4 + reg (+fr*16).
+  if (row0 + 128 <= m && col0 + 128 <= n && (ldo & 3) == 0) {
+    // interior tile: stage through padded LDS ([128][132] fp32, 67.5 KiB —
+    // fits 2 blocks/CU) and dump as coalesced NONTEMPORAL dwordx4 rows.
+    // The direct path issues 64 scalar 4 B stores/thread each with 64-bit
+    // row*ldo address math — measured 12:1 VALU:MFMA, epilogue-dominated
+    // (418 Gdist/s vs the ~1.6 Tdist/s write roofline).
+    __syncthreads();  // K-loop LDS is dead; reuse
+    float* tile = reinterpret_cast<float*>(smem);  // [128][132]
+#pragma unroll
+    for (int fr = 0; fr < 4; fr++) {
+#pragma unroll
+      for (int reg = 0; reg < 4; reg++) {
+        const int rl = wr * 64 + fr * 16 + (lane >> 4) * 4 + reg;
+        const float xv = xn[row0 + rl];
+#pragma unroll
+        for (int fc = 0; fc < 4; fc++) {
+          const int cl = wc * 64 + fc * 16 + (lane & 15);
+          float v = fmaxf(xv + yn[col0 + cl] - 2.f * acc[fr][fc][reg], 0.f);
+          if (sqrt_out) v = sqrtf(v);
+          tile[rl * 132 + cl] = v;
+        }
+      }
+    }
+    __syncthreads();
+    // 256 threads x float4: each round writes 8 rows (32 els/row = 8 thr)
+    const int tr = threadIdx.x >> 5;        // 0..7: row within round
+    const int tc = (threadIdx.x & 31) * 4;  // 0..124: col (float4)
+    const float* src = tile + tr * 132 + tc;
+    float* dst = out + (row0 + tr) * ldo + col0 + tc;
+    const long long dstep = 8 * ldo;
+#pragma unroll
+    for (int rnd = 0; rnd < 16; rnd++) {
+      const f32x4 v4 = *reinterpret_cast<const f32x4*>(src + rnd * 8 * 132);
+      __builtin_nontemporal_store(
+          v4, reinterpret_cast<f32x4*>(dst + rnd * dstep));
+    }
+    return;
+  }
 #pragma unroll
   for (int fr = 0; fr < 4; fr++) {
 #pragma unroll
@@ -205,10 +243,20 @@ void launch_pairwise_l2_mfma(const void** xsl, const void** csl, const float* xn
                              const float* yn, float* out, long long m, long long n,
                              int d, long long ldo, int nslice, bool sqrt_out,
                              hipStream_t stream) {
-  const int rg = (int)((m + 1023) / 1024);       // ceil(R/8), R=ceil(m/128)
+  int rg = (int)((m + 1023) / 1024);             // ceil(R/8), R=ceil(m/128)
   const int cg = (int)((n + 1023) / 1024);
   dim3 grid((unsigned)((long long)rg * cg * 64));
-  const size_t lds = (size_t)nslice * 2 * 8192 * sizeof(__bf16);
+  // column-major supertile slot order (adjacent output segments written by
+  // temporally adjacent blocks — DRAM page locality); rg<0 flags it to
+  // xcd_supertile_decode. RAFT_AMD_PW_RM=1 restores row-major for A/B.
+  static const bool rowmajor = [] {
+    const char* e = getenv("RAFT_AMD_PW_RM");
+    return e && e[0] == '1';
+  }();
+  if (!rowmajor) rg = -rg;
+  const size_t lds_k = (size_t)nslice * 2 * 8192 * sizeof(__bf16);
+  const size_t lds_epi = 128 * 132 * 4;  // padded fp32 staging tile
+  const size_t lds = lds_k > lds_epi ? lds_k : lds_epi;
   const __bf16* x0 = (const __bf16*)xsl[0];
   const __bf16* x1 = (const __bf16*)(nslice > 1 ? xsl[1] : xsl[0]);
   const __bf16* x2 = (const __bf16*)(nslice > 2 ? xsl[2] : xsl[0]);
@@ -273,7 +321,7 @@ __device__ __forceinline__ void pw_stage_rows(const __bf16* __restrict__ g,
 }
 
 template <int NSLICE>
-__launch_bounds__(512, 2)
+__launch_bounds__(512, 1)
 __global__ void pairwise_l2_256_kernel(const __bf16* __restrict__ x0,
                                        const __bf16* __restrict__ x1,
                                        const __bf16* __restrict__ x2,
@@ -287,15 +335,6 @@ __global__ void pairwise_l2_256_kernel(const __bf16* __restrict__ x0,
                                        long long ldo, int sqrt_out, int rg) {
   constexpr int BLOCK = 512;
   extern __shared__ __bf16 smem[];
-  const __bf16* const xg[3] = {x0, x1, x2};
-  const __bf16* const cg[3] = {c0, c1, c2};
-  __bf16* xs[NSLICE];
-  __bf16* cs[NSLICE];
-#pragma unroll
-  for (int s = 0; s < NSLICE; s++) {
-    xs[s] = smem + s * 16384;                    // [256][64]
-    cs[s] = smem + NSLICE * 16384 + s * 16384;   // [256][64]
-  }
 
   const int lane = threadIdx.x % RAFT_AMD_WAVE;
   const int w = threadIdx.x / RAFT_AMD_WAVE;
@@ -312,51 +351,64 @@ __global__ void pairwise_l2_256_kernel(const __bf16* __restrict__ x0,
 #pragma unroll
     for (int b = 0; b < 4; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int k_tiles = d / 64;
-  for (int kt = 0; kt < k_tiles; kt++) {
-#pragma unroll
-    for (int s = 0; s < NSLICE; s++) {
-      pw_stage_rows<256, BLOCK>(xg[s], xs[s], row0, (long long)kt * 64, d, m - 1);
-      pw_stage_rows<256, BLOCK>(cg[s], cs[s], col0, (long long)kt * 64, d, n - 1);
-    }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
+  if constexpr (NSLICE <= 2) {
+    // BK=32 product-phase counted-vmcnt K-loop (mfma_common.h): the old
+    // per-K-step `vmcnt(0)+barrier` full drain ran at 1 block/CU with
+    // nothing to hide it — at d=128 (2 K-steps/tile) the drains dominated.
+    Mfma256BK32 st;
+    mfma256_bk32_setup(st, row0, col0, d, m - 1, n - 1, wr, wc, lane);
+    mfma256_bk32_kloop<NSLICE>(x0, x1, c0, c1, smem, st, acc, d / 32);
+  }
+  static_assert(NSLICE <= 2, "256^2 pairwise kernel: nslice 3 routes to the 128^2 kernel");
 
+  if (row0 + 256 <= m && col0 + 256 <= n && (ldo & 3) == 0) {
+    // interior tile: stage each 128-row half through padded LDS
+    // ([128][260] fp32, 130 KiB) and dump as coalesced f32x4 rows —
+    // the per-element path issues 32 scalar 4 B stores/thread with 64-bit
+    // row*ldo address math each. sqrt_out bit 1 selects nontemporal vs
+    // cached stores (A/B: write-combine behavior differs).
+    float* tile = reinterpret_cast<float*>(smem);  // [128][260]
+    const int tid = threadIdx.x;
+    const int tr = tid >> 6;                // 0..7
+    const int tc = (tid & 63) * 4;          // 0..252
+    const bool nt = (sqrt_out & 2) != 0;
+    const bool do_sqrt = (sqrt_out & 1) != 0;
 #pragma unroll
-    for (int kf = 0; kf < 2; kf++) {
-      bf16x8 a_frag[NSLICE][8], b_frag[NSLICE][4];
-      const int kbyte = (kf * 32 + (lane >> 4) * 8) * 2;
+    for (int half = 0; half < 2; half++) {
+      __syncthreads();
+      if (wr == half) {
 #pragma unroll
-      for (int fr = 0; fr < 8; fr++) {
-        const int r = wr * 128 + fr * 16 + (lane & 15);
-        const int byte = mfma_swz(r * 128 + kbyte);
+        for (int fr = 0; fr < 8; fr++) {
 #pragma unroll
-        for (int s = 0; s < NSLICE; s++)
-          a_frag[s][fr] = *reinterpret_cast<const bf16x8*>((const char*)xs[s] + byte);
-      }
+          for (int reg = 0; reg < 4; reg++) {
+            const int rl = fr * 16 + (lane >> 4) * 4 + reg;
+            const float xv = xn[row0 + half * 128 + rl];
 #pragma unroll
-      for (int fc = 0; fc < 4; fc++) {
-        const int c = wc * 64 + fc * 16 + (lane & 15);
-        const int byte = mfma_swz(c * 128 + kbyte);
-#pragma unroll
-        for (int s = 0; s < NSLICE; s++)
-          b_frag[s][fc] = *reinterpret_cast<const bf16x8*>((const char*)cs[s] + byte);
-      }
-#pragma unroll
-      for (int fr = 0; fr < 8; fr++)
-#pragma unroll
-        for (int fc = 0; fc < 4; fc++) {
-#pragma unroll
-          for (int p = 0; p < mfma_n_products<NSLICE>(); p++) {
-            acc[fr][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                a_frag[MFMA_PROD_A[p]][fr], b_frag[MFMA_PROD_B[p]][fc],
-                acc[fr][fc], 0, 0, 0);
+            for (int fc = 0; fc < 4; fc++) {
+              const int cl = wc * 64 + fc * 16 + (lane & 15);
+              float v =
+                  fmaxf(xv + yn[col0 + cl] - 2.f * acc[fr][fc][reg], 0.f);
+              if (do_sqrt) v = sqrtf(v);
+              tile[rl * 260 + cl] = v;
+            }
           }
         }
+      }
+      __syncthreads();
+      float* dst0 = out + (row0 + half * 128 + tr) * ldo + col0 + tc;
+      const float* src0 = tile + tr * 260 + tc;
+#pragma unroll
+      for (int rnd = 0; rnd < 16; rnd++) {
+        const f32x4 v4 = *reinterpret_cast<const f32x4*>(src0 + rnd * 8 * 260);
+        f32x4* dp = reinterpret_cast<f32x4*>(dst0 + (long long)rnd * 8 * ldo);
+        if (nt)
+          __builtin_nontemporal_store(v4, dp);
+        else
+          *dp = v4;
+      }
     }
-    __syncthreads();
+    return;
   }
-
 #pragma unroll
   for (int fr = 0; fr < 8; fr++) {
 #pragma unroll
@@ -369,7 +421,7 @@ __global__ void pairwise_l2_256_kernel(const __bf16* __restrict__ x0,
         const long long col = col0 + wc * 64 + fc * 16 + (lane & 15);
         if (col < n) {
           float v = fmaxf(xv + yn[col] - 2.f * acc[fr][fc][reg], 0.f);
-          if (sqrt_out) v = sqrtf(v);
+          if (sqrt_out & 1) v = sqrtf(v);
           // streaming output (never re-read): non-temporal keeps the L2
           // clear for the operand tiles
           __builtin_nontemporal_store(v, &out[row * ldo + col]);
@@ -386,7 +438,16 @@ void launch_pairwise_l2_mfma256(const void** xsl, const void** csl, const float*
   const int rg = (int)((m + 2047) / 2048);       // ceil(R/8), R=ceil(m/256)
   const int cg = (int)((n + 2047) / 2048);
   dim3 grid((unsigned)((long long)rg * cg * 64));
-  const size_t lds = (size_t)nslice * 2 * 16384 * sizeof(__bf16);
+  // K-loop: 8 x 16 KiB BK=32 dbuf regions (128 KiB); epilogue staging:
+  // [128][260] fp32 (130 KiB) — allocate the max
+  const size_t lds_k = 8 * 16384;
+  const size_t lds_epi = 128 * 260 * 4;
+  const size_t lds = lds_k > lds_epi ? lds_k : lds_epi;
+  static const bool nt_store = [] {
+    const char* e = getenv("RAFT_AMD_PW_NT");
+    return e && e[0] == '1';  // default: cached (write-combining via L2)
+  }();
+  const int so_bits = (sqrt_out ? 1 : 0) | (nt_store ? 2 : 0);
   const __bf16* x0 = (const __bf16*)xsl[0];
   const __bf16* x1 = (const __bf16*)(nslice > 1 ? xsl[1] : xsl[0]);
   const __bf16* x2 = (const __bf16*)(nslice > 2 ? xsl[2] : xsl[0]);
@@ -402,23 +463,18 @@ void launch_pairwise_l2_mfma256(const void** xsl, const void** csl, const float*
       static bool a1 = (set_attr((const void*)&pairwise_l2_256_kernel<1>), true);
       (void)a1;
       hipLaunchKernelGGL((pairwise_l2_256_kernel<1>), grid, dim3(512), lds, stream,
-                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out, rg);
+                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, so_bits, rg);
       break;
     }
     case 2: {
       static bool a2 = (set_attr((const void*)&pairwise_l2_256_kernel<2>), true);
       (void)a2;
       hipLaunchKernelGGL((pairwise_l2_256_kernel<2>), grid, dim3(512), lds, stream,
-                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out, rg);
+                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, so_bits, rg);
       break;
     }
-    default: {
-      static bool a3 = (set_attr((const void*)&pairwise_l2_256_kernel<3>), true);
-      (void)a3;
-      hipLaunchKernelGGL((pairwise_l2_256_kernel<3>), grid, dim3(512), lds, stream,
-                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out, rg);
-      break;
-    }
+    default:
+      throw std::runtime_error("pairwise 256^2 kernel: nslice must be 1 or 2");
   }
 }
 
