@@ -172,27 +172,31 @@ __device__ __forceinline__ void mfma_tile_kloop(
 //   B_ELE(buf, s) = (buf*4 + 2 + s) * 8192
 // ---------------------------------------------------------------------------
 
-#define RAFT_MFMA256_A_ELE(buf, s) (((buf) * 4 + (s)) * 8192)
-#define RAFT_MFMA256_B_ELE(buf, s) (((buf) * 4 + 2 + (s)) * 8192)
+// region size in bf16 elements for a [ROWS][32] tile (ROWS = 256 or 128)
+#define RAFT_MFMA256_REG(ROWS) ((ROWS) * 32)
+#define RAFT_MFMA256_A_ELE(buf, s) (((buf) * 4 + (s)) * RAFT_MFMA256_REG(ROWS))
+#define RAFT_MFMA256_B_ELE(buf, s) (((buf) * 4 + 2 + (s)) * RAFT_MFMA256_REG(ROWS))
 
 struct Mfma256BK32 {
   long long bx[2], bc[2];  // per-thread global staging offsets (2 rounds)
   int ldst[2];             // LDS dest offsets (bf16 elements)
-  int a_off[8], b_off[4];  // hoisted ds-read byte offsets
+  int a_off[8], b_off[4];  // hoisted ds-read byte offsets (a_off[ROWS/32])
 };
 
 // Balanced ADD-rotation swizzle for 64 B LDS rows: row r's k-slot s (16 B
 // units) stored at slot (s + (r>>1)) & 3 — a 16-row b128 column read covers
 // every 128 B bank window exactly 2x (the minimum; the XOR swizzle measures
 // 1.0 conflicts/MFMA here).
+template <int ROWS = 256>
 __device__ __forceinline__ void mfma256_bk32_setup(
     Mfma256BK32& st, long long row0, long long col0, int d, long long m_max,
     long long n_max, int wm, int wn, int lane) {
+  constexpr int BLOCK = ROWS * 2;   // threads: 512 (256^2) / 256 (128^2)
   const int tid = threadIdx.x;
   const int w = tid >> 6;
 #pragma unroll
   for (int j = 0; j < 2; j++) {
-    const int o = j * 8192 + tid * 16;   // 16-aligned linear dest byte
+    const int o = j * BLOCK * 16 + tid * 16;   // 16-aligned linear dest byte
     const int rr = o >> 6;               // dest row 0..255
     const int sd = (o >> 4) & 3;         // dest slot
     const int kk = (((sd - (rr >> 1)) & 3) << 3);  // source k (bf16 elems)
@@ -202,12 +206,12 @@ __device__ __forceinline__ void mfma256_bk32_setup(
     long long rc = col0 + rr;
     if (rc > n_max) rc = n_max;
     st.bc[j] = rc * (long long)d + kk;
-    st.ldst[j] = (j * 8192 + w * 1024) / 2;
+    st.ldst[j] = (j * BLOCK * 16 + w * 1024) / 2;
   }
   const int ks = lane >> 4;            // K slot 0..3 (8 bf16 each)
 #pragma unroll
-  for (int fr = 0; fr < 8; fr++) {
-    const int rr = wm * 128 + fr * 16 + (lane & 15);
+  for (int fr = 0; fr < ROWS / 32; fr++) {
+    const int rr = wm * (ROWS / 2) + fr * 16 + (lane & 15);
     st.a_off[fr] = rr * 64 + (((ks + (rr >> 1)) & 3) << 4);
   }
 #pragma unroll
@@ -217,12 +221,14 @@ __device__ __forceinline__ void mfma256_bk32_setup(
   }
 }
 
-template <int NSLICE>
+template <int NSLICE, int ROWS = 256>
 __device__ __forceinline__ void mfma256_bk32_kloop(
     const __bf16* __restrict__ x0, const __bf16* __restrict__ x1,
     const __bf16* __restrict__ c0, const __bf16* __restrict__ c1,
-    __bf16* smem, const Mfma256BK32& st, f32x4 (&acc)[8][4], int kt_tiles) {
+    __bf16* smem, const Mfma256BK32& st, f32x4 (&acc)[ROWS / 32][4],
+    int kt_tiles) {
   static_assert(NSLICE <= 2, "mfma256_bk32_kloop: NSLICE 1 or 2");
+  constexpr int FR = ROWS / 32;
 #define M256_GA(slice, buf, koff)                                              \
   do {                                                                         \
     GLOAD_LDS((slice == 0 ? x0 : x1) + st.bx[0] + (koff),                      \
@@ -250,7 +256,7 @@ __device__ __forceinline__ void mfma256_bk32_kloop(
   auto mfma32 = [&](int buf_a_ele, const bf16x8(&b_frag)[4]) {
     const char* abase = lds_base + buf_a_ele * 2;
 #pragma unroll
-    for (int qd = 0; qd < 4; qd++) {
+    for (int qd = 0; qd < FR / 2; qd++) {
       bf16x8 a2[2];
 #pragma unroll
       for (int fi = 0; fi < 2; fi++)

@@ -16,7 +16,7 @@
 
 namespace raft_amd {
 
-template <int NSLICE>
+template <int NSLICE, bool BK32 = false>
 __launch_bounds__(256, 2)
 __global__ void pairwise_l2_kernel(const __bf16* __restrict__ x0,
                                    const __bf16* __restrict__ x1,
@@ -55,8 +55,16 @@ __global__ void pairwise_l2_kernel(const __bf16* __restrict__ x0,
 #pragma unroll
     for (int b = 0; b < 4; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  mfma_tile_kloop<NSLICE>(xg, cg, xs, cs, acc, row0, col0, d, m - 1, n - 1,
-                          wr, wc, lane);
+  if constexpr (BK32 && NSLICE <= 2) {
+    // counted-vmcnt BK=32 dbuf K-loop at 2 blocks/CU: no full drains AND
+    // cross-block overlap (A/B RAFT_AMD_PW_BK32=1)
+    Mfma256BK32 st;
+    mfma256_bk32_setup<128>(st, row0, col0, d, m - 1, n - 1, wr, wc, lane);
+    mfma256_bk32_kloop<NSLICE, 128>(x0, x1, c0, c1, smem, st, acc, d / 32);
+  } else {
+    mfma_tile_kloop<NSLICE>(xg, cg, xs, cs, acc, row0, col0, d, m - 1, n - 1,
+                            wr, wc, lane);
+  }
 
   // fused epilogue: d2 = max(xn[r] + yn[c] - 2 acc, 0).
   // C/D layout: col = lane&15 (+fc*16), row = (lane>>4)*4 + reg (+fr*16).
@@ -254,6 +262,10 @@ void launch_pairwise_l2_mfma(const void** xsl, const void** csl, const float* xn
     return e && e[0] == '1';
   }();
   if (!rowmajor) rg = -rg;
+  static const bool bk32 = [] {
+    const char* e = getenv("RAFT_AMD_PW_BK32");
+    return e && e[0] == '1';
+  }();
   const size_t lds_k = (size_t)nslice * 2 * 8192 * sizeof(__bf16);
   const size_t lds_epi = 128 * 132 * 4;  // padded fp32 staging tile
   const size_t lds = lds_k > lds_epi ? lds_k : lds_epi;
@@ -265,12 +277,22 @@ void launch_pairwise_l2_mfma(const void** xsl, const void** csl, const float* xn
   const __bf16* c2 = (const __bf16*)(nslice > 2 ? csl[2] : csl[0]);
   switch (nslice) {
     case 1:
-      hipLaunchKernelGGL((pairwise_l2_kernel<1>), grid, dim3(256), lds, stream,
-                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out, rg);
+      if (bk32 && d % 32 == 0)
+        hipLaunchKernelGGL((pairwise_l2_kernel<1, true>), grid, dim3(256), lds,
+                           stream, x0, x1, x2, c0, c1, c2, xn, yn, out, m, n,
+                           d, ldo, sqrt_out, rg);
+      else
+        hipLaunchKernelGGL((pairwise_l2_kernel<1>), grid, dim3(256), lds, stream,
+                           x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out, rg);
       break;
     case 2:
-      hipLaunchKernelGGL((pairwise_l2_kernel<2>), grid, dim3(256), lds, stream,
-                         x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out, rg);
+      if (bk32 && d % 32 == 0)
+        hipLaunchKernelGGL((pairwise_l2_kernel<2, true>), grid, dim3(256), lds,
+                           stream, x0, x1, x2, c0, c1, c2, xn, yn, out, m, n,
+                           d, ldo, sqrt_out, rg);
+      else
+        hipLaunchKernelGGL((pairwise_l2_kernel<2>), grid, dim3(256), lds, stream,
+                           x0, x1, x2, c0, c1, c2, xn, yn, out, m, n, d, ldo, sqrt_out, rg);
       break;
     case 3: {
       static bool attr_set = false;
@@ -526,48 +548,58 @@ __global__ void pairwise_l2_filter256_kernel(const __bf16* __restrict__ x0,
 #pragma unroll
     for (int b = 0; b < 4; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int k_tiles = d / 64;
-  for (int kt = 0; kt < k_tiles; kt++) {
+  if constexpr (NSLICE <= 2) {
+    // counted-vmcnt BK=32 dbuf K-loop (mfma_common.h): the old full
+    // per-K-step drain ran at 1 block/CU with nothing to hide it — this
+    // kernel has no tile writes, so the K-loop IS the kernel
+    // (RAFT_AMD_KNN_DRAIN=1 in the launcher restores the old loop for A/B)
+    Mfma256BK32 st;
+    mfma256_bk32_setup<256>(st, row0, col0, d, m - 1, n - 1, wr, wc, lane);
+    mfma256_bk32_kloop<NSLICE, 256>(x0, x1, c0, c1, smem, st, acc, d / 32);
+  } else {
+    const int k_tiles = d / 64;
+    for (int kt = 0; kt < k_tiles; kt++) {
 #pragma unroll
-    for (int s = 0; s < NSLICE; s++) {
-      pw_stage_rows<256, BLOCK>(xg[s], xs[s], row0, (long long)kt * 64, d, m - 1);
-      pw_stage_rows<256, BLOCK>(cg[s], cs[s], col0, (long long)kt * 64, d, n - 1);
-    }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
-#pragma unroll
-    for (int kf = 0; kf < 2; kf++) {
-      bf16x8 a_frag[NSLICE][8], b_frag[NSLICE][4];
-      const int kbyte = (kf * 32 + (lane >> 4) * 8) * 2;
-#pragma unroll
-      for (int fr = 0; fr < 8; fr++) {
-        const int r = wr * 128 + fr * 16 + (lane & 15);
-        const int byte = mfma_swz(r * 128 + kbyte);
-#pragma unroll
-        for (int s = 0; s < NSLICE; s++)
-          a_frag[s][fr] = *reinterpret_cast<const bf16x8*>((const char*)xs[s] + byte);
+      for (int s = 0; s < NSLICE; s++) {
+        pw_stage_rows<256, BLOCK>(xg[s], xs[s], row0, (long long)kt * 64, d, m - 1);
+        pw_stage_rows<256, BLOCK>(cg[s], cs[s], col0, (long long)kt * 64, d, n - 1);
       }
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
 #pragma unroll
-      for (int fc = 0; fc < 4; fc++) {
-        const int c = wc * 64 + fc * 16 + (lane & 15);
-        const int byte = mfma_swz(c * 128 + kbyte);
+      for (int kf = 0; kf < 2; kf++) {
+        bf16x8 a_frag[NSLICE][8], b_frag[NSLICE][4];
+        const int kbyte = (kf * 32 + (lane >> 4) * 8) * 2;
 #pragma unroll
-        for (int s = 0; s < NSLICE; s++)
-          b_frag[s][fc] = *reinterpret_cast<const bf16x8*>((const char*)cs[s] + byte);
-      }
+        for (int fr = 0; fr < 8; fr++) {
+          const int r = wr * 128 + fr * 16 + (lane & 15);
+          const int byte = mfma_swz(r * 128 + kbyte);
 #pragma unroll
-      for (int fr = 0; fr < 8; fr++)
+          for (int s = 0; s < NSLICE; s++)
+            a_frag[s][fr] = *reinterpret_cast<const bf16x8*>((const char*)xs[s] + byte);
+        }
 #pragma unroll
         for (int fc = 0; fc < 4; fc++) {
+          const int c = wc * 64 + fc * 16 + (lane & 15);
+          const int byte = mfma_swz(c * 128 + kbyte);
 #pragma unroll
-          for (int p = 0; p < mfma_n_products<NSLICE>(); p++) {
-            acc[fr][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                a_frag[MFMA_PROD_A[p]][fr], b_frag[MFMA_PROD_B[p]][fc],
-                acc[fr][fc], 0, 0, 0);
-          }
+          for (int s = 0; s < NSLICE; s++)
+            b_frag[s][fc] = *reinterpret_cast<const bf16x8*>((const char*)cs[s] + byte);
         }
+#pragma unroll
+        for (int fr = 0; fr < 8; fr++)
+#pragma unroll
+          for (int fc = 0; fc < 4; fc++) {
+#pragma unroll
+            for (int p = 0; p < mfma_n_products<NSLICE>(); p++) {
+              acc[fr][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  a_frag[MFMA_PROD_A[p]][fr], b_frag[MFMA_PROD_B[p]][fc],
+                  acc[fr][fc], 0, 0, 0);
+            }
+          }
+      }
+      __syncthreads();
     }
-    __syncthreads();
   }
 
 #pragma unroll
@@ -604,7 +636,9 @@ void launch_pairwise_l2_filter256(const void** xsl, const void** csl, const floa
   const int rg = (int)((m + 2047) / 2048);       // ceil(R/8), R=ceil(m/256)
   const int cg = (int)((n + 2047) / 2048);
   dim3 grid((unsigned)((long long)rg * cg * 64));
-  const size_t lds = (size_t)nslice * 2 * 16384 * sizeof(__bf16);
+  // nslice<=2: BK=32 counted-vmcnt loop needs the 8 x 16 KiB dbuf regions
+  const size_t lds = nslice <= 2 ? (size_t)(8 * 16384)
+                                 : (size_t)nslice * 2 * 16384 * sizeof(__bf16);
   const __bf16* x0 = (const __bf16*)xsl[0];
   const __bf16* x1 = (const __bf16*)(nslice > 1 ? xsl[1] : xsl[0]);
   const __bf16* x2 = (const __bf16*)(nslice > 2 ? xsl[2] : xsl[0]);
