@@ -12,6 +12,9 @@
 
 namespace raft_amd {
 
+// from solver_kernels.hip
+void launch_cholesky_r1_update_f32(float*, float*, int, long long, hipStream_t);
+void launch_cholesky_r1_update_f64(double*, double*, int, long long, hipStream_t);
 // from reductions.hip
 template <int OP, typename T>
 void launch_reduce_rows(const T*, T*, long long, long long, hipStream_t);
@@ -610,6 +613,23 @@ torch::Tensor gemm_f32(torch::Tensor a, torch::Tensor b) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reduce_rows", &reduce_rows, "rowwise reduction (op code)");
+  m.def("cholesky_r1_update_", [](torch::Tensor l, torch::Tensor x) {
+    TORCH_CHECK(l.is_cuda() && l.dim() == 2 && l.size(0) == l.size(1) &&
+                l.is_contiguous() && x.is_contiguous() &&
+                x.numel() == l.size(0));
+    const int n = (int)l.size(0);
+    if (l.scalar_type() == torch::kFloat32) {
+      raft_amd::launch_cholesky_r1_update_f32(l.data_ptr<float>(),
+                                              x.data_ptr<float>(), n, n,
+                                              cur_stream());
+    } else if (l.scalar_type() == torch::kFloat64) {
+      raft_amd::launch_cholesky_r1_update_f64(l.data_ptr<double>(),
+                                              x.data_ptr<double>(), n, n,
+                                              cur_stream());
+    } else {
+      TORCH_CHECK(false, "cholesky_r1_update_: fp32/fp64 only");
+    }
+  }, "in-place rank-1 Cholesky update (single-kernel hyperbolic rotations)");
   m.def("reduce_cols", &reduce_cols, "columnwise reduction (op code)");
   m.def("row_argmin", &row_argmin, "rowwise argmin");
   m.def("rows_sqnorm_bf16", &rows_sqnorm_bf16, "bf16 row squared norms -> f32");

@@ -69,11 +69,18 @@ def cholesky_r1_update(l: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
     """Rank-1 update of a Cholesky factor: chol(A + x x^T) from L=chol(A).
 
     Reference parity: raft/linalg/cholesky_r1_update.cuh. Classic hyperbolic
-    rotation scheme, O(n^2).
+    rotation scheme, O(n^2). On GPU this is ONE kernel launch
+    (csrc/solver_kernels.hip) — the per-k Python loop cost n x ~6 launch
+    round-trips.
     """
     l = l.clone()
     x = x.clone().to(l.dtype)
     n = l.shape[0]
+    if l.is_cuda and l.dtype in (torch.float32, torch.float64) and n > 1:
+        from raft_amd._ext import require_ext
+        lc = l.contiguous()
+        require_ext().cholesky_r1_update_(lc, x.contiguous())
+        return lc
     for k in range(n):
         lkk = l[k, k]
         r = torch.sqrt(lkk * lkk + x[k] * x[k])

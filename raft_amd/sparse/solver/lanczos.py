@@ -114,9 +114,48 @@ def lanczos_min_eigenpairs(a, k: int | None = None, config: LanczosConfig | None
                 beta_last = b
                 v_next = u / b.clamp_min(1e-300)
 
+    # hipGraph capture of the restart cycle (the per-step chain
+    # of ~10 small launches made the solver host-dispatch bound — 58-161
+    # steps/s run-to-run). The _extend(k) cycle is shape-static and
+    # sync-free, so one graph replays the whole ncv-k step chain as a
+    # single launch. The rare degeneracy redo and the restart math stay
+    # eager. Opt out with RAFT_AMD_LANCZOS_GRAPH=0.
+    import os
+    use_graph = (device.type == "cuda"
+                 and os.environ.get("RAFT_AMD_LANCZOS_GRAPH", "1") != "0")
+    graph_state = {}
+
+    def _extend_graphed(start: int) -> bool:
+        nonlocal v_next, beta_last
+        if not use_graph:
+            return False
+        try:
+            if "graph" not in graph_state:
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    _extend(start, careful=False)  # allocator warmup
+                torch.cuda.current_stream().wait_stream(side)
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    _extend(start, careful=False)
+                # capture does not execute: replay below produces the values.
+                # v_next/beta_last now reference capture-owned tensors that
+                # every replay refreshes in place.
+                graph_state["graph"] = g
+                graph_state["outs"] = (v_next, beta_last)
+            graph_state["graph"].replay()
+            v_next, beta_last = graph_state["outs"]
+            return True
+        except Exception:
+            graph_state["graph"] = None
+            return False
+
     def _extend_checked(start: int):
         """Sync-free extend + one degeneracy check per cycle (rare redo)."""
-        _extend(start, careful=False)
+        if not (start == k and graph_state.get("graph", True) is not None
+                and _extend_graphed(start)):
+            _extend(start, careful=False)
         betas = torch.diagonal(t_mat, 1)[max(start - 1, 0):]
         if bool((betas.abs() < 1e-30).any()) or bool(beta_last.abs() < 1e-30):
             _extend(start, careful=True)

@@ -92,3 +92,35 @@ class TestSpectral:
         good = torch.tensor([0] * 5 + [1] * 5)
         bad = torch.tensor([0, 1] * 5)
         assert analyze_modularity(g, good) > analyze_modularity(g, bad)
+
+
+class TestLAPBatched:
+    def test_batched_matches_scipy(self):
+        from scipy.optimize import linear_sum_assignment
+        from raft_amd.solver.lap import linear_assignment_batched
+        torch.manual_seed(0)
+        costs = torch.rand(6, 40, 40) * 10
+        assign, totals = linear_assignment_batched(costs)
+        for b in range(6):
+            r, c = linear_sum_assignment(costs[b].numpy())
+            ref = float(costs[b].numpy()[r, c].sum())
+            assert abs(float(totals[b]) - ref) < 1e-4 * max(1.0, abs(ref))
+            assert sorted(assign[b].tolist()) == list(range(40))
+
+    def test_batched_class_wrapper(self):
+        from raft_amd.solver.lap import LinearAssignmentProblem
+        torch.manual_seed(1)
+        costs = torch.rand(3, 16, 16)
+        lap = LinearAssignmentProblem(16)
+        a = lap.solve(costs)
+        assert a.shape == (3, 16)
+        assert lap.get_primal_objective().shape == (3,)
+
+    def test_integer_costs_exact(self):
+        from scipy.optimize import linear_sum_assignment
+        from raft_amd.solver.lap import linear_assignment
+        torch.manual_seed(2)
+        ci = torch.randint(0, 5, (30, 30)).double()
+        a, t = linear_assignment(ci)
+        r, c = linear_sum_assignment(ci.numpy())
+        assert abs(t - ci.numpy()[r, c].sum()) < 1e-9
