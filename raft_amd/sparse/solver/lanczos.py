@@ -114,15 +114,15 @@ def lanczos_min_eigenpairs(a, k: int | None = None, config: LanczosConfig | None
                 beta_last = b
                 v_next = u / b.clamp_min(1e-300)
 
-    # hipGraph capture of the restart cycle (the per-step chain
-    # of ~10 small launches made the solver host-dispatch bound — 58-161
-    # steps/s run-to-run). The _extend(k) cycle is shape-static and
-    # sync-free, so one graph replays the whole ncv-k step chain as a
-    # single launch. The rare degeneracy redo and the restart math stay
-    # eager. Opt out with RAFT_AMD_LANCZOS_GRAPH=0.
+    # EXPERIMENTAL hipGraph capture of the restart cycle (opt-in:
+    # RAFT_AMD_LANCZOS_GRAPH=1). Measured on MI355X (round 2): capture of
+    # the torch-op chain through hipGraph REGRESSED the 10M-row bench
+    # (105 vs 153 steps/s — replays tripping the degeneracy redo) and hung
+    # outright on a 100k-row case, so the default stays the eager sync-free
+    # loop. Kept for future ROCm versions.
     import os
     use_graph = (device.type == "cuda"
-                 and os.environ.get("RAFT_AMD_LANCZOS_GRAPH", "1") != "0")
+                 and os.environ.get("RAFT_AMD_LANCZOS_GRAPH", "0") == "1")
     graph_state = {}
 
     def _extend_graphed(start: int) -> bool:

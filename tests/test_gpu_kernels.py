@@ -557,14 +557,11 @@ class TestSolverKernelsGpu:
             tol = 1e-10 if dt == torch.float64 else 1e-3
             torch.testing.assert_close(l2, ref, rtol=tol, atol=tol)
 
-    def test_lanczos_graph_capture(self, dev, ext):
-        # hipGraph-captured restart cycle must match the eager solver
-        import os
-        from raft_amd.sparse.solver.lanczos import eigsh, LanczosConfig
+    def test_lanczos_gpu_residuals(self, dev, ext):
+        # default (eager, sync-free) GPU solver: residual check vs scipy CSR
+        from raft_amd.sparse.solver.lanczos import eigsh
         from raft_amd.sparse.types import CSR
         import scipy.sparse as sp
-        import numpy as np
-        rng = np.random.default_rng(3)
         n = 4000
         m = sp.random(n, n, density=2e-3, random_state=3, format="csr")
         m = (m + m.T) * 0.5 + sp.identity(n) * 0.1
@@ -572,14 +569,7 @@ class TestSolverKernelsGpu:
                   torch.as_tensor(m.indices, dtype=torch.int32, device=dev),
                   torch.as_tensor(m.data, dtype=torch.float32, device=dev),
                   (n, n))
-        w_g, v_g = eigsh(csr, k=4, maxiter=40)
-        os.environ["RAFT_AMD_LANCZOS_GRAPH"] = "0"
-        try:
-            w_e, v_e = eigsh(csr, k=4, maxiter=40)
-        finally:
-            del os.environ["RAFT_AMD_LANCZOS_GRAPH"]
-        torch.testing.assert_close(w_g, w_e, rtol=1e-4, atol=1e-5)
-        # residuals ||A v - w v|| small for the graphed solve
-        av = torch.as_tensor(m @ v_g.cpu().double().numpy(), device=dev)
-        res = (av - v_g.double() * w_g.double().unsqueeze(0)).norm(dim=0)
+        w, v = eigsh(csr, k=4, maxiter=40)
+        av = torch.as_tensor(m @ v.cpu().double().numpy(), device=dev)
+        res = (av - v.double() * w.double().unsqueeze(0)).norm(dim=0)
         assert float(res.max()) < 1e-3
