@@ -60,6 +60,10 @@ void launch_select_k(const float*, float*, int*, void*, long long, long long, in
                      bool, bool, hipStream_t);
 void launch_select_k_warpsort(const float*, float*, int*, long long, long long, int,
                               bool, hipStream_t);
+template <typename T>
+void launch_select_k_generic_t(const T*, const long long*, long long, long long,
+                               T*, long long*, long long, long long, bool,
+                               hipStream_t);
 // from fused_l2nn.hip
 void launch_fused_l2nn_split(const void**, const void**, const float*, const float*,
                              float*, int*, float*, long long, int, int, int,
@@ -336,6 +340,59 @@ torch::Tensor reduce_rows_by_key(torch::Tensor x, torch::Tensor keys, int64_t n_
                                       x.size(0), x.size(1), n_keys, replicas,
                                       cur_stream());
   return sums;
+}
+
+std::tuple<torch::Tensor, torch::Tensor> select_k_generic(
+    torch::Tensor x, c10::optional<torch::Tensor> row_off, int64_t k,
+    bool select_min) {
+  // x: [batch, len] dense, or flat [nnz] values with row_off [batch+1]
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  long long batch, stride, len_fixed;
+  const long long* ro = nullptr;
+  torch::Tensor ro_t;
+  if (row_off.has_value()) {
+    ro_t = row_off.value();
+    TORCH_CHECK(ro_t.is_cuda() && ro_t.scalar_type() == torch::kInt64 &&
+                ro_t.is_contiguous() && x.dim() == 1);
+    batch = ro_t.numel() - 1;
+    stride = 0;
+    len_fixed = 0;
+    ro = ro_t.data_ptr<long long>();
+  } else {
+    TORCH_CHECK(x.dim() == 2);
+    batch = x.size(0);
+    stride = len_fixed = x.size(1);
+  }
+  auto out_v = torch::empty({batch, k}, x.options());
+  auto out_i = torch::empty({batch, k}, x.options().dtype(torch::kInt64));
+  auto* oi = out_i.data_ptr<long long>();
+  switch (x.scalar_type()) {
+    case torch::kFloat32:
+      raft_amd::launch_select_k_generic_t<float>(
+          x.data_ptr<float>(), ro, stride, len_fixed, out_v.data_ptr<float>(),
+          oi, batch, k, select_min, cur_stream());
+      break;
+    case torch::kFloat64:
+      raft_amd::launch_select_k_generic_t<double>(
+          x.data_ptr<double>(), ro, stride, len_fixed,
+          out_v.data_ptr<double>(), oi, batch, k, select_min, cur_stream());
+      break;
+    case torch::kBFloat16:
+      raft_amd::launch_select_k_generic_t<__bf16>(
+          reinterpret_cast<const __bf16*>(x.data_ptr<at::BFloat16>()), ro,
+          stride, len_fixed, reinterpret_cast<__bf16*>(out_v.data_ptr<at::BFloat16>()),
+          oi, batch, k, select_min, cur_stream());
+      break;
+    case torch::kHalf:
+      raft_amd::launch_select_k_generic_t<_Float16>(
+          reinterpret_cast<const _Float16*>(x.data_ptr<at::Half>()), ro,
+          stride, len_fixed, reinterpret_cast<_Float16*>(out_v.data_ptr<at::Half>()),
+          oi, batch, k, select_min, cur_stream());
+      break;
+    default:
+      TORCH_CHECK(false, "select_k_generic: fp32/fp64/bf16/fp16 only");
+  }
+  return {out_v, out_i};
 }
 
 std::tuple<torch::Tensor, torch::Tensor> select_k(torch::Tensor x, int64_t k,
@@ -653,6 +710,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kmeans_update_centroids", &kmeans_update_centroids,
         "centroids = counts>0 ? sums/counts : centroids");
   m.def("select_k", &select_k, "batched top-k (radix)");
+  m.def("select_k_generic", &select_k_generic,
+        "generic top-k: any dtype, unbounded k, int64 idx, CSR row offsets",
+        pybind11::arg("x"), pybind11::arg("row_off") = pybind11::none(),
+        pybind11::arg("k"), pybind11::arg("select_min") = true);
   m.def("pairwise_l2_filter", &pairwise_l2_filter,
         "threshold-filtered pairwise L2 candidate emission (fused kNN)");
   m.def("pairwise_l2_mfma", &pairwise_l2_mfma,

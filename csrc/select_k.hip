@@ -380,6 +380,187 @@ void launch_select_k_warpsort(const float* x, float* out_v, int* out_i,
                      stream, x, out_v, out_i, batch, len, k, select_min);
 }
 
+// ---------------------------------------------------------------------------
+// Generic select-k: any value dtype (fp32/bf16/fp16 via 32-bit ordinals,
+// fp64 via 64-bit ordinals), UNBOUNDED k, int64 indices (rows may exceed
+// 2^31 elements) and per-row VARIABLE lengths — the CSR adapter that does
+// NOT densify. Block-per-row threshold+filter:
+//   1. byte-wise ordinal histogram passes narrow to the k-th ordinal
+//      (4 passes for 32-bit ordinals, 8 for fp64) — no candidate storage,
+//      so k is unbounded;
+//   2. two filter passes write ordinals < kth, then the equality backfill.
+// Rows shorter than k pad value with +inf (select_min; -inf otherwise) and
+// index -1 (DOCUMENTED sentinel of the variable-length path). Output is
+// UNSORTED; callers sort the [batch, k] slab when requested (k can exceed
+// any LDS sort capacity).
+// Reference parity: matrix/detail/select_radix.cuh per-row len_i
+// (:722-725) + sparse/matrix/detail/select_k-inl.cuh CSR layout adapter
+// (:63-96); covers the MATRIX_SELECT_LARGE_TEST len>2^31 case via int64
+// indices and long long scans.
+// ---------------------------------------------------------------------------
+
+template <typename T>
+struct SelOrd;
+template <>
+struct SelOrd<float> {
+  using ord = uint32_t;
+  static __device__ __forceinline__ uint32_t to_ord(float f, bool mn) {
+    return f32_to_ord(f, mn);
+  }
+};
+template <>
+struct SelOrd<__bf16> {
+  using ord = uint32_t;
+  static __device__ __forceinline__ uint32_t to_ord(__bf16 h, bool mn) {
+    return f32_to_ord((float)h, mn);
+  }
+};
+template <>
+struct SelOrd<_Float16> {
+  using ord = uint32_t;
+  static __device__ __forceinline__ uint32_t to_ord(_Float16 h, bool mn) {
+    return f32_to_ord((float)h, mn);
+  }
+};
+template <>
+struct SelOrd<double> {
+  using ord = uint64_t;
+  static __device__ __forceinline__ uint64_t to_ord(double f, bool mn) {
+    if (f != f) return ~0ull;
+    uint64_t u = __double_as_longlong(f);
+    u = (u & 0x8000000000000000ull) ? ~u : (u | 0x8000000000000000ull);
+    return mn ? u : ~u;
+  }
+};
+
+template <typename T, int BLOCK = 256>
+__global__ void select_k_generic_kernel(const T* __restrict__ vals,
+                                        const long long* __restrict__ row_off,
+                                        long long stride, long long len_fixed,
+                                        T* __restrict__ out_v,
+                                        long long* __restrict__ out_i,
+                                        long long batch, long long k,
+                                        bool select_min) {
+  using OrdT = typename SelOrd<T>::ord;
+  constexpr int PASSES = (int)sizeof(OrdT);
+  constexpr int NW = BLOCK / RAFT_AMD_WAVE;
+  __shared__ unsigned long long hist[NW][256];
+  __shared__ unsigned long long sh_prefix, sh_below, sh_lt, sh_eq;
+  const int wid = threadIdx.x / RAFT_AMD_WAVE;
+
+  for (long long row = blockIdx.x; row < batch; row += gridDim.x) {
+    const long long base = row_off ? row_off[row] : row * stride;
+    const long long len = row_off ? row_off[row + 1] - row_off[row] : len_fixed;
+    const T* rp = vals + base;
+    const long long kk = k < len ? k : len;
+    if (kk <= 0) {
+      for (long long j = threadIdx.x; j < k; j += BLOCK) {
+        out_v[row * k + j] = (T)(select_min ? INFINITY : -INFINITY);
+        out_i[row * k + j] = -1;
+      }
+      continue;
+    }
+    OrdT prefix = 0, prefix_mask = 0;
+    long long remaining = kk;
+#pragma unroll
+    for (int pass = 0; pass < PASSES; pass++) {
+      const int shift = 8 * (PASSES - 1 - pass);
+      for (int b = threadIdx.x; b < NW * 256; b += BLOCK)
+        reinterpret_cast<unsigned long long*>(hist)[b] = 0;
+      __syncthreads();
+      for (long long j = threadIdx.x; j < len; j += BLOCK) {
+        const OrdT u = SelOrd<T>::to_ord(rp[j], select_min);
+        if ((u & prefix_mask) == prefix)
+          atomicAdd(&hist[wid][(u >> shift) & 0xFF], 1ull);
+      }
+      __syncthreads();
+      for (int b = threadIdx.x; b < 256; b += BLOCK) {
+        unsigned long long s = 0;
+#pragma unroll
+        for (int w = 0; w < NW; w++) s += hist[w][b];
+        hist[0][b] = s;
+      }
+      __syncthreads();
+      if (threadIdx.x == 0) {
+        unsigned long long cum = 0;
+        int bucket = 255;
+        for (int b = 0; b < 256; b++) {
+          const unsigned long long c = hist[0][b];
+          if (cum + c >= (unsigned long long)remaining) { bucket = b; break; }
+          cum += c;
+        }
+        sh_prefix = (unsigned long long)bucket;
+        sh_below = cum;
+      }
+      __syncthreads();
+      prefix |= ((OrdT)sh_prefix) << shift;
+      prefix_mask |= ((OrdT)0xFF) << shift;
+      remaining -= (long long)sh_below;
+      __syncthreads();
+    }
+    const OrdT kth = prefix;
+    if (threadIdx.x == 0) { sh_lt = 0; sh_eq = 0; }
+    __syncthreads();
+    // filter pass 1: strict winners
+    for (long long j = threadIdx.x; j < len; j += BLOCK) {
+      const OrdT u = SelOrd<T>::to_ord(rp[j], select_min);
+      if (u < kth) {
+        const unsigned long long pos = atomicAdd(&sh_lt, 1ull);
+        out_v[row * k + (long long)pos] = rp[j];
+        out_i[row * k + (long long)pos] = j;
+      }
+    }
+    __syncthreads();
+    const long long n_lt = (long long)sh_lt;
+    // filter pass 2: equality backfill (exactly kk - n_lt slots)
+    for (long long j = threadIdx.x; j < len; j += BLOCK) {
+      const OrdT u = SelOrd<T>::to_ord(rp[j], select_min);
+      if (u == kth) {
+        const unsigned long long e = atomicAdd(&sh_eq, 1ull);
+        if ((long long)e < kk - n_lt) {
+          out_v[row * k + n_lt + (long long)e] = rp[j];
+          out_i[row * k + n_lt + (long long)e] = j;
+        }
+      }
+    }
+    // pad short rows
+    for (long long j = kk + threadIdx.x; j < k; j += BLOCK) {
+      out_v[row * k + j] = (T)(select_min ? INFINITY : -INFINITY);
+      out_i[row * k + j] = -1;
+    }
+    __syncthreads();
+  }
+}
+
+template <typename T>
+void launch_select_k_generic_t(const T* vals, const long long* row_off,
+                               long long stride, long long len_fixed, T* out_v,
+                               long long* out_i, long long batch, long long k,
+                               bool select_min, hipStream_t stream) {
+  const int grid = (int)(batch < 4096 ? batch : 4096);
+  hipLaunchKernelGGL((select_k_generic_kernel<T, 256>), dim3(grid), dim3(256),
+                     0, stream, vals, row_off, stride, len_fixed, out_v, out_i,
+                     batch, k, select_min);
+}
+
+template void launch_select_k_generic_t<float>(const float*, const long long*,
+                                               long long, long long, float*,
+                                               long long*, long long, long long,
+                                               bool, hipStream_t);
+template void launch_select_k_generic_t<double>(const double*, const long long*,
+                                                long long, long long, double*,
+                                                long long*, long long,
+                                                long long, bool, hipStream_t);
+template void launch_select_k_generic_t<__bf16>(const __bf16*, const long long*,
+                                                long long, long long, __bf16*,
+                                                long long*, long long,
+                                                long long, bool, hipStream_t);
+template void launch_select_k_generic_t<_Float16>(const _Float16*,
+                                                  const long long*, long long,
+                                                  long long, _Float16*,
+                                                  long long*, long long,
+                                                  long long, bool, hipStream_t);
+
 int select_k_grid(long long batch) { return (int)(batch < 4096 ? batch : 4096); }
 
 long long select_k_workspace_bytes(long long batch) {

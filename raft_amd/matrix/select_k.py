@@ -46,6 +46,21 @@ def select_k(x: torch.Tensor, k: int, select_min: bool = True,
     k = int(k)
     assert 0 < k <= n, f"k={k} out of range for row length {n}"
 
+    if (on_gpu(x) and algo != SelectAlgo.TORCH
+            and (x.dtype in (torch.float64, torch.bfloat16, torch.float16)
+                 or (x.dtype == torch.float32 and k > 2048))):
+        # generic native engine: any dtype (64-bit ordinals for fp64),
+        # unbounded k, int64 indices (rows may exceed 2^31 elements).
+        # Output is unsorted; sort the [batch, k] slab here when asked.
+        ext = require_ext()
+        vals, idx = ext.select_k_generic(x.contiguous(), None, k=k,
+                                         select_min=bool(select_min))
+        if sorted:
+            order = torch.argsort(vals, dim=1, descending=not select_min)
+            vals = torch.gather(vals, 1, order)
+            idx = torch.gather(idx, 1, order)
+        return vals, idx
+
     if (on_gpu(x) and x.dtype == torch.float32 and algo != SelectAlgo.TORCH
             and k <= 64 and n <= 4096
             and algo in (SelectAlgo.AUTO, SelectAlgo.WARPSORT)):

@@ -573,3 +573,56 @@ class TestSolverKernelsGpu:
         av = torch.as_tensor(m @ v.cpu().double().numpy(), device=dev)
         res = (av - v.double() * w.double().unsqueeze(0)).norm(dim=0)
         assert float(res.max()) < 1e-3
+
+
+class TestSelectKGeneric:
+    """Generic select engine: dtypes, unbounded k, int64 idx, CSR rows
+    (reference select test matrix: select_radix len_i + CSR adapter)."""
+
+    @pytest.mark.parametrize("dt", [torch.float64, torch.bfloat16, torch.float16])
+    @pytest.mark.parametrize("select_min", [True, False])
+    def test_dtypes_vs_topk(self, dev, ext, dt, select_min):
+        from raft_amd.matrix import select_k
+        torch.manual_seed(0)
+        x = (torch.randn(64, 3000, device=dev) * 10).to(dt)
+        vals, idx = select_k(x, 16, select_min=select_min)
+        ref_v, _ = torch.topk(x.float(), 16, dim=1, largest=not select_min)
+        torch.testing.assert_close(vals.float(), ref_v, rtol=0, atol=0)
+        g = torch.gather(x, 1, idx)
+        torch.testing.assert_close(g.float(), vals.float(), rtol=0, atol=0)
+
+    def test_large_k_fp32(self, dev, ext):
+        from raft_amd.matrix import select_k
+        torch.manual_seed(1)
+        x = torch.randn(8, 100000, device=dev)
+        vals, idx = select_k(x, 5000, select_min=True)   # k > 2048
+        ref_v, _ = torch.topk(x, 5000, dim=1, largest=False)
+        torch.testing.assert_close(vals, ref_v, rtol=0, atol=0)
+        assert idx.dtype == torch.int64
+        torch.testing.assert_close(torch.gather(x, 1, idx), vals)
+
+    def test_csr_no_densify_power_law(self, dev, ext):
+        # ragged rows incl. empty and shorter-than-k; compare vs the CPU
+        # densify oracle
+        from raft_amd.sparse.types import CSR
+        from raft_amd.sparse.select_k import csr_select_k
+        torch.manual_seed(2)
+        lens = torch.tensor([0, 3, 5000, 17, 1, 900, 64, 2], dtype=torch.int64)
+        indptr = torch.cat([torch.zeros(1, dtype=torch.int64),
+                            lens.cumsum(0)])
+        nnz = int(indptr[-1])
+        vals = torch.randn(nnz)
+        cols = torch.randint(0, 100000, (nnz,))
+        a_gpu = CSR(indptr.to(dev).to(torch.int32), cols.to(dev).to(torch.int32),
+                    vals.to(dev), (8, 100000))
+        a_cpu = CSR(indptr.to(torch.int32), cols.to(torch.int32), vals,
+                    (8, 100000))
+        k = 8
+        gv, gi = csr_select_k(a_gpu, k)
+        cv, ci = csr_select_k(a_cpu, k)
+        torch.testing.assert_close(gv.cpu(), cv, rtol=0, atol=0)
+        # where padded (inf), index must be -1; real slots: same VALUES
+        # (column ids may differ on duplicate values)
+        pad = torch.isinf(cv)
+        assert (gi.cpu()[pad] == -1).all()
+        assert (gi.cpu()[~pad] >= 0).all()
