@@ -357,7 +357,7 @@ std::tuple<torch::Tensor, torch::Tensor> select_k_generic(
     batch = ro_t.numel() - 1;
     stride = 0;
     len_fixed = 0;
-    ro = ro_t.data_ptr<long long>();
+    ro = reinterpret_cast<const long long*>(ro_t.data_ptr<int64_t>());
   } else {
     TORCH_CHECK(x.dim() == 2);
     batch = x.size(0);
@@ -365,7 +365,7 @@ std::tuple<torch::Tensor, torch::Tensor> select_k_generic(
   }
   auto out_v = torch::empty({batch, k}, x.options());
   auto out_i = torch::empty({batch, k}, x.options().dtype(torch::kInt64));
-  auto* oi = out_i.data_ptr<long long>();
+  auto* oi = reinterpret_cast<long long*>(out_i.data_ptr<int64_t>());
   switch (x.scalar_type()) {
     case torch::kFloat32:
       raft_amd::launch_select_k_generic_t<float>(
