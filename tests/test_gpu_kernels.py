@@ -568,7 +568,7 @@ class TestSolverKernelsGpu:
         csr = CSR(torch.as_tensor(m.indptr, dtype=torch.int32, device=dev),
                   torch.as_tensor(m.indices, dtype=torch.int32, device=dev),
                   torch.as_tensor(m.data, dtype=torch.float32, device=dev),
-                  (n, n))
+                  n, n)
         w, v = eigsh(csr, k=4, maxiter=40)
         av = torch.as_tensor(m @ v.cpu().double().numpy(), device=dev)
         res = (av - v.double() * w.double().unsqueeze(0)).norm(dim=0)
@@ -614,9 +614,9 @@ class TestSelectKGeneric:
         vals = torch.randn(nnz)
         cols = torch.randint(0, 100000, (nnz,))
         a_gpu = CSR(indptr.to(dev).to(torch.int32), cols.to(dev).to(torch.int32),
-                    vals.to(dev), (8, 100000))
+                    vals.to(dev), 8, 100000)
         a_cpu = CSR(indptr.to(torch.int32), cols.to(torch.int32), vals,
-                    (8, 100000))
+                    8, 100000)
         k = 8
         gv, gi = csr_select_k(a_gpu, k)
         cv, ci = csr_select_k(a_cpu, k)
