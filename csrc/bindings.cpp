@@ -12,6 +12,17 @@
 
 namespace raft_amd {
 
+// from linewise.hip
+void launch_linewise(const float*, float*, const float*, const float*,
+                     long long, long long, bool, int, int, hipStream_t);
+// from histogram.hip
+void launch_histogram(const float*, long long, long long, int, float, float,
+                      unsigned long long*, hipStream_t);
+void launch_bitset_set(unsigned int*, const long long*, long long, int, hipStream_t);
+void launch_bitset_test(const unsigned int*, const long long*, bool*, long long,
+                        hipStream_t);
+void launch_bitset_count(const unsigned int*, long long, unsigned long long*,
+                         hipStream_t);
 // from solver_kernels.hip
 void launch_cholesky_r1_update_f32(float*, float*, int, long long, hipStream_t);
 void launch_cholesky_r1_update_f64(double*, double*, int, long long, hipStream_t);
@@ -710,6 +721,69 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kmeans_update_centroids", &kmeans_update_centroids,
         "centroids = counts>0 ? sums/counts : centroids");
   m.def("select_k", &select_k, "batched top-k (radix)");
+  m.def("linewise", [](torch::Tensor x, torch::Tensor v1,
+                       c10::optional<torch::Tensor> v2, bool along_rows,
+                       int64_t op1, int64_t op2) {
+    TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous() &&
+                x.scalar_type() == torch::kFloat32);
+    const long long n = x.size(0), d = x.size(1);
+    TORCH_CHECK(v1.numel() == (along_rows ? d : n) && v1.is_contiguous());
+    const float* v2p = nullptr;
+    torch::Tensor v2t;
+    if (v2.has_value()) {
+      v2t = v2.value().contiguous();
+      TORCH_CHECK(v2t.numel() == v1.numel());
+      v2p = v2t.data_ptr<float>();
+    }
+    auto out = torch::empty_like(x);
+    raft_amd::launch_linewise(x.data_ptr<float>(), out.data_ptr<float>(),
+                              v1.data_ptr<float>(), v2p, n, d, along_rows,
+                              (int)op1, (int)op2, cur_stream());
+    return out;
+  }, "fused linewise broadcast: out = (x op1 v1) [op2 v2]",
+        pybind11::arg("x"), pybind11::arg("v1"),
+        pybind11::arg("v2") = pybind11::none(),
+        pybind11::arg("along_rows") = true, pybind11::arg("op1") = 0,
+        pybind11::arg("op2") = 0);
+  m.def("histogram_f32", [](torch::Tensor x, int64_t n_bins, double lo, double hi) {
+    TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous() &&
+                x.scalar_type() == torch::kFloat32);
+    auto out = torch::zeros({n_bins, x.size(1)},
+                            x.options().dtype(torch::kInt64));
+    raft_amd::launch_histogram(
+        x.data_ptr<float>(), x.size(0), x.size(1), (int)n_bins, (float)lo,
+        (float)hi,
+        reinterpret_cast<unsigned long long*>(out.data_ptr<int64_t>()),
+        cur_stream());
+    return out;
+  }, "per-column histograms (LDS-multi/LDS/gmem strategies by n_bins)");
+  m.def("bitset_set_", [](torch::Tensor words, torch::Tensor idx, bool value) {
+    TORCH_CHECK(words.is_cuda() && words.scalar_type() == torch::kInt32 &&
+                words.is_contiguous() && idx.scalar_type() == torch::kInt64);
+    raft_amd::launch_bitset_set(
+        reinterpret_cast<unsigned int*>(words.data_ptr<int>()),
+        reinterpret_cast<const long long*>(idx.contiguous().data_ptr<int64_t>()),
+        idx.numel(), value ? 1 : 0, cur_stream());
+  }, "O(k) bitset scatter set/clear");
+  m.def("bitset_test", [](torch::Tensor words, torch::Tensor idx) {
+    TORCH_CHECK(words.is_cuda() && words.scalar_type() == torch::kInt32);
+    auto out = torch::empty({idx.numel()}, words.options().dtype(torch::kBool));
+    raft_amd::launch_bitset_test(
+        reinterpret_cast<const unsigned int*>(words.data_ptr<int>()),
+        reinterpret_cast<const long long*>(idx.contiguous().data_ptr<int64_t>()),
+        out.data_ptr<bool>(), idx.numel(), cur_stream());
+    return out;
+  }, "bitset membership test");
+  m.def("bitset_count", [](torch::Tensor words) {
+    TORCH_CHECK(words.is_cuda() && words.scalar_type() == torch::kInt32);
+    auto out = torch::zeros({1}, words.options().dtype(torch::kInt64));
+    raft_amd::launch_bitset_count(
+        reinterpret_cast<const unsigned int*>(words.data_ptr<int>()),
+        words.numel(),
+        reinterpret_cast<unsigned long long*>(out.data_ptr<int64_t>()),
+        cur_stream());
+    return out;
+  }, "bitset popcount");
   m.def("select_k_generic", &select_k_generic,
         "generic top-k: any dtype, unbounded k, int64 idx, CSR row offsets",
         pybind11::arg("x"), pybind11::arg("row_off") = pybind11::none(),

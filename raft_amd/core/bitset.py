@@ -43,11 +43,23 @@ class Bitset:
 
     # -- mutation ------------------------------------------------------------
     def set(self, idx: torch.Tensor, value: bool = True) -> None:
-        # scatter through a dense bool view: index_put with duplicate words
-        # would drop updates (no bitwise-or scatter primitive)
-        dense = self.to_dense()
-        dense[idx.to(self.device, torch.int64)] = bool(value)
-        self.words = Bitset.from_dense(dense).words
+        """O(k) scatter set/clear. GPU: atomicOr/And kernel
+        (csrc/histogram.hip); CPU: numpy bitwise-or.at — the round-1 version
+        materialized the whole dense mask per call."""
+        idx = idx.to(self.device, torch.int64)
+        if self.words.is_cuda:
+            from raft_amd._ext import require_ext
+            self.words = self.words.contiguous()
+            require_ext().bitset_set_(self.words, idx, bool(value))
+            return
+        import numpy as np
+        w = self.words.numpy().view(np.uint32)
+        word_idx = (idx // self.WORD_BITS).numpy()
+        masks = (np.uint32(1) << (idx % self.WORD_BITS).numpy().astype(np.uint32))
+        if value:
+            np.bitwise_or.at(w, word_idx, masks)
+        else:
+            np.bitwise_and.at(w, word_idx, ~masks)
 
     def flip(self) -> None:
         self.words = ~self.words
@@ -61,6 +73,9 @@ class Bitset:
     # -- queries -------------------------------------------------------------
     def test(self, idx: torch.Tensor) -> torch.Tensor:
         idx = idx.to(self.device, torch.int64)
+        if self.words.is_cuda:
+            from raft_amd._ext import require_ext
+            return require_ext().bitset_test(self.words.contiguous(), idx)
         word = idx // self.WORD_BITS
         bit = (idx % self.WORD_BITS).to(torch.int32)
         return ((self.words[word] >> bit) & 1).to(torch.bool)
@@ -89,8 +104,14 @@ class Bitset:
         return out
 
     def count(self) -> int:
-        """Population count (reference: bitset count via detail::popc)."""
-        return int(self.to_dense().sum().item())
+        """Population count (reference: bitset count via detail::popc).
+        GPU: wave-reduced __popc kernel; CPU: byte popcount over the packed
+        words (no dense expansion)."""
+        if self.words.is_cuda:
+            from raft_amd._ext import require_ext
+            return int(require_ext().bitset_count(self.words.contiguous()).item())
+        import numpy as np
+        return int(np.unpackbits(self.words.numpy().view(np.uint8)).sum())
 
     def sparsity(self) -> float:
         return 1.0 - self.count() / max(self.n_bits, 1)

@@ -12,7 +12,7 @@ from .map import (
     map_then_reduce, map_reduce,
 )
 from .norm import norm, normalize, row_norm, col_norm, NormType
-from .matrix_vector import matrix_vector_op, linewise_op
+from .matrix_vector import matrix_vector_op, linewise_fused, linewise_op
 from .gemm import gemm, gemv, dot, axpy, gemm_bf16_f32, gemm_fp32_emulated
 from .reduce_by_key import reduce_rows_by_key, reduce_cols_by_key
 from .decomp import eig, eigh, eig_jacobi, svd, qr, cholesky, cholesky_r1_update, lstsq
@@ -26,7 +26,7 @@ __all__ = [
     "add", "subtract", "multiply", "divide", "power", "sqrt", "eltwise",
     "map_then_reduce", "map_reduce",
     "norm", "normalize", "row_norm", "col_norm", "NormType",
-    "matrix_vector_op", "linewise_op",
+    "matrix_vector_op", "linewise_op", "linewise_fused",
     "gemm", "gemv", "dot", "axpy", "gemm_bf16_f32", "gemm_fp32_emulated",
     "reduce_rows_by_key", "reduce_cols_by_key",
     "eig", "eigh", "eig_jacobi", "svd", "qr", "cholesky", "cholesky_r1_update", "lstsq",

@@ -626,3 +626,75 @@ class TestSelectKGeneric:
         pad = torch.isinf(cv)
         assert (gi.cpu()[pad] == -1).all()
         assert (gi.cpu()[~pad] >= 0).all()
+
+
+class TestHistogramBitsetGpu:
+    @pytest.mark.parametrize("n,d,bins", [(100000, 8, 256), (50000, 3, 4096),
+                                          (20000, 2, 20000)])
+    def test_histogram_strategies(self, dev, ext, n, d, bins):
+        # each shape hits a different strategy (LDS-multi / LDS / gmem)
+        torch.manual_seed(0)
+        x = torch.rand(n, d, device=dev) * 10 - 5
+        out = ext.histogram_f32(x.contiguous(), bins, -5.0, 5.0)
+        # reference with the SAME fp32 binning formula
+        scale = bins / 10.0
+        b = ((x - (-5.0)) * scale).floor().clamp_(0, bins - 1).to(torch.int64)
+        ref = torch.zeros(bins, d, dtype=torch.int64, device=dev)
+        for j in range(d):
+            ref[:, j] = torch.bincount(b[:, j], minlength=bins)
+        assert torch.equal(out, ref)
+
+    def test_histogram_python_api(self, dev):
+        from raft_amd.stats import histogram
+        x = torch.rand(10000, 4, device=dev)
+        h = histogram(x, 64)
+        assert h.shape == (64, 4)
+        assert int(h.sum()) == 40000
+
+    def test_bitset_set_test_count(self, dev, ext):
+        from raft_amd.core import Bitset
+        n = 1_000_000
+        bs = Bitset(n, device=dev, default=False)
+        torch.manual_seed(1)
+        idx = torch.randint(0, n, (50000,), device=dev)
+        bs.set(idx, True)
+        uniq = int(idx.unique().numel())
+        assert bs.count() == uniq
+        assert bool(bs.test(idx).all())
+        # clear half
+        half = idx[:25000]
+        bs.set(half, False)
+        assert not bool(bs.test(half).any())
+        expect = int(torch.cat([idx.unique(),
+                                half.unique()]).unique().numel()) \
+            - int(half.unique().numel())
+        # remaining = uniq minus cleared uniq that were set
+        remaining = int((~torch.isin(idx.unique(), half.unique())).sum())
+        assert bs.count() == remaining
+
+
+class TestLinewiseGpu:
+    @pytest.mark.parametrize("along_rows", [True, False])
+    @pytest.mark.parametrize("op", ["add", "sub", "mul", "div"])
+    def test_matrix_vector_native(self, dev, ext, along_rows, op):
+        from raft_amd.linalg.matrix_vector import matrix_vector_op
+        torch.manual_seed(0)
+        x = torch.randn(333, 129, device=dev)   # odd d -> scalar tail path
+        v = torch.rand((129 if along_rows else 333,), device=dev) + 0.5
+        out = matrix_vector_op(x, v, op, along_rows)
+        ref = getattr(torch, op if op != "div" else "div")(
+            x, v.unsqueeze(0 if along_rows else 1))
+        torch.testing.assert_close(out, ref)
+
+    @pytest.mark.parametrize("along_rows", [True, False])
+    def test_linewise_fused_standardize(self, dev, ext, along_rows):
+        from raft_amd.linalg.matrix_vector import linewise_fused
+        torch.manual_seed(1)
+        x = torch.randn(1000, 256, device=dev)
+        n = 256 if along_rows else 1000
+        mu = torch.randn(n, device=dev)
+        sig = torch.rand(n, device=dev) + 0.5
+        out = linewise_fused(x, mu, "sub", sig, "div", along_rows)
+        mu_b = mu.unsqueeze(0 if along_rows else 1)
+        sig_b = sig.unsqueeze(0 if along_rows else 1)
+        torch.testing.assert_close(out, (x - mu_b) / sig_b)
