@@ -12,6 +12,7 @@
 #pragma once
 
 #include "cluster.hpp"
+#include "core/mdspan.hpp"
 #include "core.hpp"
 #include "distance.hpp"
 #include "linalg.hpp"
@@ -20,3 +21,4 @@
 #include "random.hpp"
 #include "reductions.hpp"
 #include "sparse.hpp"
+#include "mdspan_api.hpp"

@@ -10,6 +10,15 @@ python scripts/include_checker.py
 echo "== native build (gfx950 cross-compile) =="
 python build_ext.py
 
+echo "== C++ library + out-of-tree consumer (mdspan API) =="
+cmake -S . -B build/cpp -G Ninja > /dev/null
+cmake --build build/cpp -j 16 > /dev/null
+cmake --install build/cpp --prefix build/install > /dev/null
+cmake -S tests/cpp/consumer -B build/consumer -G Ninja \
+  -Draft_amd_DIR="$PWD/build/install/lib/cmake/raft_amd" > /dev/null
+cmake --build build/consumer > /dev/null
+echo "consumer built against installed package"
+
 echo "== CPU test suite =="
 python -m pytest tests/ -q -m "not gpu"
 

@@ -83,3 +83,49 @@ def test_umbrella_header_links(tmp_path):
                         "-lhipblaslt", "-o", str(tmp_path / "consumer")],
                        capture_output=True, timeout=300)
     assert r.returncode == 0, r.stderr.decode()
+
+
+class TestMdspanHeader:
+    def test_mdspan_host_semantics(self, tmp_path):
+        """Compile + run the host-side mdspan/mdarray semantics test
+        (extents/layouts/strides/mdarray round-trip/memory tagging)."""
+        import subprocess
+        exe = tmp_path / "mdspan_host"
+        r = subprocess.run(
+            ["/opt/rocm/lib/llvm/bin/clang++", "-std=c++17", "-I", "include",
+             "-D__HIP_PLATFORM_AMD__=1", "-I/opt/rocm/include",
+             "tests/cpp/test_mdspan_host.cpp", "-o", str(exe)],
+            cwd=ROOT, capture_output=True, text=True, timeout=300)
+        assert r.returncode == 0, r.stderr[-2000:]
+        r = subprocess.run([str(exe)], capture_output=True, text=True, timeout=60)
+        assert r.returncode == 0 and "MDSPAN_HOST_OK" in r.stdout
+
+    def test_consumer_sources_configure(self, tmp_path):
+        """The out-of-tree consumer configures against the installed package
+        when a build tree exists (full build exercised in scripts/ci.sh and
+        the GPU run test)."""
+        import os
+        import subprocess
+        pkg = os.path.join(ROOT, "build", "install", "lib", "cmake", "raft_amd")
+        if not os.path.isdir(pkg):
+            import pytest
+            pytest.skip("no installed build tree (run scripts/ci.sh)")
+        r = subprocess.run(
+            ["cmake", "-S", "tests/cpp/consumer", "-B", str(tmp_path / "b"),
+             f"-Draft_amd_DIR={pkg}"],
+            cwd=ROOT, capture_output=True, text=True, timeout=300)
+        assert r.returncode == 0, r.stderr[-2000:]
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(not torch.cuda.is_available(), reason="no GPU")
+def test_consumer_binary_runs():
+    """The out-of-tree consumer (mdspan/mdarray API end to end: host->device
+    copies, pairwise L1, row_argmin, select_k, device->host, host-verified)
+    built against the INSTALLED package must run on the GPU."""
+    exe = os.path.join(ROOT, "build", "consumer", "consumer")
+    if not os.path.exists(exe):
+        pytest.skip("consumer binary not built (run scripts/ci.sh)")
+    r = subprocess.run([exe], capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, (r.stdout, r.stderr)
+    assert "CONSUMER_OK" in r.stdout
