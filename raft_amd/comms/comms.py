@@ -65,6 +65,14 @@ class Comms:
     def allreduce(self, t: torch.Tensor, op: ReduceOp = ReduceOp.SUM) -> torch.Tensor:
         raise NotImplementedError
 
+    def allreduce_async(self, t: torch.Tensor, op: ReduceOp = ReduceOp.SUM):
+        """Start an allreduce and return a waitable handle (or None).
+        On RCCL the collective runs on the communicator stream and overlaps
+        compute issued afterwards on the current stream; `handle.wait()`
+        inserts the stream dependency."""
+        self.allreduce(t, op)
+        return None
+
     def bcast(self, t: torch.Tensor, root: int = 0) -> torch.Tensor:
         raise NotImplementedError
 
@@ -194,6 +202,10 @@ class TorchDistComms(Comms):
     def allreduce(self, t, op=ReduceOp.SUM):
         dist.all_reduce(t, op=op.to_dist(), group=self.group)
         return t
+
+    def allreduce_async(self, t, op=ReduceOp.SUM):
+        return dist.all_reduce(t, op=op.to_dist(), group=self.group,
+                               async_op=True)
 
     def bcast(self, t, root=0):
         dist.broadcast(t, src=root, group=self.group)
