@@ -12,6 +12,8 @@
 
 namespace raft_amd {
 
+// from rng.hip (device sampling test driver)
+void launch_device_sample_test(const float*, int*, int, uint64_t, hipStream_t);
 // from linewise.hip
 void launch_linewise(const float*, float*, const float*, const float*,
                      long long, long long, bool, int, int, hipStream_t);
@@ -41,8 +43,8 @@ void launch_l2nn_epilogue(const float*, const float*, const float*, float*, int*
 void launch_pairwise_unexpanded(const float*, const float*, float*, long long, long long,
                                 long long, int, float, hipStream_t);
 // from rng.hip
-void launch_rng_uniform(float*, long long, uint64_t, uint64_t, hipStream_t);
-void launch_rng_normal(float*, long long, uint64_t, uint64_t, hipStream_t);
+void launch_rng_uniform(float*, long long, uint64_t, uint64_t, hipStream_t, int);
+void launch_rng_normal(float*, long long, uint64_t, uint64_t, hipStream_t, int);
 void launch_make_blobs(float*, int*, const float*, long long, long long, int, float,
                        uint64_t, uint64_t, hipStream_t);
 // from spmv.hip
@@ -223,17 +225,19 @@ torch::Tensor pairwise_unexpanded(torch::Tensor x, torch::Tensor y, int64_t code
   return out;
 }
 
-torch::Tensor rng_uniform(int64_t n, int64_t seed, int64_t subseq, int64_t device) {
+torch::Tensor rng_uniform(int64_t n, int64_t seed, int64_t subseq, int64_t device,
+                          int64_t gen = 0) {
   auto out = torch::empty({n}, torch::dtype(torch::kFloat32).device(torch::kCUDA, device));
   raft_amd::launch_rng_uniform(out.data_ptr<float>(), n, (uint64_t)seed,
-                               (uint64_t)subseq, cur_stream());
+                               (uint64_t)subseq, cur_stream(), (int)gen);
   return out;
 }
 
-torch::Tensor rng_normal(int64_t n, int64_t seed, int64_t subseq, int64_t device) {
+torch::Tensor rng_normal(int64_t n, int64_t seed, int64_t subseq, int64_t device,
+                         int64_t gen = 0) {
   auto out = torch::empty({n}, torch::dtype(torch::kFloat32).device(torch::kCUDA, device));
   raft_amd::launch_rng_normal(out.data_ptr<float>(), n, (uint64_t)seed,
-                              (uint64_t)subseq, cur_stream());
+                              (uint64_t)subseq, cur_stream(), (int)gen);
   return out;
 }
 
@@ -705,8 +709,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("l2_epilogue_", &l2_epilogue_, "in-place L2 distance epilogue");
   m.def("l2nn_epilogue", &l2nn_epilogue, "fused argmin epilogue over GEMM tile");
   m.def("pairwise_unexpanded", &pairwise_unexpanded, "tiled unexpanded distances");
-  m.def("rng_uniform", &rng_uniform, "PCG32 uniform [0,1)");
-  m.def("rng_normal", &rng_normal, "PCG32 + Box-Muller standard normal");
+  m.def("rng_uniform", &rng_uniform, "counter-based uniform [0,1)",
+        pybind11::arg("n"), pybind11::arg("seed"), pybind11::arg("subseq"),
+        pybind11::arg("device"), pybind11::arg("gen") = 0);
+  m.def("rng_normal", &rng_normal, "counter-based Box-Muller standard normal",
+        pybind11::arg("n"), pybind11::arg("seed"), pybind11::arg("subseq"),
+        pybind11::arg("device"), pybind11::arg("gen") = 0);
   m.def("make_blobs", &make_blobs, "fused gaussian blob generator");
   m.def("csr_spmv", &csr_spmv, "CSR SpMV (sub-wave per row)");
   m.def("reduce_rows_by_key", &reduce_rows_by_key, "keyed row accumulation");
@@ -721,6 +729,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kmeans_update_centroids", &kmeans_update_centroids,
         "centroids = counts>0 ? sums/counts : centroids");
   m.def("select_k", &select_k, "batched top-k (radix)");
+  m.def("device_sample_test", [](torch::Tensor weights, int64_t n_draws,
+                                 int64_t seed) {
+    TORCH_CHECK(weights.is_cuda() && weights.numel() == 256 &&
+                weights.scalar_type() == torch::kFloat32);
+    auto out = torch::empty({n_draws}, weights.options().dtype(torch::kInt32));
+    raft_amd::launch_device_sample_test(weights.contiguous().data_ptr<float>(),
+                                        out.data_ptr<int>(), (int)n_draws,
+                                        (uint64_t)seed, cur_stream());
+    return out;
+  }, "block_random_sample test driver (weighted in-kernel selection)");
   m.def("linewise", [](torch::Tensor x, torch::Tensor v1,
                        c10::optional<torch::Tensor> v2, bool along_rows,
                        int64_t op1, int64_t op2) {

@@ -311,13 +311,47 @@ class TorchDistComms(Comms):
             w.wait()
 
 
+_MPI_RANK_VARS = ["OMPI_COMM_WORLD_RANK", "PMI_RANK", "SLURM_PROCID",
+                  "MV2_COMM_WORLD_RANK"]
+_MPI_SIZE_VARS = ["OMPI_COMM_WORLD_SIZE", "PMI_SIZE", "SLURM_NTASKS",
+                  "MV2_COMM_WORLD_SIZE"]
+_MPI_LOCAL_VARS = ["OMPI_COMM_WORLD_LOCAL_RANK", "MPI_LOCALRANKID",
+                   "SLURM_LOCALID", "MV2_COMM_WORLD_LOCAL_RANK"]
+
+
+def _adopt_mpi_env() -> bool:
+    """mpi_comms-style bootstrap (reference comms/mpi_comms.hpp:50): when
+    launched under mpirun/srun instead of torchrun, adopt the MPI rank
+    variables into the torchrun contract so the same RCCL process-group
+    rendezvous works. MASTER_ADDR defaults to 127.0.0.1 (single node);
+    multi-node mpirun exports it explicitly."""
+    if "RANK" in os.environ and "WORLD_SIZE" in os.environ:
+        return False
+    for rv, sv in zip(_MPI_RANK_VARS, _MPI_SIZE_VARS):
+        if rv in os.environ and sv in os.environ:
+            os.environ["RANK"] = os.environ[rv]
+            os.environ["WORLD_SIZE"] = os.environ[sv]
+            for lv in _MPI_LOCAL_VARS:
+                if lv in os.environ:
+                    os.environ.setdefault("LOCAL_RANK", os.environ[lv])
+                    break
+            os.environ.setdefault("LOCAL_RANK", os.environ[rv])
+            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+            os.environ.setdefault("MASTER_PORT", "29741")
+            return True
+    return False
+
+
 def init(backend: Optional[str] = None, timeout_s: int = 900) -> TorchDistComms:
-    """Bootstrap from torchrun-style env (RANK/WORLD_SIZE/MASTER_ADDR/PORT).
+    """Bootstrap from torchrun-style env (RANK/WORLD_SIZE/MASTER_ADDR/PORT)
+    or, when absent, from MPI launcher env (OMPI/PMI/SLURM rank variables —
+    the mpi_comms path: `mpirun -np N python app.py` works with no torchrun).
 
     The MNMG analog of raft_dask Comms.init() (comms.py:161): rendezvous, RCCL
     communicator creation, and handle injection collapse into process-group init.
     """
     if not dist.is_initialized():
+        _adopt_mpi_env()
         if backend is None:
             backend = "nccl" if torch.cuda.is_available() else "gloo"
         if backend == "nccl":

@@ -30,9 +30,12 @@ _PCG_INC = 1442695040888963407
 
 @dataclass
 class RngState:
-    """seed + base_subsequence (rng_state.hpp:21-43)."""
+    """seed + base_subsequence + generator type (rng_state.hpp:21-43:
+    GeneratorType {PCG, Philox}). Both generators are counter-based and
+    bitwise CPU/GPU identical."""
     seed: int = 0
     base_subsequence: int = 0
+    gen_type: str = "pcg"      # "pcg" | "philox"
 
     def advance(self, n: int = 1) -> None:
         self.base_subsequence = (self.base_subsequence + n) & _M64
@@ -68,10 +71,48 @@ def _ror32(x: torch.Tensor, rot: torch.Tensor) -> torch.Tensor:
     return ((x >> rot) | (x << ((32 - rot) & 31))) & _M32
 
 
+_PHILOX_M0 = 0xD2511F53
+_PHILOX_M1 = 0xCD9E8D57
+_PHILOX_W0 = 0x9E3779B9
+_PHILOX_W1 = 0xBB67AE85
+
+
+def _philox_block(seed: int, subsequence: int, idx: torch.Tensor) -> torch.Tensor:
+    """Vectorized Philox4x32-10: one 32-bit draw per element of idx.
+
+    Counter = (idx_lo, idx_hi, subseq_lo, subseq_hi), key = (seed_lo,
+    seed_hi); 10 rounds of the standard Philox round function. Mirrors
+    csrc/rng.hip philox_hash() EXACTLY (uint32 mulhi/mullo emulated in
+    int64), so CPU and GPU draws are bitwise identical.
+    """
+    c0 = idx & _M32
+    c1 = _lshr(idx, 32) & _M32
+    c2 = torch.full_like(idx, subsequence & 0xFFFFFFFF)
+    c3 = torch.full_like(idx, (subsequence >> 32) & 0xFFFFFFFF)
+    k0, k1 = seed & 0xFFFFFFFF, (seed >> 32) & 0xFFFFFFFF
+    for _ in range(10):
+        p0 = c0 * _PHILOX_M0          # exact 32x32 -> 64 in int64
+        p1 = c2 * _PHILOX_M1
+        n0 = (_lshr(p1, 32) ^ c1 ^ k0) & _M32
+        n1 = p1 & _M32
+        n2 = (_lshr(p0, 32) ^ c3 ^ k1) & _M32
+        n3 = p0 & _M32
+        c0, c1, c2, c3 = n0, n1, n2, n3
+        k0 = (k0 + _PHILOX_W0) & 0xFFFFFFFF
+        k1 = (k1 + _PHILOX_W1) & 0xFFFFFFFF
+    return c0
+
+
+def _u32_block(state: RngState, subsequence: int, idx: torch.Tensor) -> torch.Tensor:
+    if state.gen_type == "philox":
+        return _philox_block(state.seed, subsequence, idx)
+    return _pcg32_block(state.seed, subsequence, idx)
+
+
 def _draw_u32(n: int, state: RngState, device, n_draws: int = 1) -> torch.Tensor:
     """[n_draws, n] uint32 draws (as int64) at subsequence offsets 0..n_draws-1."""
     idx = torch.arange(n, dtype=torch.int64, device=device)
-    outs = [_pcg32_block(state.seed, state.base_subsequence + d, idx) for d in range(n_draws)]
+    outs = [_u32_block(state, state.base_subsequence + d, idx) for d in range(n_draws)]
     state.advance(n_draws)
     return torch.stack(outs, dim=0)
 
@@ -83,7 +124,8 @@ def _gpu_or_cpu_uniform01(shape, state: RngState, device, dtype) -> torch.Tensor
     device = torch.device(device) if device is not None else torch.device("cpu")
     if device.type == "cuda":
         ext = require_ext()
-        out = ext.rng_uniform(n, int(state.seed), int(state.base_subsequence), device.index or 0)
+        out = ext.rng_uniform(n, int(state.seed), int(state.base_subsequence), device.index or 0,
+                              gen=1 if state.gen_type == "philox" else 0)
         state.advance(1)
         return out.reshape(shape).to(dtype)
     u = _draw_u32(n, state, device, n_draws=1)[0]
@@ -115,7 +157,8 @@ def uniform_int(shape, low: int, high: int, state: RngState | None = None, devic
     device = torch.device(device) if device is not None else torch.device("cpu")
     if device.type == "cuda":
         ext = require_ext()
-        u = ext.rng_uniform(n, int(state.seed), int(state.base_subsequence), device.index or 0)
+        u = ext.rng_uniform(n, int(state.seed), int(state.base_subsequence), device.index or 0,
+                            gen=1 if state.gen_type == "philox" else 0)
         state.advance(1)
         draw = (u.double() * (high - low)).long() + low
         return draw.reshape(shape).to(dtype)
@@ -134,7 +177,8 @@ def normal(shape, mu: float = 0.0, sigma: float = 1.0, state: RngState | None = 
     device = torch.device(device) if device is not None else torch.device("cpu")
     if device.type == "cuda":
         ext = require_ext()
-        out = ext.rng_normal(n, int(state.seed), int(state.base_subsequence), device.index or 0)
+        out = ext.rng_normal(n, int(state.seed), int(state.base_subsequence), device.index or 0,
+                             gen=1 if state.gen_type == "philox" else 0)
         state.advance(2)
         return (out.reshape(shape) * sigma + mu).to(dtype)
     us = _draw_u32(n, state, device, n_draws=2)

@@ -131,3 +131,41 @@ def test_collectives_gloo_world3():
     from tests.conftest import spawn_gloo
 
     spawn_gloo(_collective_worker, world_size=3)
+
+
+def _mpi_style_entry(rank, world, port):
+    """Simulate an mpirun-launched process: only OMPI_* vars set."""
+    import os
+    for k in ("RANK", "WORLD_SIZE", "LOCAL_RANK"):
+        os.environ.pop(k, None)
+    os.environ["OMPI_COMM_WORLD_RANK"] = str(rank)
+    os.environ["OMPI_COMM_WORLD_SIZE"] = str(world)
+    os.environ["OMPI_COMM_WORLD_LOCAL_RANK"] = str(rank)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    from raft_amd.comms import init_comms
+    c = init_comms(backend="gloo")
+    assert c.get_size() == world and c.get_rank() == rank
+    t = torch.ones(3)
+    c.allreduce(t)
+    assert torch.equal(t, torch.full((3,), float(world)))
+    import torch.distributed as dist
+    dist.destroy_process_group()
+
+
+def test_mpi_launcher_bootstrap():
+    """mpi_comms parity (reference comms/mpi_comms.hpp:50): processes with
+    only MPI rank env vars rendezvous through the same init()."""
+    import torch.multiprocessing as mp
+    from tests.conftest import _free_port
+
+    port = _free_port()
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_mpi_style_entry, args=(r, 2, port))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(180)
+    for p in procs:
+        assert p.exitcode == 0, f"worker exited with {p.exitcode}"

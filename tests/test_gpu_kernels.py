@@ -698,3 +698,27 @@ class TestLinewiseGpu:
         mu_b = mu.unsqueeze(0 if along_rows else 1)
         sig_b = sig.unsqueeze(0 if along_rows else 1)
         torch.testing.assert_close(out, (x - mu_b) / sig_b)
+
+
+class TestPhiloxGpu:
+    def test_philox_bitwise_cpu_gpu(self, dev, ext):
+        from raft_amd.random.rng import RngState, uniform
+        sg = RngState(seed=77, gen_type="philox")
+        sc = RngState(seed=77, gen_type="philox")
+        g = uniform((40000,), state=sg, device=dev)
+        c = uniform((40000,), state=sc, device="cpu")
+        assert torch.equal(g.cpu(), c.float())
+
+    def test_device_sampling_proportional(self, dev, ext):
+        # block_random_sample: empirical pick frequency must track weights
+        w = torch.zeros(256, device=dev)
+        w[10] = 1.0
+        w[20] = 2.0
+        w[200] = 4.0
+        draws = ext.device_sample_test(w, 70000, 123)
+        counts = torch.bincount(draws.long(), minlength=256).float()
+        assert counts[[10, 20, 200]].sum() == 70000  # zero-weight never picked
+        p = counts / 70000
+        assert abs(p[10] - 1 / 7) < 0.02
+        assert abs(p[20] - 2 / 7) < 0.02
+        assert abs(p[200] - 4 / 7) < 0.02
