@@ -6,8 +6,11 @@
 
 namespace raft_amd {
 
-void launch_rng_uniform(float* out, long long n, uint64_t seed, uint64_t subseq, hipStream_t s);
-void launch_rng_normal(float* out, long long n, uint64_t seed, uint64_t subseq, hipStream_t s);
+// gen: 0 = PCG32 (default), 1 = Philox4x32-10
+void launch_rng_uniform(float* out, long long n, uint64_t seed, uint64_t subseq, hipStream_t s,
+                        int gen = 0);
+void launch_rng_normal(float* out, long long n, uint64_t seed, uint64_t subseq, hipStream_t s,
+                       int gen = 0);
 void launch_make_blobs(float* x, int* labels, const float* centers, long long n_rows,
                        long long d, int k, float cluster_std, uint64_t seed,
                        uint64_t subseq, hipStream_t s);
