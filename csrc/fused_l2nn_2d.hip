@@ -354,6 +354,16 @@ static int l2nn_2d_gt() {
   return gt;
 }
 
+// effective col-tiles-per-block: drop to the largest of {gt, 2, 1} dividing
+// the col-tile count, so odd tile counts (e.g. n=384) keep the 2D engine's
+// L2 win instead of falling back to v1 (round-1 NOTES item 15)
+static int l2nn_2d_gt_for(int n) {
+  int gt = l2nn_2d_gt();
+  const int tiles = n / 128;
+  while (gt > 1 && tiles % gt != 0) gt >>= 1;
+  return gt;
+}
+
 bool fused_l2nn_2d_supported(int nslice, long long m, int n, int d) {
   static const char mode = [] {
     const char* e = getenv("RAFT_AMD_L2NN_2D");
@@ -361,7 +371,7 @@ bool fused_l2nn_2d_supported(int nslice, long long m, int n, int d) {
   }();
   if (mode == '0') return false;
   if (nslice > 2) return false;  // 3-slice LDS (96 KiB) drops to 1 block/CU
-  if ((n / 128) % l2nn_2d_gt() != 0) return false;
+  if (n % 128 != 0) return false;
   if (mode == '1') return true;
   // auto: pays off when the col-tile loop is long enough that X re-reads
   // dominate and m is large enough that partial traffic amortizes
@@ -375,7 +385,7 @@ void launch_fused_l2nn_2d(const void** xsl, const void** csl, const float* xn,
                           float* dmin, int* amin, float* dmin2,
                           long long m, int n, int d, int nslice,
                           hipStream_t stream) {
-  const int gt = l2nn_2d_gt();
+  const int gt = l2nn_2d_gt_for(n);
   const int n_row_tiles = (int)((m + 127) / 128);
   const int n_groups = n / 128 / gt;
   const int grid = n_row_tiles * n_groups;

@@ -4,6 +4,7 @@ import torch
 
 from raft_amd import linalg
 from raft_amd.linalg import Apply, NormType
+from raft_amd.linalg.decomp import eig_jacobi, eig_selective
 
 
 class TestReduce:
@@ -227,3 +228,42 @@ class TestDecompCrossValidation:
         _, s, _ = rsvd(a, k=6, n_iter=4, seed=0)
         ref = np.linalg.svd(a.numpy(), compute_uv=False)[:6]
         np.testing.assert_allclose(s.numpy(), ref, rtol=1e-3)
+
+
+class TestEigVariants:
+    """eig_jacobi is a REAL cyclic-Jacobi solver and eig_selective a real
+    partial solver (syevj/syevdx parity, not eigh aliases)."""
+
+    @pytest.mark.parametrize("n", [1, 2, 7, 32, 65])
+    def test_jacobi_matches_eigh(self, n):
+        torch.manual_seed(n)
+        a = torch.randn(n, n, dtype=torch.float64)
+        a = (a + a.t()) / 2
+        w_j, v_j = eig_jacobi(a, tol=1e-12)
+        w_r = torch.linalg.eigvalsh(a)
+        torch.testing.assert_close(w_j, w_r, rtol=1e-8, atol=1e-8)
+        # eigenvector residuals
+        res = (a @ v_j - v_j * w_j.unsqueeze(0)).norm()
+        assert float(res) < 1e-7 * max(1.0, float(a.norm()))
+        # orthonormal
+        torch.testing.assert_close(v_j.t() @ v_j, torch.eye(n, dtype=a.dtype),
+                                   rtol=1e-8, atol=1e-8)
+
+    def test_selective_lobpcg_path(self):
+        torch.manual_seed(3)
+        n, k = 256, 5
+        a = torch.randn(n, n, dtype=torch.float64)
+        a = (a + a.t()) / 2
+        w_full = torch.linalg.eigvalsh(a)
+        w_top, v_top = eig_selective(a, k, largest=True, method="lobpcg")
+        torch.testing.assert_close(w_top, w_full[-k:], rtol=1e-6, atol=1e-6)
+        w_bot, v_bot = eig_selective(a, k, largest=False, method="lobpcg")
+        torch.testing.assert_close(w_bot, w_full[:k], rtol=1e-6, atol=1e-6)
+
+    def test_selective_full_fallback(self):
+        torch.manual_seed(4)
+        a = torch.randn(16, 16)
+        a = (a + a.t()) / 2
+        w, v = eig_selective(a, 12, largest=True)   # wide range -> full
+        torch.testing.assert_close(w, torch.linalg.eigvalsh(a)[-12:],
+                                   rtol=1e-4, atol=1e-4)
