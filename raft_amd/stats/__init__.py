@@ -6,7 +6,7 @@ from .contingency import contingency_matrix
 from .clustering import (
     adjusted_rand_index, rand_index, mutual_info_score, entropy,
     homogeneity_score, completeness_score, v_measure, kl_divergence,
-    dispersion, silhouette_score,
+    dispersion, silhouette_score, silhouette_score_batched,
 )
 from .regression import r2_score, regression_metrics, information_criterion
 from .classification import accuracy_score
@@ -17,7 +17,7 @@ __all__ = [
     "mean_center", "mean_add", "cov", "histogram", "contingency_matrix",
     "adjusted_rand_index", "rand_index", "mutual_info_score", "entropy",
     "homogeneity_score", "completeness_score", "v_measure", "kl_divergence",
-    "dispersion", "silhouette_score", "r2_score", "regression_metrics",
+    "dispersion", "silhouette_score", "silhouette_score_batched", "r2_score", "regression_metrics",
     "information_criterion", "accuracy_score", "neighborhood_recall",
     "trustworthiness_score",
 ]

@@ -138,3 +138,12 @@ def silhouette_score(x: torch.Tensor, labels: torch.Tensor, n_clusters: int | No
                           torch.zeros_like(a))
         s_total += float(sil.sum())
     return s_total / n
+
+
+def silhouette_score_batched(x: torch.Tensor, labels: torch.Tensor,
+                             n_clusters: int | None = None,
+                             batch_size: int = 4096) -> float:
+    """Batched silhouette (reference stats/detail/batched/silhouette_score):
+    identical result to silhouette_score, with the row-batch size exposed so
+    the [batch, n] distance block is bounded on huge inputs."""
+    return silhouette_score(x, labels, n_clusters=n_clusters, chunk=batch_size)

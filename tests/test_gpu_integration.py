@@ -330,3 +330,33 @@ class TestKMeansWideDim:
         # the fast loop's result is a valid EM trajectory: inertia decreases
         _, inertia_v2 = kmeans_iterate(x, cv, 1, fp32_mode="bf16x2v")
         assert inertia_v2 <= inertia_v * 1.0001
+
+
+class TestKnnFp32Exactness:
+    """the fp32 filtered path re-ranks candidates by EXACT fp32
+    distances with a per-row margin proof — results must match the exact
+    expanded-fp32 top-k, and fp32_mode='native' must skip the MFMA filter."""
+
+    def test_fp32_filtered_rerank_exact(self, dev):
+        from raft_amd.neighbors import knn
+        torch.manual_seed(0)
+        x = torch.randn(60000, 128, device=dev)
+        q = torch.randn(2000, 128, device=dev)
+        d, i = knn(x, q, k=16)                       # auto -> filtered+rerank
+        xn = (x * x).sum(1)
+        qn = (q * q).sum(1)
+        ref = (qn.unsqueeze(1) + xn.unsqueeze(0) - 2.0 * (q @ x.t())).clamp_min(0)
+        rd, ri = torch.topk(ref, 16, dim=1, largest=False)
+        agree = (i == ri).float().mean()
+        assert float(agree) > 0.9995, float(agree)   # near-ties may reorder
+        torch.testing.assert_close(d, rd, rtol=1e-4, atol=1e-3)
+
+    def test_fp32_native_mode_honored(self, dev):
+        from raft_amd.neighbors import knn
+        torch.manual_seed(1)
+        x = torch.randn(20000, 64, device=dev)
+        q = torch.randn(500, 64, device=dev)
+        d_nat, i_nat = knn(x, q, k=8, fp32_mode="native")
+        d_auto, i_auto = knn(x, q, k=8)
+        assert (i_nat == i_auto).float().mean() > 0.999
+        torch.testing.assert_close(d_nat, d_auto, rtol=1e-3, atol=1e-3)

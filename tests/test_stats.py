@@ -217,3 +217,30 @@ class TestSklearnCrossValidation:
         counts = torch.bincount(labels).numpy()
         assert entropy(labels) == pytest.approx(
             float(sp_entropy(counts / counts.sum())), abs=1e-6)
+
+
+class TestWeightedCovBatchedSilhouette:
+    def test_weighted_cov_matches_repetition(self):
+        # integer weights == row repetition
+        from raft_amd.stats import cov
+        torch.manual_seed(0)
+        x = torch.randn(50, 4, dtype=torch.float64)
+        w = torch.randint(1, 4, (50,)).double()
+        rep = torch.repeat_interleave(x, w.long(), dim=0)
+        torch.testing.assert_close(cov(x, weights=w), cov(rep),
+                                   rtol=1e-10, atol=1e-10)
+        torch.testing.assert_close(cov(x, sample=False, weights=w),
+                                   cov(rep, sample=False),
+                                   rtol=1e-10, atol=1e-10)
+
+    def test_batched_silhouette_equals_unbatched(self):
+        from raft_amd.stats import silhouette_score, silhouette_score_batched
+        from raft_amd.random import make_blobs, RngState
+        x, y, _ = make_blobs(600, 5, n_clusters=4, cluster_std=0.4,
+                             state=RngState(seed=2))
+        full = silhouette_score(x, y)
+        batched = silhouette_score_batched(x, y, batch_size=100)
+        assert abs(full - batched) < 1e-9
+        # sklearn cross-check
+        from sklearn.metrics import silhouette_score as sk
+        assert abs(full - sk(x.numpy(), y.numpy())) < 1e-4
