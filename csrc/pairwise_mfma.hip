@@ -203,11 +203,174 @@ __global__ void pairwise_l2_filter_kernel(const __bf16* __restrict__ x0,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Persistent-X filter (bf16 index, d <= 128): the X row panel (full depth,
+// <= 32 KiB) stages ONCE per block and stays LDS-resident while the block
+// walks CT col-tiles, double-buffering ONLY the 16 KiB C chunks with counted
+// vmcnt (flight depth 1 chunk, no full drains). Halves the gloads per
+// K-step and removes the X re-stage entirely; 64 KiB LDS keeps 2 blocks/CU
+// (the l2nn persistent-X experiment lost to occupancy at 1 block/CU — this
+// configuration keeps both the residency AND the 2-block overlap).
+// Block decode groups same-col-group blocks on one XCD so the C panels are
+// the L2-shared working set.
+// ---------------------------------------------------------------------------
+template <int CT = 8>
+__launch_bounds__(256, 2)
+__global__ void pairwise_l2_filter_px_kernel(
+    const __bf16* __restrict__ x0g, const __bf16* __restrict__ c0g,
+    const float* __restrict__ xn, const float* __restrict__ yn,
+    const float* __restrict__ thr, float* __restrict__ out_d,
+    int* __restrict__ out_i, int* __restrict__ cnt, int cap,
+    long long col_offset, long long m, long long n, int d, int n_row_tiles) {
+  extern __shared__ __bf16 smem[];
+  const int kts = d / 64;
+
+  // bijective XCD-contiguous remap, COLUMN-group-major (blocks sharing a C
+  // group land on one XCD)
+  const int nwg = gridDim.x;
+  const int bid = blockIdx.x;
+  const int q = nwg >> 3, r = nwg & 7;
+  const int xcd = bid & 7, slot = bid >> 3;
+  const int t = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + slot;
+  const long long row0 = (long long)(t % n_row_tiles) * 128;
+  const int ctg = t / n_row_tiles;
+  const long long col_base = (long long)ctg * CT * 128;
+  if (row0 >= m || col_base >= n) return;
+  const int nct = (int)((n - col_base + 127) / 128) < CT
+                      ? (int)((n - col_base + 127) / 128)
+                      : CT;
+
+  const int lane = threadIdx.x % RAFT_AMD_WAVE;
+  const int w = threadIdx.x / RAFT_AMD_WAVE;
+  const int wr = w >> 1, wc = w & 1;
+
+  // stage the X panel once: kts chunks of [128][64] at kt*8192 elements
+  for (int kt = 0; kt < kts; kt++)
+    mfma_stage_tile128(x0g, smem + kt * 8192, row0, (long long)kt * 64, d,
+                       m - 1);
+  __bf16* cb0 = smem + kts * 8192;   // two 16 KiB C chunk buffers
+  __bf16* cb1 = cb0 + 8192;
+  const int total = nct * kts;
+  auto stage_c = [&](int s, __bf16* buf) {
+    const long long c0 = col_base + (long long)(s / kts) * 128;
+    mfma_stage_tile128(c0g, buf, c0, (long long)(s % kts) * 64, d, n - 1);
+  };
+  stage_c(0, cb0);
+  if (total > 1) stage_c(1, cb1);
+  // X (4*kts gloads) + chunk 0 must land; chunk 1 stays in flight
+  if (total > 1) {
+    asm volatile("s_waitcnt vmcnt(4)\n\ts_barrier" ::: "memory");
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)\n\ts_barrier" ::: "memory");
+  }
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int a = 0; a < 4; a++)
+#pragma unroll
+    for (int b = 0; b < 4; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  for (int s = 0; s < total; s++) {
+    const int kt = s % kts;
+    __bf16* cbuf = (s & 1) ? cb1 : cb0;
+    const __bf16* xchunk = smem + kt * 8192;
+    // ALL fragments for this chunk first (a/b for both kf), then a barrier
+    // confirming every wave's reads are done, THEN the refill of this
+    // buffer with chunk s+2 — a mid-read refill races other waves' ds_reads
+    bf16x8 a_frag[2][4], b_frag[2][4];
+#pragma unroll
+    for (int kf = 0; kf < 2; kf++) {
+      const int kbyte = (kf * 32 + (lane >> 4) * 8) * 2;
+#pragma unroll
+      for (int fr = 0; fr < 4; fr++) {
+        const int rr = wr * 64 + fr * 16 + (lane & 15);
+        a_frag[kf][fr] = *reinterpret_cast<const bf16x8*>(
+            (const char*)xchunk + mfma_swz(rr * 128 + kbyte));
+      }
+#pragma unroll
+      for (int fc = 0; fc < 4; fc++) {
+        const int cc = wc * 64 + fc * 16 + (lane & 15);
+        b_frag[kf][fc] = *reinterpret_cast<const bf16x8*>(
+            (const char*)cbuf + mfma_swz(cc * 128 + kbyte));
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)\n\ts_barrier" ::: "memory");
+    if (s + 2 < total) stage_c(s + 2, cbuf);
+#pragma unroll
+    for (int kf = 0; kf < 2; kf++)
+#pragma unroll
+      for (int fr = 0; fr < 4; fr++)
+#pragma unroll
+        for (int fc = 0; fc < 4; fc++)
+          acc[fr][fc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[kf][fr], b_frag[kf][fc], acc[fr][fc], 0, 0, 0);
+    if (kt == kts - 1) {
+      // filter epilogue for this col tile (registers + global atomics only)
+      const long long col0 = col_base + (long long)(s / kts) * 128;
+#pragma unroll
+      for (int fr = 0; fr < 4; fr++) {
+#pragma unroll
+        for (int reg = 0; reg < 4; reg++) {
+          const long long row = row0 + wr * 64 + fr * 16 + (lane >> 4) * 4 + reg;
+          if (row >= m) continue;
+          const float xv = xn[row];
+          const float tv = thr[row];
+#pragma unroll
+          for (int fc = 0; fc < 4; fc++) {
+            const long long col = col0 + wc * 64 + fc * 16 + (lane & 15);
+            if (col < n) {
+              const float d2 = fmaxf(xv + yn[col] - 2.f * acc[fr][fc][reg], 0.f);
+              if (d2 <= tv) {
+                const int pos = atomicAdd(&cnt[row], 1);
+                if (pos < cap) {
+                  out_d[row * cap + pos] = d2;
+                  out_i[row * cap + pos] = (int)(col + col_offset);
+                }
+              }
+            }
+          }
+        }
+      }
+#pragma unroll
+      for (int a = 0; a < 4; a++)
+#pragma unroll
+        for (int b = 0; b < 4; b++) acc[a][b] = f32x4{0.f, 0.f, 0.f, 0.f};
+    }
+    // chunk s+1 must have landed before the next iteration reads it (leave
+    // the 4 loads issued for s+2 in flight)
+    if (s + 2 < total) {
+      asm volatile("s_waitcnt vmcnt(4)\n\ts_barrier" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)\n\ts_barrier" ::: "memory");
+    }
+  }
+}
+
 void launch_pairwise_l2_filter(const void** xsl, const void** csl, const float* xn,
                                const float* yn, const float* thr, float* out_d,
                                int* out_i, int* cnt, int cap, long long col_offset,
                                long long m, long long n, int d, int nslice,
                                hipStream_t stream) {
+  // Persistent-X A/B (RAFT_AMD_KNN_PX=1): 7.5k vs 9.9k q/s at the 100M
+  // headline — the 1-chunk flight depth can't hide HBM latency in the
+  // streaming regime and the per-chunk lockstep loses to the independent
+  // full-drain blocks. OFF by default; kept as a recorded experiment.
+  static const bool px = [] {
+    const char* e = getenv("RAFT_AMD_KNN_PX");
+    return e && e[0] == '1';
+  }();
+  if (px && nslice == 1 && d % 64 == 0 && d <= 128 && n >= 1024) {
+    constexpr int CT = 8;
+    const int n_row_tiles = (int)((m + 127) / 128);
+    const int n_groups = (int)((n + (long long)CT * 128 - 1) / ((long long)CT * 128));
+    dim3 grid((unsigned)((long long)n_row_tiles * n_groups));
+    const size_t lds = (size_t)(d / 64) * 8192 * 2 + 2 * 16384;  // X + 2 C bufs
+    hipLaunchKernelGGL((pairwise_l2_filter_px_kernel<CT>), grid, dim3(256), lds,
+                       stream, (const __bf16*)xsl[0], (const __bf16*)csl[0],
+                       xn, yn, thr, out_d, out_i, cnt, cap, col_offset, m, n,
+                       d, n_row_tiles);
+    return;
+  }
   const int rg = (int)((m + 1023) / 1024);       // ceil(R/8), R=ceil(m/128)
   const int cg = (int)((n + 1023) / 1024);
   dim3 grid((unsigned)((long long)rg * cg * 64));
