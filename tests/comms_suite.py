@@ -22,6 +22,19 @@ def collective_suite(c, device):
     expected = sum(r + 1 for r in range(world))
     assert torch.equal(t, torch.full((4,), float(expected), device=dev))
 
+    # async allreduce (the kmeans split-batch overlap path): two outstanding
+    # collectives issued back to back, waited in order
+    ta = torch.full((8,), float(rank + 1), device=dev)
+    tb = torch.full((8,), 2.0 * (rank + 1), device=dev)
+    wa = c.allreduce_async(ta)
+    wb = c.allreduce_async(tb)
+    if wa is not None:
+        wa.wait()
+    if wb is not None:
+        wb.wait()
+    assert torch.equal(ta, torch.full((8,), float(expected), device=dev))
+    assert torch.equal(tb, torch.full((8,), 2.0 * expected, device=dev))
+
     # allreduce MAX (the bench elapsed-time reduction path)
     t = torch.full((1,), float(rank), device=dev, dtype=torch.float64)
     c.allreduce(t, op=ReduceOp.MAX)
