@@ -7,12 +7,70 @@
 // type-tagged (core/mdspan.hpp): passing a host view is a compile error.
 #pragma once
 
+#include "cluster.hpp"
 #include "core/mdspan.hpp"
 #include "distance.hpp"
 #include "matrix.hpp"
+#include "neighbors.hpp"
 #include "reductions.hpp"
 
 namespace raft_amd {
+
+// ---- fused L2 nearest-neighbor (the k-means assignment step) --------------
+// x [m, d] fp32, c [n, d] fp32 -> (dmin[m], amin[m]) with PROVABLY
+// fp32-exact argmin: the split-bf16 (bf16x2) MFMA engine + the in-kernel
+// second-best margin test + exact-fp32 rescan/repair of inconclusive rows —
+// the flagship engine, reachable from pure C++ (scratch comes from the
+// caller's workspace uvector, grown on demand). Requires d % 64 == 0.
+inline void fused_l2nn(device_matrix_view<const float> x,
+                       device_matrix_view<const float> c,
+                       device_vector_view<float> dmin,
+                       device_vector_view<int> amin,
+                       device_uvector<char>& workspace,
+                       hipStream_t stream = nullptr) {
+  const long long m = x.extent(0), d = x.extent(1), n = c.extent(0);
+  if (c.extent(1) != d || dmin.extent(0) != m || amin.extent(0) != m)
+    throw std::invalid_argument("fused_l2nn: extents mismatch");
+  if (d % 64 != 0) throw std::invalid_argument("fused_l2nn: d % 64 != 0");
+  constexpr int kSlices = 2;
+  // scratch layout: x slices | c slices | xn | cn | dmin2 | cn_max
+  const std::size_t xs_b = static_cast<std::size_t>(m) * d * 2;
+  const std::size_t cs_b = static_cast<std::size_t>(n) * d * 2;
+  const std::size_t xn_b = static_cast<std::size_t>(m) * 4;
+  const std::size_t cn_b = static_cast<std::size_t>(n) * 4;
+  const std::size_t need = kSlices * (xs_b + cs_b) + xn_b + cn_b + xn_b + 4;
+  if (workspace.size() < need) workspace = device_uvector<char>(need);
+  char* p = workspace.data();
+  void* xs[3] = {p, p + xs_b, nullptr};
+  p += kSlices * xs_b;
+  void* cs[3] = {p, p + cs_b, nullptr};
+  p += kSlices * cs_b;
+  float* xn = reinterpret_cast<float*>(p);
+  p += xn_b;
+  float* cn = reinterpret_cast<float*>(p);
+  p += cn_b;
+  float* dmin2 = reinterpret_cast<float*>(p);
+  p += xn_b;
+  float* cn_max = reinterpret_cast<float*>(p);
+
+  launch_split_bf16_norms(x.data_handle(), xs[0], xs[1], nullptr, xn, kSlices,
+                          m, d, stream);
+  launch_split_bf16_norms(c.data_handle(), cs[0], cs[1], nullptr, cn, kSlices,
+                          n, d, stream);
+  const void* xsl[3] = {xs[0], xs[1], nullptr};
+  const void* csl[3] = {cs[0], cs[1], nullptr};
+  launch_fused_l2nn_split(xsl, csl, xn, cn, dmin.data_handle(),
+                          amin.data_handle(), dmin2, m, static_cast<int>(n),
+                          static_cast<int>(d), kSlices, stream);
+  // exact-fp32 verification/repair (needs max(cn) on device: one kMax
+  // row-reduction over cn viewed as a [1, n] matrix)
+  launch_reduce_rows<static_cast<int>(ReduceOpCode::kMax), float>(cn, cn_max,
+                                                                  1, n, stream);
+  launch_l2nn_verify_repair(x.data_handle(), c.data_handle(), xn,
+                            dmin.data_handle(), amin.data_handle(), dmin2,
+                            cn_max, m, static_cast<int>(n),
+                            static_cast<int>(d), stream);
+}
 
 // ---- matrix::select_k -----------------------------------------------------
 inline void select_k(device_matrix_view<const float> in,

@@ -78,11 +78,56 @@ int main() {
     if (std::fabs(h_vals.view()(i, 0) - best) > 1e-3f) bad++;
     (void)second;
   }
+  // fused L2-NN through the mdspan overload (the flagship engine from
+  // pure C++): x [m, 64] fp32 vs c [n2, 64]
+  const std::int64_t d2 = 64, n2 = 128;
+  auto hx2 = make_host_matrix<float>(m, d2);
+  auto hc2 = make_host_matrix<float>(n2, d2);
+  for (std::size_t i = 0; i < hx2.size(); i++) hx2.data_handle()[i] = u(rng);
+  for (std::size_t i = 0; i < hc2.size(); i++) hc2.data_handle()[i] = u(rng);
+  auto dx2 = make_device_matrix<float>(m, d2);
+  auto dc2 = make_device_matrix<float>(n2, d2);
+  copy(dx2.view(), {hx2.data_handle(), dextents<std::int64_t, 2>(m, d2)},
+       nullptr);
+  copy(dc2.view(), {hc2.data_handle(), dextents<std::int64_t, 2>(n2, d2)},
+       nullptr);
+  auto dmin = make_device_vector<float>(m);
+  auto damin = make_device_vector<int>(m);
+  device_uvector<char> l2ws;
+  fused_l2nn({dx2.data_handle(), dextents<std::int64_t, 2>(m, d2)},
+             {dc2.data_handle(), dextents<std::int64_t, 2>(n2, d2)},
+             dmin.view(), damin.view(), l2ws);
+  auto h_dmin = make_host_vector<float>(m);
+  auto h_damin = make_host_vector<int>(m);
+  copy(h_dmin.view(),
+       {static_cast<const float*>(dmin.data_handle()),
+        dextents<std::int64_t, 1>(m)},
+       nullptr);
+  copy(h_damin.view(),
+       {static_cast<const int*>(damin.data_handle()),
+        dextents<std::int64_t, 1>(m)},
+       nullptr);
+  check_hip_(hipDeviceSynchronize(), "sync2");
+  for (std::int64_t i = 0; i < m; i++) {
+    float best = 1e30f;
+    int besti = -1;
+    for (std::int64_t j = 0; j < n2; j++) {
+      float s = 0.f;
+      for (std::int64_t cidx = 0; cidx < d2; cidx++) {
+        const float dd = hx2.view()(i, cidx) - hc2.view()(j, cidx);
+        s += dd * dd;
+      }
+      if (s < best) { best = s; besti = (int)j; }
+    }
+    if (h_damin.view()(i) != besti) bad++;
+    if (std::fabs(h_dmin.view()(i) - best) > 1e-2f) bad++;
+  }
+
   if (bad) {
     std::printf("CONSUMER_FAIL bad=%d\n", bad);
     return 1;
   }
-  std::printf("CONSUMER_OK m=%lld n=%lld k=%lld\n", (long long)m, (long long)n,
-              (long long)k);
+  std::printf("CONSUMER_OK m=%lld n=%lld k=%lld (pairwise+select_k+fused_l2nn)\n",
+              (long long)m, (long long)n, (long long)k);
   return 0;
 }
