@@ -27,6 +27,13 @@ void launch_bitset_count(const unsigned int*, long long, unsigned long long*,
                          hipStream_t);
 // from solver_kernels.hip
 void launch_cholesky_r1_update_f32(float*, float*, int, long long, hipStream_t);
+void launch_lanczos_pre(float*, const float*, const float*, const float*,
+                        double*, long long, hipStream_t);
+void launch_lanczos_sub_alpha(float*, const float*, const double*, float*,
+                              long long, hipStream_t);
+void launch_lanczos_norm2(const float*, double*, long long, hipStream_t);
+void launch_lanczos_normalize(const float*, float*, const double*, float*,
+                              float*, float*, long long, hipStream_t);
 void launch_cholesky_r1_update_f64(double*, double*, int, long long, hipStream_t);
 // from reductions.hip
 template <int OP, typename T>
@@ -685,6 +692,42 @@ torch::Tensor gemm_f32(torch::Tensor a, torch::Tensor b) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reduce_rows", &reduce_rows, "rowwise reduction (op code)");
+  m.def("lanczos_pre_", [](torch::Tensor u, torch::Tensor v_i,
+                           c10::optional<torch::Tensor> v_prev,
+                           c10::optional<torch::Tensor> beta,
+                           torch::Tensor alpha_out) {
+    raft_amd::launch_lanczos_pre(
+        u.data_ptr<float>(), v_i.data_ptr<float>(),
+        v_prev.has_value() ? v_prev->data_ptr<float>() : nullptr,
+        beta.has_value() ? beta->data_ptr<float>() : nullptr,
+        alpha_out.data_ptr<double>(), u.numel(), cur_stream());
+  }, "fused: u -= beta*v_prev; alpha_out = dot(v_i, u)");
+  m.def("lanczos_sub_alpha_", [](torch::Tensor u, torch::Tensor v_i,
+                                 torch::Tensor alpha, torch::Tensor t_diag) {
+    raft_amd::launch_lanczos_sub_alpha(u.data_ptr<float>(),
+                                       v_i.data_ptr<float>(),
+                                       alpha.data_ptr<double>(),
+                                       t_diag.data_ptr<float>(), u.numel(),
+                                       cur_stream());
+  }, "fused: u -= alpha*v_i; t_diag[0] = alpha");
+  m.def("lanczos_norm2_", [](torch::Tensor u, torch::Tensor norm2_out) {
+    raft_amd::launch_lanczos_norm2(u.data_ptr<float>(),
+                                   norm2_out.data_ptr<double>(), u.numel(),
+                                   cur_stream());
+  }, "norm2_out[0] = ||u||^2 (fp64 accumulate)");
+  m.def("lanczos_normalize_", [](torch::Tensor u, torch::Tensor v_next,
+                                 torch::Tensor norm2,
+                                 c10::optional<torch::Tensor> t_up,
+                                 c10::optional<torch::Tensor> t_dn,
+                                 c10::optional<torch::Tensor> beta_out) {
+    raft_amd::launch_lanczos_normalize(
+        u.data_ptr<float>(), v_next.data_ptr<float>(),
+        norm2.data_ptr<double>(),
+        t_up.has_value() ? t_up->data_ptr<float>() : nullptr,
+        t_dn.has_value() ? t_dn->data_ptr<float>() : nullptr,
+        beta_out.has_value() ? beta_out->data_ptr<float>() : nullptr,
+        u.numel(), cur_stream());
+  }, "fused: v_next = u/||u||; t couplings = ||u||");
   m.def("cholesky_r1_update_", [](torch::Tensor l, torch::Tensor x) {
     TORCH_CHECK(l.is_cuda() && l.dim() == 2 && l.size(0) == l.size(1) &&
                 l.is_contiguous() && x.is_contiguous() &&

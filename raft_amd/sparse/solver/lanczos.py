@@ -80,6 +80,19 @@ def lanczos_min_eigenpairs(a, k: int | None = None, config: LanczosConfig | None
         u = u - basis.t() @ (basis @ u)
         return u
 
+    # fused-step scratch (GPU fp32): device-resident alpha/norm2 scalars and
+    # the carried (v_next, beta_last) buffers — the eager loop is ~16 host
+    # dispatches per step; the fused kernels cut it to ~8 (hipGraph capture
+    # regressed on ROCm 7.2, so dispatch count is the lever)
+    fused_ok = device.type == "cuda" and dtype == torch.float32
+    if fused_ok:
+        from raft_amd._ext import require_ext as _re
+        _ext = _re()
+        _alpha = torch.zeros(1, dtype=torch.float64, device=device)
+        _norm2 = torch.zeros(1, dtype=torch.float64, device=device)
+        _v_next_buf = torch.empty(n, dtype=dtype, device=device)
+        _beta_buf = torch.zeros(1, dtype=dtype, device=device)
+
     def _extend(start: int, careful: bool = False):
         """Run the three-term recurrence from index `start` to ncv-1, filling
         t_mat tridiagonally below/right of `start` (lanczos_aux).
@@ -89,6 +102,37 @@ def lanczos_min_eigenpairs(a, k: int | None = None, config: LanczosConfig | None
         careful=True (per-step checks) in that rare case.
         """
         nonlocal v_next, beta_last
+        if fused_ok and not careful:
+            for i in range(start, ncv):
+                u = op(v[i])
+                if not u.is_contiguous():
+                    u = u.contiguous()
+                if i == start and start > 0:
+                    # arrowhead couplings (once per cycle): eager
+                    u = torch.addmv(u, v[:start].t(), t_mat[start, :start],
+                                    alpha=-1.0)
+                    _ext.lanczos_pre_(u, v[i], None, None, _alpha)
+                elif i > start:
+                    _ext.lanczos_pre_(u, v[i], v[i - 1], t_mat[i, i - 1:i],
+                                      _alpha)
+                else:
+                    _ext.lanczos_pre_(u, v[i], None, None, _alpha)
+                _ext.lanczos_sub_alpha_(u, v[i], _alpha, t_mat[i, i:i + 1])
+                basis = v[: i + 1]
+                for _pass in range(2):   # 2-pass CGS, sub fused into addmv
+                    w = basis @ u
+                    u = torch.addmv(u, basis.t(), w, alpha=-1.0)
+                _ext.lanczos_norm2_(u, _norm2)
+                if i + 1 < ncv:
+                    _ext.lanczos_normalize_(u, v[i + 1], _norm2,
+                                            t_mat[i, i + 1:i + 2],
+                                            t_mat[i + 1, i:i + 1], None)
+                else:
+                    _ext.lanczos_normalize_(u, _v_next_buf, _norm2, None,
+                                            None, _beta_buf)
+                    v_next = _v_next_buf
+                    beta_last = _beta_buf.reshape(())
+            return
         for i in range(start, ncv):
             u = op(v[i])
             if i == start and start > 0:
