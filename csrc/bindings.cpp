@@ -14,6 +14,9 @@ namespace raft_amd {
 
 // from rng.hip (device sampling test driver)
 void launch_device_sample_test(const float*, int*, int, uint64_t, hipStream_t);
+// from spmv.hip (sddmm)
+void launch_sddmm(const float*, const float*, const int*, const int*, float*,
+                  long long, long long, hipStream_t);
 // from linewise.hip
 void launch_linewise(const float*, float*, const float*, const float*,
                      long long, long long, bool, int, int, hipStream_t);
@@ -782,6 +785,21 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                                         (uint64_t)seed, cur_stream());
     return out;
   }, "block_random_sample test driver (weighted in-kernel selection)");
+  m.def("sddmm", [](torch::Tensor a, torch::Tensor b, torch::Tensor rows,
+                    torch::Tensor cols) {
+    TORCH_CHECK(a.is_cuda() && a.is_contiguous() && b.is_contiguous() &&
+                a.scalar_type() == torch::kFloat32 &&
+                a.size(1) == b.size(1) &&
+                rows.scalar_type() == torch::kInt32 &&
+                cols.scalar_type() == torch::kInt32);
+    auto vals = torch::empty({rows.numel()}, a.options());
+    raft_amd::launch_sddmm(a.data_ptr<float>(), b.data_ptr<float>(),
+                           rows.contiguous().data_ptr<int>(),
+                           cols.contiguous().data_ptr<int>(),
+                           vals.data_ptr<float>(), rows.numel(), a.size(1),
+                           cur_stream());
+    return vals;
+  }, "sampled dense-dense matmul: vals[e] = dot(a[rows[e]], b[cols[e]])");
   m.def("linewise", [](torch::Tensor x, torch::Tensor v1,
                        c10::optional<torch::Tensor> v2, bool along_rows,
                        int64_t op1, int64_t op2) {

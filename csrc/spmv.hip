@@ -61,3 +61,50 @@ template void launch_csr_spmv<double>(const int*, const int*, const double*, con
                                       double*, long long, long long, hipStream_t);
 
 }  // namespace raft_amd
+
+namespace raft_amd {
+
+// SDDMM: vals[e] = dot(a[rows[e]], b[cols[e]]) — sampled dense-dense matmul
+// (reference: cusparse SDDMM, sddmm.hpp:43). Sub-wave per edge (SW lanes
+// cover d); no [nnz, d] gather temporaries (the composition path
+// materializes 2*nnz*d floats — 100 GB at nnz=1e8, d=128).
+template <int SW>
+__global__ void sddmm_kernel(const float* __restrict__ a,
+                             const float* __restrict__ b,
+                             const int* __restrict__ rows,
+                             const int* __restrict__ cols,
+                             float* __restrict__ vals, long long nnz,
+                             long long d) {
+  const long long edges_per_block = blockDim.x / SW;
+  const int lane = threadIdx.x % SW;
+  long long e = (long long)blockIdx.x * edges_per_block + threadIdx.x / SW;
+  const long long stride = (long long)gridDim.x * edges_per_block;
+  for (; e < nnz; e += stride) {
+    const float* ar = a + (long long)rows[e] * d;
+    const float* br = b + (long long)cols[e] * d;
+    float acc = 0.f;
+    for (long long j = lane; j < d; j += SW) acc += ar[j] * br[j];
+    for (int off = SW >> 1; off > 0; off >>= 1)
+      acc += __shfl_xor(acc, off, RAFT_AMD_WAVE);
+    if (lane == 0) vals[e] = acc;
+  }
+}
+
+void launch_sddmm(const float* a, const float* b, const int* rows,
+                  const int* cols, float* vals, long long nnz, long long d,
+                  hipStream_t stream) {
+  constexpr int BLOCK = 256;
+  auto launch = [&](auto swc) {
+    constexpr int SW = decltype(swc)::value;
+    long long blocks = (nnz * SW + BLOCK - 1) / BLOCK;
+    if (blocks > 1048576) blocks = 1048576;
+    hipLaunchKernelGGL((sddmm_kernel<SW>), dim3((unsigned)blocks), dim3(BLOCK),
+                       0, stream, a, b, rows, cols, vals, nnz, d);
+  };
+  if (d <= 8) launch(std::integral_constant<int, 4>{});
+  else if (d <= 32) launch(std::integral_constant<int, 16>{});
+  else if (d <= 128) launch(std::integral_constant<int, 32>{});
+  else launch(std::integral_constant<int, 64>{});
+}
+
+}  // namespace raft_amd

@@ -49,8 +49,17 @@ def spmm(a, b: torch.Tensor) -> torch.Tensor:
 
 
 def sddmm(a: torch.Tensor, b: torch.Tensor, mask: CSR) -> CSR:
-    """Sampled dense-dense matmul: (A @ B^T) restricted to mask's pattern."""
+    """Sampled dense-dense matmul: (A @ B^T) restricted to mask's pattern.
+
+    GPU: native sub-wave-per-edge kernel (csrc/spmv.hip) — no [nnz, d]
+    gather temporaries (the CPU composition materializes 2*nnz*d floats)."""
     coo = csr_to_coo(mask)
+    if a.is_cuda and a.dtype == torch.float32 and b.dtype == torch.float32:
+        from raft_amd._ext import require_ext
+        vals = require_ext().sddmm(a.contiguous(), b.contiguous(),
+                                   coo.rows.to(torch.int32),
+                                   coo.cols.to(torch.int32))
+        return CSR(mask.indptr, mask.indices, vals, mask.n_rows, mask.n_cols)
     vals = (a[coo.rows.to(torch.int64)] * b[coo.cols.to(torch.int64)]).sum(dim=1)
     return CSR(mask.indptr, mask.indices, vals, mask.n_rows, mask.n_cols)
 

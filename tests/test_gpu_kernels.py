@@ -722,3 +722,26 @@ class TestPhiloxGpu:
         assert abs(p[10] - 1 / 7) < 0.02
         assert abs(p[20] - 2 / 7) < 0.02
         assert abs(p[200] - 4 / 7) < 0.02
+
+
+class TestSddmmGpu:
+    @pytest.mark.parametrize("d", [8, 32, 100, 256])
+    def test_sddmm_vs_dense(self, dev, ext, d):
+        import scipy.sparse as sp
+        from raft_amd.sparse.types import CSR
+        from raft_amd.sparse.linalg import sddmm
+        torch.manual_seed(d)
+        m, n = 500, 400
+        mask_sp = sp.random(m, n, density=0.02, random_state=d, format="csr")
+        mask = CSR(torch.as_tensor(mask_sp.indptr, dtype=torch.int32, device=dev),
+                   torch.as_tensor(mask_sp.indices, dtype=torch.int32, device=dev),
+                   torch.ones(mask_sp.nnz, device=dev), m, n)
+        a = torch.randn(m, d, device=dev)
+        b = torch.randn(n, d, device=dev)
+        out = sddmm(a, b, mask)
+        dense = (a @ b.t())
+        rows = torch.repeat_interleave(
+            torch.arange(m, device=dev),
+            (mask.indptr[1:] - mask.indptr[:-1]).to(torch.int64))
+        ref = dense[rows, mask.indices.to(torch.int64)]
+        torch.testing.assert_close(out.values, ref, rtol=1e-4, atol=1e-4)
