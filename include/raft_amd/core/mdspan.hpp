@@ -19,6 +19,7 @@
 
 #include <cstddef>
 #include <cstdint>
+#include <memory>
 #include <stdexcept>
 #include <type_traits>
 #include <utility>
@@ -402,6 +403,60 @@ template <class T, class IndexType = std::int64_t>
 auto make_host_vector(IndexType n) {
   return host_mdarray<T, dextents<IndexType, 1>>(dextents<IndexType, 1>(n));
 }
+
+// ---------------------------------------------------------------------------
+// mdbuffer — location-polymorphic owning buffer (core/mdbuffer.cuh parity):
+// holds EITHER a host or a device mdarray and serves views in a REQUESTED
+// memory kind, copying lazily on first cross-location request. The C++
+// analog of the reference's std::variant machinery, sized to the two
+// locations raft_amd uses.
+// ---------------------------------------------------------------------------
+template <class T, class Extents, class Layout = layout_right>
+class mdbuffer {
+ public:
+  explicit mdbuffer(const Extents& e, memory_kind where = memory_kind::device)
+      : ext_(e), kind_(where) {
+    if (where == memory_kind::device)
+      dev_ = std::make_unique<device_mdarray<T, Extents, Layout>>(e);
+    else
+      host_ = std::make_unique<host_mdarray<T, Extents, Layout>>(e);
+  }
+
+  memory_kind kind() const noexcept { return kind_; }
+  const Extents& extents() const noexcept { return ext_; }
+
+  // view in the requested location; copies across lazily (both copies stay
+  // alive afterwards — the caller owns coherence, like the reference)
+  mdspan<T, Extents, Layout, device_accessor<T>> device_view(
+      hipStream_t stream = nullptr) {
+    if (!dev_) {
+      dev_ = std::make_unique<device_mdarray<T, Extents, Layout>>(ext_);
+      check_hip_(hipMemcpyAsync(dev_->data_handle(), host_->data_handle(),
+                                dev_->size() * sizeof(T),
+                                hipMemcpyHostToDevice, stream),
+                 "mdbuffer H2D");
+    }
+    return dev_->view();
+  }
+  mdspan<T, Extents, Layout, host_accessor<T>> host_view(
+      hipStream_t stream = nullptr) {
+    if (!host_) {
+      host_ = std::make_unique<host_mdarray<T, Extents, Layout>>(ext_);
+      check_hip_(hipMemcpyAsync(host_->data_handle(), dev_->data_handle(),
+                                host_->size() * sizeof(T),
+                                hipMemcpyDeviceToHost, stream),
+                 "mdbuffer D2H");
+      check_hip_(hipStreamSynchronize(stream), "mdbuffer D2H sync");
+    }
+    return host_->view();
+  }
+
+ private:
+  Extents ext_;
+  memory_kind kind_;
+  std::unique_ptr<device_mdarray<T, Extents, Layout>> dev_;
+  std::unique_ptr<host_mdarray<T, Extents, Layout>> host_;
+};
 
 // host<->device copies for mdarray/mdspan pairs (contiguous layouts)
 template <class T, class E, class L>

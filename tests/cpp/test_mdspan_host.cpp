@@ -46,6 +46,14 @@ int main() {
                                 device_matrix_view<float>>);
   static_assert(device_matrix_view<float>::kind() == memory_kind::device);
 
+  // mdbuffer: host-only round-trip (device paths exercised on GPU)
+  mdbuffer<float, dextents<std::int64_t, 2>> buf(
+      dextents<std::int64_t, 2>(3, 3), memory_kind::host);
+  auto bv = buf.host_view();
+  bv(2, 2) = 9.f;
+  assert(buf.host_view()(2, 2) == 9.f);
+  assert(buf.kind() == memory_kind::host);
+
   std::printf("MDSPAN_HOST_OK\n");
   return 0;
 }
