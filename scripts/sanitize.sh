@@ -15,7 +15,7 @@ echo "== host ASan+UBSan: mdspan/mdarray semantics =="
 mkdir -p build
 /opt/rocm/lib/llvm/bin/clang++ -std=c++17 -g -fsanitize=address,undefined \
   -fno-omit-frame-pointer -I include -D__HIP_PLATFORM_AMD__=1 \
-  -I/opt/rocm/include tests/cpp/test_mdspan_host.cpp -o build/mdspan_host_asan
+  -I/opt/rocm/include tests/cpp/test_mdspan_host.cpp -L/opt/rocm/lib -lamdhip64 -o build/mdspan_host_asan
 ./build/mdspan_host_asan
 
 if python -c 'import torch, sys; sys.exit(0 if torch.cuda.is_available() else 1)' 2>/dev/null; then

@@ -94,7 +94,8 @@ class TestMdspanHeader:
         r = subprocess.run(
             ["/opt/rocm/lib/llvm/bin/clang++", "-std=c++17", "-I", "include",
              "-D__HIP_PLATFORM_AMD__=1", "-I/opt/rocm/include",
-             "tests/cpp/test_mdspan_host.cpp", "-o", str(exe)],
+             "tests/cpp/test_mdspan_host.cpp", "-L/opt/rocm/lib",
+             "-lamdhip64", "-o", str(exe)],
             cwd=ROOT, capture_output=True, text=True, timeout=300)
         assert r.returncode == 0, r.stderr[-2000:]
         r = subprocess.run([str(exe)], capture_output=True, text=True, timeout=60)
