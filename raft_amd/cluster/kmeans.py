@@ -431,8 +431,13 @@ def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify):
                 for _ in range(nslice)]
     cn = torch.empty(k, dtype=torch.float32, device=dev)
     world = comms.get_size()
+    # overlap pays when the halves keep the GPU busy longer than the doubled
+    # per-iteration dispatch (~0.5 ms) it costs — gate on shard size
+    # (RAFT_AMD_KMEANS_OVERLAP=1 forces on, =0 off)
+    ov_env = os.environ.get("RAFT_AMD_KMEANS_OVERLAP", "auto")
     overlap = (world > 1 and x.shape[0] >= 2
-               and os.environ.get("RAFT_AMD_KMEANS_OVERLAP", "1") != "0")
+               and (ov_env == "1" or (ov_env != "0"
+                                      and x.shape[0] >= 4_000_000)))
     if overlap:
         h = x.shape[0] // 2
         halves = [(x[:h], [s[:h] for s in x_slices], xn[:h]),

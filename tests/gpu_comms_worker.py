@@ -53,6 +53,9 @@ def main():
     assert torch.allclose(packed, ref * world, rtol=1e-5, atol=1e-5)
 
     # ---- distributed k-means on device: the bench.py --gpus N inner loop ----
+    # force the split-batch allreduce overlap so the async path is exercised
+    # on multi-GPU nodes regardless of shard size
+    os.environ["RAFT_AMD_KMEANS_OVERLAP"] = "1"
     from raft_amd.cluster.kmeans import kmeans_iterate
     from raft_amd.comms import LoopbackComms
     from raft_amd.random import make_blobs, RngState
