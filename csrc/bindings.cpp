@@ -65,9 +65,11 @@ void launch_csr_spmv(const int*, const int*, const T*, const T*, T*, long long,
 void launch_reduce_rows_by_key(const float*, const int*, float*, float*, long long,
                                long long, long long, int, hipStream_t);
 void launch_reduce_rows_by_key_sorted(const float*, const int*, const int*, float*,
-                                      float*, long long, long long, hipStream_t);
+                                      float*, const float*, float*, long long,
+                                      long long, hipStream_t);
 void launch_split_bf16_norms(const float*, void*, void*, void*, float*, int,
-                             long long, long long, hipStream_t);
+                             long long, long long, hipStream_t,
+                             float* = nullptr);
 void launch_kmeans_update_centroids(const float*, const float*, float*, long long,
                                     long long, hipStream_t);
 void launch_l2nn_verify_repair(const float*, const float*, const float*, float*, int*,
@@ -75,8 +77,8 @@ void launch_l2nn_verify_repair(const float*, const float*, const float*, float*,
                                hipStream_t);
 void launch_kmeans_update_verify(const float*, const int*, const int*, const float*,
                                  const float*, float*, int*, const float*,
-                                 const float*, float*, float*, long long, long long,
-                                 int, hipStream_t);
+                                 const float*, float*, float*, float*, long long,
+                                 long long, int, hipStream_t);
 // from select_k.hip
 long long select_k_workspace_bytes(long long batch);
 void launch_select_k(const float*, float*, int*, void*, long long, long long, int,
@@ -292,25 +294,30 @@ torch::Tensor reduce_rows_by_key_sorted(torch::Tensor x, torch::Tensor perm,
   raft_amd::launch_reduce_rows_by_key_sorted(x.data_ptr<float>(), perm.data_ptr<int>(),
                                              keys_sorted.data_ptr<int>(),
                                              sums.data_ptr<float>(), nullptr,
+                                             nullptr, nullptr,
                                              x.size(0), x.size(1), cur_stream());
   return sums;
 }
 
 void reduce_rows_by_key_sorted_into(torch::Tensor x, torch::Tensor perm,
                                     torch::Tensor keys_sorted, torch::Tensor sums,
-                                    torch::Tensor counts) {
+                                    torch::Tensor counts,
+                                    c10::optional<torch::Tensor> dmin,
+                                    c10::optional<torch::Tensor> inertia_acc) {
   check_f32_2d(x, "x");
   TORCH_CHECK(perm.scalar_type() == torch::kInt32 && keys_sorted.scalar_type() == torch::kInt32);
   TORCH_CHECK(sums.is_contiguous() && counts.is_contiguous());
-  raft_amd::launch_reduce_rows_by_key_sorted(x.data_ptr<float>(), perm.data_ptr<int>(),
-                                             keys_sorted.data_ptr<int>(),
-                                             sums.data_ptr<float>(),
-                                             counts.data_ptr<float>(), x.size(0),
-                                             x.size(1), cur_stream());
+  raft_amd::launch_reduce_rows_by_key_sorted(
+      x.data_ptr<float>(), perm.data_ptr<int>(), keys_sorted.data_ptr<int>(),
+      sums.data_ptr<float>(), counts.data_ptr<float>(),
+      dmin.has_value() ? dmin->data_ptr<float>() : nullptr,
+      inertia_acc.has_value() ? inertia_acc->data_ptr<float>() : nullptr,
+      x.size(0), x.size(1), cur_stream());
 }
 
 void split_bf16_norms(torch::Tensor c, std::vector<torch::Tensor> slices,
-                      torch::Tensor cn) {
+                      torch::Tensor cn,
+                      c10::optional<torch::Tensor> cn_max = c10::nullopt) {
   check_f32_2d(c, "c");
   const int nslice = (int)slices.size();
   TORCH_CHECK(nslice >= 1 && nslice <= 3);
@@ -320,16 +327,18 @@ void split_bf16_norms(torch::Tensor c, std::vector<torch::Tensor> slices,
   void* p0 = slices[0].data_ptr();
   void* p1 = nslice > 1 ? slices[1].data_ptr() : p0;
   void* p2 = nslice > 2 ? slices[2].data_ptr() : p0;
-  raft_amd::launch_split_bf16_norms(c.data_ptr<float>(), p0, p1, p2,
-                                    cn.data_ptr<float>(), nslice, c.size(0),
-                                    c.size(1), cur_stream());
+  raft_amd::launch_split_bf16_norms(
+      c.data_ptr<float>(), p0, p1, p2, cn.data_ptr<float>(), nslice, c.size(0),
+      c.size(1), cur_stream(),
+      cn_max.has_value() ? cn_max->data_ptr<float>() : nullptr);
 }
 
 void kmeans_update_verify(torch::Tensor x, torch::Tensor perm,
                           torch::Tensor keys_sorted, torch::Tensor c,
                           torch::Tensor xn, torch::Tensor dmin, torch::Tensor amin,
                           torch::Tensor dmin2, torch::Tensor cn_max,
-                          torch::Tensor sums, torch::Tensor counts) {
+                          torch::Tensor sums, torch::Tensor counts,
+                          c10::optional<torch::Tensor> inertia_acc) {
   check_f32_2d(x, "x");
   check_f32_2d(c, "c");
   TORCH_CHECK(perm.scalar_type() == torch::kInt32 && keys_sorted.scalar_type() == torch::kInt32);
@@ -338,8 +347,9 @@ void kmeans_update_verify(torch::Tensor x, torch::Tensor perm,
       x.data_ptr<float>(), perm.data_ptr<int>(), keys_sorted.data_ptr<int>(),
       c.data_ptr<float>(), xn.data_ptr<float>(), dmin.data_ptr<float>(),
       amin.data_ptr<int>(), dmin2.data_ptr<float>(), cn_max.data_ptr<float>(),
-      sums.data_ptr<float>(), counts.data_ptr<float>(), x.size(0), x.size(1),
-      (int)c.size(0), cur_stream());
+      sums.data_ptr<float>(), counts.data_ptr<float>(),
+      inertia_acc.has_value() ? inertia_acc->data_ptr<float>() : nullptr,
+      x.size(0), x.size(1), (int)c.size(0), cur_stream());
 }
 
 void kmeans_update_centroids(torch::Tensor sums, torch::Tensor counts,
@@ -767,10 +777,21 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reduce_rows_by_key_sorted", &reduce_rows_by_key_sorted,
         "keyed row accumulation over a key-sorted permutation");
   m.def("reduce_rows_by_key_sorted_into", &reduce_rows_by_key_sorted_into,
+        pybind11::arg("x"), pybind11::arg("perm"), pybind11::arg("keys_sorted"),
+        pybind11::arg("sums"), pybind11::arg("counts"),
+        pybind11::arg("dmin") = pybind11::none(),
+        pybind11::arg("inertia_acc") = pybind11::none(),
         "keyed row accumulation + counts into caller buffers");
   m.def("split_bf16_norms", &split_bf16_norms,
+        pybind11::arg("c"), pybind11::arg("slices"), pybind11::arg("cn"),
+        pybind11::arg("cn_max") = pybind11::none(),
         "fused fp32->bf16 slice split + row sq-norms");
   m.def("kmeans_update_verify", &kmeans_update_verify,
+        pybind11::arg("x"), pybind11::arg("perm"), pybind11::arg("keys_sorted"),
+        pybind11::arg("c"), pybind11::arg("xn"), pybind11::arg("dmin"),
+        pybind11::arg("amin"), pybind11::arg("dmin2"), pybind11::arg("cn_max"),
+        pybind11::arg("sums"), pybind11::arg("counts"),
+        pybind11::arg("inertia_acc") = pybind11::none(),
         "fused centroid-sum accumulation + exact-fp32 verify/refine (one X pass)");
   m.def("kmeans_update_centroids", &kmeans_update_centroids,
         "centroids = counts>0 ? sums/counts : centroids");

@@ -54,6 +54,8 @@ __global__ void reduce_rows_by_key_sorted_kernel(const float* __restrict__ x,
                                                  const int* __restrict__ keys_sorted,
                                                  float* __restrict__ sums,
                                                  float* __restrict__ counts,
+                                                 const float* __restrict__ dmin,
+                                                 float* __restrict__ inertia_acc,
                                                  long long n_rows, long long d,
                                                  long long chunk) {
   const int lane = threadIdx.x % RAFT_AMD_WAVE;
@@ -69,8 +71,10 @@ __global__ void reduce_rows_by_key_sorted_kernel(const float* __restrict__ x,
 
   int cur_key = keys_sorted[start];
   int run_len = 0;
+  double local_inertia = 0.0;
   for (long long i = start; i < stop; i++) {
     const int key = keys_sorted[i];
+    if (inertia_acc && lane == 0) local_inertia += (double)dmin[perm[i]];
     if (key != cur_key) {
       float* sp = sums + (long long)cur_key * d;
       for (int j = 0; j < dreg; j++) {
@@ -95,6 +99,7 @@ __global__ void reduce_rows_by_key_sorted_kernel(const float* __restrict__ x,
     if (col < d) atomicAdd(&sp[col], acc[j]);
   }
   if (counts && lane == 0) atomicAdd(&counts[cur_key], (float)run_len);
+  if (inertia_acc && lane == 0) atomicAdd(inertia_acc, (float)local_inertia);
 }
 
 // Fused centroid-update + exact-fp32 verify/refine in ONE X pass.
@@ -111,8 +116,8 @@ __global__ void kmeans_update_verify_kernel(
     const float* __restrict__ xn, float* __restrict__ dmin,
     int* __restrict__ amin, const float* __restrict__ dmin2,
     const float* __restrict__ cn_max_p, float* __restrict__ sums,
-    float* __restrict__ counts, long long n_rows, long long d, int n_centroids,
-    long long chunk) {
+    float* __restrict__ counts, float* __restrict__ inertia_acc,
+    long long n_rows, long long d, int n_centroids, long long chunk) {
   const int lane = threadIdx.x % RAFT_AMD_WAVE;
   const long long wave_id =
       ((long long)blockIdx.x * blockDim.x + threadIdx.x) / RAFT_AMD_WAVE;
@@ -126,6 +131,7 @@ __global__ void kmeans_update_verify_kernel(
   for (int j = 0; j < MAX_DREG; j++) acc[j] = 0.f;
   int cur_key = -1;
   int run_len = 0;
+  double local_inertia = 0.0;
 
   auto flush = [&]() {
     if (cur_key < 0) return;
@@ -185,6 +191,7 @@ __global__ void kmeans_update_verify_kernel(
         dmin[row] = bestv;
         amin[row] = besti;
       }
+      local_inertia += (double)bestv;
       if (besti == cur_key) {
         #pragma unroll
     for (int j = 0; j < MAX_DREG; j++) acc[j] += xv[j];
@@ -209,19 +216,22 @@ __global__ void kmeans_update_verify_kernel(
       }
       a = wave_reduce_sum(a);
       if (lane == 0) dmin[row] = a;
+      local_inertia += (double)a;
       #pragma unroll
     for (int j = 0; j < MAX_DREG; j++) acc[j] += xv[j];
       run_len++;
     }
   }
   flush();
+  if (inertia_acc && lane == 0) atomicAdd(inertia_acc, (float)local_inertia);
 }
 
 void launch_kmeans_update_verify(const float* x, const int* perm,
                                  const int* keys_sorted, const float* c,
                                  const float* xn, float* dmin, int* amin,
                                  const float* dmin2, const float* cn_max_dev,
-                                 float* sums, float* counts, long long n_rows,
+                                 float* sums, float* counts, float* inertia_acc,
+                                 long long n_rows,
                                  long long d, int n_centroids, hipStream_t stream) {
   const long long n_waves_target = 2048 * 4;
   long long chunk = (n_rows + n_waves_target - 1) / n_waves_target;
@@ -231,11 +241,13 @@ void launch_kmeans_update_verify(const float* x, const int* perm,
   if (d <= 256) {
     hipLaunchKernelGGL((kmeans_update_verify_kernel<4>), dim3(grid), dim3(256), 0,
                        stream, x, perm, keys_sorted, c, xn, dmin, amin, dmin2,
-                       cn_max_dev, sums, counts, n_rows, d, n_centroids, chunk);
+                       cn_max_dev, sums, counts, inertia_acc, n_rows, d,
+                       n_centroids, chunk);
   } else {
     hipLaunchKernelGGL((kmeans_update_verify_kernel<16>), dim3(grid), dim3(256), 0,
                        stream, x, perm, keys_sorted, c, xn, dmin, amin, dmin2,
-                       cn_max_dev, sums, counts, n_rows, d, n_centroids, chunk);
+                       cn_max_dev, sums, counts, inertia_acc, n_rows, d,
+                       n_centroids, chunk);
   }
 }
 
@@ -245,7 +257,9 @@ __global__ void split_bf16_norms_kernel(const float* __restrict__ c,
                                         __bf16* __restrict__ s0,
                                         __bf16* __restrict__ s1,
                                         __bf16* __restrict__ s2,
-                                        float* __restrict__ cn, int nslice,
+                                        float* __restrict__ cn,
+                                        float* __restrict__ cn_max,
+                                        int nslice,
                                         long long n_rows, long long d) {
   __shared__ float lds[256 / RAFT_AMD_WAVE];
   for (long long row = blockIdx.x; row < n_rows; row += gridDim.x) {
@@ -271,6 +285,8 @@ __global__ void split_bf16_norms_kernel(const float* __restrict__ c,
       float s = 0.f;
       for (int w = 0; w < (int)(blockDim.x / RAFT_AMD_WAVE); w++) s += lds[w];
       cn[row] = s;
+      // fused max(cn) (norms are non-negative: int-ordered atomicMax works)
+      if (cn_max) atomicMax(reinterpret_cast<int*>(cn_max), __float_as_int(s));
     }
     __syncthreads();
   }
@@ -278,10 +294,12 @@ __global__ void split_bf16_norms_kernel(const float* __restrict__ c,
 
 void launch_split_bf16_norms(const float* c, void* s0, void* s1, void* s2, float* cn,
                              int nslice, long long n_rows, long long d,
-                             hipStream_t stream) {
+                             hipStream_t stream, float* cn_max) {
+  if (cn_max) (void)hipMemsetAsync(cn_max, 0, sizeof(float), stream);
   int grid = (int)(n_rows < 2048 ? n_rows : 2048);
   hipLaunchKernelGGL(split_bf16_norms_kernel, dim3(grid), dim3(256), 0, stream, c,
-                     (__bf16*)s0, (__bf16*)s1, (__bf16*)s2, cn, nslice, n_rows, d);
+                     (__bf16*)s0, (__bf16*)s1, (__bf16*)s2, cn, cn_max, nslice,
+                     n_rows, d);
 }
 
 // centroid update: c[k] = counts[k] > 0 ? sums[k]/counts[k] : c[k]
@@ -390,8 +408,9 @@ void launch_l2nn_verify_repair(const float* x, const float* c, const float* xn,
 
 void launch_reduce_rows_by_key_sorted(const float* x, const int* perm,
                                       const int* keys_sorted, float* sums,
-                                      float* counts, long long n_rows, long long d,
-                                      hipStream_t stream) {
+                                      float* counts, const float* dmin,
+                                      float* inertia_acc, long long n_rows,
+                                      long long d, hipStream_t stream) {
   // chunk sized so the grid fills the chip (~2048 blocks * 4 waves)
   const long long n_waves_target = 2048 * 4;
   long long chunk = (n_rows + n_waves_target - 1) / n_waves_target;
@@ -400,13 +419,16 @@ void launch_reduce_rows_by_key_sorted(const float* x, const int* perm,
   const int grid = (int)((n_waves * RAFT_AMD_WAVE + 255) / 256);
   if (d <= 256) {
     hipLaunchKernelGGL((reduce_rows_by_key_sorted_kernel<4>), dim3(grid), dim3(256),
-                       0, stream, x, perm, keys_sorted, sums, counts, n_rows, d, chunk);
+                       0, stream, x, perm, keys_sorted, sums, counts, dmin,
+                       inertia_acc, n_rows, d, chunk);
   } else if (d <= 1024) {
     hipLaunchKernelGGL((reduce_rows_by_key_sorted_kernel<16>), dim3(grid), dim3(256),
-                       0, stream, x, perm, keys_sorted, sums, counts, n_rows, d, chunk);
+                       0, stream, x, perm, keys_sorted, sums, counts, dmin,
+                       inertia_acc, n_rows, d, chunk);
   } else {
     hipLaunchKernelGGL((reduce_rows_by_key_sorted_kernel<64>), dim3(grid), dim3(256),
-                       0, stream, x, perm, keys_sorted, sums, counts, n_rows, d, chunk);
+                       0, stream, x, perm, keys_sorted, sums, counts, dmin,
+                       inertia_acc, n_rows, d, chunk);
   }
 }
 

@@ -430,6 +430,7 @@ def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify):
     c_slices = [torch.empty((k, d), dtype=torch.bfloat16, device=dev)
                 for _ in range(nslice)]
     cn = torch.empty(k, dtype=torch.float32, device=dev)
+    cn_max = torch.zeros(1, dtype=torch.float32, device=dev)
     world = comms.get_size()
     # overlap pays when the halves keep the GPU busy longer than the doubled
     # per-iteration dispatch (~0.5 ms) it costs — gate on shard size
@@ -453,19 +454,21 @@ def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify):
         sums = packed[: k * d].view(k, d)
         counts = packed[k * d: k * d + k]
         if verify:
-            # ONE X pass: centroid-sum accumulation + exact-fp32 verify/refine
-            cn_max = cn.max().reshape(1)
+            # ONE X pass: centroid-sum accumulation + exact-fp32
+            # verify/refine + the inertia fold (sum of the FINAL repaired
+            # distances accumulates into packed[-1] in-kernel); cn_max comes
+            # fused out of split_bf16_norms
             ext.kmeans_update_verify(xh, perm.to(torch.int32), keys_sorted,
                                      centroids, xnh, dmin, amin, dmin2, cn_max,
-                                     sums, counts)
+                                     sums, counts, packed[-1:])
         else:
             ext.reduce_rows_by_key_sorted_into(xh, perm.to(torch.int32),
-                                               keys_sorted, sums, counts)
-        packed[-1] = torch.sum(dmin, dtype=torch.float64).float()
+                                               keys_sorted, sums, counts,
+                                               dmin, packed[-1:])
         return packed
 
     for _ in range(n_iters):
-        ext.split_bf16_norms(centroids, c_slices, cn)
+        ext.split_bf16_norms(centroids, c_slices, cn, cn_max)
         if overlap:
             packed_a = _local_update(*halves[0])
             work_a = comms.allreduce_async(packed_a, op=ReduceOp.SUM)
