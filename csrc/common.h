@@ -126,4 +126,104 @@ inline int grid_1d(long long total_threads, int block) {
   return (int)(blocks < cap ? blocks : cap);
 }
 
+// ---------------------------------------------------------------------------
+// integer utilities (reference parity: raft/util/pow2_utils.cuh,
+// fast_int_div.cuh, integer_utils.hpp)
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__host__ __device__ __forceinline__ constexpr T ceildiv(T a, T b) {
+  return (a + b - 1) / b;
+}
+
+template <typename T>
+__host__ __device__ __forceinline__ constexpr T round_up(T a, T b) {
+  return ceildiv(a, b) * b;
+}
+
+// compile-time power-of-two helper: masks/shifts instead of div/mod
+template <long long V>
+struct Pow2 {
+  static_assert((V & (V - 1)) == 0 && V > 0, "Pow2: V must be a power of 2");
+  static constexpr long long value = V;
+  static constexpr long long mask = V - 1;
+  static constexpr int log2 = (V == 1) ? 0 : 1 + Pow2<V / 2>::log2;
+  template <typename T>
+  __host__ __device__ __forceinline__ static constexpr T div(T x) {
+    return x >> log2;
+  }
+  template <typename T>
+  __host__ __device__ __forceinline__ static constexpr T mod(T x) {
+    return x & (T)mask;
+  }
+  template <typename T>
+  __host__ __device__ __forceinline__ static constexpr T round_up_(T x) {
+    return (x + (T)mask) & ~(T)mask;
+  }
+};
+template <>
+struct Pow2<1> {
+  static constexpr long long value = 1, mask = 0;
+  static constexpr int log2 = 0;
+  template <typename T>
+  __host__ __device__ __forceinline__ static constexpr T div(T x) { return x; }
+  template <typename T>
+  __host__ __device__ __forceinline__ static constexpr T mod(T) { return 0; }
+  template <typename T>
+  __host__ __device__ __forceinline__ static constexpr T round_up_(T x) {
+    return x;
+  }
+};
+
+// runtime fast division by an invariant divisor (magic-number method,
+// Granlund-Montgomery): one 32x32 mulhi + shift instead of v_div
+struct FastIntDiv {
+  uint32_t d, magic;
+  int shift;
+  __host__ __device__ explicit FastIntDiv(uint32_t divisor) : d(divisor) {
+    shift = 0;
+    uint32_t t = divisor - 1;
+    while (t >>= 1) shift++;
+    shift += 1;  // ceil(log2(d))
+    if ((divisor & (divisor - 1)) == 0) {
+      magic = 0;  // power of two: pure shift
+      shift = 0;
+      uint32_t v = divisor;
+      while (v >>= 1) shift++;
+    } else {
+      const uint64_t m = ((1ull << (32 + shift)) + divisor - 1) / divisor;
+      magic = (uint32_t)m;
+    }
+  }
+  __host__ __device__ __forceinline__ uint32_t div(uint32_t n) const {
+    if (magic == 0) return n >> shift;
+    return (uint32_t)(((uint64_t)n * magic) >> 32) >> shift;
+  }
+  __host__ __device__ __forceinline__ uint32_t mod(uint32_t n) const {
+    return n - div(n) * d;
+  }
+};
+
+// ---------------------------------------------------------------------------
+// float atomic max/min via int-ordered CAS-free atomics (reference parity:
+// raft/util/device_atomics.cuh; the minmax.cuh encode_traits trick) —
+// NON-NEGATIVE floats only for the int-ordered fast path; general values
+// use the ordered-uint encoding.
+// ---------------------------------------------------------------------------
+
+// classic sign-split trick on PLAIN float storage: non-negative IEEE floats
+// order as signed ints, negative ones reverse-order as unsigned ints
+__device__ __forceinline__ void atomic_max_float(float* p, float v) {
+  if (v >= 0.f)
+    atomicMax(reinterpret_cast<int*>(p), __float_as_int(v));
+  else
+    atomicMin(reinterpret_cast<unsigned int*>(p), __float_as_uint(v));
+}
+__device__ __forceinline__ void atomic_min_float(float* p, float v) {
+  if (v >= 0.f)
+    atomicMin(reinterpret_cast<int*>(p), __float_as_int(v));
+  else
+    atomicMax(reinterpret_cast<unsigned int*>(p), __float_as_uint(v));
+}
+
 }  // namespace raft_amd
