@@ -1,3 +1,3 @@
-from .lap import linear_assignment, LinearAssignmentProblem
+from .lap import linear_assignment, linear_assignment_batched, LinearAssignmentProblem
 
-__all__ = ["linear_assignment", "LinearAssignmentProblem"]
+__all__ = ["linear_assignment", "linear_assignment_batched", "LinearAssignmentProblem"]
