@@ -74,11 +74,11 @@ void launch_kmeans_update_centroids(const float*, const float*, float*, long lon
                                     long long, hipStream_t);
 void launch_l2nn_verify_repair(const float*, const float*, const float*, float*, int*,
                                const float*, const float*, long long, int, int,
-                               hipStream_t);
+                               hipStream_t, float, float);
 void launch_kmeans_update_verify(const float*, const int*, const int*, const float*,
                                  const float*, float*, int*, const float*,
                                  const float*, float*, float*, float*, long long,
-                                 long long, int, hipStream_t);
+                                 long long, int, hipStream_t, float, float);
 // from select_k.hip
 long long select_k_workspace_bytes(long long batch);
 void launch_select_k(const float*, float*, int*, void*, long long, long long, int,
@@ -338,7 +338,8 @@ void kmeans_update_verify(torch::Tensor x, torch::Tensor perm,
                           torch::Tensor xn, torch::Tensor dmin, torch::Tensor amin,
                           torch::Tensor dmin2, torch::Tensor cn_max,
                           torch::Tensor sums, torch::Tensor counts,
-                          c10::optional<torch::Tensor> inertia_acc) {
+                          c10::optional<torch::Tensor> inertia_acc,
+                          double lead, double tail) {
   check_f32_2d(x, "x");
   check_f32_2d(c, "c");
   TORCH_CHECK(perm.scalar_type() == torch::kInt32 && keys_sorted.scalar_type() == torch::kInt32);
@@ -349,7 +350,8 @@ void kmeans_update_verify(torch::Tensor x, torch::Tensor perm,
       amin.data_ptr<int>(), dmin2.data_ptr<float>(), cn_max.data_ptr<float>(),
       sums.data_ptr<float>(), counts.data_ptr<float>(),
       inertia_acc.has_value() ? inertia_acc->data_ptr<float>() : nullptr,
-      x.size(0), x.size(1), (int)c.size(0), cur_stream());
+      x.size(0), x.size(1), (int)c.size(0), cur_stream(), (float)lead,
+      (float)tail);
 }
 
 void kmeans_update_centroids(torch::Tensor sums, torch::Tensor counts,
@@ -575,7 +577,7 @@ torch::Tensor pairwise_l2_mfma(std::vector<torch::Tensor> x_slices,
 
 void l2nn_verify_repair(torch::Tensor x, torch::Tensor c, torch::Tensor xn,
                         torch::Tensor dmin, torch::Tensor amin, torch::Tensor dmin2,
-                        torch::Tensor cn_max) {
+                        torch::Tensor cn_max, double lead, double tail) {
   check_f32_2d(x, "x");
   check_f32_2d(c, "c");
   TORCH_CHECK(cn_max.scalar_type() == torch::kFloat32 && cn_max.numel() >= 1);
@@ -583,7 +585,8 @@ void l2nn_verify_repair(torch::Tensor x, torch::Tensor c, torch::Tensor xn,
                                       xn.data_ptr<float>(), dmin.data_ptr<float>(),
                                       amin.data_ptr<int>(), dmin2.data_ptr<float>(),
                                       cn_max.data_ptr<float>(), x.size(0),
-                                      (int)c.size(0), (int)x.size(1), cur_stream());
+                                      (int)c.size(0), (int)x.size(1), cur_stream(),
+                                      (float)lead, (float)tail);
 }
 
 void pairwise_l2_filter(std::vector<torch::Tensor> x_slices,
@@ -792,6 +795,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("amin"), pybind11::arg("dmin2"), pybind11::arg("cn_max"),
         pybind11::arg("sums"), pybind11::arg("counts"),
         pybind11::arg("inertia_acc") = pybind11::none(),
+        pybind11::arg("lead") = 0x1p-13, pybind11::arg("tail") = 0x1p-18,
         "fused centroid-sum accumulation + exact-fp32 verify/refine (one X pass)");
   m.def("kmeans_update_centroids", &kmeans_update_centroids,
         "centroids = counts>0 ? sums/counts : centroids");
@@ -896,6 +900,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("yn"), pybind11::arg("out") = pybind11::none(),
         pybind11::arg("sqrt_out") = false);
   m.def("l2nn_verify_repair", &l2nn_verify_repair,
+        pybind11::arg("x"), pybind11::arg("c"), pybind11::arg("xn"),
+        pybind11::arg("dmin"), pybind11::arg("amin"), pybind11::arg("dmin2"),
+        pybind11::arg("cn_max"),
+        pybind11::arg("lead") = 0x1p-13, pybind11::arg("tail") = 0x1p-18,
         "exact-fp32 verification/repair of split-bf16 L2-NN results");
   m.def("fused_l2nn_split", &fused_l2nn_split,
         "fused split-bf16 MFMA L2-NN (distance + argmin, no materialization)");

@@ -117,7 +117,8 @@ __global__ void kmeans_update_verify_kernel(
     int* __restrict__ amin, const float* __restrict__ dmin2,
     const float* __restrict__ cn_max_p, float* __restrict__ sums,
     float* __restrict__ counts, float* __restrict__ inertia_acc,
-    long long n_rows, long long d, int n_centroids, long long chunk) {
+    long long n_rows, long long d, int n_centroids, long long chunk,
+    float lead, float tail) {
   const int lane = threadIdx.x % RAFT_AMD_WAVE;
   const long long wave_id =
       ((long long)blockIdx.x * blockDim.x + threadIdx.x) / RAFT_AMD_WAVE;
@@ -167,8 +168,8 @@ __global__ void kmeans_update_verify_kernel(
     }
     const float xnr = xn[row];
     const float margin = dmin2[row] - dmin[row];
-    const float bound = 2.f * (exp2f(-13.f) * sqrtf(fmaxf(xnr * cn_max, 0.f)) +
-                               exp2f(-18.f) * (xnr + cn_max));
+    const float bound = 2.f * (lead * sqrtf(fmaxf(xnr * cn_max, 0.f)) +
+                               tail * (xnr + cn_max));
     if (margin < bound) {
       // exact rescan over all centroids (rare: near-ties only)
       float bestv = INFINITY;
@@ -232,7 +233,8 @@ void launch_kmeans_update_verify(const float* x, const int* perm,
                                  const float* dmin2, const float* cn_max_dev,
                                  float* sums, float* counts, float* inertia_acc,
                                  long long n_rows,
-                                 long long d, int n_centroids, hipStream_t stream) {
+                                 long long d, int n_centroids, hipStream_t stream,
+                                 float lead, float tail) {
   const long long n_waves_target = 2048 * 4;
   long long chunk = (n_rows + n_waves_target - 1) / n_waves_target;
   if (chunk < 8) chunk = 8;
@@ -242,12 +244,12 @@ void launch_kmeans_update_verify(const float* x, const int* perm,
     hipLaunchKernelGGL((kmeans_update_verify_kernel<4>), dim3(grid), dim3(256), 0,
                        stream, x, perm, keys_sorted, c, xn, dmin, amin, dmin2,
                        cn_max_dev, sums, counts, inertia_acc, n_rows, d,
-                       n_centroids, chunk);
+                       n_centroids, chunk, lead, tail);
   } else {
     hipLaunchKernelGGL((kmeans_update_verify_kernel<16>), dim3(grid), dim3(256), 0,
                        stream, x, perm, keys_sorted, c, xn, dmin, amin, dmin2,
                        cn_max_dev, sums, counts, inertia_acc, n_rows, d,
-                       n_centroids, chunk);
+                       n_centroids, chunk, lead, tail);
   }
 }
 
@@ -329,9 +331,17 @@ void launch_kmeans_update_centroids(const float* sums, const float* counts, floa
 // is inside the provable split-emulation error bound rescan ALL centroids
 // in exact fp32 and repair the argmin. Device-side only — no host sync.
 //
-// Error bound (NSLICE=2, per-element bf16 rounding |r_i| <= 2^-9|x_i| twice):
-// |score error| <= 2*(||x1||*||c1|| + ...) <= ~2^-14.5*sqrt(xn*cn); we use
-// E = 2^-13*sqrt(xn*cn_max) + 2^-18*(xn+cn_max) (3x headroom + accumulate).
+// Error bound, parametrized by (lead, tail) per split mode; the dot-product
+// emulation error satisfies |Δdot| <= lead*sqrt(xn*cn) + tail*(xn+cn):
+//   bf16x2v (2 slices, 3 products; |x - x0 - x1| <= 2^-18|x|):
+//     dropped p11 + rounding <= ~2^-14.5*sqrt(xn*cn)
+//     -> lead = 2^-13, tail = 2^-18 (~3x headroom + fp32 accumulation)
+//   bf16x1v (1 slice, 1 product; |x - bf16(x)| <= 2^-9|x| on both sides):
+//     |x.c - a.b| <= |δa.c| + |a.δb| <= 2.01*2^-9*||x||*||c||  (Cauchy-Schwarz)
+//     plus MFMA fp32 accumulation (~d*2^-24, <= 2^-16 at d=256)
+//     -> lead = 2^-7, tail = 2^-12 (~2.3x headroom)
+// The kernel bound is 2*E because best may be over- and runner-up
+// under-estimated by E each; margin >= 2E proves the emulated argmin exact.
 __global__ void l2nn_verify_repair_kernel(const float* __restrict__ x,
                                           const float* __restrict__ c,
                                           const float* __restrict__ xn,
@@ -339,7 +349,8 @@ __global__ void l2nn_verify_repair_kernel(const float* __restrict__ x,
                                           int* __restrict__ amin,
                                           const float* __restrict__ dmin2,
                                           const float* __restrict__ cn_max_p,
-                                          long long m, int n, int d) {
+                                          long long m, int n, int d,
+                                          float lead, float tail) {
   const float cn_max = *cn_max_p;
   const long long waves_per_block = blockDim.x / RAFT_AMD_WAVE;
   const int lane = threadIdx.x % RAFT_AMD_WAVE;
@@ -349,8 +360,8 @@ __global__ void l2nn_verify_repair_kernel(const float* __restrict__ x,
     const float* rp = x + row * d;
     const float xnr = xn[row];
     const float margin = dmin2[row] - dmin[row];
-    const float bound = 2.f * (exp2f(-13.f) * sqrtf(fmaxf(xnr * cn_max, 0.f)) +
-                               exp2f(-18.f) * (xnr + cn_max));
+    const float bound = 2.f * (lead * sqrtf(fmaxf(xnr * cn_max, 0.f)) +
+                               tail * (xnr + cn_max));
     int a = amin[row];
     if (margin < bound) {
       // full exact rescan (rare: near-ties only)
@@ -398,12 +409,12 @@ __global__ void l2nn_verify_repair_kernel(const float* __restrict__ x,
 void launch_l2nn_verify_repair(const float* x, const float* c, const float* xn,
                                float* dmin, int* amin, const float* dmin2,
                                const float* cn_max_dev, long long m, int n, int d,
-                               hipStream_t stream) {
+                               hipStream_t stream, float lead, float tail) {
   // uncapped: one wave per row (row-serial grid-stride was latency-bound)
   long long blocks = (m * RAFT_AMD_WAVE + 255) / 256;
   int grid = (int)(blocks > 2147483647ll ? 2147483647ll : blocks);
   hipLaunchKernelGGL(l2nn_verify_repair_kernel, dim3(grid), dim3(256), 0, stream,
-                     x, c, xn, dmin, amin, dmin2, cn_max_dev, m, n, d);
+                     x, c, xn, dmin, amin, dmin2, cn_max_dev, m, n, d, lead, tail);
 }
 
 void launch_reduce_rows_by_key_sorted(const float* x, const int* perm,

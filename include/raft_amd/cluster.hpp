@@ -22,13 +22,17 @@ void launch_reduce_rows_by_key_sorted(const float* x, const int* perm,
 // fused one-X-pass centroid-sum accumulation + exact-fp32 verify/refine of
 // rows whose split-error margin is inconclusive (the bf16x2v engine's
 // update step)
+// lead/tail: margin-bound constants of the split mode that produced
+// dmin/dmin2 (defaults = the bf16x2v engine; bf16x1v uses 2^-7 / 2^-12 —
+// see csrc/kmeans.hip l2nn_verify_repair_kernel for the derivation)
 void launch_kmeans_update_verify(const float* x, const int* perm,
                                  const int* keys_sorted, const float* c,
                                  const float* xn, float* dmin, int* amin,
                                  const float* dmin2, const float* cn_max_dev,
                                  float* sums, float* counts, float* inertia_acc,
                                  long long n_rows, long long d, int n_centroids,
-                                 hipStream_t s);
+                                 hipStream_t s, float lead = 0x1p-13f,
+                                 float tail = 0x1p-18f);
 // split fp32 -> nslice bf16 slices + squared row norms in one pass
 // cn_max (optional): fused max(cn) into one device float (zeroed first)
 void launch_split_bf16_norms(const float* c, void* s0, void* s1, void* s2, float* cn,

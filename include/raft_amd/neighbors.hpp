@@ -23,10 +23,13 @@ bool fused_l2nn_w8_supported(int nslice, int n, int d);
 void launch_fused_l2nn_w8(const void** xsl, const void** csl, const float* xn,
                           const float* cn, float* dmin, int* amin, float* dmin2,
                           long long m, int n, int d, int nslice, hipStream_t s);
-// exact-fp32 rescan/repair of rows whose margin is inside the split bound
+// exact-fp32 rescan/repair of rows whose margin is inside the split bound;
+// lead/tail select the bound for the split mode that produced dmin/dmin2
+// (defaults = bf16x2v; bf16x1v passes 2^-7 / 2^-12)
 void launch_l2nn_verify_repair(const float* x, const float* c, const float* xn,
                                float* dmin, int* amin, const float* dmin2,
                                const float* cn_max_dev, long long m, int n, int d,
-                               hipStream_t s);
+                               hipStream_t s, float lead = 0x1p-13f,
+                               float tail = 0x1p-18f);
 
 }  // namespace raft_amd

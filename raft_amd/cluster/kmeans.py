@@ -358,13 +358,15 @@ def kmeans_iterate(x: torch.Tensor, centroids: torch.Tensor, n_iters: int,
     (fused L2-NN over all rows) + update (keyed reduction, counts, ONE packed
     allreduce, centroid recompute). Returns (centroids, inertia).
     """
-    from raft_amd.neighbors.fused_l2nn import (_MODE_NSLICE, _VERIFY_MODES,
+    from raft_amd.neighbors.fused_l2nn import (_DEFAULT_BOUND, _MODE_BOUND,
+                                               _MODE_NSLICE, _VERIFY_MODES,
                                                fused_l2nn_presplit,
                                                split_bf16_slices)
 
     comms = comms or LoopbackComms()
     k, d = centroids.shape
     inertia = float("inf")
+    bound = _MODE_BOUND.get(fp32_mode, _DEFAULT_BOUND)
     # X is iteration-invariant: pre-split the bf16 slices and row norms ONCE
     # (the same caching the reference does for row norms in its kmeans)
     use_fused = (x.is_cuda and x.dtype == torch.float32
@@ -376,14 +378,15 @@ def kmeans_iterate(x: torch.Tensor, centroids: torch.Tensor, n_iters: int,
     if use_fused and k % 128 == 0:
         return _fast_iterate(x, x_slices, xn, centroids.contiguous().clone(),
                              n_iters, comms, _MODE_NSLICE[fp32_mode],
-                             fp32_mode in _VERIFY_MODES)
+                             fp32_mode in _VERIFY_MODES, bound)
 
     inertia_t = None
     for it in range(n_iters):
         if use_fused:
             vx = x if fp32_mode in _VERIFY_MODES else None
             dmin, labels = fused_l2nn_presplit(x_slices, xn, centroids,
-                                               int32_labels=True, verify_x=vx)
+                                               int32_labels=True, verify_x=vx,
+                                               bound=bound)
         else:
             dmin, labels = fused_l2nn(x, centroids, fp32_mode=fp32_mode,
                                       chunk_rows=chunk_rows)
@@ -408,7 +411,8 @@ def kmeans_iterate(x: torch.Tensor, centroids: torch.Tensor, n_iters: int,
     return centroids, inertia
 
 
-def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify):
+def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify,
+                  bound=(2.0 ** -13, 2.0 ** -18)):
     """Minimal-dispatch EM loop: every per-iteration stage is ONE kernel
     (centroid split+norms, fused assignment, verify/repair, keyed reduction
     with counts, centroid update) + the rocPRIM label sort + ONE packed
@@ -460,7 +464,8 @@ def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify):
             # fused out of split_bf16_norms
             ext.kmeans_update_verify(xh, perm.to(torch.int32), keys_sorted,
                                      centroids, xnh, dmin, amin, dmin2, cn_max,
-                                     sums, counts, packed[-1:])
+                                     sums, counts, packed[-1:],
+                                     lead=bound[0], tail=bound[1])
         else:
             ext.reduce_rows_by_key_sorted_into(xh, perm.to(torch.int32),
                                                keys_sorted, sums, counts,

@@ -23,11 +23,20 @@ from raft_amd._ext import ext_or_none, require_ext
 from raft_amd.utils import on_gpu, row_chunks
 from raft_amd.linalg.gemm import gemm_fp32_emulated
 
-_MODE_NSLICE = {"bf16x2": 2, "bf16x3": 3, "bf16x2v": 2, "auto": 2, "fused": 3}
+_MODE_NSLICE = {"bf16x2": 2, "bf16x3": 3, "bf16x2v": 2, "bf16x1v": 1,
+                "auto": 2, "fused": 3}
 #: modes that run the exact-fp32 verification/repair pass (provably exact
 #: argmin: rows inside the split-error margin are rescanned in fp32, and the
 #: chosen distance is recomputed exactly for every row)
-_VERIFY_MODES = {"bf16x2v", "auto"}
+_VERIFY_MODES = {"bf16x2v", "bf16x1v", "auto"}
+#: (lead, tail) margin-bound constants per verified mode — the provable
+#: |Δdot| <= lead*sqrt(xn*cn) + tail*(xn+cn) envelope of the split emulation
+#: (derivation: csrc/kmeans.hip l2nn_verify_repair_kernel). bf16x1v does a
+#: single MFMA product (1/3 the matrix work of bf16x2v) against a 2^6-wider
+#: bound; correctness is identical — only the rescan fraction grows on data
+#: with near-tied neighbors.
+_MODE_BOUND = {"bf16x1v": (2.0 ** -7, 2.0 ** -12)}
+_DEFAULT_BOUND = (2.0 ** -13, 2.0 ** -18)
 
 
 def split_bf16_slices(t: torch.Tensor, nslice: int):
@@ -55,7 +64,8 @@ def _pad_cols(y: torch.Tensor, yn: torch.Tensor, mult: int = 128):
 
 def fused_l2nn_presplit(x_slices, xn: torch.Tensor, y: torch.Tensor,
                         sqrt: bool = False, int32_labels: bool = False,
-                        verify_x: torch.Tensor | None = None):
+                        verify_x: torch.Tensor | None = None,
+                        bound: tuple[float, float] = _DEFAULT_BOUND):
     """Fused kernel entry with precomputed X slices (k-means hot loop).
 
     verify_x: the original fp32 matrix — enables the exact-fp32
@@ -75,7 +85,8 @@ def fused_l2nn_presplit(x_slices, xn: torch.Tensor, y: torch.Tensor,
     if verify_x is not None:
         cn_max = yn.max().reshape(1)   # device scalar: no host sync
         ext.l2nn_verify_repair(verify_x.contiguous(), y.contiguous(),
-                               xn.contiguous(), dmin, amin, dmin2, cn_max)
+                               xn.contiguous(), dmin, amin, dmin2, cn_max,
+                               lead=bound[0], tail=bound[1])
     if sqrt:
         dmin = dmin.clamp_min(0).sqrt()
     return dmin, (amin if int32_labels else amin.to(torch.int64))
@@ -109,7 +120,9 @@ def fused_l2nn(x: torch.Tensor, y: torch.Tensor, sqrt: bool = False,
                 xs = split_bf16_slices(x, nslice)
                 xn = (x * x).sum(dim=1)
                 vx = x if fp32_mode in _VERIFY_MODES else None
-                return fused_l2nn_presplit(xs, xn, y, sqrt=sqrt, verify_x=vx)
+                return fused_l2nn_presplit(
+                    xs, xn, y, sqrt=sqrt, verify_x=vx,
+                    bound=_MODE_BOUND.get(fp32_mode, _DEFAULT_BOUND))
             return _chunked_gpu(x, y, sqrt, fp32_mode, chunk_rows)
 
     # CPU oracle

@@ -187,16 +187,19 @@ class TestFusedL2NNMfma:
         rd, ra = ref.min(dim=1)
         assert (amin == ra).float().mean() > 0.99
 
-    def test_verified_mode_exact_on_adversarial_ties(self, dev, ext):
-        """bf16x2v must produce the exact fp32 argmin even with near-duplicate
-        centroids (margins far inside the split-emulation error)."""
+    @pytest.mark.parametrize("vmode", ["bf16x2v", "bf16x1v"])
+    def test_verified_mode_exact_on_adversarial_ties(self, dev, ext, vmode):
+        """Verified modes must produce the exact fp32 argmin even with
+        near-duplicate centroids (margins far inside the split-emulation
+        error). bf16x1v's emulation is 2^6 coarser — same guarantee, its
+        wider bound just routes more rows through the exact rescan."""
         torch.manual_seed(5)
         x = torch.randn(8192, 128, device=dev) * 10
         y = torch.randn(256, 128, device=dev) * 10
         # make half the centroids near-duplicates of the other half
         y[128:] = y[:128] + torch.randn(128, 128, device=dev) * 1e-4
         from raft_amd.neighbors.fused_l2nn import fused_l2nn
-        _, av = fused_l2nn(x, y, fp32_mode="bf16x2v")
+        _, av = fused_l2nn(x, y, fp32_mode=vmode)
         # the meaningful guarantee: the chosen centroid's TRUE distance is
         # fp32-indistinguishable from the optimum for every row (near-tie
         # winners may differ between any two fp32 summation orders, including
@@ -210,12 +213,13 @@ class TestFusedL2NNMfma:
                      .squeeze(1) > 1.0)
         assert (av[margin_ok] == ref.argmin(dim=1)[margin_ok]).all()
 
-    def test_verified_mode_exact_distances(self, dev, ext):
+    @pytest.mark.parametrize("vmode", ["bf16x2v", "bf16x1v"])
+    def test_verified_mode_exact_distances(self, dev, ext, vmode):
         from raft_amd.neighbors.fused_l2nn import fused_l2nn
         torch.manual_seed(6)
         x = torch.randn(4096, 256, device=dev)
         y = torch.randn(512, 256, device=dev)
-        dv, av = fused_l2nn(x, y, fp32_mode="bf16x2v")
+        dv, av = fused_l2nn(x, y, fp32_mode=vmode)
         ref = torch.cdist(x.double(), y.double()) ** 2
         rd = ref[torch.arange(4096, device=dev), av]
         # distances are exact-fp32 recomputed: error = fp32 rounding only
@@ -250,7 +254,8 @@ class TestFusedL2NNMfma:
                                    state=RngState(seed=7), device=dev)
         c = centers + 0.3
         _, an = fused_l2nn(x, c, fp32_mode="native")
-        for mode, bar in (("bf16x2", 0.9999), ("bf16x3", 0.9999), ("bf16x2v", 1.0)):
+        for mode, bar in (("bf16x2", 0.9999), ("bf16x3", 0.9999),
+                          ("bf16x2v", 1.0), ("bf16x1v", 1.0)):
             _, am = fused_l2nn(x, c, fp32_mode=mode)
             agree = float((am == an).float().mean())
             assert agree >= bar, (mode, agree)
@@ -416,14 +421,16 @@ class TestKMeansGpu:
         d = torch.cdist(centers, model.centroids)
         assert d.min(dim=1).values.max() < 1.0
 
-    def test_fast_iterate_matches_cpu(self, dev, ext):
-        """the minimal-dispatch fused EM loop (k % 128 == 0) vs CPU oracle."""
+    @pytest.mark.parametrize("vmode", ["bf16x2v", "bf16x1v"])
+    def test_fast_iterate_matches_cpu(self, dev, ext, vmode):
+        """the minimal-dispatch fused EM loop (k % 128 == 0) vs CPU oracle —
+        both verified engines (2-slice tight bound, 1-slice wide bound)."""
         from raft_amd.cluster.kmeans import kmeans_iterate
         from raft_amd.random import make_blobs, RngState
         x, _, centers = make_blobs(20000, 64, n_clusters=128, cluster_std=0.4,
                                    state=RngState(seed=9), device=dev)
         c0 = centers + 0.2
-        cg, ig = kmeans_iterate(x, c0.clone(), 3, fp32_mode="bf16x2v")
+        cg, ig = kmeans_iterate(x, c0.clone(), 3, fp32_mode=vmode)
         cc, ic = kmeans_iterate(x.cpu(), c0.cpu().clone(), 3)
         torch.testing.assert_close(cg.cpu(), cc, rtol=1e-3, atol=1e-3)
         assert abs(ig - ic) / ic < 1e-3
