@@ -35,9 +35,13 @@ def parse_args():
     p.add_argument("--rows", type=int, default=10_000_000, help="total rows (global)")
     p.add_argument("--dim", type=int, default=256)
     p.add_argument("--k", type=int, default=1024)
-    p.add_argument("--fp32-mode", default="bf16x2v",
-                   choices=["bf16x2v", "bf16x3", "bf16x2", "native"],
-                   help="fp32 GEMM engine for the assignment step")
+    p.add_argument("--fp32-mode", default="auto",
+                   choices=["auto", "bf16x2v", "bf16x1v", "bf16x3", "bf16x2",
+                            "native"],
+                   help="fp32 GEMM engine for the assignment step; auto = "
+                        "adaptive verified engine (starts 1-product bf16x1v, "
+                        "widens to bf16x2v if >2%% of rows hit the exact "
+                        "rescan) — exact fp32 argmin either way")
     p.add_argument("--chunk-rows", type=int, default=262144)
     p.add_argument("--seed", type=int, default=42)
     p.add_argument("--check", action="store_true",

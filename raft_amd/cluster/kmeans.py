@@ -378,7 +378,8 @@ def kmeans_iterate(x: torch.Tensor, centroids: torch.Tensor, n_iters: int,
     if use_fused and k % 128 == 0:
         return _fast_iterate(x, x_slices, xn, centroids.contiguous().clone(),
                              n_iters, comms, _MODE_NSLICE[fp32_mode],
-                             fp32_mode in _VERIFY_MODES, bound)
+                             fp32_mode in _VERIFY_MODES, bound,
+                             adaptive=(fp32_mode == "auto"))
 
     inertia_t = None
     for it in range(n_iters):
@@ -411,8 +412,13 @@ def kmeans_iterate(x: torch.Tensor, centroids: torch.Tensor, n_iters: int,
     return centroids, inertia
 
 
+#: engine chosen by the last adaptive (fp32_mode="auto") _fast_iterate run
+#: (1 = stayed on bf16x1v, 2 = widened to bf16x2v) — test observability
+_LAST_ADAPTIVE_NSLICE = None
+
+
 def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify,
-                  bound=(2.0 ** -13, 2.0 ** -18)):
+                  bound=(2.0 ** -13, 2.0 ** -18), adaptive=False):
     """Minimal-dispatch EM loop: every per-iteration stage is ONE kernel
     (centroid split+norms, fused assignment, verify/repair, keyed reduction
     with counts, centroid update) + the rocPRIM label sort + ONE packed
@@ -425,12 +431,23 @@ def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify,
     the collective hides under compute. allreduce(A)+allreduce(B) equals
     allreduce(A+B) by linearity, so the update is bitwise the same modulo
     fp32 summation order. Opt out with RAFT_AMD_KMEANS_OVERLAP=0.
+
+    adaptive=True (fp32_mode="auto"): start on the 1-product bf16x1v engine
+    (1/3 the MFMA work), measure the provable-rescan fraction of the FIRST
+    iteration (rows whose emulated margin falls inside the wide 2^-7 bound),
+    and widen to the 2-slice engine if >2% of rows would rescan. Both
+    engines produce the exact fp32 argmin, so the switch is purely a
+    performance guard — per-rank decisions need not agree.
     """
     import os
     from raft_amd._ext import require_ext
+    from raft_amd.neighbors.fused_l2nn import _MODE_BOUND
     ext = require_ext()
     k, d = centroids.shape
     dev = x.device
+    cur_nslice = 1 if (adaptive and verify) else nslice
+    cur_bound = _MODE_BOUND["bf16x1v"] if (adaptive and verify) else bound
+    frac_t = [None]  # first-iteration provable-rescan fraction (device)
     c_slices = [torch.empty((k, d), dtype=torch.bfloat16, device=dev)
                 for _ in range(nslice)]
     cn = torch.empty(k, dtype=torch.float32, device=dev)
@@ -451,8 +468,15 @@ def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify,
 
     def _local_update(xh, xh_slices, xnh):
         """assignment + keyed reduction for one row range -> packed buffer"""
-        dmin, amin, dmin2 = ext.fused_l2nn_split(list(xh_slices), c_slices,
-                                                 xnh, cn)
+        dmin, amin, dmin2 = ext.fused_l2nn_split(list(xh_slices[:cur_nslice]),
+                                                 c_slices[:cur_nslice], xnh, cn)
+        if adaptive and frac_t[0] is None:
+            # pre-repair margin stats (dmin is overwritten by the verify
+            # kernel below): fraction of rows the exact rescan will touch
+            lead_, tail_ = cur_bound
+            b = 2.0 * (lead_ * (xnh * cn_max).clamp_min(0).sqrt()
+                       + tail_ * (xnh + cn_max))
+            frac_t[0] = ((dmin2 - dmin) < b).float().mean()
         keys_sorted, perm = torch.sort(amin)
         packed = torch.zeros(k * d + k + 1, dtype=torch.float32, device=dev)
         sums = packed[: k * d].view(k, d)
@@ -465,15 +489,15 @@ def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify,
             ext.kmeans_update_verify(xh, perm.to(torch.int32), keys_sorted,
                                      centroids, xnh, dmin, amin, dmin2, cn_max,
                                      sums, counts, packed[-1:],
-                                     lead=bound[0], tail=bound[1])
+                                     lead=cur_bound[0], tail=cur_bound[1])
         else:
             ext.reduce_rows_by_key_sorted_into(xh, perm.to(torch.int32),
                                                keys_sorted, sums, counts,
                                                dmin, packed[-1:])
         return packed
 
-    for _ in range(n_iters):
-        ext.split_bf16_norms(centroids, c_slices, cn, cn_max)
+    for it in range(n_iters):
+        ext.split_bf16_norms(centroids, c_slices[:cur_nslice], cn, cn_max)
         if overlap:
             packed_a = _local_update(*halves[0])
             work_a = comms.allreduce_async(packed_a, op=ReduceOp.SUM)
@@ -493,6 +517,14 @@ def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify,
         counts = packed[k * d: k * d + k]
         ext.kmeans_update_centroids(sums, counts, centroids)
         inertia_t = packed[-1]
+        if adaptive and it == 0 and frac_t[0] is not None:
+            # one host sync, once: widen to the tight-bound 2-slice engine
+            # if the wide bf16x1v bound would rescan >2% of rows each iter
+            if nslice >= 2 and float(frac_t[0].item()) > 0.02:
+                cur_nslice, cur_bound = nslice, bound
+            adaptive = False
+            global _LAST_ADAPTIVE_NSLICE
+            _LAST_ADAPTIVE_NSLICE = cur_nslice
     inertia = float(inertia_t.item()) if inertia_t is not None else float("inf")
     return centroids, inertia
 

@@ -421,10 +421,11 @@ class TestKMeansGpu:
         d = torch.cdist(centers, model.centroids)
         assert d.min(dim=1).values.max() < 1.0
 
-    @pytest.mark.parametrize("vmode", ["bf16x2v", "bf16x1v"])
+    @pytest.mark.parametrize("vmode", ["bf16x2v", "bf16x1v", "auto"])
     def test_fast_iterate_matches_cpu(self, dev, ext, vmode):
         """the minimal-dispatch fused EM loop (k % 128 == 0) vs CPU oracle —
-        both verified engines (2-slice tight bound, 1-slice wide bound)."""
+        both verified engines (2-slice tight bound, 1-slice wide bound) plus
+        the adaptive auto engine (starts 1-product, widens on rescan rate)."""
         from raft_amd.cluster.kmeans import kmeans_iterate
         from raft_amd.random import make_blobs, RngState
         x, _, centers = make_blobs(20000, 64, n_clusters=128, cluster_std=0.4,
@@ -434,6 +435,30 @@ class TestKMeansGpu:
         cc, ic = kmeans_iterate(x.cpu(), c0.cpu().clone(), 3)
         torch.testing.assert_close(cg.cpu(), cc, rtol=1e-3, atol=1e-3)
         assert abs(ig - ic) / ic < 1e-3
+
+    def test_adaptive_auto_engine_decisions(self, dev, ext):
+        """fp32_mode="auto" stays on the 1-product engine on well-separated
+        data and widens to the 2-slice engine on near-tie-heavy data; both
+        paths stay exact (every verified engine rescans uncertain rows in
+        exact fp32, so results agree across engines)."""
+        import raft_amd.cluster.kmeans as km
+        from raft_amd.random import make_blobs, RngState
+        # clear margins -> stays on bf16x1v
+        x, _, centers = make_blobs(20000, 64, n_clusters=128, cluster_std=0.3,
+                                   state=RngState(seed=12), device=dev)
+        km.kmeans_iterate(x, (centers + 0.2).clone(), 2, fp32_mode="auto")
+        assert km._LAST_ADAPTIVE_NSLICE == 1
+        # near-duplicate centroids -> margins collapse -> widens to 2 slices
+        torch.manual_seed(11)
+        xd = torch.randn(20000, 64, device=dev) * 5
+        c0 = torch.randn(128, 64, device=dev) * 5
+        c0[64:] = c0[:64] + 1e-5
+        ca, ia = km.kmeans_iterate(xd, c0.clone(), 3, fp32_mode="auto")
+        assert km._LAST_ADAPTIVE_NSLICE == 2
+        # the widened run matches the always-2-slice verified engine
+        cv, iv = km.kmeans_iterate(xd, c0.clone(), 3, fp32_mode="bf16x2v")
+        assert abs(ia - iv) / iv < 1e-5
+        torch.testing.assert_close(ca, cv, rtol=1e-4, atol=1e-4)
 
     def test_split_norms_and_update_kernels(self, dev, ext):
         torch.manual_seed(10)
