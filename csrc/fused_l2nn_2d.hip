@@ -345,20 +345,25 @@ void launch_l2nn_combine(const float* pd, const float* pd2, const int* pi,
                      n_groups);
 }
 
-static int l2nn_2d_gt() {
+static int l2nn_2d_gt_env() {
   static const int gt = [] {
     const char* e = getenv("RAFT_AMD_L2NN_GT");
-    const int v = e ? atoi(e) : 2;
-    return (v == 1 || v == 2 || v == 4) ? v : 2;
+    const int v = e ? atoi(e) : 0;
+    return (v == 1 || v == 2 || v == 4 || v == 8) ? v : 0;
   }();
   return gt;
 }
 
-// effective col-tiles-per-block: drop to the largest of {gt, 2, 1} dividing
-// the col-tile count, so odd tile counts (e.g. n=384) keep the 2D engine's
-// L2 win instead of falling back to v1 (round-1 NOTES item 15)
-static int l2nn_2d_gt_for(int n) {
-  int gt = l2nn_2d_gt();
+// effective col-tiles-per-block: nslice-aware default (measured at 10M x 256
+// k=1024: 1-slice GT8 9.55 / GT4 9.69 / GT2 11.30 ms — the epilogue is a
+// bigger share of the cheaper 1-product k-loop, so amortizing it over more
+// tiles wins; 2-slice GT2 remains the optimum). Halve until it divides the
+// col-tile
+// count, so odd tile counts (e.g. n=384) keep the 2D engine's L2 win
+// instead of falling back to v1 (round-1 NOTES item 15).
+static int l2nn_2d_gt_for(int n, int nslice) {
+  int gt = l2nn_2d_gt_env();
+  if (gt == 0) gt = nslice == 1 ? 8 : 2;
   const int tiles = n / 128;
   while (gt > 1 && tiles % gt != 0) gt >>= 1;
   return gt;
@@ -385,7 +390,7 @@ void launch_fused_l2nn_2d(const void** xsl, const void** csl, const float* xn,
                           float* dmin, int* amin, float* dmin2,
                           long long m, int n, int d, int nslice,
                           hipStream_t stream) {
-  const int gt = l2nn_2d_gt_for(n);
+  const int gt = l2nn_2d_gt_for(n, nslice);
   const int n_row_tiles = (int)((m + 127) / 128);
   const int n_groups = n / 128 / gt;
   const int grid = n_row_tiles * n_groups;
@@ -444,7 +449,8 @@ void launch_fused_l2nn_2d(const void** xsl, const void** csl, const float* xn,
   if (nslice == 1) {
     if (gt == 1) L2NN2D_LAUNCH(1, 1, false);
     else if (gt == 2) L2NN2D_LAUNCH(1, 2, false);
-    else L2NN2D_LAUNCH(1, 4, false);
+    else if (gt == 4) L2NN2D_LAUNCH(1, 4, false);
+    else L2NN2D_LAUNCH(1, 8, false);
   } else if (nslice == 2) {
     if (gt == 1) { if (ph) L2NN2D_LAUNCH(2, 1, true); else L2NN2D_LAUNCH(2, 1, false); }
     else if (gt == 2) { if (ph) L2NN2D_LAUNCH(2, 2, true); else L2NN2D_LAUNCH(2, 2, false); }
