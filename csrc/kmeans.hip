@@ -109,7 +109,10 @@ __global__ void reduce_rows_by_key_sorted_kernel(const float* __restrict__ x,
 // inside the split-error margin rescan all centroids exactly (rare); a
 // rescan that CHANGES the assignment routes that row's contribution to the
 // new cluster via direct atomics (and fixes amin/dmin).
-template <int MAX_DREG>
+// VEC4: lane-contiguous column map col(j) = (j/4)*256 + lane*4 + j%4 so each
+// lane's 4-column chunk loads as ONE dwordx4 (wave covers 1024 B/chunk);
+// requires d % 4 == 0. Otherwise the classic col(j) = j*64 + lane scalar map.
+template <int MAX_DREG, bool VEC4>
 __global__ void kmeans_update_verify_kernel(
     const float* __restrict__ x, const int* __restrict__ perm,
     const int* __restrict__ keys_sorted, const float* __restrict__ c,
@@ -134,12 +137,43 @@ __global__ void kmeans_update_verify_kernel(
   int run_len = 0;
   double local_inertia = 0.0;
 
+  auto colmap = [&](int j) -> long long {
+    if constexpr (VEC4)
+      return (long long)(j >> 2) * (RAFT_AMD_WAVE * 4) + lane * 4 + (j & 3);
+    else
+      return (long long)j * RAFT_AMD_WAVE + lane;
+  };
+  auto load_row = [&](const float* __restrict__ rp, float* dst) {
+    if constexpr (VEC4) {
+      #pragma unroll
+      for (int q = 0; q < (MAX_DREG + 3) / 4; q++) {
+        const long long base = (long long)q * (RAFT_AMD_WAVE * 4) + lane * 4;
+        if (base + 3 < d) {
+          const float4 t = *reinterpret_cast<const float4*>(rp + base);
+          dst[q * 4 + 0] = t.x; dst[q * 4 + 1] = t.y;
+          dst[q * 4 + 2] = t.z; dst[q * 4 + 3] = t.w;
+        } else {
+          for (int jj = 0; jj < 4 && q * 4 + jj < MAX_DREG; jj++) {
+            const long long col = base + jj;
+            dst[q * 4 + jj] = col < d ? rp[col] : 0.f;
+          }
+        }
+      }
+    } else {
+      #pragma unroll
+      for (int j = 0; j < MAX_DREG; j++) {
+        const long long col = colmap(j);
+        dst[j] = col < d ? rp[col] : 0.f;
+      }
+    }
+  };
+
   auto flush = [&]() {
     if (cur_key < 0) return;
     float* sp = sums + (long long)cur_key * d;
     #pragma unroll
     for (int j = 0; j < MAX_DREG; j++) {
-      const long long col = (long long)j * RAFT_AMD_WAVE + lane;
+      const long long col = colmap(j);
       if (col < d) atomicAdd(&sp[col], acc[j]);
       acc[j] = 0.f;
     }
@@ -153,19 +187,10 @@ __global__ void kmeans_update_verify_kernel(
     if (key != cur_key) {
       flush();
       cur_key = key;
-      const float* cp = c + (long long)key * d;
-      #pragma unroll
-    for (int j = 0; j < MAX_DREG; j++) {
-        const long long col = (long long)j * RAFT_AMD_WAVE + lane;
-        creg[j] = col < d ? cp[col] : 0.f;
-      }
+      load_row(c + (long long)key * d, creg);
     }
     const float* rp = x + row * d;
-    #pragma unroll
-    for (int j = 0; j < MAX_DREG; j++) {
-      const long long col = (long long)j * RAFT_AMD_WAVE + lane;
-      xv[j] = col < d ? rp[col] : 0.f;
-    }
+    load_row(rp, xv);
     const float xnr = xn[row];
     const float margin = dmin2[row] - dmin[row];
     const float bound = 2.f * (lead * sqrtf(fmaxf(xnr * cn_max, 0.f)) +
@@ -179,7 +204,7 @@ __global__ void kmeans_update_verify_kernel(
         float a = 0.f;
         #pragma unroll
     for (int j = 0; j < MAX_DREG; j++) {
-          const long long col = (long long)j * RAFT_AMD_WAVE + lane;
+          const long long col = colmap(j);
           if (col < d) {
             const float diff = xv[j] - cp[col];
             a += diff * diff;
@@ -202,7 +227,7 @@ __global__ void kmeans_update_verify_kernel(
         float* sp = sums + (long long)besti * d;
         #pragma unroll
     for (int j = 0; j < MAX_DREG; j++) {
-          const long long col = (long long)j * RAFT_AMD_WAVE + lane;
+          const long long col = colmap(j);
           if (col < d) atomicAdd(&sp[col], xv[j]);
         }
         if (lane == 0) atomicAdd(&counts[besti], 1.f);
@@ -240,17 +265,18 @@ void launch_kmeans_update_verify(const float* x, const int* perm,
   if (chunk < 8) chunk = 8;
   const long long n_waves = (n_rows + chunk - 1) / chunk;
   const int grid = (int)((n_waves * RAFT_AMD_WAVE + 255) / 256);
+#define KMUV_LAUNCH(MD, V4)                                                   \
+  hipLaunchKernelGGL((kmeans_update_verify_kernel<MD, V4>), dim3(grid),       \
+                     dim3(256), 0, stream, x, perm, keys_sorted, c, xn, dmin, \
+                     amin, dmin2, cn_max_dev, sums, counts, inertia_acc,      \
+                     n_rows, d, n_centroids, chunk, lead, tail)
+  const bool v4 = (d % 4 == 0);
   if (d <= 256) {
-    hipLaunchKernelGGL((kmeans_update_verify_kernel<4>), dim3(grid), dim3(256), 0,
-                       stream, x, perm, keys_sorted, c, xn, dmin, amin, dmin2,
-                       cn_max_dev, sums, counts, inertia_acc, n_rows, d,
-                       n_centroids, chunk, lead, tail);
+    if (v4) KMUV_LAUNCH(4, true); else KMUV_LAUNCH(4, false);
   } else {
-    hipLaunchKernelGGL((kmeans_update_verify_kernel<16>), dim3(grid), dim3(256), 0,
-                       stream, x, perm, keys_sorted, c, xn, dmin, amin, dmin2,
-                       cn_max_dev, sums, counts, inertia_acc, n_rows, d,
-                       n_centroids, chunk, lead, tail);
+    if (v4) KMUV_LAUNCH(16, true); else KMUV_LAUNCH(16, false);
   }
+#undef KMUV_LAUNCH
 }
 
 // fused centroid prep: split fp32 centroids into bf16 slices + row sq-norms
