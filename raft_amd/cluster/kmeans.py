@@ -341,7 +341,10 @@ def kmeans_iter_state(x: torch.Tensor, fp32_mode: str = "auto"):
     if not (x.is_cuda and x.dtype == torch.float32 and fp32_mode in _MODE_NSLICE
             and x.shape[1] % 64 == 0):
         return None
-    nslice = _MODE_NSLICE[fp32_mode]
+    # "auto" starts on the 1-product engine: materialize only slice 0 here
+    # (halves the split memory and the one-time split pass); the adaptive
+    # loop appends the residual slice lazily if it widens to 2-slice
+    nslice = 1 if fp32_mode == "auto" else _MODE_NSLICE[fp32_mode]
     ext = require_ext()
     slices = [torch.empty_like(x, dtype=torch.bfloat16) for _ in range(nslice)]
     xn = torch.empty(x.shape[0], dtype=torch.float32, device=x.device)
@@ -380,6 +383,14 @@ def kmeans_iterate(x: torch.Tensor, centroids: torch.Tensor, n_iters: int,
                              n_iters, comms, _MODE_NSLICE[fp32_mode],
                              fp32_mode in _VERIFY_MODES, bound,
                              adaptive=(fp32_mode == "auto"))
+
+    if use_fused and len(x_slices) < _MODE_NSLICE[fp32_mode]:
+        # the generic loop below has no adaptive widen: materialize the
+        # full split upfront ("auto" state carries only slice 0 initially)
+        from raft_amd._ext import require_ext
+        s_new = torch.empty_like(x_slices[0])
+        require_ext().split_bf16_norms(x.contiguous(), [x_slices[0], s_new], xn)
+        x_slices.append(s_new)
 
     inertia_t = None
     for it in range(n_iters):
@@ -521,6 +532,17 @@ def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify,
             # one host sync, once: widen to the tight-bound 2-slice engine
             # if the wide bf16x1v bound would rescan >2% of rows each iter
             if nslice >= 2 and float(frac_t[0].item()) > 0.02:
+                while len(x_slices) < nslice:
+                    # lazily materialize the residual slice (auto state
+                    # starts 1-slice); recomputes slice 0/xn with identical
+                    # values in the same fused pass
+                    s_new = torch.empty_like(x_slices[0])
+                    ext.split_bf16_norms(x.contiguous(),
+                                         [x_slices[0], s_new], xn)
+                    x_slices.append(s_new)
+                if overlap:
+                    halves = [(x[:h], [s[:h] for s in x_slices], xn[:h]),
+                              (x[h:], [s[h:] for s in x_slices], xn[h:])]
                 cur_nslice, cur_bound = nslice, bound
             adaptive = False
             global _LAST_ADAPTIVE_NSLICE
