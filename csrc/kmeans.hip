@@ -270,7 +270,9 @@ void launch_kmeans_update_verify(const float* x, const int* perm,
                      dim3(256), 0, stream, x, perm, keys_sorted, c, xn, dmin, \
                      amin, dmin2, cn_max_dev, sums, counts, inertia_acc,      \
                      n_rows, d, n_centroids, chunk, lead, tail)
-  const bool v4 = (d % 4 == 0);
+  // VEC4 keeps only d/4 lanes loading (4 cols per lane): require d >= 256
+  // so all 64 lanes stay active
+  const bool v4 = (d % 4 == 0) && d >= 256;
   if (d <= 256) {
     if (v4) KMUV_LAUNCH(4, true); else KMUV_LAUNCH(4, false);
   } else {
