@@ -9,10 +9,12 @@
 
 #include "cluster.hpp"
 #include "core/mdspan.hpp"
+#include "core/sparse_types.hpp"
 #include "distance.hpp"
 #include "matrix.hpp"
 #include "neighbors.hpp"
 #include "reductions.hpp"
+#include "sparse.hpp"
 
 namespace raft_amd {
 
@@ -123,6 +125,31 @@ inline void row_argmin(device_matrix_view<const float> in,
     throw std::invalid_argument("row_argmin: output extent mismatch");
   launch_row_argmin(in.data_handle(), out.data_handle(), in.extent(0),
                     in.extent(1), stream);
+}
+
+// ---- sparse ---------------------------------------------------------------
+// y = A x over a CSR view (reference parity: raft sparse/linalg/spmv.cuh);
+// the sub-wave-per-row kernel picks 2..64 lanes/row from mean nnz/row
+template <typename T>
+inline void spmv(device_csr_view<T> a, device_vector_view<const T> x,
+                 device_vector_view<T> y, hipStream_t stream = nullptr) {
+  if (x.extent(0) != a.n_cols || y.extent(0) != a.n_rows)
+    throw std::invalid_argument("spmv: extents mismatch");
+  launch_csr_spmv<T>(a.indptr, a.indices, a.values, x.data_handle(),
+                     y.data_handle(), a.n_rows, a.nnz, stream);
+}
+
+// SDDMM: out_vals[e] = dot(a[rows[e]], b[cols[e]]) for a COO sampling
+// pattern (reference parity: raft sparse/linalg/sddmm.cuh)
+inline void sddmm(device_matrix_view<const float> a,
+                  device_matrix_view<const float> b,
+                  device_coo_view<float> pattern,
+                  device_vector_view<float> out_vals,
+                  hipStream_t stream = nullptr) {
+  if (a.extent(1) != b.extent(1) || out_vals.extent(0) != pattern.nnz)
+    throw std::invalid_argument("sddmm: extents mismatch");
+  launch_sddmm(a.data_handle(), b.data_handle(), pattern.rows, pattern.cols,
+               out_vals.data_handle(), pattern.nnz, a.extent(1), stream);
 }
 
 }  // namespace raft_amd
