@@ -35,6 +35,8 @@ void launch_lanczos_pre(float*, const float*, const float*, const float*,
 void launch_lanczos_sub_alpha(float*, const float*, const double*, float*,
                               long long, hipStream_t);
 void launch_lanczos_norm2(const float*, double*, long long, hipStream_t);
+void sgemv_rowmajor(const float*, const float*, float*, long long, long long,
+                    bool, float, float, void*);
 void launch_lanczos_normalize(const float*, float*, const double*, float*,
                               float*, float*, long long, hipStream_t);
 void launch_cholesky_r1_update_f64(double*, double*, int, long long, hipStream_t);
@@ -744,6 +746,70 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         beta_out.has_value() ? beta_out->data_ptr<float>() : nullptr,
         u.numel(), cur_stream());
   }, "fused: v_next = u/||u||; t couplings = ||u||");
+  m.def("lanczos_cycle_", [](torch::Tensor indptr, torch::Tensor indices,
+                             torch::Tensor values, torch::Tensor v,
+                             torch::Tensor t_mat, torch::Tensor u,
+                             torch::Tensor w, torch::Tensor v_next,
+                             torch::Tensor beta_out, torch::Tensor alpha_scal,
+                             torch::Tensor norm_scal, int64_t start,
+                             int64_t ncv) {
+    // Whole ncv-step extension cycle driven from C++: the per-step python
+    // dispatch (~1 ms/step measured at 10M rows) disappears — one host call
+    // per restart cycle. Mirrors sparse/solver/lanczos.py _extend exactly:
+    // SpMV, (arrowhead | beta-recurrence) + alpha dot, alpha subtract,
+    // 2-pass CGS reorth (rocBLAS sgemv pairs), norm, normalize.
+    TORCH_CHECK(v.is_cuda() && v.is_contiguous() &&
+                v.scalar_type() == torch::kFloat32);
+    TORCH_CHECK(t_mat.is_contiguous() && t_mat.size(0) == ncv &&
+                t_mat.size(1) == ncv);
+    TORCH_CHECK(indptr.scalar_type() == torch::kInt32 &&
+                indices.scalar_type() == torch::kInt32 &&
+                values.scalar_type() == torch::kFloat32);
+    TORCH_CHECK(u.numel() == v.size(1) && v_next.numel() == v.size(1) &&
+                w.numel() >= ncv);
+    const long long n = v.size(1);
+    const long long nnz = indices.numel();
+    float* vp = v.data_ptr<float>();
+    float* tp = t_mat.data_ptr<float>();
+    float* up = u.data_ptr<float>();
+    float* wp = w.data_ptr<float>();
+    double* ap = alpha_scal.data_ptr<double>();
+    double* np2 = norm_scal.data_ptr<double>();
+    hipStream_t s = cur_stream();
+    for (long long i = start; i < ncv; i++) {
+      raft_amd::launch_csr_spmv<float>(
+          indptr.data_ptr<int>(), indices.data_ptr<int>(),
+          values.data_ptr<float>(), vp + i * n, up, n, nnz, s);
+      if (i == start && start > 0) {
+        // arrowhead couplings: u -= V[:start]^T t_mat[start, :start]
+        raft_amd::sgemv_rowmajor(vp, tp + start * ncv, up, start, n,
+                                 /*trans=*/true, -1.f, 1.f, s);
+        raft_amd::launch_lanczos_pre(up, vp + i * n, nullptr, nullptr, ap, n, s);
+      } else if (i > start) {
+        raft_amd::launch_lanczos_pre(up, vp + i * n, vp + (i - 1) * n,
+                                     tp + i * ncv + (i - 1), ap, n, s);
+      } else {
+        raft_amd::launch_lanczos_pre(up, vp + i * n, nullptr, nullptr, ap, n, s);
+      }
+      raft_amd::launch_lanczos_sub_alpha(up, vp + i * n, ap, tp + i * ncv + i,
+                                         n, s);
+      for (int pass = 0; pass < 2; pass++) {
+        raft_amd::sgemv_rowmajor(vp, up, wp, i + 1, n, false, 1.f, 0.f, s);
+        raft_amd::sgemv_rowmajor(vp, wp, up, i + 1, n, true, -1.f, 1.f, s);
+      }
+      raft_amd::launch_lanczos_norm2(up, np2, n, s);
+      if (i + 1 < ncv) {
+        raft_amd::launch_lanczos_normalize(up, vp + (i + 1) * n, np2,
+                                           tp + i * ncv + (i + 1),
+                                           tp + (i + 1) * ncv + i, nullptr, n,
+                                           s);
+      } else {
+        raft_amd::launch_lanczos_normalize(up, v_next.data_ptr<float>(), np2,
+                                           nullptr, nullptr,
+                                           beta_out.data_ptr<float>(), n, s);
+      }
+    }
+  }, "full Lanczos extension cycle (SpMV + CGS reorth + fused steps) in C++");
   m.def("cholesky_r1_update_", [](torch::Tensor l, torch::Tensor x) {
     TORCH_CHECK(l.is_cuda() && l.dim() == 2 && l.size(0) == l.size(1) &&
                 l.is_contiguous() && x.is_contiguous() &&

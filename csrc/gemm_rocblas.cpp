@@ -68,6 +68,21 @@ void gemm_bf16_f32_nt_rowmajor(const void* a, const void* b, float* c, long long
       rocblas_datatype_f32_r, rocblas_gemm_algo_standard, 0, 0));
 }
 
+// fp32 SGEMV over a row-major basis V [rows, n] (== column-major [n, rows]):
+//   trans=false: y[rows] = alpha * V x[n]   + beta * y   (projection)
+//   trans=true:  y[n]    = alpha * V^T x[rows] + beta * y  (reconstruction)
+// The two calls form one classical Gram-Schmidt pass (Lanczos reorth).
+void sgemv_rowmajor(const float* V, const float* x, float* y, long long rows,
+                    long long n, bool trans, float alpha, float beta,
+                    void* stream) {
+  rocblas_handle h = get_handle();
+  ROCBLAS_CHECK(rocblas_set_stream(h, (hipStream_t)stream));
+  ROCBLAS_CHECK(rocblas_sgemv(
+      h, trans ? rocblas_operation_none : rocblas_operation_transpose,
+      (rocblas_int)n, (rocblas_int)rows, &alpha, V, (rocblas_int)n, x, 1,
+      &beta, y, 1));
+}
+
 // fp32 SGEMM (row-major) — the native fp32 vector-ALU path for comparison.
 void gemm_f32_rowmajor(const float* a, const float* b, float* c, long long m,
                        long long n, long long k, float beta, void* stream) {
