@@ -85,6 +85,17 @@ def main():
     print(f"[rank {rank}] distributed kmeans OK "
           f"(inertia {inertia_d:.6e} vs single {inertia_s:.6e})", flush=True)
 
+    # the adaptive auto engine under real comms: per-rank engine decisions
+    # are independent but every verified engine is exact, so the distributed
+    # result must still match the single-process run
+    ca, inertia_a = kmeans_iterate(x_local, c0.clone(), 3, comms=c,
+                                   fp32_mode="auto")
+    torch.cuda.synchronize()
+    rel_a = abs(inertia_a - inertia_s) / max(abs(inertia_s), 1e-30)
+    assert rel_a < 1e-4, (inertia_a, inertia_s)
+    assert torch.allclose(ca, cs, rtol=1e-4, atol=1e-4)
+    print(f"[rank {rank}] distributed kmeans (auto engine) OK", flush=True)
+
     import torch.distributed as dist
     dist.barrier()
     dist.destroy_process_group()

@@ -217,6 +217,28 @@ class TestLanczosGpu:
         ref = torch.linalg.eigvalsh(dense)[:4]
         torch.testing.assert_close(w.to(ref.dtype), ref, atol=1e-3, rtol=1e-3)
 
+    def test_cpp_cycle_matches_per_step_path(self, dev):
+        """plain-CSR input drives the one-call C++ extension cycle
+        (ext.lanczos_cycle_); a LinearOperator wrapping the SAME matrix
+        takes the per-step fused path — eigenvalues must agree."""
+        from raft_amd.sparse import CSR
+        from raft_amd.sparse.solver import eigsh
+        from raft_amd.sparse.solver.linear_operator import LinearOperator
+        from raft_amd.sparse.linalg import spmv
+        torch.manual_seed(5)
+        n = 600
+        dense = torch.randn(n, n, device=dev)
+        dense = (dense + dense.T) / 2
+        dense = dense * (torch.rand(n, n, device=dev) < 0.08)
+        dense = (dense + dense.T) / 2
+        dense += torch.diag(torch.rand(n, device=dev))
+        a = CSR.from_dense(dense)
+        w_cycle, _ = eigsh(a, k=5, maxiter=150)
+        op = LinearOperator((n, n), lambda x: spmv(a, x), device=a.device,
+                            dtype=a.values.dtype)
+        w_step, _ = eigsh(op, k=5, maxiter=150)
+        torch.testing.assert_close(w_cycle, w_step, atol=1e-4, rtol=1e-4)
+
 
 class TestMstGpu:
     def test_total_weight_vs_scipy(self, dev):
