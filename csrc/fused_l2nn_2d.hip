@@ -109,31 +109,22 @@ __global__ void fused_l2nn_2d_kernel(const __bf16* __restrict__ x0,
                               wr, wc, lane);
     }
 
-    // lane-local running top-2 across the group's tiles (disjoint columns).
-    // cn[col] depends only on fc -> hoisted to 4 loads per tile (was 64 per
-    // lane: a measured VALU/vmem-issue hot spot, profiles/pmc_l2nn_x1v_gt8
-    // .txt shows 8.7 VALU/MFMA); the second-best update is one v_med3_f32
-    // (new best2 = middle of {s, best, best2} given best <= best2).
+    // lane-local running top-2 across the group's tiles (disjoint columns)
     const int col_base = (int)col0 + wc * 64;
-    float cn_r[4];
-    int col_r[4];
-#pragma unroll
-    for (int fc = 0; fc < 4; fc++) {
-      col_r[fc] = col_base + fc * 16 + (lane & 15);
-      cn_r[fc] = cn[col_r[fc]];
-    }
 #pragma unroll
     for (int fr = 0; fr < 4; fr++) {
 #pragma unroll
       for (int reg = 0; reg < 4; reg++) {
 #pragma unroll
         for (int fc = 0; fc < 4; fc++) {
-          const float s = cn_r[fc] - 2.f * acc[fr][fc][reg];
-          best2[fr][reg] = __builtin_amdgcn_fmed3f(s, best[fr][reg],
-                                                   best2[fr][reg]);
+          const int col = col_base + fc * 16 + (lane & 15);
+          const float s = cn[col] - 2.f * acc[fr][fc][reg];
           if (s < best[fr][reg]) {
+            best2[fr][reg] = best[fr][reg];
             best[fr][reg] = s;
-            bidx[fr][reg] = col_r[fc];
+            bidx[fr][reg] = col;
+          } else if (s < best2[fr][reg]) {
+            best2[fr][reg] = s;
           }
         }
       }
