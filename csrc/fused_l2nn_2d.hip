@@ -390,7 +390,14 @@ void launch_fused_l2nn_2d(const void** xsl, const void** csl, const float* xn,
                           float* dmin, int* amin, float* dmin2,
                           long long m, int n, int d, int nslice,
                           hipStream_t stream) {
-  const int gt = l2nn_2d_gt_for(n, nslice);
+  static const bool bk32_env = [] {
+    const char* e = getenv("RAFT_AMD_L2NN_BK32");
+    return e && e[0] == '1';
+  }();
+  int gt = l2nn_2d_gt_for(n, nslice);
+  // the BK32 variant is only instantiated up to GT=4: clamp BEFORE n_groups
+  // so grid coverage matches the kernel's per-block tile count
+  if (bk32_env && gt > 4) gt = 4;
   const int n_row_tiles = (int)((m + 127) / 128);
   const int n_groups = n / 128 / gt;
   const int grid = n_row_tiles * n_groups;
@@ -419,11 +426,7 @@ void launch_fused_l2nn_2d(const void** xsl, const void** csl, const float* xn,
                          dim3(256), lds, stream, x0, x1, x2, c0, c1, c2, cn,   \
                          pd, pd2, pi, m, n, d, n_groups);                      \
   } while (0)
-  static const bool bk32 = [] {
-    const char* e = getenv("RAFT_AMD_L2NN_BK32");
-    return e && e[0] == '1';
-  }();
-  if (bk32 && (nslice == 1 || nslice == 2) && d % 32 == 0) {
+  if (bk32_env && (nslice == 1 || nslice == 2) && d % 32 == 0) {
     const size_t lds32 = (size_t)nslice * 2 * 4096 * sizeof(__bf16);
 #define L2NN2D_BK32(NS, G)                                                     \
   hipLaunchKernelGGL((fused_l2nn_2d_bk32_kernel<NS, G>), dim3(grid),           \
