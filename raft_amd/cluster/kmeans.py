@@ -379,6 +379,15 @@ def kmeans_iterate(x: torch.Tensor, centroids: torch.Tensor, n_iters: int,
             state = kmeans_iter_state(x, fp32_mode)
         x_slices, xn = state
     if use_fused and k % 128 == 0:
+        if fp32_mode != "auto" and len(x_slices) < _MODE_NSLICE[fp32_mode]:
+            # a 1-slice "auto" state reused with an explicit 2-slice mode:
+            # materialize the residual slice (the adaptive path does this
+            # lazily itself)
+            from raft_amd._ext import require_ext
+            s_new = torch.empty_like(x_slices[0])
+            require_ext().split_bf16_norms(x.contiguous(),
+                                           [x_slices[0], s_new], xn)
+            x_slices.append(s_new)
         return _fast_iterate(x, x_slices, xn, centroids.contiguous().clone(),
                              n_iters, comms, _MODE_NSLICE[fp32_mode],
                              fp32_mode in _VERIFY_MODES, bound,
