@@ -89,7 +89,9 @@ __global__ void fused_l2nn_2d_kernel(const __bf16* __restrict__ x0,
       bidx[a][b] = 0;
     }
 
-#pragma unroll
+  // NOT unrolled: the k-loop body is large; unrolling it GT times (8 at the
+  // 1-slice default) bloats I-cache for zero register benefit
+#pragma unroll 1
   for (int g = 0; g < GT; g++) {
     const long long col0 = ((long long)grp * GT + g) * 128;
     f32x4 acc[4][4];
@@ -224,7 +226,9 @@ __global__ void fused_l2nn_2d_bk32_kernel(const __bf16* __restrict__ x0,
       bidx[a][b] = 0;
     }
 
-#pragma unroll
+  // NOT unrolled: the k-loop body is large; unrolling it GT times (8 at the
+  // 1-slice default) bloats I-cache for zero register benefit
+#pragma unroll 1
   for (int g = 0; g < GT; g++) {
     const long long col0 = ((long long)grp * GT + g) * 128;
     f32x4 acc[4][4];
