@@ -111,8 +111,12 @@ __global__ void fused_l2nn_2d_kernel(const __bf16* __restrict__ x0,
                               wr, wc, lane);
     }
 
-    // lane-local running top-2 across the group's tiles (disjoint columns)
+    // lane-local running top-2 across the group's tiles (disjoint columns);
+    // cn[col] depends only on fc -> 4 loads per tile instead of 64
     const int col_base = (int)col0 + wc * 64;
+    float cn_r[4];
+#pragma unroll
+    for (int fc = 0; fc < 4; fc++) cn_r[fc] = cn[col_base + fc * 16 + (lane & 15)];
 #pragma unroll
     for (int fr = 0; fr < 4; fr++) {
 #pragma unroll
@@ -120,7 +124,7 @@ __global__ void fused_l2nn_2d_kernel(const __bf16* __restrict__ x0,
 #pragma unroll
         for (int fc = 0; fc < 4; fc++) {
           const int col = col_base + fc * 16 + (lane & 15);
-          const float s = cn[col] - 2.f * acc[fr][fc][reg];
+          const float s = cn_r[fc] - 2.f * acc[fr][fc][reg];
           // new second-best = middle of {s, best, best2} (invariant
           // best <= best2): one v_med3_f32 instead of a cndmask chain
           best2[fr][reg] = __builtin_amdgcn_fmed3f(s, best[fr][reg],
