@@ -108,12 +108,12 @@ __global__ void fused_l2nn_kernel(const __bf16* __restrict__ x0,
         for (int fc = 0; fc < 4; fc++) {
           const int col = col_base + fc * 16 + (lane & 15);
           const float s = cn[col] - 2.f * acc[fr][fc][reg];
+          // second-best via one v_med3_f32 (invariant best <= best2)
+          best2[fr][reg] = __builtin_amdgcn_fmed3f(s, best[fr][reg],
+                                                   best2[fr][reg]);
           if (s < best[fr][reg]) {
-            best2[fr][reg] = best[fr][reg];
             best[fr][reg] = s;
             bidx[fr][reg] = col;
-          } else if (s < best2[fr][reg]) {
-            best2[fr][reg] = s;
           }
         }
       }

@@ -72,12 +72,10 @@ __global__ void fused_l2nn_256_kernel(const __bf16* __restrict__ x0,
       for (int fc = 0; fc < 4; fc++) {
         const int col = (int)col0 + wn * 64 + fc * 16 + (lane & 15);
         const float s = cn[col] - 2.f * acc[fr][fc][reg];
+        v2 = __builtin_amdgcn_fmed3f(s, v, v2);  // second-best in one op
         if (s < v) {
-          v2 = v;
           v = s;
           vi = col;
-        } else if (s < v2) {
-          v2 = s;
         }
       }
       best[fr][reg] = v;
