@@ -177,6 +177,19 @@ __global__ void pairwise_l2_filter_kernel(const __bf16* __restrict__ x0,
   mfma_tile_kloop<NSLICE>(xg, cg, xs, cs, acc, row0, col0, d, m - 1, n - 1,
                           wr, wc, lane);
 
+  // Epilogue VALU diet (same recipe as the l2nn 2d kernel, measured there
+  // via profiles/pmc_l2nn_x1v_gt8.txt): yn[col] and the col<n guard depend
+  // only on fc — hoisted to 4 registers (out-of-range columns carry +inf,
+  // so they can never pass the threshold); the fmax moves into the rare
+  // admitted-candidate store, with thr clamped to >= 0 once per row so the
+  // unclamped compare admits the same set.
+  const long long colb = col0 + wc * 64 + (lane & 15);
+  float yn_r[4];
+#pragma unroll
+  for (int fc = 0; fc < 4; fc++) {
+    const long long col = colb + fc * 16;
+    yn_r[fc] = col < n ? yn[col] : INFINITY;
+  }
 #pragma unroll
   for (int fr = 0; fr < 4; fr++) {
 #pragma unroll
@@ -184,18 +197,15 @@ __global__ void pairwise_l2_filter_kernel(const __bf16* __restrict__ x0,
       const long long row = row0 + wr * 64 + fr * 16 + (lane >> 4) * 4 + reg;
       if (row >= m) continue;
       const float xv = xn[row];
-      const float t = thr[row];
+      const float t = fmaxf(thr[row], 0.f);
 #pragma unroll
       for (int fc = 0; fc < 4; fc++) {
-        const long long col = col0 + wc * 64 + fc * 16 + (lane & 15);
-        if (col < n) {
-          const float d2 = fmaxf(xv + yn[col] - 2.f * acc[fr][fc][reg], 0.f);
-          if (d2 <= t) {
-            const int pos = atomicAdd(&cnt[row], 1);
-            if (pos < cap) {
-              out_d[row * cap + pos] = d2;
-              out_i[row * cap + pos] = (int)(col + col_offset);
-            }
+        const float s = xv + yn_r[fc] - 2.f * acc[fr][fc][reg];
+        if (s <= t) {
+          const int pos = atomicAdd(&cnt[row], 1);
+          if (pos < cap) {
+            out_d[row * cap + pos] = fmaxf(s, 0.f);
+            out_i[row * cap + pos] = (int)(colb + fc * 16 + col_offset);
           }
         }
       }
@@ -307,6 +317,15 @@ __global__ void pairwise_l2_filter_px_kernel(
     if (kt == kts - 1) {
       // filter epilogue for this col tile (registers + global atomics only)
       const long long col0 = col_base + (long long)(s / kts) * 128;
+      // hoisted epilogue (see pairwise_l2_filter_kernel): yn + col guard
+      // live in 4 regs; +inf padding makes out-of-range columns inadmissible
+      const long long colb = col0 + wc * 64 + (lane & 15);
+      float yn_r[4];
+#pragma unroll
+      for (int fc = 0; fc < 4; fc++) {
+        const long long col = colb + fc * 16;
+        yn_r[fc] = col < n ? yn[col] : INFINITY;
+      }
 #pragma unroll
       for (int fr = 0; fr < 4; fr++) {
 #pragma unroll
@@ -314,18 +333,15 @@ __global__ void pairwise_l2_filter_px_kernel(
           const long long row = row0 + wr * 64 + fr * 16 + (lane >> 4) * 4 + reg;
           if (row >= m) continue;
           const float xv = xn[row];
-          const float tv = thr[row];
+          const float tv = fmaxf(thr[row], 0.f);
 #pragma unroll
           for (int fc = 0; fc < 4; fc++) {
-            const long long col = col0 + wc * 64 + fc * 16 + (lane & 15);
-            if (col < n) {
-              const float d2 = fmaxf(xv + yn[col] - 2.f * acc[fr][fc][reg], 0.f);
-              if (d2 <= tv) {
-                const int pos = atomicAdd(&cnt[row], 1);
-                if (pos < cap) {
-                  out_d[row * cap + pos] = d2;
-                  out_i[row * cap + pos] = (int)(col + col_offset);
-                }
+            const float sd = xv + yn_r[fc] - 2.f * acc[fr][fc][reg];
+            if (sd <= tv) {
+              const int pos = atomicAdd(&cnt[row], 1);
+              if (pos < cap) {
+                out_d[row * cap + pos] = fmaxf(sd, 0.f);
+                out_i[row * cap + pos] = (int)(colb + fc * 16 + col_offset);
               }
             }
           }
@@ -765,6 +781,13 @@ __global__ void pairwise_l2_filter256_kernel(const __bf16* __restrict__ x0,
     }
   }
 
+  const long long colb = col0 + wc * 64 + (lane & 15);
+  float yn_r[4];
+#pragma unroll
+  for (int fc = 0; fc < 4; fc++) {
+    const long long col = colb + fc * 16;
+    yn_r[fc] = col < n ? yn[col] : INFINITY;
+  }
 #pragma unroll
   for (int fr = 0; fr < 8; fr++) {
 #pragma unroll
@@ -772,18 +795,15 @@ __global__ void pairwise_l2_filter256_kernel(const __bf16* __restrict__ x0,
       const long long row = row0 + wr * 128 + fr * 16 + (lane >> 4) * 4 + reg;
       if (row >= m) continue;
       const float xv = xn[row];
-      const float t = thr[row];
+      const float t = fmaxf(thr[row], 0.f);
 #pragma unroll
       for (int fc = 0; fc < 4; fc++) {
-        const long long col = col0 + wc * 64 + fc * 16 + (lane & 15);
-        if (col < n) {
-          const float d2 = fmaxf(xv + yn[col] - 2.f * acc[fr][fc][reg], 0.f);
-          if (d2 <= t) {
-            const int pos = atomicAdd(&cnt[row], 1);
-            if (pos < cap) {
-              out_d[row * cap + pos] = d2;
-              out_i[row * cap + pos] = (int)(col + col_offset);
-            }
+        const float sd = xv + yn_r[fc] - 2.f * acc[fr][fc][reg];
+        if (sd <= t) {
+          const int pos = atomicAdd(&cnt[row], 1);
+          if (pos < cap) {
+            out_d[row * cap + pos] = fmaxf(sd, 0.f);
+            out_i[row * cap + pos] = (int)(colb + fc * 16 + col_offset);
           }
         }
       }
