@@ -76,6 +76,10 @@ __global__ void pairwise_l2_kernel(const __bf16* __restrict__ x0,
     // (418 Gdist/s vs the ~1.6 Tdist/s write roofline).
     __syncthreads();  // K-loop LDS is dead; reuse
     float* tile = reinterpret_cast<float*>(smem);  // [128][132]
+    float yn_r[4];
+#pragma unroll
+    for (int fc = 0; fc < 4; fc++)
+      yn_r[fc] = yn[col0 + wc * 64 + fc * 16 + (lane & 15)];
 #pragma unroll
     for (int fr = 0; fr < 4; fr++) {
 #pragma unroll
@@ -85,7 +89,7 @@ __global__ void pairwise_l2_kernel(const __bf16* __restrict__ x0,
 #pragma unroll
         for (int fc = 0; fc < 4; fc++) {
           const int cl = wc * 64 + fc * 16 + (lane & 15);
-          float v = fmaxf(xv + yn[col0 + cl] - 2.f * acc[fr][fc][reg], 0.f);
+          float v = fmaxf(xv + yn_r[fc] - 2.f * acc[fr][fc][reg], 0.f);
           if (sqrt_out) v = sqrtf(v);
           tile[rl * 132 + cl] = v;
         }
