@@ -77,8 +77,15 @@ def _slices_of(t: torch.Tensor, fp32_mode: str):
     from raft_amd.neighbors.fused_l2nn import split_bf16_slices
     if t.dtype == torch.bfloat16:
         return [t.contiguous()]
-    nsl = 3 if fp32_mode == "bf16x3" else 2
+    nsl = {"bf16x3": 3, "bf16x1v": 1}.get(fp32_mode, 2)
     return split_bf16_slices(t, nsl)
+
+
+#: provable |Δdot| <= lead*sqrt(qn*xn) + tail*(qn+xn) envelopes per slice
+#: count (same family as the verified L2NN engine, csrc/kmeans.hip)
+_KNN_BOUND = {1: (2.0 ** -7, 2.0 ** -12),
+              2: (2.0 ** -13, 2.0 ** -18),
+              3: (2.0 ** -21, 2.0 ** -24)}
 
 
 def _norms(t: torch.Tensor) -> torch.Tensor:
@@ -147,10 +154,14 @@ def _knn_gpu_filtered_block(x, queries, k, fp32_mode, index_chunk, res):
     # high). bf16 input (1 slice) has no split error. Same bound family as
     # the verified L2NN engine (csrc/kmeans.hip l2nn verify).
     nslice = len(q_slices)
-    if nslice >= 2:
+    # bf16 input is exact in its own dtype (single slice, no split error);
+    # fp32 input is EMULATED at any slice count — including the 1-slice
+    # bf16x1v mode, whose wider 2^-7 bound still inflates thresholds by far
+    # less than the inter-candidate spacing on realistic data
+    emulated = queries.dtype != torch.bfloat16
+    if emulated:
         xm = float(xn_full.max())
-        lead = 2.0 ** -13 if nslice == 2 else 2.0 ** -21
-        tail = 2.0 ** -18 if nslice == 2 else 2.0 ** -24
+        lead, tail = _KNN_BOUND[nslice]
         thr = thr + 4.0 * (lead * torch.sqrt(qn.clamp_min(0) * xm)
                            + tail * (qn + xm))
 
@@ -172,7 +183,7 @@ def _knn_gpu_filtered_block(x, queries, k, fp32_mode, index_chunk, res):
 
     # ---- 3. select over candidates -----------------------------------------
     bad = (cnt < k) | (cnt > cap)
-    if nslice >= 2:
+    if emulated:
         # fp32 input: candidate distances are split-bf16 EMULATED — re-rank
         # the top 2k candidates by EXACT fp32 distances, then prove per row
         # that no candidate outside the re-ranked set can be a true neighbor:

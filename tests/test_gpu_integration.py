@@ -359,12 +359,16 @@ class TestKnnFp32Exactness:
     distances with a per-row margin proof — results must match the exact
     expanded-fp32 top-k, and fp32_mode='native' must skip the MFMA filter."""
 
-    def test_fp32_filtered_rerank_exact(self, dev):
+    @pytest.mark.parametrize("mode", ["auto", "bf16x1v"])
+    def test_fp32_filtered_rerank_exact(self, dev, mode):
+        """auto = 2-slice filter; bf16x1v = 1-slice filter with the wider
+        2^-7 threshold inflation — both must land the exact fp32 top-k via
+        the re-rank + margin proof."""
         from raft_amd.neighbors import knn
         torch.manual_seed(0)
         x = torch.randn(60000, 128, device=dev)
         q = torch.randn(2000, 128, device=dev)
-        d, i = knn(x, q, k=16)                       # auto -> filtered+rerank
+        d, i = knn(x, q, k=16, fp32_mode=mode)       # filtered+rerank
         xn = (x * x).sum(1)
         qn = (q * q).sum(1)
         ref = (qn.unsqueeze(1) + xn.unsqueeze(0) - 2.0 * (q @ x.t())).clamp_min(0)
