@@ -219,8 +219,11 @@ def _knn_gpu_filtered_block(x, queries, k, fp32_mode, index_chunk, res):
     n_bad = int(bad.sum().item())
     if n_bad:
         rows = bad.nonzero(as_tuple=True)[0]
+        # the tiled fallback runs the GEMM layer, which has no 1-slice mode
+        # (it exists only as a filter inflation choice) — fall back at auto
+        fb_mode = "auto" if fp32_mode == "bf16x1v" else fp32_mode
         bd, bi = _knn_tiled(x, queries[rows], k, DistanceType.L2Expanded,
-                            8192, index_chunk, fp32_mode)
+                            8192, index_chunk, fb_mode)
         vals[rows] = bd.to(vals.dtype)
         idx[rows] = bi
     return vals, idx
