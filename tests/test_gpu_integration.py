@@ -359,11 +359,11 @@ class TestKnnFp32Exactness:
     distances with a per-row margin proof — results must match the exact
     expanded-fp32 top-k, and fp32_mode='native' must skip the MFMA filter."""
 
-    @pytest.mark.parametrize("mode", ["auto", "bf16x1v"])
+    @pytest.mark.parametrize("mode", ["auto", "bf16x1v", "bf16x2"])
     def test_fp32_filtered_rerank_exact(self, dev, mode):
-        """auto = 2-slice filter; bf16x1v = 1-slice filter with the wider
-        2^-7 threshold inflation — both must land the exact fp32 top-k via
-        the re-rank + margin proof."""
+        """auto/bf16x1v = 1-slice filter with the wider 2^-7 threshold
+        inflation; bf16x2 = the tighter 2-slice filter — all must land the
+        exact fp32 top-k via the re-rank + margin proof."""
         from raft_amd.neighbors import knn
         torch.manual_seed(0)
         x = torch.randn(60000, 128, device=dev)
