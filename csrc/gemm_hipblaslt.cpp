@@ -10,6 +10,7 @@
 #include <hip/hip_runtime.h>
 #include <hipblaslt/hipblaslt.h>
 
+#include <algorithm>
 #include <map>
 #include <mutex>
 #include <stdexcept>
@@ -117,6 +118,17 @@ void lt_matmul(const LtPlan& p, const void* a, const void* b, float* c,
 void gemm_bf16_f32_rowmajor_lt(const void* a, const void* b, float* c,
                                long long m, long long n, long long k,
                                float beta, void* stream) {
+  // vendor 32-bit C-element overflow guard (see gemm_rocblas.cpp): outputs
+  // >= 2^31 elements corrupt tail rows on this stack — row-chunk below it
+  constexpr long long kMax = 1ll << 30;
+  if (m * n > kMax && m > 1) {
+    const long long rows = std::max(1ll, kMax / n);
+    for (long long r0 = 0; r0 < m; r0 += rows)
+      gemm_bf16_f32_rowmajor_lt(
+          static_cast<const unsigned short*>(a) + r0 * k, b, c + r0 * n,
+          std::min(rows, m - r0), n, k, beta, stream);
+    return;
+  }
   auto& p = get_plan(n, m, k, /*trans_a=*/false);
   lt_matmul(p, b, a, c, beta, (hipStream_t)stream);
 }
@@ -126,6 +138,15 @@ void gemm_bf16_f32_rowmajor_lt(const void* a, const void* b, float* c,
 void gemm_bf16_f32_nt_rowmajor_lt(const void* a, const void* b, float* c,
                                   long long m, long long n, long long k,
                                   float beta, void* stream) {
+  constexpr long long kMax = 1ll << 30;  // see gemm_rocblas.cpp guard
+  if (m * n > kMax && m > 1) {
+    const long long rows = std::max(1ll, kMax / n);
+    for (long long r0 = 0; r0 < m; r0 += rows)
+      gemm_bf16_f32_nt_rowmajor_lt(
+          static_cast<const unsigned short*>(a) + r0 * k, b, c + r0 * n,
+          std::min(rows, m - r0), n, k, beta, stream);
+    return;
+  }
   auto& p = get_plan(n, m, k, /*trans_a=*/true);
   lt_matmul(p, b, a, c, beta, (hipStream_t)stream);
 }

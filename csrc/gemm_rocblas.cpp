@@ -8,6 +8,7 @@
 
 #include <rocblas/rocblas.h>
 
+#include <algorithm>
 #include <mutex>
 #include <stdexcept>
 #include <string>
@@ -31,12 +32,27 @@ static rocblas_handle get_handle() {
   return h;
 }
 
+// rocblas_gemm_ex corrupts outputs with >= 2^31 elements on this stack
+// (measured 2026-09: a [256, 30M] fp32 C has exact rows 0..~200 and garbage
+// near the tail — a 32-bit C-element index in the vendor epilogue;
+// rocblas_sgemm is NOT affected). Every gemm_ex wrapper row-chunks below
+// the boundary; our own HIP kernels index 64-bit and never hit this.
+static constexpr long long kMaxGemmExElems = 1ll << 30;
+
 // C[m,n] (row-major, fp32) = A[m,k] (row-major bf16) @ B[k,n] (row-major bf16)
 //                            + beta * C
 // Row-major is expressed as the transposed column-major problem:
 // C^T = B^T A^T with everything column-major.
 void gemm_bf16_f32_rowmajor(const void* a, const void* b, float* c, long long m,
                             long long n, long long k, float beta, void* stream) {
+  if (m * n > kMaxGemmExElems && m > 1) {
+    const long long rows = std::max(1ll, kMaxGemmExElems / n);
+    for (long long r0 = 0; r0 < m; r0 += rows)
+      gemm_bf16_f32_rowmajor(
+          static_cast<const unsigned short*>(a) + r0 * k, b, c + r0 * n,
+          std::min(rows, m - r0), n, k, beta, stream);
+    return;
+  }
   rocblas_handle h = get_handle();
   ROCBLAS_CHECK(rocblas_set_stream(h, (hipStream_t)stream));
   const float alpha = 1.0f;
@@ -55,6 +71,15 @@ void gemm_bf16_f32_rowmajor(const void* a, const void* b, float* c, long long m,
 // Column-major: C_cm[n,m] = B_cm^T[n,k] * A_cm[k,m].
 void gemm_bf16_f32_nt_rowmajor(const void* a, const void* b, float* c, long long m,
                                long long n, long long k, float beta, void* stream) {
+  if (m * n > kMaxGemmExElems && m > 1) {
+    // see kMaxGemmExElems: vendor gemm_ex 32-bit C-index overflow guard
+    const long long rows = std::max(1ll, kMaxGemmExElems / n);
+    for (long long r0 = 0; r0 < m; r0 += rows)
+      gemm_bf16_f32_nt_rowmajor(
+          static_cast<const unsigned short*>(a) + r0 * k, b, c + r0 * n,
+          std::min(rows, m - r0), n, k, beta, stream);
+    return;
+  }
   rocblas_handle h = get_handle();
   ROCBLAS_CHECK(rocblas_set_stream(h, (hipStream_t)stream));
   const float alpha = 1.0f;

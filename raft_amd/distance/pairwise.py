@@ -100,6 +100,18 @@ def _metric_from_str(s: str) -> DistanceType:
 def _gemm_xyt(x, y, fp32_mode):
     if x.dtype == torch.float32 and fp32_mode not in ("native", "auto"):
         return gemm_fp32_emulated(x, y.t(), mode=fp32_mode)
+    m, n = x.shape[0], y.shape[0]
+    if x.is_cuda and m * n > (1 << 30) and m > 1:
+        # vendor GEMM (hipBLASLt via torch.matmul, rocblas_gemm_ex) corrupts
+        # outputs >= 2^31 elements on this stack (measured: [256 x 30M] fp32
+        # C exact for rows 0..~200, garbage near the tail — 32-bit C-element
+        # indexing). Row-chunk below the boundary; the
+        # raft_amd HIP kernels index 64-bit and are unaffected.
+        rows = max(1, (1 << 30) // n)
+        out = torch.empty((m, n), dtype=x.dtype, device=x.device)
+        for r0 in range(0, m, rows):
+            torch.matmul(x[r0:r0 + rows], y.t(), out=out[r0:r0 + rows])
+        return out
     return x @ y.t()
 
 

@@ -31,7 +31,17 @@ def gemm(a: torch.Tensor, b: torch.Tensor, alpha: float = 1.0, beta: float = 0.0
         a = a.t()
     if trans_b:
         b = b.t()
-    out = torch.matmul(a, b)
+    if (a.is_cuda and a.dim() == 2 and b.dim() == 2
+            and a.shape[0] * b.shape[1] > (1 << 30) and a.shape[0] > 1):
+        # vendor GEMM corrupts outputs >= 2^31 elements on this stack
+        # (32-bit C-element indexing, measured — see BASELINE.md): row-chunk
+        m, n = a.shape[0], b.shape[1]
+        rows = max(1, (1 << 30) // n)
+        out = torch.empty((m, n), dtype=a.dtype, device=a.device)
+        for r0 in range(0, m, rows):
+            torch.matmul(a[r0:r0 + rows], b, out=out[r0:r0 + rows])
+    else:
+        out = torch.matmul(a, b)
     if alpha != 1.0:
         out = out * alpha
     if c is not None and beta != 0.0:

@@ -137,6 +137,10 @@ def _chunked_gpu(x, y, sqrt, fp32_mode, chunk_rows):
     """GEMM + fused argmin-epilogue, chunked so the distance tile stays small."""
     ext = require_ext()
     m = x.shape[0]
+    # keep each GEMM output under 2^30 elements: vendor GEMM corrupts
+    # >= 2^31-element outputs on this stack (32-bit C-element indexing,
+    # measured — BASELINE.md)
+    chunk_rows = max(1, min(chunk_rows, (1 << 30) // max(y.shape[0], 1)))
     yn = (y * y).sum(dim=1)
     xn = (x * x).sum(dim=1)
     dmin = torch.empty(m, dtype=x.dtype, device=x.device)
@@ -152,6 +156,9 @@ def _chunked_gpu(x, y, sqrt, fp32_mode, chunk_rows):
 
 
 def _gemm_xyt(x, y, fp32_mode):
+    # _chunked_gpu clamps chunk_rows so the output stays < 2^30 elements
+    # (vendor GEMM 32-bit C-index overflow); the emulated
+    # path's rocblas wrappers additionally self-chunk.
     if x.dtype == torch.float32 and fp32_mode in ("bf16x3", "bf16x2"):
         return gemm_fp32_emulated(x, y.t(), mode=fp32_mode)
     return x @ y.t()

@@ -521,6 +521,26 @@ class TestGemmGpu:
         c2 = ext.gemm_bf16_f32(a, b, c, 1.0)
         torch.testing.assert_close(c2, (a.float() @ b.float()) * 2, rtol=2e-2, atol=2e-2)
 
+    def test_huge_output_gemm_guard(self, dev, ext):
+        """Vendor gemm_ex corrupts >= 2^31-element outputs on this stack
+        (32-bit C-element indexing — measured). The raft_amd
+        wrappers row-chunk below the boundary: a [256 x 20M] C (5.1e9
+        elements) must be exact everywhere, INCLUDING the tail rows where
+        the unguarded vendor call returns garbage."""
+        torch.manual_seed(0)
+        a = torch.randn(256, 128, device=dev).bfloat16()
+        b = torch.randn(20_000_000, 128, device=dev).bfloat16()
+        out = ext.gemm_bf16_f32_nt(a, b)
+        cols = torch.randint(0, 20_000_000, (512,),
+                             generator=torch.Generator(dev).manual_seed(1),
+                             device=dev)
+        for r in (0, 128, 255):
+            exact = (a[r].float().double().unsqueeze(0)
+                     * b[cols].float().double()).sum(1)
+            err = float((out[r, cols].double() - exact).abs().max())
+            assert err < 1e-3, (r, err)
+        del out
+
     def test_fp32_emulation_gpu_accuracy(self, dev, ext):
         from raft_amd.linalg import gemm_fp32_emulated
         torch.manual_seed(1)
