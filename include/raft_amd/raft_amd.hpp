@@ -13,6 +13,7 @@
 
 #include "cluster.hpp"
 #include "core/mdspan.hpp"
+#include "core/resources.hpp"
 #include "core.hpp"
 #include "distance.hpp"
 #include "linalg.hpp"

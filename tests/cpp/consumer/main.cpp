@@ -41,9 +41,15 @@ int main() {
 
   auto vals = make_device_matrix<float>(m, k);
   auto idxs = make_device_matrix<int>(m, k);
-  device_uvector<char> ws;
+  // the C++ resources handle owns the stream + grow-on-demand workspace
+  device_resources res(/*device_id=*/0, /*n_pool_streams=*/2);
   select_k({dist.data_handle(), dextents<std::int64_t, 2>(m, n)}, vals.view(),
-           idxs.view(), ws, /*select_min=*/true, /*sorted=*/true);
+           idxs.view(), res.get_workspace(), /*select_min=*/true,
+           /*sorted=*/true, res.get_stream());
+  res.sync_stream();
+  if (res.stream_pool_size() != 2 ||
+      res.get_stream_from_pool(5) == nullptr)
+    return 2;
 
   auto h_dist = make_host_matrix<float>(m, n);
   auto h_amin = make_host_vector<int>(m);
@@ -205,7 +211,7 @@ int main() {
   }
   std::printf(
       "CONSUMER_OK m=%lld n=%lld k=%lld "
-      "(pairwise+select_k+fused_l2nn+spmv+sddmm)\n",
+      "(resources+pairwise+select_k+fused_l2nn+spmv+sddmm)\n",
       (long long)m, (long long)n, (long long)k);
   return 0;
 }
