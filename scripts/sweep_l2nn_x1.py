@@ -34,7 +34,7 @@ print(f"RESULT {(time.perf_counter() - t0) / N * 1e3:.3f} ms")
 """
 
 CONFIGS = {
-    "default(GT2)": {},
+    "default(GT8)": {},
     "GT1": {"RAFT_AMD_L2NN_GT": "1"},
     "GT4": {"RAFT_AMD_L2NN_GT": "4"},
     "BK32": {"RAFT_AMD_L2NN_BK32": "1"},
