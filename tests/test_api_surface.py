@@ -175,3 +175,40 @@ class TestRound2Surface:
         assert w.shape == (6,)
         w2, _ = eig_selective(a, 2, largest=False)
         assert w2.shape == (2,)
+
+    def test_adaptive_engine_surface(self):
+        """Late-round-2 surfaces: the bf16x1v verified mode, its bound
+        constants, and the kNN per-slice inflation table."""
+        from raft_amd.neighbors.fused_l2nn import (_DEFAULT_BOUND, _MODE_BOUND,
+                                                   _MODE_NSLICE, _VERIFY_MODES)
+        assert _MODE_NSLICE["bf16x1v"] == 1 and "bf16x1v" in _VERIFY_MODES
+        lead, tail = _MODE_BOUND["bf16x1v"]
+        # the 1-slice bound is exactly 2^6 wider than the 2-slice bound
+        assert lead == _DEFAULT_BOUND[0] * 64 and tail == _DEFAULT_BOUND[1] * 64
+        from raft_amd.neighbors.brute_force import _KNN_BOUND, _slices_of
+        assert set(_KNN_BOUND) == {1, 2, 3}
+        assert _KNN_BOUND[1] == (lead, tail)
+        x = torch.randn(8, 64)
+        assert len(_slices_of(x, "auto")) == 1          # 1-slice auto filter
+        assert len(_slices_of(x, "bf16x2")) == 2
+        assert len(_slices_of(x, "bf16x3")) == 3
+        back = sum(s.float() for s in _slices_of(x, "bf16x3"))
+        assert torch.allclose(back, x, atol=1e-5)
+
+    def test_gemm_guard_shape_logic(self):
+        """linalg.gemm row-chunks >2^30-element outputs (vendor 32-bit
+        C-index overflow guard) — CPU tensors bypass the guard and small
+        GPU-shaped calls are untouched; verify the small-shape path here."""
+        from raft_amd.linalg import gemm
+        a = torch.randn(8, 16)
+        b = torch.randn(16, 4)
+        torch.testing.assert_close(gemm(a, b), a @ b)
+
+    def test_cpp_resources_header_exists(self):
+        import os
+        p = os.path.join(os.path.dirname(os.path.dirname(
+            os.path.abspath(__file__))), "include", "raft_amd", "core",
+            "resources.hpp")
+        src = open(p).read()
+        assert "class device_resources" in src
+        assert "get_stream_from_pool" in src
