@@ -58,6 +58,55 @@ def knn(x: torch.Tensor, queries: torch.Tensor, k: int,
     return _knn_tiled(x, queries, k, metric, query_chunk, index_chunk, fp32_mode)
 
 
+class BruteForceIndex:
+    """Prebuilt brute-force index: build once, search many times.
+
+    Reference parity: the build/search split of raft's historical
+    brute-force knn (index object holding the dataset + norms). Caches the
+    bf16 filter slice(s) and the row norms of the index matrix, so repeated
+    searches skip the per-call index split — at 100M x 128 fp32 the split
+    alone is a measurable fraction of a one-shot knn() call.
+    """
+
+    def __init__(self, x: torch.Tensor, fp32_mode: str = "auto", res=None):
+        self.x = x
+        self.fp32_mode = fp32_mode
+        self._filterable = (
+            x.is_cuda and x.shape[1] % 64 == 0
+            and x.dtype in (torch.bfloat16, torch.float32)
+            and not (x.dtype == torch.float32 and fp32_mode == "native"))
+        if self._filterable:
+            self.slices = _slices_of(x, fp32_mode)
+            self.xn = _norms(x)
+        else:
+            self.slices = None
+            self.xn = None
+
+    @property
+    def n_rows(self) -> int:
+        return self.x.shape[0]
+
+    @property
+    def dim(self) -> int:
+        return self.x.shape[1]
+
+    def search(self, queries: torch.Tensor, k: int, res=None):
+        """k nearest index rows per query. Returns (dists [q,k], idx [q,k])."""
+        if (self._filterable and queries.is_cuda
+                and self.x.shape[0] >= 8 * k):
+            from raft_amd.core.resources import get_resources
+            r = get_resources(res if res is not None else queries.device)
+            return _knn_gpu_filtered(self.x, queries, k, self.fp32_mode,
+                                     res=r, pre=(self.slices, self.xn))
+        return knn(self.x, queries, k, fp32_mode=self.fp32_mode, res=res)
+
+
+def brute_force_build(x: torch.Tensor, fp32_mode: str = "auto",
+                      res=None) -> BruteForceIndex:
+    """reference brute_force::build parity."""
+    return BruteForceIndex(x, fp32_mode=fp32_mode, res=res)
+
+
 def _tiles_from_budget(budget_bytes: int, q: int, n: int, d: int, k: int):
     """Size the (query_chunk x index_chunk) distance tile from the workspace
     budget: the fp32 tile is the dominant scratch of the tiled path; keep it
@@ -105,10 +154,12 @@ def _norms(t: torch.Tensor) -> torch.Tensor:
 
 
 def _knn_gpu_filtered(x, queries, k, fp32_mode, index_chunk: int = 4_000_000,
-                      res=None):
+                      res=None, pre=None):
     """Query-blocked driver: splits queries so the candidate buffers
     (cap x 8 B/row) + the exact-rerank gather fit the workspace budget, then
-    runs the sample->filter->select pipeline per block."""
+    runs the sample->filter->select pipeline per block. pre (optional):
+    (index_slices, index_norms) from a BruteForceIndex — skips the per-call
+    index split."""
     from raft_amd.core.resources import get_resources
     res = get_resources(res if res is not None else queries.device)
     m = queries.shape[0]
@@ -117,18 +168,20 @@ def _knn_gpu_filtered(x, queries, k, fp32_mode, index_chunk: int = 4_000_000,
     budget = max(res.workspace_budget(), 1 << 22)
     q_block = max(256, min(m, int(budget // (2 * per_row)) or 1))
     if q_block >= m:
-        return _knn_gpu_filtered_block(x, queries, k, fp32_mode, index_chunk, res)
+        return _knn_gpu_filtered_block(x, queries, k, fp32_mode, index_chunk,
+                                       res, pre=pre)
     out_d = torch.empty((m, k), dtype=torch.float32, device=queries.device)
     out_i = torch.empty((m, k), dtype=torch.int64, device=queries.device)
     for s0, s1 in row_chunks(m, q_block):
         dv, iv = _knn_gpu_filtered_block(x, queries[s0:s1], k, fp32_mode,
-                                         index_chunk, res)
+                                         index_chunk, res, pre=pre)
         out_d[s0:s1] = dv
         out_i[s0:s1] = iv
     return out_d, out_i
 
 
-def _knn_gpu_filtered_block(x, queries, k, fp32_mode, index_chunk, res):
+def _knn_gpu_filtered_block(x, queries, k, fp32_mode, index_chunk, res,
+                            pre=None):
     ext = require_ext()
     m, d = queries.shape
     n = x.shape[0]
@@ -136,15 +189,19 @@ def _knn_gpu_filtered_block(x, queries, k, fp32_mode, index_chunk, res):
 
     q_slices = _slices_of(queries, fp32_mode)
     qn = _norms(queries)
-    xn_full = _norms(x)
+    pre_slices = pre[0] if pre is not None else None
+    xn_full = pre[1] if pre is not None else _norms(x)
 
     # ---- 1. sample -> per-row thresholds -----------------------------------
     s = max(min(65536, n), n // 1024)
     gen = torch.Generator(device="cpu").manual_seed(0x5eed)
     sample_idx = (torch.randint(0, n, (s,), generator=gen)
                   .to(dev))
-    xs_sample = x[sample_idx].contiguous()
-    sample_slices = _slices_of(xs_sample, fp32_mode)
+    if pre_slices is not None:
+        sample_slices = [sl[sample_idx].contiguous() for sl in pre_slices]
+    else:
+        xs_sample = x[sample_idx].contiguous()
+        sample_slices = _slices_of(xs_sample, fp32_mode)
     sn = xn_full[sample_idx].contiguous()
     j = max(2, (4 * k * s) // max(n, 1))
     thr = torch.empty(m, dtype=torch.float32, device=dev)
@@ -182,8 +239,11 @@ def _knn_gpu_filtered_block(x, queries, k, fp32_mode, index_chunk, res):
     cand_i.fill_(-1)
     cnt = torch.zeros(m, dtype=torch.int32, device=dev)
     for c0, c1 in row_chunks(n, index_chunk):
-        xc = x[c0:c1]
-        ext.pairwise_l2_filter(q_slices, _slices_of(xc, fp32_mode), qn,
+        if pre_slices is not None:
+            chunk_slices = [sl[c0:c1] for sl in pre_slices]
+        else:
+            chunk_slices = _slices_of(x[c0:c1], fp32_mode)
+        ext.pairwise_l2_filter(q_slices, chunk_slices, qn,
                                xn_full[c0:c1].contiguous(), thr,
                                cand_d, cand_i, cnt, c0)
 

@@ -212,3 +212,16 @@ class TestRound2Surface:
         src = open(p).read()
         assert "class device_resources" in src
         assert "get_stream_from_pool" in src
+
+    def test_brute_force_index_surface(self):
+        """build/search split (reference brute_force::build parity) — CPU
+        falls back to the tiled path through knn()."""
+        from raft_amd.neighbors import BruteForceIndex, brute_force_build, knn
+        x = torch.randn(500, 32)
+        q = torch.randn(20, 32)
+        idx = brute_force_build(x)
+        assert idx.n_rows == 500 and idx.dim == 32
+        d, i = idx.search(q, 4)
+        dr, ir = knn(x, q, 4)
+        assert torch.equal(i, ir)
+        torch.testing.assert_close(d, dr)

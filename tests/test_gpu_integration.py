@@ -377,6 +377,22 @@ class TestKnnFp32Exactness:
         assert float(agree) > 0.9995, float(agree)   # near-ties may reorder
         torch.testing.assert_close(d, rd, rtol=1e-4, atol=1e-3)
 
+    @pytest.mark.parametrize("dt", [torch.float32, torch.bfloat16])
+    def test_brute_force_index_matches_oneshot(self, dev, dt):
+        """BruteForceIndex (build/search split, cached slices+norms) must
+        return exactly what the one-shot knn() filtered path returns."""
+        from raft_amd.neighbors import brute_force_build, knn
+        torch.manual_seed(2)
+        x = torch.randn(80000, 128, device=dev).to(dt)
+        q = torch.randn(1000, 128, device=dev).to(
+            torch.float32 if dt == torch.float32 else dt)
+        idx = brute_force_build(x)
+        d1, i1 = idx.search(q, 16)
+        d2, i2 = idx.search(q, 16)          # second search reuses the cache
+        dr, ir = knn(x, q, 16)
+        assert torch.equal(i1, ir) and torch.equal(i2, ir)
+        torch.testing.assert_close(d1, dr)
+
     def test_fp32_native_mode_honored(self, dev):
         from raft_amd.neighbors import knn
         torch.manual_seed(1)
