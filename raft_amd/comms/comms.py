@@ -199,24 +199,52 @@ class TorchDistComms(Comms):
     def barrier(self) -> None:
         dist.barrier(group=self.group)
 
+    def _gloo_cuda(self, t) -> bool:
+        """gloo groups don't support (all of) the CUDA-tensor collectives —
+        stage through the host instead of faulting (mixed setups: a gloo
+        subgroup created on a GPU node, e.g. for control-plane traffic)."""
+        return t.is_cuda and dist.get_backend(self.group) == "gloo"
+
     def allreduce(self, t, op=ReduceOp.SUM):
+        if self._gloo_cuda(t):
+            h = t.cpu()
+            dist.all_reduce(h, op=op.to_dist(), group=self.group)
+            t.copy_(h)
+            return t
         dist.all_reduce(t, op=op.to_dist(), group=self.group)
         return t
 
     def allreduce_async(self, t, op=ReduceOp.SUM):
+        if self._gloo_cuda(t):
+            self.allreduce(t, op)        # host-staged: synchronous
+            return None
         return dist.all_reduce(t, op=op.to_dist(), group=self.group,
                                async_op=True)
 
     def bcast(self, t, root=0):
+        if self._gloo_cuda(t):
+            h = t.cpu()
+            dist.broadcast(h, src=root, group=self.group)
+            t.copy_(h)
+            return t
         dist.broadcast(t, src=root, group=self.group)
         return t
 
     def reduce(self, t, root=0, op=ReduceOp.SUM):
+        if self._gloo_cuda(t):
+            h = t.cpu()
+            dist.reduce(h, dst=root, op=op.to_dist(), group=self.group)
+            t.copy_(h)
+            return t
         dist.reduce(t, dst=root, op=op.to_dist(), group=self.group)
         return t
 
     def allgather(self, t):
         world = self.get_size()
+        if self._gloo_cuda(t):
+            h = torch.empty((world,) + tuple(t.shape), dtype=t.dtype)
+            dist.all_gather(list(h.unbind(0)), t.cpu(), group=self.group)
+            return h.to(t.device)
         out = torch.empty((world,) + tuple(t.shape), dtype=t.dtype, device=t.device)
         dist.all_gather(list(out.unbind(0)), t.contiguous(), group=self.group)
         return out

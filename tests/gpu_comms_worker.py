@@ -96,7 +96,23 @@ def main():
     assert torch.allclose(ca, cs, rtol=1e-4, atol=1e-4)
     print(f"[rank {rank}] distributed kmeans (auto engine) OK", flush=True)
 
+    # mixed setup: a gloo subgroup on a GPU node with CUDA tensors must
+    # host-stage instead of faulting (control-plane traffic pattern)
     import torch.distributed as dist
+    gloo_group = dist.new_group(backend="gloo")
+    from raft_amd.comms.comms import TorchDistComms
+    gc = TorchDistComms(gloo_group)
+    t = torch.full((64,), float(rank + 1), device=dev)
+    gc.allreduce(t)
+    expect = sum(range(1, world + 1))
+    assert torch.allclose(t, torch.full_like(t, float(expect)))
+    g2 = gc.allgather(torch.tensor([float(rank)], device=dev))
+    assert g2.shape == (world, 1) and float(g2[rank, 0]) == float(rank)
+    b = torch.full((8,), float(rank), device=dev)
+    gc.bcast(b, root=0)
+    assert torch.allclose(b, torch.zeros_like(b))
+    print(f"[rank {rank}] gloo-on-GPU host-staged collectives OK", flush=True)
+
     dist.barrier()
     dist.destroy_process_group()
     print(f"[rank {rank}] GPU_COMMS_WORKER_OK", flush=True)
