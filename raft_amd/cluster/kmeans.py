@@ -378,7 +378,7 @@ def kmeans_iterate(x: torch.Tensor, centroids: torch.Tensor, n_iters: int,
         if state is None:
             state = kmeans_iter_state(x, fp32_mode)
         x_slices, xn = state
-    if use_fused and k % 128 == 0:
+    if use_fused:
         if fp32_mode != "auto" and len(x_slices) < _MODE_NSLICE[fp32_mode]:
             # a 1-slice "auto" state reused with an explicit 2-slice mode:
             # materialize the residual slice (the adaptive path does this
@@ -465,12 +465,23 @@ def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify,
     ext = require_ext()
     k, d = centroids.shape
     dev = x.device
+    # ANY k takes this path: pad the centroid rows to a 128 multiple with
+    # ZERO vectors whose cn entries are poisoned to +inf after each split —
+    # the sweep can never pick them (score = inf), the verify/rescan kernels
+    # see only the real k rows (centroids[:k] slice), and update leaves the
+    # zero rows untouched (counts 0). Removes the old k % 128 fast-path cliff.
+    pad = (-k) % 128
+    if pad:
+        centroids = torch.cat([centroids,
+                               torch.zeros(pad, d, dtype=centroids.dtype,
+                                           device=dev)])
+    kp = k + pad
     cur_nslice = 1 if (adaptive and verify) else nslice
     cur_bound = _MODE_BOUND["bf16x1v"] if (adaptive and verify) else bound
     frac_t = [None]  # first-iteration provable-rescan fraction (device)
-    c_slices = [torch.empty((k, d), dtype=torch.bfloat16, device=dev)
+    c_slices = [torch.empty((kp, d), dtype=torch.bfloat16, device=dev)
                 for _ in range(nslice)]
-    cn = torch.empty(k, dtype=torch.float32, device=dev)
+    cn = torch.empty(kp, dtype=torch.float32, device=dev)
     cn_max = torch.zeros(1, dtype=torch.float32, device=dev)
     world = comms.get_size()
     # overlap pays when the halves keep the GPU busy longer than the doubled
@@ -498,17 +509,18 @@ def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify,
                        + tail_ * (xnh + cn_max))
             frac_t[0] = ((dmin2 - dmin) < b).float().mean()
         keys_sorted, perm = torch.sort(amin)
-        packed = torch.zeros(k * d + k + 1, dtype=torch.float32, device=dev)
-        sums = packed[: k * d].view(k, d)
-        counts = packed[k * d: k * d + k]
+        packed = torch.zeros(kp * d + kp + 1, dtype=torch.float32, device=dev)
+        sums = packed[: kp * d].view(kp, d)
+        counts = packed[kp * d: kp * d + kp]
         if verify:
             # ONE X pass: centroid-sum accumulation + exact-fp32
             # verify/refine + the inertia fold (sum of the FINAL repaired
             # distances accumulates into packed[-1] in-kernel); cn_max comes
-            # fused out of split_bf16_norms
+            # fused out of split_bf16_norms. centroids[:k]: the rescan must
+            # only see the REAL rows (the zero padding rows would win it)
             ext.kmeans_update_verify(xh, perm.to(torch.int32), keys_sorted,
-                                     centroids, xnh, dmin, amin, dmin2, cn_max,
-                                     sums, counts, packed[-1:],
+                                     centroids[:k], xnh, dmin, amin, dmin2,
+                                     cn_max, sums, counts, packed[-1:],
                                      lead=cur_bound[0], tail=cur_bound[1])
         else:
             ext.reduce_rows_by_key_sorted_into(xh, perm.to(torch.int32),
@@ -518,6 +530,10 @@ def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify,
 
     for it in range(n_iters):
         ext.split_bf16_norms(centroids, c_slices[:cur_nslice], cn, cn_max)
+        if pad:
+            # poison the padding rows' norms AFTER the split (their zero
+            # vectors leave the fused cn_max untouched): sweep score = +inf
+            cn[k:].fill_(float("inf"))
         if overlap:
             packed_a = _local_update(*halves[0])
             work_a = comms.allreduce_async(packed_a, op=ReduceOp.SUM)
@@ -533,8 +549,8 @@ def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify,
             packed = _local_update(x, x_slices, xn)
             if world > 1:
                 comms.allreduce(packed, op=ReduceOp.SUM)
-        sums = packed[: k * d].view(k, d)
-        counts = packed[k * d: k * d + k]
+        sums = packed[: kp * d].view(kp, d)
+        counts = packed[kp * d: kp * d + kp]
         ext.kmeans_update_centroids(sums, counts, centroids)
         inertia_t = packed[-1]
         if adaptive and it == 0 and frac_t[0] is not None:
@@ -557,7 +573,7 @@ def _fast_iterate(x, x_slices, xn, centroids, n_iters, comms, nslice, verify,
             global _LAST_ADAPTIVE_NSLICE
             _LAST_ADAPTIVE_NSLICE = cur_nslice
     inertia = float(inertia_t.item()) if inertia_t is not None else float("inf")
-    return centroids, inertia
+    return centroids[:k] if pad else centroids, inertia
 
 
 def kmeans_predict(model_or_centroids, x: torch.Tensor, fp32_mode: str = "auto") -> torch.Tensor:

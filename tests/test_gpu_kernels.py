@@ -436,6 +436,22 @@ class TestKMeansGpu:
         torch.testing.assert_close(cg.cpu(), cc, rtol=1e-3, atol=1e-3)
         assert abs(ig - ic) / ic < 1e-3
 
+    @pytest.mark.parametrize("k", [100, 1000])
+    def test_fast_iterate_any_k_padding(self, dev, ext, k):
+        """ANY k now takes the minimal-dispatch fast path (centroids padded
+        to a 128 multiple with +inf-norm zero rows) — results must match the
+        CPU oracle exactly like the k%128==0 case."""
+        from raft_amd.cluster.kmeans import kmeans_iterate
+        from raft_amd.random import make_blobs, RngState
+        x, _, centers = make_blobs(20000, 64, n_clusters=k, cluster_std=0.3,
+                                   state=RngState(seed=13), device=dev)
+        c0 = centers + 0.2
+        cg, ig = kmeans_iterate(x, c0.clone(), 3, fp32_mode="auto")
+        assert cg.shape == (k, 64)
+        cc, ic = kmeans_iterate(x.cpu(), c0.cpu().clone(), 3)
+        torch.testing.assert_close(cg.cpu(), cc, rtol=1e-3, atol=1e-3)
+        assert abs(ig - ic) / ic < 1e-3
+
     def test_adaptive_auto_engine_decisions(self, dev, ext):
         """fp32_mode="auto" stays on the 1-product engine on well-separated
         data and widens to the 2-slice engine on near-tie-heavy data; both
