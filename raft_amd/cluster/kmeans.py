@@ -370,6 +370,18 @@ def kmeans_iterate(x: torch.Tensor, centroids: torch.Tensor, n_iters: int,
     k, d = centroids.shape
     inertia = float("inf")
     bound = _MODE_BOUND.get(fp32_mode, _DEFAULT_BOUND)
+    if (x.is_cuda and x.dtype == torch.float32 and fp32_mode in _MODE_NSLICE
+            and d % 64 != 0):
+        # zero feature columns change nothing (distances, sums, counts all
+        # identical; padded centroid columns stay 0): pad to the MFMA K
+        # granularity so ANY d takes the fused engines
+        dp = (-d) % 64
+        xp = torch.nn.functional.pad(x, (0, dp))
+        cp = torch.nn.functional.pad(centroids, (0, dp))
+        c_out, inertia = kmeans_iterate(xp, cp, n_iters, comms=comms,
+                                        fp32_mode=fp32_mode,
+                                        chunk_rows=chunk_rows)
+        return c_out[:, :d].contiguous(), inertia
     # X is iteration-invariant: pre-split the bf16 slices and row norms ONCE
     # (the same caching the reference does for row norms in its kmeans)
     use_fused = (x.is_cuda and x.dtype == torch.float32

@@ -103,6 +103,16 @@ def fused_l2nn(x: torch.Tensor, y: torch.Tensor, sqrt: bool = False,
     d = x.shape[1]
 
     if on_gpu(x, y):
+        if (d % 64 != 0 and x.dtype in (torch.float32, torch.bfloat16)
+                and (x.dtype == torch.bfloat16 or fp32_mode in _MODE_NSLICE)):
+            # zero columns leave every pairwise distance unchanged: pad to
+            # the MFMA K granularity and take the fused path (closes the
+            # round-1 "d % 64 falls back to chunked GEMM" perf cliff)
+            dp = (-d) % 64
+            xp = torch.nn.functional.pad(x, (0, dp))
+            yp = torch.nn.functional.pad(y, (0, dp))
+            return fused_l2nn(xp, yp, sqrt=sqrt, fp32_mode=fp32_mode,
+                              chunk_rows=chunk_rows)
         if x.dtype == torch.bfloat16 and d % 64 == 0:
             ext = require_ext()
             xn = x.to(torch.float32).pow(2).sum(dim=1)

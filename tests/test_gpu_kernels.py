@@ -436,6 +436,28 @@ class TestKMeansGpu:
         torch.testing.assert_close(cg.cpu(), cc, rtol=1e-3, atol=1e-3)
         assert abs(ig - ic) / ic < 1e-3
 
+    def test_fast_path_any_d_padding(self, dev, ext):
+        """d % 64 != 0 now pads feature columns with zeros and takes the
+        fused engines (distances unchanged) — kmeans and fused_l2nn must
+        match the CPU oracle at d=100."""
+        from raft_amd.cluster.kmeans import kmeans_iterate
+        from raft_amd.neighbors.fused_l2nn import fused_l2nn
+        from raft_amd.random import make_blobs, RngState
+        x, _, centers = make_blobs(20000, 100, n_clusters=96, cluster_std=0.3,
+                                   state=RngState(seed=14), device=dev)
+        c0 = centers + 0.2
+        cg, ig = kmeans_iterate(x, c0.clone(), 3, fp32_mode="auto")
+        assert cg.shape == (96, 100)
+        cc, ic = kmeans_iterate(x.cpu(), c0.cpu().clone(), 3)
+        torch.testing.assert_close(cg.cpu(), cc, rtol=1e-3, atol=1e-3)
+        assert abs(ig - ic) / ic < 1e-3
+        d1, a1 = fused_l2nn(x, c0)
+        ref = torch.cdist(x.double(), c0.double()) ** 2
+        rd, ra = ref.min(dim=1)
+        assert (a1 == ra).float().mean() > 0.999
+        rel = ((d1.double() - rd).abs() / rd.clamp_min(1e-3)).max()
+        assert float(rel) < 1e-4
+
     @pytest.mark.parametrize("k", [100, 1000])
     def test_fast_iterate_any_k_padding(self, dev, ext, k):
         """ANY k now takes the minimal-dispatch fast path (centroids padded
