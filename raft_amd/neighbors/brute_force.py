@@ -50,10 +50,18 @@ def knn(x: torch.Tensor, queries: torch.Tensor, k: int,
                                     x.shape[0], x.shape[1], k)
         query_chunk = query_chunk or qc
         index_chunk = index_chunk or ic
-    if (queries.is_cuda and metric in (DistanceType.L2Expanded, "sqeuclidean")
-            and x.shape[1] % 64 == 0 and x.shape[0] >= 8 * k
-            and x.dtype in (torch.bfloat16, torch.float32)
-            and not (x.dtype == torch.float32 and fp32_mode == "native")):
+    filterable = (queries.is_cuda
+                  and metric in (DistanceType.L2Expanded, "sqeuclidean")
+                  and x.shape[0] >= 8 * k
+                  and x.dtype in (torch.bfloat16, torch.float32)
+                  and not (x.dtype == torch.float32 and fp32_mode == "native"))
+    if filterable and x.shape[1] % 64 != 0:
+        # zero feature columns change no distance: pad to the MFMA K
+        # granularity and keep the filtered path (round-1 d-cliff)
+        dp = (-x.shape[1]) % 64
+        x = torch.nn.functional.pad(x, (0, dp))
+        queries = torch.nn.functional.pad(queries, (0, dp))
+    if filterable:
         return _knn_gpu_filtered(x, queries, k, fp32_mode, res=res)
     return _knn_tiled(x, queries, k, metric, query_chunk, index_chunk, fp32_mode)
 
@@ -69,12 +77,16 @@ class BruteForceIndex:
     """
 
     def __init__(self, x: torch.Tensor, fp32_mode: str = "auto", res=None):
-        self.x = x
+        self.dim_orig = x.shape[1]
         self.fp32_mode = fp32_mode
         self._filterable = (
-            x.is_cuda and x.shape[1] % 64 == 0
-            and x.dtype in (torch.bfloat16, torch.float32)
+            x.is_cuda and x.dtype in (torch.bfloat16, torch.float32)
             and not (x.dtype == torch.float32 and fp32_mode == "native"))
+        if self._filterable and x.shape[1] % 64 != 0:
+            # pad the MFMA K granularity ONCE at build (zero columns change
+            # no distance); search pads queries to match
+            x = torch.nn.functional.pad(x, (0, (-x.shape[1]) % 64))
+        self.x = x
         if self._filterable:
             self.slices = _slices_of(x, fp32_mode)
             self.xn = _norms(x)
@@ -88,10 +100,13 @@ class BruteForceIndex:
 
     @property
     def dim(self) -> int:
-        return self.x.shape[1]
+        return self.dim_orig
 
     def search(self, queries: torch.Tensor, k: int, res=None):
         """k nearest index rows per query. Returns (dists [q,k], idx [q,k])."""
+        if queries.shape[1] != self.x.shape[1]:
+            queries = torch.nn.functional.pad(
+                queries, (0, self.x.shape[1] - queries.shape[1]))
         if (self._filterable and queries.is_cuda
                 and self.x.shape[0] >= 8 * k):
             from raft_amd.core.resources import get_resources

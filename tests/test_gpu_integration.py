@@ -393,6 +393,24 @@ class TestKnnFp32Exactness:
         assert torch.equal(i1, ir) and torch.equal(i2, ir)
         torch.testing.assert_close(d1, dr)
 
+    def test_knn_any_d_padding(self, dev):
+        """d=100 now stays on the filtered path (zero-padded K) for both
+        one-shot knn and BruteForceIndex — must match the exact fp32 top-k."""
+        from raft_amd.neighbors import brute_force_build, knn
+        torch.manual_seed(3)
+        x = torch.randn(60000, 100, device=dev)
+        q = torch.randn(500, 100, device=dev)
+        d, i = knn(x, q, k=8)
+        idx = brute_force_build(x)
+        assert idx.dim == 100
+        d2, i2 = idx.search(q, 8)
+        ref = ((q*q).sum(1, keepdim=True) + (x*x).sum(1).unsqueeze(0)
+               - 2.0 * (q @ x.t())).clamp_min(0)
+        rd, ri = torch.topk(ref, 8, dim=1, largest=False)
+        assert (i == ri).float().mean() > 0.999
+        assert torch.equal(i, i2)
+        torch.testing.assert_close(d, rd, rtol=1e-4, atol=1e-3)
+
     def test_fp32_native_mode_honored(self, dev):
         from raft_amd.neighbors import knn
         torch.manual_seed(1)
