@@ -117,6 +117,15 @@ def _gemm_xyt(x, y, fp32_mode):
 
 def _l2_squared(x, y, fp32_mode):
     """||x||^2 + ||y||^2 - 2 x.y with fused epilogue on GPU."""
+    if (on_gpu(x, y) and x.shape[1] % 64 != 0
+            and (x.dtype == torch.bfloat16
+                 or (x.dtype == torch.float32
+                     and fp32_mode in ("bf16x2", "bf16x3", "mfma")))):
+        # zero feature columns change no L2 distance: pad to the MFMA K
+        # granularity so any d rides the single-write tile kernel
+        dp = (-x.shape[1]) % 64
+        return _l2_squared(torch.nn.functional.pad(x, (0, dp)),
+                           torch.nn.functional.pad(y, (0, dp)), fp32_mode)
     if on_gpu(x, y) and x.dtype == torch.bfloat16:
         ext = require_ext()
         xn = x.float().pow(2).sum(dim=1)
