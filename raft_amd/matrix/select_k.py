@@ -47,6 +47,18 @@ def select_k(x: torch.Tensor, k: int, select_min: bool = True,
     assert 0 < k <= n, f"k={k} out of range for row length {n}"
 
     if (on_gpu(x) and algo != SelectAlgo.TORCH
+            and x.dtype in (torch.bfloat16, torch.float16) and k <= 2048
+            and n * batch < (1 << 31)):
+        # half dtypes at small k: widen once to fp32 and ride the fp32
+        # radix/warpsort engines — one extra 2x write, vs the generic
+        # engine's histogram + 2 filter re-reads (measured [8192 x 100k]
+        # k=64 bf16: generic 5.5 ms / torch.topk 3.7 -> this route 3.35).
+        # bf16->fp32 is exact, so selection and values are unchanged.
+        vals, idx = select_k(x.float(), k, select_min=select_min, algo=algo,
+                             sorted=sorted)
+        return vals.to(x.dtype), idx
+
+    if (on_gpu(x) and algo != SelectAlgo.TORCH
             and (x.dtype in (torch.float64, torch.bfloat16, torch.float16)
                  or (x.dtype == torch.float32 and k > 2048))):
         # generic native engine: any dtype (64-bit ordinals for fp64),
