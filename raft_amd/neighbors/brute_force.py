@@ -43,6 +43,12 @@ def knn(x: torch.Tensor, queries: torch.Tensor, k: int,
     workspace budget (reference: workspace-resource-driven batching) — set a
     workspace limit via res.set_workspace_limit(nbytes) to bound scratch.
     """
+    if x.dtype == torch.float16 and x.is_cuda:
+        # fp16 -> fp32 is exact: ride the fp32 filtered path (reference
+        # supports half-precision knn; the bf16 split of an exact widening
+        # keeps the same provable-inflation guarantees)
+        return knn(x.float(), queries.float(), k, metric, query_chunk,
+                   index_chunk, fp32_mode, res)
     from raft_amd.core.resources import get_resources
     res = get_resources(res if res is not None else queries.device)
     if query_chunk is None or index_chunk is None:

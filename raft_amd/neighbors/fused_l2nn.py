@@ -103,6 +103,12 @@ def fused_l2nn(x: torch.Tensor, y: torch.Tensor, sqrt: bool = False,
     d = x.shape[1]
 
     if on_gpu(x, y):
+        if x.dtype == torch.float16:
+            # exact widening -> the fp32 engines (incl. verified modes)
+            dmin, amin = fused_l2nn(x.float(), y.float(), sqrt=sqrt,
+                                    fp32_mode=fp32_mode,
+                                    chunk_rows=chunk_rows)
+            return dmin, amin
         if (d % 64 != 0 and x.dtype in (torch.float32, torch.bfloat16)
                 and (x.dtype == torch.bfloat16 or fp32_mode in _MODE_NSLICE)):
             # zero columns leave every pairwise distance unchanged: pad to
