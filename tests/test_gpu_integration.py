@@ -393,6 +393,18 @@ class TestKnnFp32Exactness:
         assert torch.equal(i1, ir) and torch.equal(i2, ir)
         torch.testing.assert_close(d1, dr)
 
+    def test_knn_fp16_input(self, dev):
+        """fp16 index/queries widen exactly to fp32 and ride the filtered
+        path — results must match the fp32 run on the same data."""
+        from raft_amd.neighbors import knn
+        torch.manual_seed(4)
+        x = torch.randn(40000, 128, device=dev).half()
+        q = torch.randn(300, 128, device=dev).half()
+        d, i = knn(x, q, k=8)
+        dr, ir = knn(x.float(), q.float(), k=8)
+        assert torch.equal(i, ir)
+        torch.testing.assert_close(d, dr)
+
     def test_knn_any_d_padding(self, dev):
         """d=100 now stays on the filtered path (zero-padded K) for both
         one-shot knn and BruteForceIndex — must match the exact fp32 top-k."""
